@@ -1,0 +1,365 @@
+"""Parzen-window KDE for TPE, in SoA (structure-of-arrays) layout.
+
+Unlike the reference's per-dimension NamedTuple list
+(reference ``optuna/samplers/_tpe/parzen_estimator.py`` :34 and
+``probability_distributions.py`` :139), the estimator here packs all numerical
+dimensions into dense ``(K, D)`` mu/sigma matrices plus a per-dimension descriptor
+table (kind / low / high / step / adapted bounds). That layout is what the K1/K2
+HIP kernels consume directly — fitting and the S×K×D log-pdf reduction run on
+device with no repacking — and the numpy host path below is the golden reference
+for them.
+
+Behavioral parity with the reference estimator (same kernels, same magic clip,
+same prior kernel, same discrete/log handling):
+ * per-point sigma = max gap to sorted neighbors incl. domain endpoints
+   (endpoints excluded from first/last gap unless consider_endpoints)
+ * magic clip: sigma ∈ [(high-low)/min(100, 1+K), high-low], K = n_obs+1
+ * +1 prior kernel: mu = domain midpoint, sigma = domain width
+ * categorical: prior-smoothed one-hot weight matrix
+ * discrete dims integrate the kernel over [x-step/2, x+step/2]
+ * log dims model the density in log space (no Jacobian; it cancels in the
+   below/above EI ratio)
+"""
+from __future__ import annotations
+
+from typing import Callable, NamedTuple
+
+import numpy as np
+
+from optuna_amd.distributions import (
+    BaseDistribution,
+    CategoricalDistribution,
+    FloatDistribution,
+    IntDistribution,
+)
+from optuna_amd.samplers._tpe import _truncnorm_np as _tn
+
+
+EPS = 1e-12
+
+# Descriptor "kind" codes, shared with the HIP kernels (see _hip/kernels/tpe.hip).
+KIND_CONT = 0  # continuous
+KIND_LOG = 1  # continuous, log domain
+KIND_DISC = 2  # discretized (step)
+KIND_LOG_DISC = 3  # discretized, log domain
+KIND_CAT = 4  # categorical
+
+
+class _ParzenEstimatorParameters(NamedTuple):
+    consider_prior: bool
+    prior_weight: float
+    consider_magic_clip: bool
+    consider_endpoints: bool
+    weights: Callable[[int], np.ndarray]
+    multivariate: bool
+
+
+class _NumericalDims(NamedTuple):
+    """SoA view of all numerical dimensions of the KDE."""
+
+    dim_indices: np.ndarray  # (Dn,) int — position in the search-space order
+    kinds: np.ndarray  # (Dn,) int8 — KIND_* codes (non-categorical)
+    lows: np.ndarray  # (Dn,) original domain low
+    highs: np.ndarray  # (Dn,) original domain high
+    steps: np.ndarray  # (Dn,) step or 0
+    adapted_lows: np.ndarray  # (Dn,) KDE-domain low  (step-widened, logged)
+    adapted_highs: np.ndarray  # (Dn,) KDE-domain high
+    mus: np.ndarray  # (K, Dn)
+    sigmas: np.ndarray  # (K, Dn)
+
+
+class _CategoricalDim(NamedTuple):
+    dim_index: int
+    weights: np.ndarray  # (K, C)
+
+
+class _ParzenEstimator:
+    """Mixture of K product kernels over the search space (see module docstring)."""
+
+    def __init__(
+        self,
+        observations: dict[str, np.ndarray],
+        search_space: dict[str, BaseDistribution],
+        parameters: _ParzenEstimatorParameters,
+        predetermined_weights: np.ndarray | None = None,
+    ) -> None:
+        if parameters.prior_weight < 0:
+            raise ValueError(
+                "A non-negative value must be specified for prior_weight, but got "
+                f"{parameters.prior_weight}."
+            )
+        self._search_space = search_space
+        self._param_names = list(search_space.keys())
+
+        n_obs = len(next(iter(observations.values()))) if observations else 0
+        for name in self._param_names:
+            assert len(observations[name]) == n_obs
+
+        if predetermined_weights is not None:
+            assert n_obs == len(predetermined_weights)
+        weights = (
+            predetermined_weights
+            if predetermined_weights is not None
+            else self._call_weights_func(parameters.weights, n_obs)
+        )
+        if n_obs == 0:
+            weights = np.array([1.0])
+        else:
+            weights = np.append(weights, [parameters.prior_weight])
+        weights = weights / weights.sum()
+        self._weights = weights
+        self._n_kernels = n_obs + 1
+
+        # ---- build per-dimension model, packing numerical dims into SoA ------------
+        num_idx: list[int] = []
+        kinds: list[int] = []
+        lows: list[float] = []
+        highs: list[float] = []
+        steps: list[float] = []
+        alows: list[float] = []
+        ahighs: list[float] = []
+        mus_cols: list[np.ndarray] = []
+        sigmas_cols: list[np.ndarray] = []
+        categoricals: list[_CategoricalDim] = []
+
+        for i, name in enumerate(self._param_names):
+            dist = search_space[name]
+            obs = np.asarray(observations[name], dtype=np.float64)
+            if isinstance(dist, CategoricalDistribution):
+                categoricals.append(
+                    _CategoricalDim(i, self._categorical_weights(obs, len(dist.choices), parameters))
+                )
+                continue
+            assert isinstance(dist, (FloatDistribution, IntDistribution))
+            step = float(dist.step) if dist.step is not None else 0.0
+            is_log = dist.log
+            low = float(dist.low)
+            high = float(dist.high)
+            a_low = low - step / 2 if step else low
+            a_high = high + step / 2 if step else high
+            if is_log:
+                a_low, a_high = np.log(a_low), np.log(a_high)
+            mu, sigma = self._numerical_kernels(
+                np.log(obs) if is_log else obs, a_low, a_high, parameters
+            )
+            num_idx.append(i)
+            if is_log and step:
+                kinds.append(KIND_LOG_DISC)
+            elif is_log:
+                kinds.append(KIND_LOG)
+            elif step:
+                kinds.append(KIND_DISC)
+            else:
+                kinds.append(KIND_CONT)
+            lows.append(low)
+            highs.append(high)
+            steps.append(step)
+            alows.append(a_low)
+            ahighs.append(a_high)
+            mus_cols.append(mu)
+            sigmas_cols.append(sigma)
+
+        self._numerical = _NumericalDims(
+            dim_indices=np.asarray(num_idx, dtype=np.int64),
+            kinds=np.asarray(kinds, dtype=np.int8),
+            lows=np.asarray(lows, dtype=np.float64),
+            highs=np.asarray(highs, dtype=np.float64),
+            steps=np.asarray(steps, dtype=np.float64),
+            adapted_lows=np.asarray(alows, dtype=np.float64),
+            adapted_highs=np.asarray(ahighs, dtype=np.float64),
+            mus=(
+                np.stack(mus_cols, axis=1)
+                if mus_cols
+                else np.empty((self._n_kernels if n_obs else 1, 0))
+            ),
+            sigmas=(
+                np.stack(sigmas_cols, axis=1)
+                if sigmas_cols
+                else np.empty((self._n_kernels if n_obs else 1, 0))
+            ),
+        )
+        self._categoricals = categoricals
+
+    # ---- fitting helpers ------------------------------------------------------------
+
+    @staticmethod
+    def _call_weights_func(weights_func: Callable[[int], np.ndarray], n: int) -> np.ndarray:
+        w = np.asarray(weights_func(n), dtype=np.float64)[:n]
+        if np.any(w < 0):
+            raise ValueError(
+                f"The `weights` function is not allowed to return negative values {w}. "
+                f"The argument of the `weights` function is {n}."
+            )
+        if len(w) > 0 and np.sum(w) <= 0:
+            raise ValueError(
+                f"The `weights` function is not allowed to return all-zero values {w}. "
+                f"The argument of the `weights` function is {n}."
+            )
+        if not np.all(np.isfinite(w)):
+            raise ValueError(
+                f"The `weights` function is not allowed to return infinite or NaN values {w}. "
+                f"The argument of the `weights` function is {n}."
+            )
+        return w
+
+    def _categorical_weights(
+        self, observations: np.ndarray, n_choices: int, parameters: _ParzenEstimatorParameters
+    ) -> np.ndarray:
+        n_obs = len(observations)
+        if n_obs == 0:
+            return np.full((1, n_choices), 1.0 / n_choices)
+        n_kernels = n_obs + 1
+        w = np.full((n_kernels, n_choices), parameters.prior_weight / n_kernels)
+        idx = observations.astype(int)
+        w[np.arange(n_obs), idx] += 1.0
+        row_sums = w.sum(axis=1, keepdims=True)
+        return w / np.where(row_sums == 0, 1.0, row_sums)
+
+    def _numerical_kernels(
+        self,
+        mus: np.ndarray,
+        low: float,
+        high: float,
+        parameters: _ParzenEstimatorParameters,
+    ) -> tuple[np.ndarray, np.ndarray]:
+        """Per-point sigma = max gap to sorted neighbors; +1 prior kernel appended."""
+        n = len(mus)
+        if n == 0:
+            return np.array([0.5 * (low + high)]), np.array([high - low])
+
+        order = np.argsort(mus)
+        padded = np.empty(n + 2, dtype=np.float64)
+        padded[0] = low
+        padded[1:-1] = mus[order]
+        padded[-1] = high
+        gaps_left = padded[1:-1] - padded[:-2]
+        gaps_right = padded[2:] - padded[1:-1]
+        sorted_sigmas = np.maximum(gaps_left, gaps_right)
+        if not parameters.consider_endpoints and n >= 2:
+            sorted_sigmas[0] = padded[2] - padded[1]
+            sorted_sigmas[-1] = padded[-2] - padded[-3]
+        sigmas = sorted_sigmas[np.argsort(order)]
+
+        maxsigma = high - low
+        if parameters.consider_magic_clip:
+            minsigma = (high - low) / min(100.0, 1.0 + (n + 1))
+        else:
+            minsigma = EPS
+        sigmas = np.clip(sigmas, minsigma, maxsigma)
+
+        mus_out = np.append(mus, [0.5 * (low + high)])
+        sigmas_out = np.append(sigmas, [high - low])
+        return mus_out, sigmas_out
+
+    # ---- sampling / scoring ---------------------------------------------------------
+
+    @property
+    def weights(self) -> np.ndarray:
+        return self._weights
+
+    @property
+    def n_dims(self) -> int:
+        return len(self._param_names)
+
+    def sample(self, rng: np.random.RandomState, size: int) -> dict[str, np.ndarray]:
+        samples = self._sample_array(rng, size)
+        return {name: samples[:, i] for i, name in enumerate(self._param_names)}
+
+    def log_pdf(self, samples_dict: dict[str, np.ndarray]) -> np.ndarray:
+        x = np.column_stack([samples_dict[name] for name in self._param_names]) if (
+            self._param_names
+        ) else np.empty((0, 0))
+        return self._log_pdf_array(x)
+
+    # Array-level interfaces (used by the HIP dispatch and bench harness).
+
+    def _sample_array(self, rng: np.random.RandomState, size: int) -> np.ndarray:
+        active = rng.choice(len(self._weights), p=self._weights, size=size)
+        out = np.empty((size, self.n_dims), dtype=np.float64)
+
+        num = self._numerical
+        if num.mus.shape[1]:
+            mus = num.mus[active, :]  # (S, Dn)
+            sigmas = num.sigmas[active, :]
+            a = (num.adapted_lows[np.newaxis, :] - mus) / sigmas
+            b = (num.adapted_highs[np.newaxis, :] - mus) / sigmas
+            vals = _tn.rvs(a, b, loc=mus, scale=sigmas, random_state=rng)
+            is_log = (num.kinds == KIND_LOG) | (num.kinds == KIND_LOG_DISC)
+            vals[:, is_log] = np.exp(vals[:, is_log])
+            has_step = num.steps > 0
+            if np.any(has_step):
+                lo = num.lows[has_step]
+                st = num.steps[has_step]
+                hi = num.highs[has_step]
+                vals[:, has_step] = np.clip(
+                    lo + np.round((vals[:, has_step] - lo) / st) * st, lo, hi
+                )
+            out[:, num.dim_indices] = vals
+
+        for cat in self._categoricals:
+            active_w = cat.weights[active, :]  # (S, C)
+            q = rng.rand(size)
+            cum = np.cumsum(active_w, axis=-1)
+            cum[:, -1] = 1.0
+            out[:, cat.dim_index] = np.sum(cum < q[:, np.newaxis], axis=-1)
+        return out
+
+    def _log_pdf_array(self, x: np.ndarray) -> np.ndarray:
+        """(S, D) samples → (S,) log of the mixture density."""
+        S = x.shape[0]
+        K = len(self._weights)
+        if S == 0:
+            return np.empty(0)
+        log_terms = np.zeros((S, K), dtype=np.float64)  # per-kernel product over dims
+
+        num = self._numerical
+        Dn = num.mus.shape[1]
+        if Dn:
+            xv = x[:, num.dim_indices]  # (S, Dn)
+            is_log = (num.kinds == KIND_LOG) | (num.kinds == KIND_LOG_DISC)
+            xv = np.where(is_log[np.newaxis, :], np.log(np.maximum(xv, EPS)), xv)
+            is_disc = num.steps > 0
+            mus = num.mus  # (K, Dn)
+            sigmas = num.sigmas
+
+            # Continuous dims: truncnorm logpdf, summed over dims.
+            if np.any(~is_disc):
+                m = mus[:, ~is_disc]  # (K, Dc)
+                s = sigmas[:, ~is_disc]
+                a = (num.adapted_lows[~is_disc] - m) / s
+                b = (num.adapted_highs[~is_disc] - m) / s
+                # (S, K, Dc) broadcast
+                log_terms += _tn.logpdf(
+                    xv[:, np.newaxis, ~is_disc], a[np.newaxis], b[np.newaxis], loc=m, scale=s
+                ).sum(axis=-1)
+
+            # Discrete dims: integrated kernel mass over the step cell.
+            for j in np.nonzero(is_disc)[0]:
+                half = num.steps[j] / 2
+                xj_raw = x[:, num.dim_indices[j]]
+                if is_log[j]:
+                    left = np.log(xj_raw - half)
+                    right = np.log(xj_raw + half)
+                else:
+                    left = xj_raw - half
+                    right = xj_raw + half
+                m = mus[:, j]
+                s = sigmas[:, j]
+                cell = _tn._log_gauss_mass(
+                    (left[:, np.newaxis] - m) / s, (right[:, np.newaxis] - m) / s
+                )
+                total = _tn._log_gauss_mass(
+                    (num.adapted_lows[j] - m) / s, (num.adapted_highs[j] - m) / s
+                )
+                log_terms += cell - total[np.newaxis, :]
+
+        for cat in self._categoricals:
+            xi = x[:, cat.dim_index].astype(np.int64)
+            with np.errstate(divide="ignore"):
+                log_terms += np.log(cat.weights.T[xi, :])
+
+        log_terms += np.log(self._weights)[np.newaxis, :]
+        m = log_terms.max(axis=1)
+        m[np.isneginf(m)] = 0.0
+        with np.errstate(divide="ignore"):
+            return np.log(np.exp(log_terms - m[:, None]).sum(axis=1)) + m

@@ -1,0 +1,512 @@
+"""TPESampler — the default sampler and the framework's headline compute path.
+
+Algorithm (parity with reference ``optuna/samplers/_tpe/sampler.py`` :87-880):
+split history into below/above by ``gamma(n)``; fit Parzen KDEs to each; draw
+``n_ei_candidates`` samples from the below-KDE; rank by
+``log l(x) − log g(x)``; return the argmax. Multivariate by default for
+single-objective studies; ``group=True`` decomposes conditional spaces;
+``constant_liar`` treats RUNNING trials as bad and shares relative params via
+chunked system attrs (2045-char chunks, RDB-compatible).
+
+On a GPU box the KDE fit and the S×K×D log-pdf/EI reduction run in the
+``optuna_amd._hip`` K1/K2 kernels over the SoA estimator arrays; the numpy path
+in ``parzen.py`` is the host fallback and golden reference.
+"""
+from __future__ import annotations
+
+import json
+import math
+from typing import TYPE_CHECKING, Any, Callable, Sequence
+
+import numpy as np
+
+from optuna_amd import logging as _logging
+from optuna_amd.distributions import BaseDistribution
+from optuna_amd.samplers._base import (
+    _INDEPENDENT_SAMPLING_WARNING_TEMPLATE,
+    BaseSampler,
+    _process_constraints_after_trial,
+)
+from optuna_amd.samplers._lazy_random_state import LazyRandomState
+from optuna_amd.samplers._random import RandomSampler
+from optuna_amd.samplers._tpe.parzen import _ParzenEstimator, _ParzenEstimatorParameters
+from optuna_amd.search_space import IntersectionSearchSpace
+from optuna_amd.search_space.group_decomposed import (
+    _GroupDecomposedSearchSpace,
+    _SearchSpaceGroup,
+)
+from optuna_amd.study._multi_objective import _fast_non_domination_rank
+from optuna_amd.study._study_direction import StudyDirection
+from optuna_amd.trial import FrozenTrial, TrialState
+
+
+if TYPE_CHECKING:
+    from optuna_amd.study import Study
+
+EPS = 1e-12
+_logger = _logging.get_logger(__name__)
+
+_RELATIVE_PARAMS_KEY = "tpe:relative_params"
+# RDB system-attr values must stay under 2046 chars; long payloads are chunked.
+_SYSTEM_ATTR_MAX_LENGTH = 2045
+
+
+def default_gamma(x: int) -> int:
+    return min(math.ceil(0.1 * x), 25)
+
+
+def default_gamma_multiobjective(x: int) -> int:
+    return math.ceil(0.1 * x)
+
+
+def hyperopt_default_gamma(x: int) -> int:
+    return min(math.ceil(0.25 * math.sqrt(x)), 25)
+
+
+def default_weights(x: int) -> np.ndarray:
+    """Old trials ramp linearly; the most recent 25 have full weight."""
+    if x == 0:
+        return np.asarray([])
+    if x < 25:
+        return np.ones(x)
+    ramp = np.linspace(1.0 / x, 1.0, num=x - 25)
+    flat = np.ones(25)
+    return np.concatenate([ramp, flat], axis=0)
+
+
+class TPESampler(BaseSampler):
+    """Tree-structured Parzen Estimator sampler (see module docstring)."""
+
+    def __init__(
+        self,
+        consider_prior: bool = True,
+        prior_weight: float = 1.0,
+        consider_magic_clip: bool = True,
+        consider_endpoints: bool = False,
+        n_startup_trials: int = 10,
+        n_ei_candidates: int = 24,
+        gamma: Callable[[int], int] | None = None,
+        weights: Callable[[int], np.ndarray] = default_weights,
+        seed: int | None = None,
+        *,
+        multivariate: bool | None = None,
+        group: bool = False,
+        warn_independent_sampling: bool = True,
+        constant_liar: bool = False,
+        constraints_func: Callable[[FrozenTrial], Sequence[float]] | None = None,
+    ) -> None:
+        self._parzen_estimator_parameters = _ParzenEstimatorParameters(
+            consider_prior,
+            prior_weight,
+            consider_magic_clip,
+            consider_endpoints,
+            weights,
+            multivariate if multivariate is not None else False,
+        )
+        self._n_startup_trials = n_startup_trials
+        self._n_ei_candidates = n_ei_candidates
+        self._gamma = gamma
+        self._warn_independent_sampling = warn_independent_sampling
+        self._rng = LazyRandomState(seed)
+        self._random_sampler = RandomSampler(seed=seed)
+        self._multivariate = multivariate
+        self._group = group
+        self._group_decomposed_search_space: _GroupDecomposedSearchSpace | None = None
+        self._search_space_group: _SearchSpaceGroup | None = None
+        self._search_space = IntersectionSearchSpace(include_pruned=True)
+        self._constant_liar = constant_liar
+        self._constraints_func = constraints_func
+        # Overridable for customization (reference sampler.py keeps the same hook).
+        self._parzen_estimator_cls = _ParzenEstimator
+
+        if group:
+            if multivariate is False:
+                raise ValueError(
+                    "``group`` option can only be enabled when ``multivariate`` is enabled."
+                )
+            self._group_decomposed_search_space = _GroupDecomposedSearchSpace(True)
+
+    def reseed_rng(self) -> None:
+        self._rng.rng.seed()
+        self._random_sampler.reseed_rng()
+
+    def _is_multivariate(self, study: "Study") -> bool:
+        if self._multivariate is not None:
+            return self._multivariate
+        if self._group:
+            return True
+        # Multivariate for single-objective, independent for multi-objective.
+        return not study._is_multi_objective()
+
+    # ---- search-space inference -----------------------------------------------------
+
+    def infer_relative_search_space(
+        self, study: "Study", trial: FrozenTrial
+    ) -> dict[str, BaseDistribution]:
+        if not self._is_multivariate(study):
+            return {}
+
+        search_space: dict[str, BaseDistribution] = {}
+        if self._group:
+            assert self._group_decomposed_search_space is not None
+            self._search_space_group = self._group_decomposed_search_space.calculate(study)
+            for sub_space in self._search_space_group.search_spaces:
+                for name, distribution in sorted(sub_space.items()):
+                    if distribution.single():
+                        continue
+                    search_space[name] = distribution
+            return search_space
+
+        for name, distribution in self._search_space.calculate(study).items():
+            if distribution.single():
+                continue
+            search_space[name] = distribution
+        return search_space
+
+    # ---- relative sampling ----------------------------------------------------------
+
+    def sample_relative(
+        self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
+    ) -> dict[str, Any]:
+        if self._group:
+            assert self._search_space_group is not None
+            params: dict[str, Any] = {}
+            for sub_space in self._search_space_group.search_spaces:
+                _search_space = {}
+                for name, distribution in sorted(sub_space.items()):
+                    if distribution.single():
+                        continue
+                    if name not in search_space:
+                        # PartialFixedSampler may shrink the space below inference.
+                        continue
+                    _search_space[name] = distribution
+                params.update(self._sample_relative(study, trial, _search_space))
+        else:
+            params = self._sample_relative(study, trial, search_space)
+
+        if params != {} and self._constant_liar:
+            # Publish the sampled relative params so other workers' liar splits see
+            # this RUNNING trial's position in the space.
+            params_str = json.dumps(params)
+            for i in range(0, len(params_str), _SYSTEM_ATTR_MAX_LENGTH):
+                study._storage.set_trial_system_attr(
+                    trial._trial_id,
+                    f"{_RELATIVE_PARAMS_KEY}:{i // _SYSTEM_ATTR_MAX_LENGTH}",
+                    params_str[i : i + _SYSTEM_ATTR_MAX_LENGTH],
+                )
+        return params
+
+    def _sample_relative(
+        self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
+    ) -> dict[str, Any]:
+        if search_space == {}:
+            return {}
+        states = (TrialState.COMPLETE, TrialState.PRUNED)
+        trials = study._get_trials(deepcopy=False, states=states, use_cache=True)
+        if len(trials) < self._n_startup_trials:
+            return {}
+        return self._sample(study, trial, search_space)
+
+    def sample_independent(
+        self,
+        study: "Study",
+        trial: FrozenTrial,
+        param_name: str,
+        param_distribution: BaseDistribution,
+    ) -> Any:
+        states = (TrialState.COMPLETE, TrialState.PRUNED)
+        trials = study._get_trials(deepcopy=False, states=states, use_cache=True)
+        if len(trials) < self._n_startup_trials:
+            return self._random_sampler.sample_independent(
+                study, trial, param_name, param_distribution
+            )
+
+        if self._warn_independent_sampling and self._is_multivariate(study):
+            if any(param_name in t.params for t in trials):
+                _logger.warning(
+                    _INDEPENDENT_SAMPLING_WARNING_TEMPLATE.format(
+                        param_name=param_name,
+                        trial_number=trial.number,
+                        sampler_name=self.__class__.__name__,
+                        fallback_name=self._random_sampler.__class__.__name__,
+                        reason=(
+                            "multivariate=True,group=False does not support dynamic "
+                            "search spaces (multivariate=True,group=True does)"
+                        ),
+                    )
+                )
+
+        return self._sample(study, trial, {param_name: param_distribution})[param_name]
+
+    # ---- the compute core -----------------------------------------------------------
+
+    def _get_params(self, trial: FrozenTrial, study: "Study") -> dict[str, Any]:
+        """Params incl. constant-liar shared relative params of RUNNING trials."""
+        if trial.state.is_finished() or not self._is_multivariate(study):
+            return trial.params
+        params_strs = []
+        i = 0
+        while params_str_i := trial.system_attrs.get(f"{_RELATIVE_PARAMS_KEY}:{i}"):
+            params_strs.append(params_str_i)
+            i += 1
+        if len(params_strs) == 0:
+            return trial.params
+        try:
+            params = json.loads("".join(params_strs))
+        except json.JSONDecodeError:
+            # Concurrent chunk writes can race; fall back to committed params.
+            return trial.params
+        params.update(trial.params)
+        return params
+
+    def _get_internal_repr(
+        self,
+        trials: list[FrozenTrial],
+        search_space: dict[str, BaseDistribution],
+        study: "Study",
+    ) -> dict[str, np.ndarray]:
+        values: dict[str, list[float]] = {name: [] for name in search_space}
+        for trial in trials:
+            params = self._get_params(trial, study)
+            if search_space.keys() <= params.keys():
+                for name, distribution in search_space.items():
+                    values[name].append(distribution.to_internal_repr(params[name]))
+        return {k: np.asarray(v) for k, v in values.items()}
+
+    def _sample(
+        self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
+    ) -> dict[str, Any]:
+        if self._constant_liar:
+            states = [TrialState.COMPLETE, TrialState.PRUNED, TrialState.RUNNING]
+        else:
+            states = [TrialState.COMPLETE, TrialState.PRUNED]
+        use_cache = not self._constant_liar
+        trials = study._get_trials(deepcopy=False, states=states, use_cache=use_cache)
+        if self._constant_liar:
+            trials = [t for t in trials if t.number != trial.number]
+
+        if self._gamma is None:
+            self._gamma = (
+                default_gamma if len(study.directions) <= 1 else default_gamma_multiobjective
+            )
+        n = sum(t.state != TrialState.RUNNING for t in trials)
+
+        below_trials, above_trials = _split_trials(study, trials, self._gamma(n))
+
+        mpe_below = self._build_parzen_estimator(study, search_space, below_trials, True)
+        mpe_above = self._build_parzen_estimator(study, search_space, above_trials, False)
+
+        samples_below = mpe_below.sample(self._rng.rng, self._n_ei_candidates)
+        acq_func_vals = self._compute_acquisition_func(samples_below, mpe_below, mpe_above)
+        ret = TPESampler._compare(samples_below, acq_func_vals)
+
+        for param_name, dist in search_space.items():
+            ret[param_name] = dist.to_external_repr(ret[param_name])
+        return ret
+
+    def _build_parzen_estimator(
+        self,
+        study: "Study",
+        search_space: dict[str, BaseDistribution],
+        trials: list[FrozenTrial],
+        handle_below: bool,
+    ) -> _ParzenEstimator:
+        observations = self._get_internal_repr(trials, search_space, study)
+        if handle_below and study._is_multi_objective():
+            n_below = len(next(iter(observations.values()))) if observations else 0
+            weights_below = np.ones(n_below)
+            mpe = self._parzen_estimator_cls(
+                observations, search_space, self._parzen_estimator_parameters, weights_below
+            )
+        else:
+            mpe = self._parzen_estimator_cls(
+                observations, search_space, self._parzen_estimator_parameters
+            )
+        if not isinstance(mpe, _ParzenEstimator):
+            raise RuntimeError("_parzen_estimator_cls must override _ParzenEstimator.")
+        return mpe
+
+    def _compute_acquisition_func(
+        self,
+        samples: dict[str, np.ndarray],
+        mpe_below: _ParzenEstimator,
+        mpe_above: _ParzenEstimator,
+    ) -> np.ndarray:
+        return mpe_below.log_pdf(samples) - mpe_above.log_pdf(samples)
+
+    @classmethod
+    def _compare(
+        cls, samples: dict[str, np.ndarray], acquisition_func_vals: np.ndarray
+    ) -> dict[str, int | float]:
+        sample_size = next(iter(samples.values())).size
+        if sample_size == 0:
+            raise ValueError(f"The size of `samples` must be positive, but got {sample_size}.")
+        if sample_size != acquisition_func_vals.size:
+            raise ValueError(
+                "The sizes of `samples` and `acquisition_func_vals` must be same, but got "
+                f"({sample_size}, {acquisition_func_vals.size})."
+            )
+        best_idx = np.argmax(acquisition_func_vals)
+        return {k: v[best_idx].item() for k, v in samples.items()}
+
+    @staticmethod
+    def hyperopt_parameters() -> dict[str, Any]:
+        """Default parameters of hyperopt v0.1.2 (reference sampler.py:677-720)."""
+        return {
+            "consider_prior": True,
+            "prior_weight": 1.0,
+            "consider_magic_clip": True,
+            "consider_endpoints": False,
+            "n_startup_trials": 20,
+            "n_ei_candidates": 24,
+            "gamma": hyperopt_default_gamma,
+            "weights": default_weights,
+        }
+
+    def before_trial(self, study: "Study", trial: FrozenTrial) -> None:
+        self._random_sampler.before_trial(study, trial)
+
+    def after_trial(
+        self,
+        study: "Study",
+        trial: FrozenTrial,
+        state: TrialState,
+        values: Sequence[float] | None,
+    ) -> None:
+        assert state in [TrialState.COMPLETE, TrialState.FAIL, TrialState.PRUNED]
+        if self._constraints_func is not None:
+            _process_constraints_after_trial(self._constraints_func, study, trial, state)
+        self._random_sampler.after_trial(study, trial, state, values)
+
+
+# ----------------------------------------------------------------------------------
+# History splitting (below/above)
+# ----------------------------------------------------------------------------------
+
+
+def _get_reference_point(loss_vals: np.ndarray) -> np.ndarray:
+    worst_point = np.max(loss_vals, axis=0)
+    reference_point = np.maximum(1.1 * worst_point, 0.9 * worst_point)
+    reference_point[reference_point == 0] = EPS
+    return reference_point
+
+
+def _split_trials(
+    study: "Study", trials: list[FrozenTrial], n_below: int
+) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
+    complete_trials = []
+    pruned_trials = []
+    running_trials = []
+    infeasible_trials = []
+
+    for t in trials:
+        if t.state == TrialState.RUNNING:
+            # RUNNING first: their constraint attrs are not yet written.
+            running_trials.append(t)
+        elif _get_infeasible_trial_score(t) > 0:
+            infeasible_trials.append(t)
+        elif t.state == TrialState.COMPLETE:
+            complete_trials.append(t)
+        elif t.state == TrialState.PRUNED:
+            pruned_trials.append(t)
+        else:
+            raise AssertionError(f"unexpected state {t.state}")
+
+    below_complete, above_complete = _split_complete_trials(complete_trials, study, n_below)
+    n_below = max(0, n_below - len(below_complete))
+    below_pruned, above_pruned = _split_pruned_trials(pruned_trials, study, n_below)
+    n_below = max(0, n_below - len(below_pruned))
+    below_infeasible, above_infeasible = _split_infeasible_trials(infeasible_trials, n_below)
+
+    below = below_complete + below_pruned + below_infeasible
+    above = above_complete + above_pruned + above_infeasible + running_trials
+    below.sort(key=lambda t: t.number)
+    above.sort(key=lambda t: t.number)
+    return below, above
+
+
+def _split_complete_trials(
+    trials: Sequence[FrozenTrial], study: "Study", n_below: int
+) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
+    n_below = min(n_below, len(trials))
+    if len(study.directions) <= 1:
+        reverse = study.direction == StudyDirection.MAXIMIZE
+        sorted_trials = sorted(trials, key=lambda t: t.value, reverse=reverse)  # type: ignore[arg-type,return-value]
+        return sorted_trials[:n_below], sorted_trials[n_below:]
+    return _split_complete_trials_multi_objective(trials, study, n_below)
+
+
+def _split_complete_trials_multi_objective(
+    trials: Sequence[FrozenTrial], study: "Study", n_below: int
+) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
+    from optuna_amd._hypervolume.hssp import _solve_hssp
+
+    if n_below == 0:
+        return [], list(trials)
+    if n_below == len(trials):
+        return list(trials), []
+    assert 0 < n_below < len(trials)
+
+    lvals = np.array([t.values for t in trials])
+    lvals *= np.array(
+        [-1.0 if d == StudyDirection.MAXIMIZE else 1.0 for d in study.directions]
+    )
+    nondomination_ranks = _fast_non_domination_rank(lvals, n_below=n_below)
+    ranks, rank_counts = np.unique(nondomination_ranks, return_counts=True)
+    last_rank_before_tiebreak = int(
+        np.max(ranks[np.cumsum(rank_counts) <= n_below], initial=-1)
+    )
+    indices = np.arange(len(trials))
+    indices_below = indices[nondomination_ranks <= last_rank_before_tiebreak]
+
+    if indices_below.size < n_below:
+        # Tie-break the boundary front with greedy hypervolume subset selection.
+        need_tiebreak = nondomination_ranks == last_rank_before_tiebreak + 1
+        rank_i_lvals = lvals[need_tiebreak]
+        subset_size = n_below - indices_below.size
+        selected = _solve_hssp(
+            rank_i_lvals,
+            indices[need_tiebreak],
+            subset_size,
+            _get_reference_point(rank_i_lvals),
+        )
+        indices_below = np.append(indices_below, selected)
+
+    below_set = set(indices_below.tolist())
+    below = [trials[i] for i in range(len(trials)) if i in below_set]
+    above = [trials[i] for i in range(len(trials)) if i not in below_set]
+    return below, above
+
+
+def _get_pruned_trial_score(trial: FrozenTrial, study: "Study") -> tuple[float, float]:
+    if len(trial.intermediate_values) > 0:
+        step, intermediate_value = max(trial.intermediate_values.items())
+        if math.isnan(intermediate_value):
+            return -step, float("inf")
+        if study.direction == StudyDirection.MINIMIZE:
+            return -step, intermediate_value
+        return -step, -intermediate_value
+    return 1, 0.0
+
+
+def _split_pruned_trials(
+    trials: Sequence[FrozenTrial], study: "Study", n_below: int
+) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
+    n_below = min(n_below, len(trials))
+    sorted_trials = sorted(trials, key=lambda t: _get_pruned_trial_score(t, study))
+    return sorted_trials[:n_below], sorted_trials[n_below:]
+
+
+def _get_infeasible_trial_score(trial: FrozenTrial) -> float:
+    constraints = trial.system_attrs.get("constraints")
+    if constraints is None:
+        return 0.0
+    return sum(v for v in constraints if v > 0)
+
+
+def _split_infeasible_trials(
+    trials: Sequence[FrozenTrial], n_below: int
+) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
+    n_below = min(n_below, len(trials))
+    sorted_trials = sorted(trials, key=_get_infeasible_trial_score)
+    return sorted_trials[:n_below], sorted_trials[n_below:]
