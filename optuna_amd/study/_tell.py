@@ -129,6 +129,16 @@ def _tell_with_warning(
     if values is not None:
         values = [float(v) for v in values]
 
+    if warning_message is not None:
+        # Record while the trial is still mutable (RUNNING).
+        study._storage.set_trial_system_attr(
+            frozen_trial._trial_id, "study:tell_warning", warning_message
+        )
+        if not suppress_warning:
+            import warnings
+
+            warnings.warn(warning_message)
+
     try:
         # Hyperband needs samplers to observe the bracket-filtered study.
         from optuna_amd.pruners import _filter_study
@@ -138,13 +148,4 @@ def _tell_with_warning(
     finally:
         study._storage.set_trial_state_values(frozen_trial._trial_id, state, values)
 
-    frozen_trial = copy.deepcopy(study._storage.get_trial(frozen_trial._trial_id))
-
-    if warning_message is not None and not suppress_warning:
-        import warnings
-
-        warnings.warn(warning_message)
-        study._storage.set_trial_system_attr(
-            frozen_trial._trial_id, "study:tell_warning", warning_message
-        )
-    return frozen_trial
+    return copy.deepcopy(study._storage.get_trial(frozen_trial._trial_id))

@@ -1,0 +1,330 @@
+"""Storage conformance suite: subclass and provide a ``storage`` fixture.
+
+Any backend (in-memory, RDB, journal, gRPC proxy, RCCL table) must pass every
+test here; this is the contract that makes backends interchangeable, including
+the CAS/ownership semantics the distributed path relies on.
+
+Parity (pattern): reference ``optuna/testing/pytest_storages.py`` (StorageTestCase
+:32-36 and its ~50 behavioral tests).
+"""
+from __future__ import annotations
+
+import copy
+import math
+import threading
+from datetime import datetime
+
+import pytest
+
+from optuna_amd.distributions import (
+    CategoricalDistribution,
+    FloatDistribution,
+    IntDistribution,
+)
+from optuna_amd.exceptions import DuplicatedStudyError, UpdateFinishedTrialError
+from optuna_amd.storages import BaseStorage
+from optuna_amd.study._study_direction import StudyDirection
+from optuna_amd.trial import FrozenTrial, TrialState, create_trial
+
+
+MINIMIZE = [StudyDirection.MINIMIZE]
+
+
+class StorageTestCase:
+    """Behavioral contract tests for BaseStorage implementations."""
+
+    @pytest.fixture
+    def storage(self) -> BaseStorage:
+        raise NotImplementedError
+
+    # ---- studies --------------------------------------------------------------------
+
+    def test_create_new_study(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        studies = storage.get_all_studies()
+        assert len(studies) == 1
+        assert studies[0]._study_id == study_id
+        assert storage.get_study_directions(study_id) == MINIMIZE
+
+    def test_create_new_study_with_name(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE, study_name="my-study")
+        assert storage.get_study_name_from_id(study_id) == "my-study"
+        assert storage.get_study_id_from_name("my-study") == study_id
+
+    def test_create_new_study_duplicated_name(self, storage: BaseStorage) -> None:
+        storage.create_new_study(MINIMIZE, study_name="dup")
+        with pytest.raises(DuplicatedStudyError):
+            storage.create_new_study(MINIMIZE, study_name="dup")
+
+    def test_create_new_study_unique_id(self, storage: BaseStorage) -> None:
+        id0 = storage.create_new_study(MINIMIZE)
+        id1 = storage.create_new_study(MINIMIZE)
+        storage.delete_study(id1)
+        id2 = storage.create_new_study(MINIMIZE)
+        assert len({id0, id1, id2}) == 3
+
+    def test_delete_study(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        storage.create_new_trial(study_id)
+        storage.delete_study(study_id)
+        with pytest.raises(KeyError):
+            storage.get_study_name_from_id(study_id)
+
+    def test_delete_study_invalid_id(self, storage: BaseStorage) -> None:
+        with pytest.raises(KeyError):
+            storage.delete_study(128)
+
+    def test_get_study_id_from_name_missing(self, storage: BaseStorage) -> None:
+        with pytest.raises(KeyError):
+            storage.get_study_id_from_name("no-such-study")
+
+    def test_multi_objective_directions(self, storage: BaseStorage) -> None:
+        directions = [StudyDirection.MINIMIZE, StudyDirection.MAXIMIZE]
+        study_id = storage.create_new_study(directions)
+        assert storage.get_study_directions(study_id) == directions
+
+    def test_study_user_attrs(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        storage.set_study_user_attr(study_id, "key", [1, 2, "three"])
+        assert storage.get_study_user_attrs(study_id) == {"key": [1, 2, "three"]}
+
+    def test_study_system_attrs(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        storage.set_study_system_attr(study_id, "meta", {"a": 1})
+        assert storage.get_study_system_attrs(study_id)["meta"] == {"a": 1}
+
+    # ---- trials ---------------------------------------------------------------------
+
+    def test_create_new_trial_number_sequence(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        ids = [storage.create_new_trial(study_id) for _ in range(5)]
+        numbers = [storage.get_trial_number_from_id(tid) for tid in ids]
+        assert numbers == list(range(5))
+        assert len(set(ids)) == 5
+
+    def test_create_new_trial_initial_state(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        trial = storage.get_trial(trial_id)
+        assert trial.state == TrialState.RUNNING
+        assert trial.values is None
+        assert trial.params == {}
+        assert trial.datetime_start is not None
+        assert trial.datetime_complete is None
+
+    def test_create_new_trial_template(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        template = create_trial(
+            state=TrialState.COMPLETE,
+            value=10.0,
+            params={"x": 0.5},
+            distributions={"x": FloatDistribution(0, 1)},
+            user_attrs={"ua": 1},
+            system_attrs={"sa": 2},
+            intermediate_values={0: 1.0},
+        )
+        trial_id = storage.create_new_trial(study_id, template_trial=template)
+        trial = storage.get_trial(trial_id)
+        assert trial.state == TrialState.COMPLETE
+        assert trial.value == 10.0
+        assert trial.params == {"x": 0.5}
+        assert trial.user_attrs == {"ua": 1}
+        assert trial.system_attrs == {"sa": 2}
+        assert trial.intermediate_values == {0: 1.0}
+
+    def test_set_trial_param(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        fd = FloatDistribution(0.0, 2.0)
+        cd = CategoricalDistribution(("a", "b", None))
+        idist = IntDistribution(1, 10, log=True)
+        storage.set_trial_param(trial_id, "f", 1.5, fd)
+        storage.set_trial_param(trial_id, "c", 2, cd)
+        storage.set_trial_param(trial_id, "i", 4.0, idist)
+        assert storage.get_trial_param(trial_id, "f") == 1.5
+        assert storage.get_trial_param(trial_id, "c") == 2
+        trial = storage.get_trial(trial_id)
+        assert trial.params == {"f": 1.5, "c": None, "i": 4}
+        assert trial.distributions == {"f": fd, "c": cd, "i": idist}
+
+    def test_set_trial_param_on_finished_trial(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (1.0,))
+        with pytest.raises(UpdateFinishedTrialError):
+            storage.set_trial_param(trial_id, "x", 0.5, FloatDistribution(0, 1))
+
+    def test_set_trial_state_values_complete(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        assert storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (0.5,))
+        trial = storage.get_trial(trial_id)
+        assert trial.state == TrialState.COMPLETE
+        assert trial.value == 0.5
+        assert trial.datetime_complete is not None
+
+    def test_waiting_to_running_cas(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        template = create_trial(state=TrialState.WAITING)
+        trial_id = storage.create_new_trial(study_id, template_trial=template)
+        assert storage.set_trial_state_values(trial_id, TrialState.RUNNING)
+        # Second claim must fail.
+        assert not storage.set_trial_state_values(trial_id, TrialState.RUNNING)
+
+    def test_update_finished_trial_raises(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (0.0,))
+        with pytest.raises(UpdateFinishedTrialError):
+            storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (1.0,))
+        with pytest.raises(UpdateFinishedTrialError):
+            storage.set_trial_intermediate_value(trial_id, 0, 1.0)
+        with pytest.raises(UpdateFinishedTrialError):
+            storage.set_trial_user_attr(trial_id, "k", 1)
+        with pytest.raises(UpdateFinishedTrialError):
+            storage.set_trial_system_attr(trial_id, "k", 1)
+
+    def test_set_trial_intermediate_value(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_intermediate_value(trial_id, 0, 0.3)
+        storage.set_trial_intermediate_value(trial_id, 2, 0.4)
+        storage.set_trial_intermediate_value(trial_id, 0, 0.5)  # overwrite
+        trial = storage.get_trial(trial_id)
+        assert trial.intermediate_values == {0: 0.5, 2: 0.4}
+
+    def test_intermediate_value_nan_inf(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_intermediate_value(trial_id, 0, float("inf"))
+        storage.set_trial_intermediate_value(trial_id, 1, float("-inf"))
+        storage.set_trial_intermediate_value(trial_id, 2, float("nan"))
+        values = storage.get_trial(trial_id).intermediate_values
+        assert values[0] == float("inf")
+        assert values[1] == float("-inf")
+        assert math.isnan(values[2])
+
+    def test_trial_value_inf(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (float("-inf"),))
+        assert storage.get_trial(trial_id).value == float("-inf")
+
+    def test_trial_user_and_system_attrs(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_user_attr(trial_id, "u", {"nested": [1, 2]})
+        storage.set_trial_system_attr(trial_id, "s", "v")
+        trial = storage.get_trial(trial_id)
+        assert trial.user_attrs == {"u": {"nested": [1, 2]}}
+        assert trial.system_attrs == {"s": "v"}
+        assert storage.get_trial_user_attrs(trial_id) == {"u": {"nested": [1, 2]}}
+        assert storage.get_trial_system_attrs(trial_id) == {"s": "v"}
+
+    def test_get_trial_invalid_id(self, storage: BaseStorage) -> None:
+        with pytest.raises(KeyError):
+            storage.get_trial(2**31)
+
+    def test_get_all_trials(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        other_id = storage.create_new_study(MINIMIZE)
+        for _ in range(3):
+            storage.create_new_trial(study_id)
+        storage.create_new_trial(other_id)
+        trials = storage.get_all_trials(study_id)
+        assert len(trials) == 3
+        assert all(isinstance(t, FrozenTrial) for t in trials)
+        assert [t.number for t in trials] == [0, 1, 2]
+
+    def test_get_all_trials_state_filter(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        t0 = storage.create_new_trial(study_id)
+        storage.create_new_trial(study_id)
+        storage.set_trial_state_values(t0, TrialState.COMPLETE, (0.0,))
+        complete = storage.get_all_trials(study_id, states=(TrialState.COMPLETE,))
+        running = storage.get_all_trials(study_id, states=(TrialState.RUNNING,))
+        assert len(complete) == 1 and len(running) == 1
+
+    def test_get_all_trials_deepcopy_isolation(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_user_attr(trial_id, "list", [1])
+        trials = storage.get_all_trials(study_id, deepcopy=True)
+        trials[0].user_attrs["list"].append(2)
+        assert storage.get_trial(trial_id).user_attrs["list"] == [1]
+
+    def test_get_n_trials(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        for _ in range(4):
+            storage.create_new_trial(study_id)
+        assert storage.get_n_trials(study_id) == 4
+        assert storage.get_n_trials(study_id, TrialState.RUNNING) == 4
+        assert storage.get_n_trials(study_id, TrialState.COMPLETE) == 0
+
+    def test_get_best_trial(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        with pytest.raises(ValueError):
+            storage.get_best_trial(study_id)
+        values = [2.0, 1.0, 3.0]
+        for v in values:
+            tid = storage.create_new_trial(study_id)
+            storage.set_trial_state_values(tid, TrialState.COMPLETE, (v,))
+        assert storage.get_best_trial(study_id).value == 1.0
+
+    def test_get_best_trial_maximize(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study([StudyDirection.MAXIMIZE])
+        for v in [2.0, 1.0, 3.0]:
+            tid = storage.create_new_trial(study_id)
+            storage.set_trial_state_values(tid, TrialState.COMPLETE, (v,))
+        assert storage.get_best_trial(study_id).value == 3.0
+
+    def test_get_trial_id_from_study_id_trial_number(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        assert storage.get_trial_id_from_study_id_trial_number(study_id, 0) == trial_id
+        with pytest.raises(KeyError):
+            storage.get_trial_id_from_study_id_trial_number(study_id, 10)
+
+    def test_multi_objective_trial_values(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(
+            [StudyDirection.MINIMIZE, StudyDirection.MAXIMIZE]
+        )
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (1.0, 2.0))
+        assert storage.get_trial(trial_id).values == [1.0, 2.0]
+
+    def test_concurrent_trial_creation(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        ids: list[int] = []
+        lock = threading.Lock()
+
+        def worker() -> None:
+            for _ in range(10):
+                tid = storage.create_new_trial(study_id)
+                with lock:
+                    ids.append(tid)
+
+        threads = [threading.Thread(target=worker) for _ in range(4)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        assert len(set(ids)) == 40
+        numbers = sorted(storage.get_trial_number_from_id(tid) for tid in ids)
+        assert numbers == list(range(40))
+
+    def test_pickle_storage_roundtrip_of_trial(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_param(trial_id, "x", 0.5, FloatDistribution(0, 1))
+        trial = storage.get_trial(trial_id)
+        clone = copy.deepcopy(trial)
+        assert clone == trial
+
+    def test_datetime_types(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (0.0,))
+        trial = storage.get_trial(trial_id)
+        assert isinstance(trial.datetime_start, datetime)
+        assert isinstance(trial.datetime_complete, datetime)

@@ -1,0 +1,226 @@
+from __future__ import annotations
+
+import pytest
+
+import optuna_amd
+from optuna_amd.pruners import (
+    HyperbandPruner,
+    MedianPruner,
+    NopPruner,
+    PatientPruner,
+    PercentilePruner,
+    SuccessiveHalvingPruner,
+    ThresholdPruner,
+    WilcoxonPruner,
+)
+from optuna_amd.testing.trials import _create_frozen_trial
+from optuna_amd.trial import TrialState
+
+
+optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+
+
+def _study_with_intermediates(values_per_trial: list[list[float]], direction: str = "minimize"):
+    study = optuna_amd.create_study(direction=direction)
+    for values in values_per_trial:
+        t = study.ask()
+        t.suggest_float("x", 0, 1)
+        for step, v in enumerate(values):
+            t.report(v, step)
+        study.tell(t, values[-1])
+    return study
+
+
+def test_nop_pruner() -> None:
+    study = optuna_amd.create_study(pruner=NopPruner())
+    t = study.ask()
+    t.report(float("inf"), 0)
+    assert not t.should_prune()
+
+
+def test_median_pruner_prunes_worse_trial() -> None:
+    study = _study_with_intermediates([[1.0, 1.0], [2.0, 2.0]])
+    study.pruner = MedianPruner(n_startup_trials=2, n_warmup_steps=0)
+    t = study.ask()
+    t.report(10.0, 0)
+    assert t.should_prune()
+    t2 = study.ask()
+    t2.report(0.5, 0)
+    assert not t2.should_prune()
+
+
+def test_median_pruner_startup_trials() -> None:
+    study = optuna_amd.create_study(pruner=MedianPruner(n_startup_trials=5))
+    t = study.ask()
+    t.report(100.0, 0)
+    assert not t.should_prune()  # not enough completed trials
+
+
+def test_median_pruner_warmup_steps() -> None:
+    study = _study_with_intermediates([[1.0, 1.0], [1.0, 1.0]])
+    study.pruner = MedianPruner(n_startup_trials=1, n_warmup_steps=5)
+    t = study.ask()
+    t.report(100.0, 0)
+    assert not t.should_prune()
+
+
+def test_percentile_pruner_validation() -> None:
+    with pytest.raises(ValueError):
+        PercentilePruner(-1)
+    with pytest.raises(ValueError):
+        PercentilePruner(101)
+    with pytest.raises(ValueError):
+        PercentilePruner(50, n_startup_trials=-1)
+    with pytest.raises(ValueError):
+        PercentilePruner(50, n_warmup_steps=-1)
+    with pytest.raises(ValueError):
+        PercentilePruner(50, interval_steps=0)
+
+
+def test_percentile_pruner_maximize() -> None:
+    study = _study_with_intermediates(
+        [[10.0], [20.0], [30.0]], direction="maximize"
+    )
+    study.pruner = PercentilePruner(25.0, n_startup_trials=1, n_warmup_steps=0)
+    t = study.ask()
+    t.report(5.0, 0)
+    assert t.should_prune()
+    t2 = study.ask()
+    t2.report(50.0, 0)
+    assert not t2.should_prune()
+
+
+def test_successive_halving_basic() -> None:
+    study = optuna_amd.create_study(
+        pruner=SuccessiveHalvingPruner(min_resource=1, reduction_factor=2)
+    )
+    # First trial at rung completion: promoted (smallest among itself).
+    t = study.ask()
+    t.report(1.0, 1)
+    assert not t.should_prune()
+    study.tell(t, 1.0)
+    # A worse second trial gets pruned at the first rung.
+    t2 = study.ask()
+    t2.report(2.0, 1)
+    assert t2.should_prune()
+
+
+def test_successive_halving_validation() -> None:
+    with pytest.raises(ValueError):
+        SuccessiveHalvingPruner(min_resource=0)
+    with pytest.raises(ValueError):
+        SuccessiveHalvingPruner(min_resource="bad")
+    with pytest.raises(ValueError):
+        SuccessiveHalvingPruner(reduction_factor=1)
+    with pytest.raises(ValueError):
+        SuccessiveHalvingPruner(min_early_stopping_rate=-1)
+    with pytest.raises(ValueError):
+        SuccessiveHalvingPruner(bootstrap_count=-1)
+    with pytest.raises(ValueError):
+        SuccessiveHalvingPruner(bootstrap_count=1, min_resource="auto")
+
+
+def test_successive_halving_rung_attrs() -> None:
+    study = optuna_amd.create_study(
+        pruner=SuccessiveHalvingPruner(min_resource=1, reduction_factor=2)
+    )
+    t = study.ask()
+    t.report(1.0, 1)
+    t.should_prune()
+    attrs = study._storage.get_trial(t._trial_id).system_attrs
+    assert "completed_rung_0" in attrs
+
+
+def test_hyperband_runs_end_to_end() -> None:
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.RandomSampler(seed=0),
+        pruner=HyperbandPruner(min_resource=1, max_resource=9, reduction_factor=3),
+    )
+
+    def objective(trial: optuna_amd.Trial) -> float:
+        x = trial.suggest_float("x", 0, 1)
+        for step in range(9):
+            trial.report(x + 1 / (1 + step), step)
+            if trial.should_prune():
+                raise optuna_amd.TrialPruned()
+        return x
+
+    study.optimize(objective, n_trials=20)
+    states = {t.state for t in study.trials}
+    assert TrialState.COMPLETE in states
+    assert study.pruner._n_brackets == 3
+
+
+def test_hyperband_bracket_id_stable() -> None:
+    pruner = HyperbandPruner(min_resource=1, max_resource=9, reduction_factor=3)
+    study = optuna_amd.create_study(pruner=pruner)
+    t = study.ask()
+    t.report(1.0, 0)
+    t.should_prune()
+    study.tell(t, 1.0)
+    frozen = study.trials[0]
+    b1 = pruner._get_bracket_id(study, frozen)
+    b2 = pruner._get_bracket_id(study, frozen)
+    assert b1 == b2 and 0 <= b1 < 3
+
+
+def test_threshold_pruner() -> None:
+    study = optuna_amd.create_study(pruner=ThresholdPruner(upper=1.0))
+    t = study.ask()
+    t.report(0.5, 0)
+    assert not t.should_prune()
+    t.report(1.5, 1)
+    assert t.should_prune()
+
+    study2 = optuna_amd.create_study(pruner=ThresholdPruner(lower=0.0))
+    t2 = study2.ask()
+    t2.report(-1.0, 0)
+    assert t2.should_prune()
+
+    study3 = optuna_amd.create_study(pruner=ThresholdPruner(lower=0.0, upper=1.0))
+    t3 = study3.ask()
+    t3.report(float("nan"), 0)
+    assert t3.should_prune()
+
+    with pytest.raises(TypeError):
+        ThresholdPruner()
+    with pytest.raises(ValueError):
+        ThresholdPruner(lower=1.0, upper=0.0)
+
+
+def test_patient_pruner() -> None:
+    study = optuna_amd.create_study(pruner=PatientPruner(None, patience=1))
+    t = study.ask()
+    # Improving: never prune.
+    t.report(3.0, 0)
+    t.report(2.0, 1)
+    t.report(1.0, 2)
+    assert not t.should_prune()
+    # Stagnating beyond patience: prune.
+    t.report(1.5, 3)
+    t.report(1.6, 4)
+    assert t.should_prune()
+    with pytest.raises(ValueError):
+        PatientPruner(None, patience=-1)
+    with pytest.raises(ValueError):
+        PatientPruner(None, patience=0, min_delta=-1)
+
+
+def test_wilcoxon_pruner() -> None:
+    study = optuna_amd.create_study(pruner=WilcoxonPruner(p_threshold=0.2))
+    # Best trial with many step values.
+    best = study.ask()
+    best.suggest_float("x", 0, 1)
+    for step in range(10):
+        best.report(0.1, step)
+    study.tell(best, 0.1)
+    # Clearly-worse current trial.
+    t = study.ask()
+    t.suggest_float("x", 0, 1)
+    for step in range(10):
+        t.report(5.0 + step * 0.01, step)
+    assert t.should_prune()
+    with pytest.raises(ValueError):
+        WilcoxonPruner(p_threshold=2.0)
+    with pytest.raises(ValueError):
+        WilcoxonPruner(n_startup_steps=-1)

@@ -1,0 +1,157 @@
+"""Parzen estimator behavior: kernel construction, sampling domain, log-pdf."""
+from __future__ import annotations
+
+import numpy as np
+import pytest
+
+from optuna_amd.distributions import (
+    CategoricalDistribution,
+    FloatDistribution,
+    IntDistribution,
+)
+from optuna_amd.samplers._tpe.parzen import _ParzenEstimator, _ParzenEstimatorParameters
+from optuna_amd.samplers._tpe.sampler import default_weights
+
+
+PARAMS = _ParzenEstimatorParameters(
+    consider_prior=True,
+    prior_weight=1.0,
+    consider_magic_clip=True,
+    consider_endpoints=False,
+    weights=default_weights,
+    multivariate=True,
+)
+
+
+def _mpe(observations, space, predetermined=None) -> _ParzenEstimator:
+    return _ParzenEstimator(observations, space, PARAMS, predetermined)
+
+
+def test_empty_observations_prior_only() -> None:
+    space = {"x": FloatDistribution(0.0, 2.0)}
+    mpe = _mpe({"x": np.array([])}, space)
+    assert len(mpe.weights) == 1
+    num = mpe._numerical
+    assert num.mus.shape == (1, 1)
+    assert num.mus[0, 0] == pytest.approx(1.0)  # midpoint prior
+    assert num.sigmas[0, 0] == pytest.approx(2.0)  # domain width
+
+
+def test_kernel_count_includes_prior() -> None:
+    space = {"x": FloatDistribution(0.0, 1.0)}
+    mpe = _mpe({"x": np.array([0.2, 0.4, 0.9])}, space)
+    assert mpe._numerical.mus.shape == (4, 1)
+    assert len(mpe.weights) == 4
+    assert mpe.weights.sum() == pytest.approx(1.0)
+
+
+def test_magic_clip_bounds_sigma() -> None:
+    space = {"x": FloatDistribution(0.0, 1.0)}
+    obs = np.array([0.5, 0.5000001, 0.5000002])
+    mpe = _mpe({"x": obs}, space)
+    sigmas = mpe._numerical.sigmas[:-1, 0]
+    minsigma = 1.0 / min(100.0, 1.0 + 4)
+    assert np.all(sigmas >= minsigma - 1e-15)
+
+
+def test_sample_within_domain_and_log_aware() -> None:
+    rng = np.random.RandomState(0)
+    space = {
+        "lin": FloatDistribution(-1.0, 1.0),
+        "log": FloatDistribution(1e-3, 1e3, log=True),
+        "step": FloatDistribution(0.0, 1.0, step=0.25),
+        "int": IntDistribution(1, 10),
+        "cat": CategoricalDistribution(("a", "b", "c")),
+    }
+    observations = {
+        "lin": np.array([0.0, 0.5]),
+        "log": np.array([1.0, 100.0]),
+        "step": np.array([0.25, 0.75]),
+        "int": np.array([3.0, 7.0]),
+        "cat": np.array([0.0, 2.0]),
+    }
+    mpe = _mpe(observations, space)
+    samples = mpe.sample(rng, 256)
+    assert np.all(samples["lin"] >= -1.0) and np.all(samples["lin"] <= 1.0)
+    assert np.all(samples["log"] >= 1e-3) and np.all(samples["log"] <= 1e3)
+    steps = np.round(samples["step"] / 0.25) * 0.25
+    np.testing.assert_allclose(samples["step"], steps, atol=1e-12)
+    assert np.all(samples["int"] == np.round(samples["int"]))
+    assert set(np.unique(samples["cat"])) <= {0.0, 1.0, 2.0}
+
+
+def test_log_pdf_finite_for_samples() -> None:
+    rng = np.random.RandomState(1)
+    space = {
+        "x": FloatDistribution(0.0, 1.0),
+        "c": CategoricalDistribution(("u", "v")),
+    }
+    observations = {"x": np.array([0.1, 0.9]), "c": np.array([0.0, 1.0])}
+    mpe = _mpe(observations, space)
+    samples = mpe.sample(rng, 64)
+    lp = mpe.log_pdf(samples)
+    assert lp.shape == (64,)
+    assert np.all(np.isfinite(lp))
+
+
+def test_log_pdf_mixture_integrates_to_one_1d() -> None:
+    # Continuous 1-D KDE: numerically integrate exp(log_pdf) over the domain.
+    space = {"x": FloatDistribution(0.0, 1.0)}
+    observations = {"x": np.array([0.3, 0.6, 0.62])}
+    mpe = _mpe(observations, space)
+    grid = np.linspace(1e-9, 1 - 1e-9, 20001)
+    lp = mpe.log_pdf({"x": grid})
+    integral = np.trapezoid(np.exp(lp), grid)
+    assert integral == pytest.approx(1.0, abs=1e-3)
+
+
+def test_discrete_log_pdf_sums_to_one() -> None:
+    space = {"i": IntDistribution(0, 9)}
+    observations = {"i": np.array([2.0, 5.0])}
+    mpe = _mpe(observations, space)
+    support = np.arange(10, dtype=np.float64)
+    lp = mpe.log_pdf({"i": support})
+    assert np.exp(lp).sum() == pytest.approx(1.0, abs=1e-9)
+
+
+def test_categorical_weights_prior_smoothing() -> None:
+    space = {"c": CategoricalDistribution(("a", "b", "c"))}
+    observations = {"c": np.array([0.0, 0.0])}
+    mpe = _mpe(observations, space)
+    w = mpe._categoricals[0].weights
+    assert w.shape == (3, 3)
+    np.testing.assert_allclose(w.sum(axis=1), 1.0)
+    assert w[0, 0] > w[0, 1]  # observed choice is upweighted
+    np.testing.assert_allclose(w[2], np.full(3, 1 / 3))  # prior row uniform
+
+
+def test_predetermined_weights() -> None:
+    space = {"x": FloatDistribution(0.0, 1.0)}
+    obs = {"x": np.array([0.2, 0.8])}
+    mpe = _mpe(obs, space, predetermined=np.array([1.0, 3.0]))
+    w = mpe.weights
+    assert len(w) == 3
+    assert w[1] == pytest.approx(3 * w[0])
+
+
+def test_negative_prior_weight_rejected() -> None:
+    space = {"x": FloatDistribution(0.0, 1.0)}
+    bad = PARAMS._replace(prior_weight=-1.0)
+    with pytest.raises(ValueError):
+        _ParzenEstimator({"x": np.array([0.5])}, space, bad)
+
+
+def test_weights_func_validation() -> None:
+    space = {"x": FloatDistribution(0.0, 1.0)}
+    with pytest.raises(ValueError):
+        _ParzenEstimator(
+            {"x": np.array([0.5, 0.6])},
+            space,
+            PARAMS._replace(weights=lambda n: np.array([-1.0] * n)),
+        )
+    with pytest.raises(ValueError):
+        _ParzenEstimator(
+            {"x": np.array([0.5, 0.6])},
+            space,
+            PARAMS._replace(weights=lambda n: np.array([0.0] * n)),
+        )
