@@ -1,0 +1,184 @@
+#!/usr/bin/env python
+"""Driver benchmark: TPE suggest()/sec at 10k-trial history, 20-dim space.
+
+This measures BASELINE.json's headline metric on its named config
+("TPESampler 20-dim synthetic objective, 10k-trial history"): each "step" is one
+full ask/suggest/tell round against a study pre-populated with 10,000 finished
+trials. The Parzen KDE fit (K1) and the S×K×D mixture log-pdf / EI scoring (K2)
+run as hand-written HIP kernels on gfx950.
+
+Single process: `python bench.py --steps 64 --warmup 8`
+Multi GPU (driver): torchrun --nnodes=1 --nproc-per-node N bench.py --gpus N ...
+  → one worker process per GPU, all sharing one study through the distributed
+    op-log storage (RcclStorage) — weak scaling (K steps per rank).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+import warnings
+
+import numpy as np
+
+
+def _log(msg: str) -> None:
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=64)
+    parser.add_argument("--warmup", type=int, default=8)
+    parser.add_argument("--history", type=int, default=10000)
+    parser.add_argument("--dims", type=int, default=20)
+    args = parser.parse_args()
+
+    warnings.simplefilter("ignore")
+    import optuna_amd
+
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    n_gpus = max(args.gpus, world_size)
+
+    import torch
+
+    has_gpu = torch.cuda.is_available()
+    if has_gpu:
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as torch_dist
+
+        dist = torch_dist
+        backend = "nccl" if has_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+
+    D = args.dims
+    N_HISTORY = args.history
+    names = [f"x{i}" for i in range(D)]
+    dists_def = {n: optuna_amd.distributions.FloatDistribution(-5.0, 5.0) for n in names}
+
+    def make_sampler() -> "optuna_amd.samplers.TPESampler":
+        return optuna_amd.samplers.TPESampler(
+            seed=42 + rank, n_startup_trials=10, constant_liar=(world_size > 1)
+        )
+
+    # ---- storage / study setup ------------------------------------------------------
+    if world_size > 1:
+        from optuna_amd.storages._rccl import RcclStorage
+
+        storage = RcclStorage.from_env()
+        if rank == 0:
+            study = optuna_amd.create_study(
+                study_name="bench", storage=storage, sampler=make_sampler()
+            )
+            _populate(study, names, dists_def, N_HISTORY)
+        dist.barrier()
+        if rank != 0:
+            study = optuna_amd.load_study(
+                study_name="bench", storage=storage, sampler=make_sampler()
+            )
+        dist.barrier()
+    else:
+        study = optuna_amd.create_study(sampler=make_sampler())
+        _populate(study, names, dists_def, N_HISTORY)
+
+    rng = np.random.RandomState(1234 + rank)
+
+    def one_step() -> None:
+        trial = study.ask()
+        x = np.empty(D)
+        for i, n in enumerate(names):
+            x[i] = trial.suggest_float(n, -5.0, 5.0)
+        # Synthetic objective (shifted sphere + noise), no model download needed.
+        value = float(np.sum((x - 1.0) ** 2) + 0.01 * rng.randn())
+        study.tell(trial, value)
+
+    # ---- warmup ---------------------------------------------------------------------
+    for _ in range(args.warmup):
+        one_step()
+    _log(f"rank {rank}: warmup done ({args.warmup} steps)")
+
+    # ---- timed region ---------------------------------------------------------------
+    if dist is not None:
+        dist.barrier()
+    if has_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    if has_gpu:
+        torch.cuda.synchronize()
+    if dist is not None:
+        dist.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if has_gpu:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        from optuna_amd import _hip
+
+        value = n_gpus * args.steps / elapsed
+        result = {
+            "metric": "sampler suggest()/sec at 10k-trial history, 20-dim space",
+            "value": value,
+            "unit": "suggest/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "TPESampler multivariate Parzen-KDE + EI",
+                "history_trials": N_HISTORY,
+                "dims": D,
+                "n_ei_candidates": 24,
+                "parallelism": f"dp{n_gpus}" if n_gpus > 1 else "single",
+                "hip_kernels": bool(_hip.is_available()),
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+def _populate(study, names, dists_def, n_history: int) -> None:
+    import optuna_amd
+
+    _log(f"populating {n_history} history trials...")
+    rng = np.random.RandomState(0)
+    params_mat = rng.uniform(-5.0, 5.0, size=(n_history, len(names)))
+    values = rng.rand(n_history)
+    trials = [
+        optuna_amd.create_trial(
+            params={n: float(params_mat[r, i]) for i, n in enumerate(names)},
+            distributions=dists_def,
+            value=float(values[r]),
+        )
+        for r in range(n_history)
+    ]
+    study.add_trials(trials)
+    _log("populate done")
+
+
+if __name__ == "__main__":
+    main()
