@@ -39,15 +39,19 @@ def _get_frozen_trial(study: "Study", trial: Trial | int) -> FrozenTrial:
 
 
 def _check_values_are_feasible(study: "Study", values: Sequence[float]) -> str | None:
+    """NaN and non-float-castable values are infeasible → the trial FAILs
+    (parity: reference study/_tell.py:60-77)."""
+    errors = []
     for v in values:
-        # NaN is acceptable (it means a failed evaluation that is still COMPLETE in
-        # the reference semantics); non-float-castable values are not.
-        if v is None:
-            return f"The value {v} could not be cast to float."
         try:
             float(v)
         except (ValueError, TypeError):
-            return f"The value {v} could not be cast to float."
+            errors.append(f"The value {v!r} could not be cast to float")
+            continue
+        if math.isnan(float(v)):
+            errors.append(f"The value {v} is not acceptable")
+    if errors:
+        return "; ".join(errors)
     return None
 
 

@@ -245,10 +245,10 @@ def test_multi_objective_best_trials() -> None:
         study.direction
 
 
-def test_nan_objective() -> None:
+def test_nan_objective_fails_trial() -> None:
     study = optuna_amd.create_study()
     study.optimize(lambda t: float("nan"), n_trials=2, catch=())
-    assert all(t.state == TrialState.COMPLETE for t in study.trials)
+    assert all(t.state == TrialState.FAIL for t in study.trials)
     study.optimize(_square, n_trials=1)
     assert study.best_trial.value is not None and not math.isnan(study.best_trial.value)
 
