@@ -1,0 +1,353 @@
+"""Distributed in-memory trial table for one-process-per-GPU studies.
+
+The reference coordinates distributed workers through SQL transactions or an
+fsync'd journal file (reference ``optuna/storages/journal/_storage.py``; survey
+§5.8). On an MI355X node the workers are one process per GPU launched by
+torchrun, so this backend replaces the database with a replicated in-memory
+table:
+
+* **Op log**: every write is one record in the exact journal op format
+  (``JournalOperation`` opcodes); total order comes from an atomic fetch-add
+  sequencer in a ``torch.distributed.TCPStore`` hosted by rank 0 (tiny KB-scale
+  records are latency-bound — a loopback RTT, not a collective, is the right
+  primitive; xGMI/RCCL stays free for the samplers' bulk data).
+* **Replay**: each worker applies unseen records to a local
+  ``_ReplayState`` — the same CRDT-ish replay used by JournalStorage, so CAS
+  claim semantics, trial numbering and ownership rules are identical and the
+  storage passes the same conformance suite.
+* **Device tier**: the replayed table feeds each worker's sampler history mirror
+  (``samplers/_tpe/_history.py``), which uploads increments into HBM for the
+  K1/K2 kernels — the trial history is resident on every GPU.
+
+Because the log lives in the rank-0 store, the study exists for the lifetime of
+the job; use ``to_journal()``/JournalStorage for durable checkpoints.
+"""
+from __future__ import annotations
+
+import datetime
+import json
+import os
+import threading
+import uuid
+from typing import Any, Container, Sequence
+
+from optuna_amd import logging as _logging
+from optuna_amd.distributions import BaseDistribution, distribution_to_json
+from optuna_amd.exceptions import UpdateFinishedTrialError
+from optuna_amd.storages._base import DEFAULT_STUDY_NAME_PREFIX, BaseStorage
+from optuna_amd.storages.journal._storage import (
+    NOT_FOUND_MSG,
+    UNUPDATABLE_MSG,
+    JournalOperation,
+    _ReplayState,
+    _utcnow_iso,
+)
+from optuna_amd.study._frozen import FrozenStudy
+from optuna_amd.study._study_direction import StudyDirection
+from optuna_amd.trial import FrozenTrial, TrialState
+
+
+_logger = _logging.get_logger(__name__)
+
+_SEQ_KEY = "optuna_amd/oplog/next"
+_REC_KEY = "optuna_amd/oplog/{idx}"
+
+
+class _TcpStoreLog:
+    """Append-only log over a TCPStore: fetch-add sequencer + per-index records."""
+
+    def __init__(self, store: Any) -> None:
+        self._store = store
+
+    def append(self, records: list[dict[str, Any]]) -> None:
+        n = len(records)
+        # Reserve a contiguous index range, then publish the records.
+        end = self._store.add(_SEQ_KEY, n)
+        start = end - n
+        for i, rec in enumerate(records):
+            self._store.set(
+                _REC_KEY.format(idx=start + i),
+                json.dumps(rec, separators=(",", ":")),
+            )
+
+    def read_from(self, start: int) -> list[dict[str, Any]]:
+        end = self._store.add(_SEQ_KEY, 0)
+        out = []
+        for idx in range(start, end):
+            # set() may lag the counter by a moment on another worker; get() blocks
+            # until the key appears (bounded by the store timeout).
+            payload = self._store.get(_REC_KEY.format(idx=idx))
+            out.append(json.loads(payload))
+        return out
+
+
+class RcclStorage(BaseStorage):
+    """Shared trial table for N worker processes on one node (see module docstring)."""
+
+    def __init__(self, store: Any, worker_label: str | None = None) -> None:
+        self._log = _TcpStoreLog(store)
+        self._worker_id_prefix = (worker_label or str(uuid.uuid4())) + "-" + str(uuid.uuid4())[:8] + "-"
+        self._thread_lock = threading.Lock()
+        self._replay = _ReplayState(self._worker_id_prefix)
+        with self._thread_lock:
+            self._sync()
+
+    @classmethod
+    def from_env(cls) -> "RcclStorage":
+        """Build from torchrun env (MASTER_ADDR/MASTER_PORT/RANK/WORLD_SIZE).
+
+        The TCPStore listens on MASTER_PORT+1 (override: OPTUNA_AMD_STORE_PORT) so
+        it never collides with the RCCL process-group rendezvous.
+        """
+        from torch.distributed import TCPStore
+
+        host = os.environ.get("MASTER_ADDR", "127.0.0.1")
+        port = int(
+            os.environ.get(
+                "OPTUNA_AMD_STORE_PORT", str(int(os.environ.get("MASTER_PORT", "29500")) + 1)
+            )
+        )
+        rank = int(os.environ.get("RANK", "0"))
+        world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        store = TCPStore(
+            host,
+            port,
+            world_size,
+            is_master=(rank == 0),
+            timeout=datetime.timedelta(seconds=120),
+        )
+        storage = cls(store, worker_label=f"rank{rank}")
+        storage._store = store  # keep the master store object alive
+        return storage
+
+    # ---- log plumbing ---------------------------------------------------------------
+
+    def _append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
+        self._log.append(
+            [{"op_code": int(op), "worker_id": self._replay.worker_id, **fields}]
+        )
+
+    def _sync(self) -> None:
+        records = self._log.read_from(self._replay.log_number_read)
+        self._replay.apply_logs(records)
+
+    # ---- studies --------------------------------------------------------------------
+
+    def create_new_study(
+        self, directions: Sequence[StudyDirection], study_name: str | None = None
+    ) -> int:
+        study_name = study_name or DEFAULT_STUDY_NAME_PREFIX + str(uuid.uuid4())
+        with self._thread_lock:
+            self._append(
+                JournalOperation.CREATE_STUDY,
+                {"study_name": study_name, "directions": [int(d) for d in directions]},
+            )
+            self._sync()
+            for fs in self._replay.all_studies():
+                if fs.study_name == study_name:
+                    return fs._study_id
+            raise AssertionError("unreachable")
+
+    def delete_study(self, study_id: int) -> None:
+        with self._thread_lock:
+            self._append(JournalOperation.DELETE_STUDY, {"study_id": study_id})
+            self._sync()
+
+    def set_study_user_attr(self, study_id: int, key: str, value: Any) -> None:
+        with self._thread_lock:
+            self._append(
+                JournalOperation.SET_STUDY_USER_ATTR,
+                {"study_id": study_id, "user_attr": {key: value}},
+            )
+            self._sync()
+
+    def set_study_system_attr(self, study_id: int, key: str, value: Any) -> None:
+        with self._thread_lock:
+            self._append(
+                JournalOperation.SET_STUDY_SYSTEM_ATTR,
+                {"study_id": study_id, "system_attr": {key: value}},
+            )
+            self._sync()
+
+    def get_study_id_from_name(self, study_name: str) -> int:
+        with self._thread_lock:
+            self._sync()
+            for fs in self._replay.all_studies():
+                if fs.study_name == study_name:
+                    return fs._study_id
+            raise KeyError(NOT_FOUND_MSG)
+
+    def get_study_name_from_id(self, study_id: int) -> str:
+        with self._thread_lock:
+            self._sync()
+            return self._replay.study(study_id).study_name
+
+    def get_study_directions(self, study_id: int) -> list[StudyDirection]:
+        with self._thread_lock:
+            self._sync()
+            return self._replay.study(study_id).directions
+
+    def get_study_user_attrs(self, study_id: int) -> dict[str, Any]:
+        with self._thread_lock:
+            self._sync()
+            return self._replay.study(study_id).user_attrs
+
+    def get_study_system_attrs(self, study_id: int) -> dict[str, Any]:
+        with self._thread_lock:
+            self._sync()
+            return self._replay.study(study_id).system_attrs
+
+    def get_all_studies(self) -> list[FrozenStudy]:
+        import copy
+
+        with self._thread_lock:
+            self._sync()
+            return copy.deepcopy(self._replay.all_studies())
+
+    # ---- trials ---------------------------------------------------------------------
+
+    def create_new_trial(self, study_id: int, template_trial: FrozenTrial | None = None) -> int:
+        from datetime import timezone
+
+        log: dict[str, Any] = {"study_id": study_id, "datetime_start": _utcnow_iso()}
+        if template_trial is not None:
+            log["state"] = int(template_trial.state)
+            if template_trial.values is not None and len(template_trial.values) > 1:
+                log["value"] = None
+                log["values"] = template_trial.values
+            else:
+                log["value"] = template_trial.value
+                log["values"] = None
+            if template_trial.datetime_start:
+                log["datetime_start"] = template_trial.datetime_start.astimezone(
+                    timezone.utc
+                ).isoformat(timespec="microseconds")
+            else:
+                log["datetime_start"] = None
+            if template_trial.datetime_complete:
+                log["datetime_complete"] = template_trial.datetime_complete.astimezone(
+                    timezone.utc
+                ).isoformat(timespec="microseconds")
+            log["distributions"] = {
+                k: distribution_to_json(d) for k, d in template_trial.distributions.items()
+            }
+            log["params"] = {
+                k: template_trial.distributions[k].to_internal_repr(v)
+                for k, v in template_trial.params.items()
+            }
+            log["user_attrs"] = template_trial.user_attrs
+            log["system_attrs"] = template_trial.system_attrs
+            log["intermediate_values"] = template_trial.intermediate_values
+        with self._thread_lock:
+            self._append(JournalOperation.CREATE_TRIAL, log)
+            self._sync()
+            return self._replay.last_created_trial_id
+
+    def set_trial_param(
+        self,
+        trial_id: int,
+        param_name: str,
+        param_value_internal: float,
+        distribution: BaseDistribution,
+    ) -> None:
+        with self._thread_lock:
+            self._append(
+                JournalOperation.SET_TRIAL_PARAM,
+                {
+                    "trial_id": trial_id,
+                    "param_name": param_name,
+                    "param_value_internal": param_value_internal,
+                    "distribution": distribution_to_json(distribution),
+                },
+            )
+            self._sync()
+
+    def get_trial_id_from_study_id_trial_number(self, study_id: int, trial_number: int) -> int:
+        with self._thread_lock:
+            self._sync()
+            trial_ids = self._replay._study_id_to_trial_ids.get(study_id)
+            if trial_ids is None or len(trial_ids) <= trial_number:
+                raise KeyError(
+                    f"No trial with trial number {trial_number} exists in study with "
+                    f"study_id {study_id}."
+                )
+            return trial_ids[trial_number]
+
+    def set_trial_state_values(
+        self, trial_id: int, state: TrialState, values: Sequence[float] | None = None
+    ) -> bool:
+        log: dict[str, Any] = {
+            "trial_id": trial_id,
+            "state": int(state),
+            "values": list(values) if values is not None else None,
+        }
+        if state == TrialState.RUNNING:
+            log["datetime_start"] = _utcnow_iso()
+        elif state.is_finished():
+            log["datetime_complete"] = _utcnow_iso()
+        with self._thread_lock:
+            if state == TrialState.RUNNING:
+                self._sync()
+                existing = self._replay._trials.get(trial_id)
+                if existing is None:
+                    raise KeyError(NOT_FOUND_MSG)
+                if existing.state.is_finished():
+                    raise UpdateFinishedTrialError(
+                        UNUPDATABLE_MSG.format(trial_number=existing.number)
+                    )
+                if existing.state != TrialState.WAITING:
+                    return False
+            self._append(JournalOperation.SET_TRIAL_STATE_VALUES, log)
+            self._sync()
+            return state != TrialState.RUNNING or trial_id == self._replay.owned_trial_id
+
+    def set_trial_intermediate_value(
+        self, trial_id: int, step: int, intermediate_value: float
+    ) -> None:
+        with self._thread_lock:
+            self._append(
+                JournalOperation.SET_TRIAL_INTERMEDIATE_VALUE,
+                {"trial_id": trial_id, "step": step, "intermediate_value": intermediate_value},
+            )
+            self._sync()
+
+    def set_trial_user_attr(self, trial_id: int, key: str, value: Any) -> None:
+        with self._thread_lock:
+            self._append(
+                JournalOperation.SET_TRIAL_USER_ATTR,
+                {"trial_id": trial_id, "user_attr": {key: value}},
+            )
+            self._sync()
+
+    def set_trial_system_attr(self, trial_id: int, key: str, value: Any) -> None:
+        with self._thread_lock:
+            self._append(
+                JournalOperation.SET_TRIAL_SYSTEM_ATTR,
+                {"trial_id": trial_id, "system_attr": {key: value}},
+            )
+            self._sync()
+
+    def get_trial(self, trial_id: int) -> FrozenTrial:
+        with self._thread_lock:
+            self._sync()
+            return self._replay.trial(trial_id)
+
+    def get_all_trials(
+        self,
+        study_id: int,
+        deepcopy: bool = True,
+        states: Container[TrialState] | None = None,
+    ) -> list[FrozenTrial]:
+        import copy
+
+        with self._thread_lock:
+            self._sync()
+            trials = self._replay.all_trials(study_id, states)
+            if deepcopy:
+                trials = copy.deepcopy(trials)
+            return trials
+
+
+# Descriptive alias: the class name advertises the MI355X deployment (one rank per
+# GPU over RCCL/xGMI); the mechanism is the distributed op-log table.
+DistributedTrialTableStorage = RcclStorage

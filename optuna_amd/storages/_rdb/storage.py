@@ -585,6 +585,53 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
                 for t in trials
             ]
 
+    def _get_trials_delta(
+        self,
+        study_id: int,
+        included_trial_ids: Container[int],
+        trial_id_greater_than: int,
+    ) -> list[FrozenTrial]:
+        """Trials with id > frontier, plus the explicitly-watched unfinished ids
+        (the _CachedStorage delta protocol; reference storage.py:858-923)."""
+        with _create_scoped_session(self.scoped_session) as session:
+            self._get_study(session, study_id)
+            included = list(included_trial_ids)  # type: ignore[arg-type]
+            cond = models.TrialModel.trial_id > trial_id_greater_than
+            if included:
+                cond = sqlalchemy.or_(cond, models.TrialModel.trial_id.in_(included))
+            trials = (
+                session.query(models.TrialModel)
+                .filter(models.TrialModel.study_id == study_id, cond)
+                .order_by(models.TrialModel.trial_id)
+                .all()
+            )
+            trial_ids = [t.trial_id for t in trials]
+
+            def _bulk(model: Any) -> dict[int, list[Any]]:
+                out: dict[int, list[Any]] = {}
+                if not trial_ids:
+                    return out
+                for row in session.query(model).filter(model.trial_id.in_(trial_ids)).all():
+                    out.setdefault(row.trial_id, []).append(row)
+                return out
+
+            params = _bulk(models.TrialParamModel)
+            values = _bulk(models.TrialValueModel)
+            ivalues = _bulk(models.TrialIntermediateValueModel)
+            uattrs = _bulk(models.TrialUserAttributeModel)
+            sattrs = _bulk(models.TrialSystemAttributeModel)
+            return [
+                self._assemble_frozen_trial(
+                    t,
+                    params.get(t.trial_id, []),
+                    values.get(t.trial_id, []),
+                    ivalues.get(t.trial_id, []),
+                    uattrs.get(t.trial_id, []),
+                    sattrs.get(t.trial_id, []),
+                )
+                for t in trials
+            ]
+
     # ---- heartbeat ------------------------------------------------------------------
 
     def record_heartbeat(self, trial_id: int) -> None:

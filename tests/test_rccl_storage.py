@@ -1,0 +1,146 @@
+"""Distributed trial-table storage: conformance + multi-process (gloo-style) tests."""
+from __future__ import annotations
+
+import datetime
+import multiprocessing
+import socket
+
+import pytest
+
+import optuna_amd
+from optuna_amd.storages import BaseStorage
+from optuna_amd.testing.pytest_storages import StorageTestCase
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _make_store(port: int, world_size: int, is_master: bool):
+    from torch.distributed import TCPStore
+
+    return TCPStore(
+        "127.0.0.1",
+        port,
+        world_size,
+        is_master=is_master,
+        timeout=datetime.timedelta(seconds=60),
+    )
+
+
+class TestRcclStorageConformance(StorageTestCase):
+    @pytest.fixture
+    def storage(self) -> BaseStorage:
+        from optuna_amd.storages._rccl import RcclStorage
+
+        store = _make_store(_free_port(), 1, True)
+        s = RcclStorage(store)
+        s._store = store
+        return s
+
+
+def _worker(rank: int, world_size: int, port: int, queue) -> None:
+    try:
+        import warnings
+
+        warnings.simplefilter("ignore")
+        from optuna_amd.storages._rccl import RcclStorage
+
+        optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+        store = _make_store(port, world_size, rank == 0)
+        storage = RcclStorage(store, worker_label=f"rank{rank}")
+        if rank == 0:
+            study = optuna_amd.create_study(
+                study_name="dist", storage=storage,
+                sampler=optuna_amd.samplers.TPESampler(seed=rank, n_startup_trials=3),
+            )
+            store.set("study_ready", "1")
+        else:
+            store.get("study_ready")  # blocks until created
+            study = optuna_amd.load_study(
+                study_name="dist", storage=storage,
+                sampler=optuna_amd.samplers.TPESampler(seed=rank, n_startup_trials=3),
+            )
+
+        def objective(trial):
+            x = trial.suggest_float("x", -5, 5)
+            y = trial.suggest_float("y", -5, 5)
+            return x * x + y * y
+
+        study.optimize(objective, n_trials=8)
+        trials = study.get_trials(deepcopy=False)
+        queue.put((rank, len(trials), [t.number for t in trials]))
+    except Exception as e:  # surface the failure to the parent
+        import traceback
+
+        queue.put((rank, -1, traceback.format_exc()))
+
+
+def test_two_process_shared_study() -> None:
+    """world_size=2: both workers optimize one study; numbering stays dense."""
+    port = _free_port()
+    ctx = multiprocessing.get_context("spawn")
+    queue = ctx.Manager().Queue()
+    procs = [
+        ctx.Process(target=_worker, args=(rank, 2, port, queue)) for rank in range(2)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    results = [queue.get(timeout=10) for _ in range(2)]
+    for rank, n, payload in results:
+        assert n != -1, f"rank {rank} failed:\n{payload}"
+        assert n == 16  # both workers see all 16 trials
+        assert payload == list(range(16))  # dense, consecutive numbering
+
+
+def _claim_worker(rank: int, world_size: int, port: int, trial_id: int, queue) -> None:
+    try:
+        from optuna_amd.storages._rccl import RcclStorage
+        from optuna_amd.trial import TrialState
+
+        store = _make_store(port, world_size, rank == 0)
+        storage = RcclStorage(store, worker_label=f"claim{rank}")
+        if rank == 0:
+            study_id = storage.create_new_study(
+                [optuna_amd.study.StudyDirection.MINIMIZE], study_name="cas"
+            )
+            tid = storage.create_new_trial(
+                study_id, template_trial=optuna_amd.create_trial(state=TrialState.WAITING)
+            )
+            store.set("trial_id", str(tid))
+        tid = int(store.get("trial_id"))
+        store.add("barrier", 1)
+        while store.add("barrier", 0) < world_size:
+            pass
+        won = storage.set_trial_state_values(tid, TrialState.RUNNING)
+        queue.put((rank, bool(won)))
+    except Exception:
+        import traceback
+
+        queue.put((rank, traceback.format_exc()))
+
+
+def test_waiting_claim_cas_across_processes() -> None:
+    """Exactly one of N processes wins the WAITING→RUNNING claim."""
+    port = _free_port()
+    ctx = multiprocessing.get_context("spawn")
+    queue = ctx.Manager().Queue()
+    world = 3
+    procs = [
+        ctx.Process(target=_claim_worker, args=(rank, world, port, 0, queue))
+        for rank in range(world)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    results = [queue.get(timeout=10) for _ in range(world)]
+    wins = []
+    for rank, won in results:
+        assert isinstance(won, bool), f"rank {rank} failed:\n{won}"
+        wins.append(won)
+    assert sum(wins) == 1
