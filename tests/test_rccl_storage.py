@@ -70,6 +70,12 @@ def _worker(rank: int, world_size: int, port: int, queue) -> None:
             return x * x + y * y
 
         study.optimize(objective, n_trials=8)
+        # Barrier: wait until every worker finished its 8 trials before counting.
+        store.add("done", 1)
+        import time
+
+        while store.add("done", 0) < world_size:
+            time.sleep(0.01)
         trials = study.get_trials(deepcopy=False)
         queue.put((rank, len(trials), [t.number for t in trials]))
     except Exception as e:  # surface the failure to the parent
