@@ -78,6 +78,12 @@ def _worker(rank: int, world_size: int, port: int, queue) -> None:
             time.sleep(0.01)
         trials = study.get_trials(deepcopy=False)
         queue.put((rank, len(trials), [t.number for t in trials]))
+        # The rank-0 process hosts the TCPStore: keep it alive until every
+        # worker is done with store operations.
+        store.add("exit", 1)
+        if rank == 0:
+            while store.add("exit", 0) < world_size:
+                time.sleep(0.01)
     except Exception as e:  # surface the failure to the parent
         import traceback
 
@@ -124,6 +130,12 @@ def _claim_worker(rank: int, world_size: int, port: int, trial_id: int, queue) -
             pass
         won = storage.set_trial_state_values(tid, TrialState.RUNNING)
         queue.put((rank, bool(won)))
+        store.add("exit", 1)
+        if rank == 0:
+            import time
+
+            while store.add("exit", 0) < world_size:
+                time.sleep(0.01)
     except Exception:
         import traceback
 
