@@ -24,6 +24,7 @@ from optuna_amd.distributions import (
     check_distribution_compatibility,
 )
 from optuna_amd.trial._base import BaseTrial
+from optuna_amd.trial._frozen import FrozenTrial
 from optuna_amd.trial._state import TrialState
 
 
@@ -35,6 +36,68 @@ _logger = _logging.get_logger(__name__)
 _FIXED_PARAMS_KEY = "fixed_params"
 
 
+class _LazyTrialSystemAttrs(dict):
+    """Trial system-attrs view that loads from storage on first read.
+
+    Parity: reference trial/_trial.py:815-827 (_LazyTrialSystemAttrs).
+    """
+
+    def __init__(self, trial_id: int, storage: Any) -> None:
+        super().__init__()
+        object.__setattr__(self, "_trial_id_for_lazy", trial_id)
+        object.__setattr__(self, "_storage_for_lazy", storage)
+        object.__setattr__(self, "_lazy_initialized", False)
+
+    def _fetch(self) -> None:
+        if not object.__getattribute__(self, "_lazy_initialized"):
+            object.__setattr__(self, "_lazy_initialized", True)
+            storage = object.__getattribute__(self, "_storage_for_lazy")
+            trial_id = object.__getattribute__(self, "_trial_id_for_lazy")
+            super().update(storage.get_trial_system_attrs(trial_id))
+
+    def __getitem__(self, key: Any) -> Any:
+        self._fetch()
+        return super().__getitem__(key)
+
+    def __contains__(self, key: Any) -> bool:
+        self._fetch()
+        return super().__contains__(key)
+
+    def __iter__(self) -> Any:
+        self._fetch()
+        return super().__iter__()
+
+    def __len__(self) -> int:
+        self._fetch()
+        return super().__len__()
+
+    def get(self, key: Any, default: Any = None) -> Any:
+        self._fetch()
+        return super().get(key, default)
+
+    def items(self) -> Any:
+        self._fetch()
+        return super().items()
+
+    def keys(self) -> Any:
+        self._fetch()
+        return super().keys()
+
+    def values(self) -> Any:
+        self._fetch()
+        return super().values()
+
+    def __eq__(self, other: Any) -> bool:
+        self._fetch()
+        return super().__eq__(other)
+
+    def __ne__(self, other: Any) -> bool:
+        return not self.__eq__(other)
+
+    def __hash__(self) -> int:  # dicts are unhashable; keep that behavior
+        raise TypeError("unhashable type: '_LazyTrialSystemAttrs'")
+
+
 class Trial(BaseTrial):
     """A single in-progress evaluation of the objective function."""
 
@@ -43,15 +106,28 @@ class Trial(BaseTrial):
         self._trial_id = trial_id
         self.storage = study._storage
         self._cached_frozen_trial = self.storage.get_trial(trial_id)
+        # Enqueued fixed params are snapshotted once at creation (reference :71).
+        self._fixed_params = self._cached_frozen_trial.system_attrs.get("fixed_params", {})
         self._relative_params: dict[str, Any] | None = None
         study.sampler.before_trial(study, self._cached_frozen_trial)
+
+    def _get_latest_trial(self) -> FrozenTrial:
+        """Shallow copy of the cached trial whose system attrs re-read storage lazily.
+
+        Samplers receive this view so attrs written after the snapshot (e.g.
+        GridSampler's grid_id from before_trial) are visible without an eager
+        storage round trip per suggest (reference trial/_trial.py:700-703).
+        """
+        latest = copy.copy(self._cached_frozen_trial)
+        latest.system_attrs = _LazyTrialSystemAttrs(self._trial_id, self.storage)
+        return latest
 
     # ---- relative (joint) sampling, computed lazily on first suggest ----------------
 
     @property
     def relative_params(self) -> dict[str, Any]:
         if self._relative_params is None:
-            trial = self._cached_frozen_trial
+            trial = self._get_latest_trial()
             study = self.study
             self._relative_search_space = study.sampler.infer_relative_search_space(study, trial)
             self._relative_params = study.sampler.sample_relative(
@@ -175,14 +251,16 @@ class Trial(BaseTrial):
             )
 
         if self._is_fixed_param(name, distribution):
-            param_value = trial.system_attrs[_FIXED_PARAMS_KEY][name]
+            param_value = self._fixed_params[name]
         elif distribution.single():
             param_value = _get_single_value(distribution)
         elif self._is_relative_param(name, distribution):
             param_value = self._relative_params[name]  # type: ignore[index]
         else:
             study = self.study
-            param_value = study.sampler.sample_independent(study, trial, name, distribution)
+            param_value = study.sampler.sample_independent(
+                study, self._get_latest_trial(), name, distribution
+            )
 
         param_value_in_internal_repr = distribution.to_internal_repr(param_value)
         storage.set_trial_param(trial_id, name, param_value_in_internal_repr, distribution)
@@ -191,12 +269,9 @@ class Trial(BaseTrial):
         return param_value
 
     def _is_fixed_param(self, name: str, distribution: BaseDistribution) -> bool:
-        system_attrs = self._cached_frozen_trial.system_attrs
-        if _FIXED_PARAMS_KEY not in system_attrs:
+        if name not in self._fixed_params:
             return False
-        if name not in system_attrs[_FIXED_PARAMS_KEY]:
-            return False
-        param_value = system_attrs[_FIXED_PARAMS_KEY][name]
+        param_value = self._fixed_params[name]
         param_value_in_internal_repr = distribution.to_internal_repr(param_value)
         contained = distribution._contains(param_value_in_internal_repr)
         if not contained:
