@@ -1,0 +1,470 @@
+"""Acquisition functions over fitted GPRegressors (K5 target).
+
+Parity: reference ``optuna/_gp/acqf.py`` (standard_logei :96, logei :121, LogEI
+:156, qLogEI :180, LogPI :215 with Kriging-Believer append, qLogPI :257,
+UCB/LCB :288/:304, LogCEI :320, qLogCEI :353, LogEHVI :404 with non-dominated
+box decomposition, qLogEHVI :473, LogCEHVI :534). All math in fp64.
+"""
+from __future__ import annotations
+
+import math
+from abc import ABC, abstractmethod
+from typing import TYPE_CHECKING
+
+import numpy as np
+
+from optuna_amd._gp.gp import ConditionalGPRegressor, GPRegressor
+from optuna_amd._gp.qmc import sample_from_normal_sobol
+from optuna_amd._hypervolume.box_decomposition import get_non_dominated_box_bounds
+from optuna_amd.study._multi_objective import _is_pareto_front
+
+
+if TYPE_CHECKING:
+    import torch
+
+    from optuna_amd._gp.search_space import SearchSpace
+else:
+    from optuna_amd._imports import _LazyImport
+
+    torch = _LazyImport("torch")
+
+_SQRT_HALF = math.sqrt(0.5)
+_INV_SQRT_2PI = 1 / math.sqrt(2 * math.pi)
+_SQRT_HALF_PI = math.sqrt(0.5 * math.pi)
+_LOG_SQRT_2PI = math.log(math.sqrt(2 * math.pi))
+_EPS = 1e-12  # zero eps makes gradients NaN
+
+
+def standard_logei(z: "torch.Tensor") -> "torch.Tensor":
+    """log E_{x~N(0,1)}[max(0, x+z)], tail-stable (erfcx in the far left tail)."""
+    out = (
+        (z_half := 0.5 * z) * torch.special.erfc(-_SQRT_HALF * z)
+        + (-z_half * z).exp() * _INV_SQRT_2PI
+    ).log()
+    small = z < -25
+    if (z_small := z[small]).numel():
+        out[small] = (
+            -0.5 * z_small**2
+            - _LOG_SQRT_2PI
+            + (1 + _SQRT_HALF_PI * z_small * torch.special.erfcx(-_SQRT_HALF * z_small)).log()
+        )
+    return out
+
+
+def logei(mean: "torch.Tensor", var: "torch.Tensor", f0: float) -> "torch.Tensor":
+    sigma = var.sqrt_()
+    return standard_logei((mean - f0) / sigma) + sigma.log()
+
+
+def logehvi(
+    Y_post: "torch.Tensor",  # (..., n_qmc_samples, n_objectives)
+    non_dominated_box_lower_bounds: "torch.Tensor",
+    non_dominated_box_intervals: "torch.Tensor",
+) -> "torch.Tensor":
+    log_n = float(np.log(Y_post.shape[-2]))
+    diff = Y_post.unsqueeze(-2) - non_dominated_box_lower_bounds
+    diff.clamp_(min=torch.tensor(_EPS, dtype=torch.float64), max=non_dominated_box_intervals)
+    return torch.special.logsumexp(diff.log().sum(dim=-1), dim=(-2, -1)) - log_n
+
+
+def _per_sample_log_hvi(
+    Y_post: "torch.Tensor",
+    non_dominated_box_lower_bounds: "torch.Tensor",
+    non_dominated_box_intervals: "torch.Tensor",
+) -> "torch.Tensor":
+    diff = Y_post.unsqueeze(-2) - non_dominated_box_lower_bounds
+    diff.clamp_(min=torch.tensor(_EPS, dtype=torch.float64), max=non_dominated_box_intervals)
+    return torch.special.logsumexp(diff.log().sum(dim=-1), dim=-1)
+
+
+def _get_reference_point(Y: "torch.Tensor") -> np.ndarray:
+    loss_vals = -Y.numpy()
+    ref = np.max(loss_vals, axis=0)
+    return np.nextafter(np.maximum(1.1 * ref, 0.9 * ref), np.inf)
+
+
+def _get_boxes(Y: "torch.Tensor", ref_point: np.ndarray) -> tuple["torch.Tensor", "torch.Tensor"]:
+    loss_vals = -Y.numpy()
+    loss_vals = loss_vals[np.all(loss_vals < ref_point, axis=-1)]
+    pareto = loss_vals[_is_pareto_front(loss_vals, assume_unique_lexsorted=False)]
+    lbs, ubs = get_non_dominated_box_bounds(pareto, ref_point)
+    return torch.from_numpy(-ubs), torch.from_numpy(-lbs)
+
+
+def _mean_max_log_utility(log_utils: "torch.Tensor") -> "torch.Tensor":
+    max_in_q = torch.amax(log_utils, dim=-1)
+    return torch.special.logsumexp(max_in_q, dim=-1) - math.log(max_in_q.shape[-1])
+
+
+class BaseAcquisitionFunc(ABC):
+    def __init__(self, length_scales: np.ndarray, search_space: "SearchSpace") -> None:
+        self.length_scales = length_scales
+        self.search_space = search_space
+
+    @abstractmethod
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        raise NotImplementedError
+
+    def eval_acqf_no_grad(self, x: np.ndarray) -> np.ndarray:
+        with torch.no_grad():
+            return self.eval_acqf(torch.from_numpy(x)).detach().numpy()
+
+    def eval_acqf_with_grad(self, x: np.ndarray) -> tuple[float, np.ndarray]:
+        assert x.ndim == 1
+        x_tensor = torch.from_numpy(x).requires_grad_(True)
+        val = self.eval_acqf(x_tensor)
+        val.backward()
+        return val.item(), x_tensor.grad.detach().numpy()  # type: ignore[union-attr]
+
+
+class LogEI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr: GPRegressor,
+        search_space: "SearchSpace",
+        threshold: float,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        self._gpr = gpr
+        self._stabilizing_noise = stabilizing_noise
+        self._threshold = threshold
+        super().__init__(gpr.length_scales, search_space)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        if np.isneginf(self._threshold):
+            return torch.zeros(x.shape[:-1], dtype=torch.float64)
+        mean, var = self._gpr.posterior(x)
+        return logei(mean=mean, var=var + self._stabilizing_noise, f0=self._threshold)
+
+
+class qLogEI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr: GPRegressor,
+        search_space: "SearchSpace",
+        threshold: float,
+        n_qmc_samples: int,
+        qmc_seed: int,
+        normalized_params_of_running_trials: np.ndarray,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        self._threshold = threshold
+        self._cond_gpr = ConditionalGPRegressor(
+            gpr=gpr,
+            X_running=torch.from_numpy(normalized_params_of_running_trials),
+            n_qmc_samples=n_qmc_samples,
+            qmc_seed=qmc_seed,
+            stabilizing_noise=stabilizing_noise,
+        )
+        n_running = len(normalized_params_of_running_trials)
+        self._per_sample_shape = (n_qmc_samples, n_running + 1)
+        super().__init__(gpr.length_scales, search_space)
+
+    def compute_per_sample_log_utility(self, x: "torch.Tensor") -> "torch.Tensor":
+        if np.isneginf(self._threshold):
+            return torch.zeros(x.shape[:-1] + self._per_sample_shape, dtype=torch.float64)
+        y_post = self._cond_gpr.sample_joint_posterior(x)
+        return (y_post - self._threshold).clamp_min_(_EPS).log()
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        return _mean_max_log_utility(self.compute_per_sample_log_utility(x))
+
+
+class LogPI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr: GPRegressor,
+        search_space: "SearchSpace",
+        threshold: float,
+        normalized_params_of_running_trials: np.ndarray | None = None,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        self._gpr = gpr
+        self._stabilizing_noise = stabilizing_noise
+        self._threshold = threshold
+        if normalized_params_of_running_trials is not None:
+            # Kriging Believer: append running points at their posterior mean.
+            X_running = torch.from_numpy(normalized_params_of_running_trials)
+            self._gpr.append_running_data(X_running, gpr.posterior(X_running)[0])
+        super().__init__(gpr.length_scales, search_space)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        mean, var = self._gpr.posterior(x)
+        sigma = torch.sqrt(var + self._stabilizing_noise)
+        return torch.special.log_ndtr((mean - self._threshold) / sigma)
+
+
+class qLogPI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr: GPRegressor,
+        search_space: "SearchSpace",
+        threshold: float,
+        n_qmc_samples: int,
+        qmc_seed: int,
+        normalized_params_of_running_trials: np.ndarray,
+        stabilizing_noise: float = 1e-12,
+        tau: float = 1e-2,
+    ) -> None:
+        self._threshold = threshold
+        self._tau = tau
+        self._cond_gpr = ConditionalGPRegressor(
+            gpr=gpr,
+            X_running=torch.from_numpy(normalized_params_of_running_trials),
+            n_qmc_samples=n_qmc_samples,
+            qmc_seed=qmc_seed,
+            stabilizing_noise=stabilizing_noise,
+        )
+        super().__init__(gpr.length_scales, search_space)
+
+    def compute_per_sample_log_utility(self, x: "torch.Tensor") -> "torch.Tensor":
+        y_post = self._cond_gpr.sample_joint_posterior(x)
+        return torch.nn.functional.logsigmoid((y_post - self._threshold) / self._tau)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        return _mean_max_log_utility(self.compute_per_sample_log_utility(x))
+
+
+class UCB(BaseAcquisitionFunc):
+    def __init__(self, gpr: GPRegressor, search_space: "SearchSpace", beta: float) -> None:
+        self._gpr = gpr
+        self._beta = beta
+        super().__init__(gpr.length_scales, search_space)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        mean, var = self._gpr.posterior(x)
+        return mean + torch.sqrt(self._beta * var)
+
+
+class LCB(BaseAcquisitionFunc):
+    def __init__(self, gpr: GPRegressor, search_space: "SearchSpace", beta: float) -> None:
+        self._gpr = gpr
+        self._beta = beta
+        super().__init__(gpr.length_scales, search_space)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        mean, var = self._gpr.posterior(x)
+        return mean - torch.sqrt(self._beta * var)
+
+
+class LogCEI(BaseAcquisitionFunc):
+    """log(EI × ∏ feasibility-PI of each constraint)."""
+
+    def __init__(
+        self,
+        gpr: GPRegressor,
+        search_space: "SearchSpace",
+        threshold: float,
+        constraints_gpr_list: list[GPRegressor],
+        constraints_threshold_list: list[float],
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        assert constraints_gpr_list and len(constraints_gpr_list) == len(
+            constraints_threshold_list
+        )
+        self._acqf = LogEI(gpr, search_space, threshold, stabilizing_noise)
+        self._constraints_acqf_list = [
+            LogPI(c_gpr, search_space, c_threshold, None, stabilizing_noise)
+            for c_gpr, c_threshold in zip(constraints_gpr_list, constraints_threshold_list)
+        ]
+        super().__init__(gpr.length_scales, search_space)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        return self._acqf.eval_acqf(x) + sum(
+            acqf.eval_acqf(x) for acqf in self._constraints_acqf_list
+        )
+
+
+class qLogCEI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr: GPRegressor,
+        search_space: "SearchSpace",
+        threshold: float,
+        n_qmc_samples: int,
+        qmc_seed: int,
+        constraints_gpr_list: list[GPRegressor],
+        constraints_threshold_list: list[float],
+        normalized_params_of_running_trials: np.ndarray,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        assert constraints_gpr_list and len(constraints_gpr_list) == len(
+            constraints_threshold_list
+        )
+        self._acqf = qLogEI(
+            gpr,
+            search_space,
+            threshold,
+            n_qmc_samples,
+            qmc_seed,
+            normalized_params_of_running_trials,
+            stabilizing_noise,
+        )
+        self._constraints_acqf_list = [
+            qLogPI(
+                gpr=c_gpr,
+                search_space=search_space,
+                threshold=c_threshold,
+                n_qmc_samples=n_qmc_samples,
+                qmc_seed=qmc_seed + i + 1,
+                normalized_params_of_running_trials=normalized_params_of_running_trials,
+                stabilizing_noise=stabilizing_noise,
+            )
+            for i, (c_gpr, c_threshold) in enumerate(
+                zip(constraints_gpr_list, constraints_threshold_list)
+            )
+        ]
+        super().__init__(gpr.length_scales, search_space)
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        log_feasible_improvement = self._acqf.compute_per_sample_log_utility(x) + sum(
+            acqf.compute_per_sample_log_utility(x) for acqf in self._constraints_acqf_list
+        )
+        return _mean_max_log_utility(log_feasible_improvement)
+
+
+class LogEHVI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr_list: list[GPRegressor],
+        search_space: "SearchSpace",
+        Y_train: "torch.Tensor",
+        n_qmc_samples: int,
+        qmc_seed: int,
+        normalized_params_of_running_trials: np.ndarray | None = None,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        self._stabilizing_noise = stabilizing_noise
+        self._gpr_list = gpr_list
+        if normalized_params_of_running_trials is not None:
+            X_running = torch.from_numpy(normalized_params_of_running_trials)
+            for gpr in self._gpr_list:
+                gpr.append_running_data(X_running, gpr.posterior(X_running)[0])
+        self._fixed_samples = sample_from_normal_sobol(
+            dim=Y_train.shape[-1], n_samples=n_qmc_samples, seed=qmc_seed
+        )
+        ref_point = _get_reference_point(Y_train)
+        self._box_lower, box_upper = _get_boxes(Y_train, ref_point)
+        self._box_intervals = (box_upper - self._box_lower).clamp_min_(_EPS)
+        super().__init__(
+            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space
+        )
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        Y_post = []
+        for i, gpr in enumerate(self._gpr_list):
+            mean, var = gpr.posterior(x)
+            stdev = torch.sqrt(var + self._stabilizing_noise)
+            Y_post.append(mean[..., None] + stdev[..., None] * self._fixed_samples[..., i])
+        return logehvi(
+            Y_post=torch.stack(Y_post, dim=-1),
+            non_dominated_box_lower_bounds=self._box_lower,
+            non_dominated_box_intervals=self._box_intervals,
+        )
+
+
+class qLogEHVI(BaseAcquisitionFunc):
+    def __init__(
+        self,
+        gpr_list: list[GPRegressor],
+        search_space: "SearchSpace",
+        Y_train: "torch.Tensor",
+        n_qmc_samples: int,
+        qmc_seed: int,
+        normalized_params_of_running_trials: np.ndarray,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        self._Y_train = Y_train
+        self._cond_gpr_list = [
+            ConditionalGPRegressor(
+                gpr=gpr,
+                X_running=torch.from_numpy(normalized_params_of_running_trials),
+                n_qmc_samples=n_qmc_samples,
+                qmc_seed=qmc_seed + i,
+                stabilizing_noise=stabilizing_noise,
+            )
+            for i, gpr in enumerate(gpr_list)
+        ]
+        ref_point = _get_reference_point(Y_train)
+        lower_list, interval_list = [], []
+        for fantasy in torch.stack(
+            [cg.get_fantasy_samples() for cg in self._cond_gpr_list], dim=-1
+        ):
+            Y_fantasy = torch.cat([Y_train, fantasy], dim=0)
+            lower, upper = _get_boxes(Y_fantasy, ref_point)
+            lower_list.append(lower)
+            interval_list.append((upper - lower).clamp_min_(_EPS))
+        self._box_lower = torch.nn.utils.rnn.pad_sequence(lower_list, batch_first=True)
+        self._box_intervals = torch.nn.utils.rnn.pad_sequence(
+            interval_list, batch_first=True, padding_value=_EPS
+        )
+        super().__init__(
+            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space
+        )
+
+    def compute_per_sample_log_utility(self, x: "torch.Tensor") -> "torch.Tensor":
+        Y_candidate_post = torch.stack(
+            [cg.sample_joint_posterior(x, return_fantasy=False) for cg in self._cond_gpr_list],
+            dim=-1,
+        )
+        return _per_sample_log_hvi(
+            Y_post=Y_candidate_post,
+            non_dominated_box_lower_bounds=self._box_lower,
+            non_dominated_box_intervals=self._box_intervals,
+        )
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        log_utils = self.compute_per_sample_log_utility(x)
+        return torch.special.logsumexp(log_utils, dim=-1) - math.log(log_utils.shape[-1])
+
+
+class LogCEHVI(BaseAcquisitionFunc):
+    """EHVI over feasible observations × feasibility-PI of each constraint."""
+
+    def __init__(
+        self,
+        gpr_list: list[GPRegressor],
+        search_space: "SearchSpace",
+        Y_feasible: "torch.Tensor | None",
+        n_qmc_samples: int,
+        qmc_seed: int,
+        constraints_gpr_list: list[GPRegressor],
+        constraints_threshold_list: list[float],
+        normalized_params_of_running_trials: np.ndarray | None = None,
+        stabilizing_noise: float = 1e-12,
+    ) -> None:
+        assert constraints_gpr_list and len(constraints_gpr_list) == len(
+            constraints_threshold_list
+        )
+        self._acqf: BaseAcquisitionFunc | None = (
+            LogEHVI(
+                gpr_list=gpr_list,
+                search_space=search_space,
+                Y_train=Y_feasible,
+                n_qmc_samples=n_qmc_samples,
+                qmc_seed=qmc_seed,
+                normalized_params_of_running_trials=normalized_params_of_running_trials,
+                stabilizing_noise=stabilizing_noise,
+            )
+            if Y_feasible is not None
+            else None
+        )
+        self._constraints_acqf_list = [
+            LogPI(
+                c_gpr,
+                search_space,
+                c_threshold,
+                normalized_params_of_running_trials,
+                stabilizing_noise,
+            )
+            for c_gpr, c_threshold in zip(constraints_gpr_list, constraints_threshold_list)
+        ]
+        super().__init__(
+            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space
+        )
+
+    def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
+        feasibility = sum(acqf.eval_acqf(x) for acqf in self._constraints_acqf_list)
+        if self._acqf is None:
+            return feasibility  # no feasible observation yet: maximize feasibility
+        return self._acqf.eval_acqf(x) + feasibility

@@ -1,0 +1,182 @@
+"""GP-internal normalized search space ([0,1]^d with per-dim scale descriptors).
+
+Parity: reference ``optuna/_gp/search_space.py`` (scale types LINEAR/LOG/
+CATEGORICAL :30, normalize/unnormalize :123-168, Sobol sampling with discrete
+rounding :171-195, discrete-choice grids :103). The per-dim descriptor table
+(scale_type, bounds, step) is the same layout the K7 HIP transform kernel
+consumes for device-resident trial tables.
+"""
+from __future__ import annotations
+
+import math
+import threading
+from enum import IntEnum
+from typing import TYPE_CHECKING, Any
+
+import numpy as np
+
+from optuna_amd.distributions import (
+    BaseDistribution,
+    CategoricalDistribution,
+    FloatDistribution,
+    IntDistribution,
+)
+
+
+if TYPE_CHECKING:
+    from optuna_amd.trial import FrozenTrial
+
+_threading_lock = threading.Lock()
+
+
+class ScaleType(IntEnum):
+    LINEAR = 0
+    LOG = 1
+    CATEGORICAL = 2
+
+
+class SearchSpace:
+    def __init__(self, optuna_search_space: dict[str, BaseDistribution]) -> None:
+        self._optuna_search_space = optuna_search_space
+        n = len(optuna_search_space)
+        self.scale_types = np.empty(n, dtype=np.int64)
+        self.bounds = np.empty((n, 2), dtype=float)
+        self.steps = np.empty(n, dtype=float)
+        for i, dist in enumerate(optuna_search_space.values()):
+            if isinstance(dist, CategoricalDistribution):
+                self.scale_types[i] = ScaleType.CATEGORICAL
+                self.bounds[i] = (0.0, len(dist.choices))
+                self.steps[i] = 1.0
+            else:
+                assert isinstance(dist, (FloatDistribution, IntDistribution))
+                self.scale_types[i] = ScaleType.LOG if dist.log else ScaleType.LINEAR
+                self.bounds[i] = (dist.low, dist.high)
+                self.steps[i] = dist.step or 0.0
+        self.dim = n
+        self.is_categorical = self.scale_types == ScaleType.CATEGORICAL
+        self.discrete_indices = np.flatnonzero(self.steps > 0).astype(int)
+        self.continuous_indices = np.flatnonzero(self.steps == 0.0).astype(int)
+
+    def get_normalized_params(
+        self,
+        trials: list["FrozenTrial"],
+        trial_params: list[dict[str, Any]] | None = None,
+    ) -> np.ndarray:
+        values = np.empty((len(trials), self.dim), dtype=float)
+        if trial_params is None:
+            trial_params = [t.params for t in trials]
+        for i, (name, dist) in enumerate(self._optuna_search_space.items()):
+            if isinstance(dist, CategoricalDistribution):
+                values[:, i] = [dist.to_internal_repr(tp[name]) for tp in trial_params]
+            else:
+                values[:, i] = normalize_one_param(
+                    np.array([tp[name] for tp in trial_params]),
+                    ScaleType(self.scale_types[i]),
+                    (self.bounds[i, 0], self.bounds[i, 1]),
+                    self.steps[i],
+                )
+        return values
+
+    def get_unnormalized_param(self, normalized_param: np.ndarray) -> dict[str, Any]:
+        out: dict[str, Any] = {}
+        for i, (name, dist) in enumerate(self._optuna_search_space.items()):
+            if isinstance(dist, CategoricalDistribution):
+                out[name] = dist.to_external_repr(normalized_param[i])
+            else:
+                assert isinstance(dist, (FloatDistribution, IntDistribution))
+                scale_type = ScaleType.LOG if dist.log else ScaleType.LINEAR
+                step = 0.0 if dist.step is None else dist.step
+                value = float(
+                    np.clip(
+                        unnormalize_one_param(
+                            normalized_param[i], scale_type, (dist.low, dist.high), step
+                        ),
+                        dist.low,
+                        dist.high,
+                    )
+                )
+                out[name] = round(value) if isinstance(dist, IntDistribution) else value
+        return out
+
+    def sample_normalized_params(
+        self, n: int, rng: np.random.RandomState | None
+    ) -> np.ndarray:
+        from scipy.stats import qmc
+
+        rng = rng or np.random.RandomState()
+        with _threading_lock:
+            engine = qmc.Sobol(
+                self.dim, scramble=True, seed=rng.randint(np.iinfo(np.int32).max)
+            )
+        values = engine.random(n)
+        for i in range(self.dim):
+            if self.scale_types[i] == ScaleType.CATEGORICAL:
+                values[:, i] = np.floor(values[:, i] * self.bounds[i, 1])
+            elif self.steps[i] != 0.0:
+                values[:, i] = round_one_normalized_param(
+                    values[:, i],
+                    ScaleType(self.scale_types[i]),
+                    (self.bounds[i, 0], self.bounds[i, 1]),
+                    self.steps[i],
+                )
+        return values
+
+    def get_choices_of_discrete_params(self) -> list[np.ndarray]:
+        return [
+            (
+                np.arange(self.bounds[i, 1])
+                if self.is_categorical[i]
+                else normalize_one_param(
+                    np.arange(
+                        self.bounds[i, 0],
+                        self.bounds[i, 1] + 0.5 * self.steps[i],
+                        self.steps[i],
+                    ),
+                    ScaleType(self.scale_types[i]),
+                    (self.bounds[i, 0], self.bounds[i, 1]),
+                    self.steps[i],
+                )
+            )
+            for i in self.discrete_indices
+        ]
+
+
+def normalize_one_param(
+    param_value: np.ndarray, scale_type: ScaleType, bounds: tuple[float, float], step: float
+) -> np.ndarray:
+    if scale_type == ScaleType.CATEGORICAL:
+        return param_value
+    low, high = bounds[0] - 0.5 * step, bounds[1] + 0.5 * step
+    if scale_type == ScaleType.LOG:
+        low, high = math.log(low), math.log(high)
+        param_value = np.log(param_value)
+    if high == low:
+        return np.full_like(param_value, 0.5)
+    return (param_value - low) / (high - low)
+
+
+def unnormalize_one_param(
+    param_value: np.ndarray, scale_type: ScaleType, bounds: tuple[float, float], step: float
+) -> np.ndarray:
+    if scale_type == ScaleType.CATEGORICAL:
+        return param_value
+    low, high = bounds[0] - 0.5 * step, bounds[1] + 0.5 * step
+    if scale_type == ScaleType.LOG:
+        low, high = math.log(low), math.log(high)
+    param_value = param_value * (high - low) + low
+    if scale_type == ScaleType.LOG:
+        param_value = np.exp(param_value)
+    return param_value
+
+
+def round_one_normalized_param(
+    param_value: np.ndarray, scale_type: ScaleType, bounds: tuple[float, float], step: float
+) -> np.ndarray:
+    assert scale_type != ScaleType.CATEGORICAL
+    if step == 0.0:
+        return param_value
+    raw = unnormalize_one_param(param_value, scale_type, bounds, step)
+    raw = np.clip(
+        (raw - bounds[0] + 0.5 * step) // step * step + bounds[0], bounds[0], bounds[1]
+    )
+    return normalize_one_param(raw, scale_type, bounds, step)

@@ -1,0 +1,433 @@
+"""GPSampler: Gaussian-process Bayesian optimization.
+
+Per suggest: fit one GP per objective (kernel-parameter warm start from the
+previous fit), pick the acquisition from the {n_objectives × constraints ×
+running-trials} matrix — {LogEI, qLogEI, LogCEI, qLogCEI, LogEHVI, qLogEHVI,
+LogCEHVI} — and optimize it with the mixed QMC+local-search optimizer. Running
+trials are handled via QMC fantasies (q-variants) and their relative params are
+shared through chunked system attrs like TPE's constant liar.
+
+Parity: reference ``optuna/samplers/_gp/sampler.py`` (GPSampler :70, acqf
+selection matrix :462-573, n_preliminary_samples=2048 / n_local_search=10 /
+n_qmc_samples=128 :272-279, score standardization :62-67).
+"""
+from __future__ import annotations
+
+import json
+from typing import TYPE_CHECKING, Any, Callable, Sequence
+
+import numpy as np
+
+from optuna_amd import logging as _logging
+from optuna_amd._gp import acqf as acqf_module
+from optuna_amd._gp import gp
+from optuna_amd._gp import optim_mixed
+from optuna_amd._gp import prior
+from optuna_amd._gp import search_space as gp_search_space
+from optuna_amd.distributions import BaseDistribution
+from optuna_amd.samplers._base import (
+    _INDEPENDENT_SAMPLING_WARNING_TEMPLATE,
+    BaseSampler,
+    _process_constraints_after_trial,
+)
+from optuna_amd.samplers._lazy_random_state import LazyRandomState
+from optuna_amd.samplers._random import RandomSampler
+from optuna_amd.search_space import IntersectionSearchSpace
+from optuna_amd.study._multi_objective import _is_pareto_front
+from optuna_amd.study._study_direction import StudyDirection
+from optuna_amd.trial import FrozenTrial, TrialState
+
+
+if TYPE_CHECKING:
+    import torch
+
+    from optuna_amd.study import Study
+else:
+    from optuna_amd._imports import _LazyImport
+
+    torch = _LazyImport("torch")
+
+_logger = _logging.get_logger(__name__)
+
+EPS = 1e-10
+_RELATIVE_PARAMS_KEY = "gp:relative_params"
+_SYSTEM_ATTR_MAX_LENGTH = 2045
+_MAX_QMC_SEED_VALUE = 2**31 - 1
+
+
+def _standardize_values(values: np.ndarray) -> tuple[np.ndarray, np.ndarray, np.ndarray]:
+    clipped = gp.warn_and_convert_inf(values)
+    means = np.mean(clipped, axis=0)
+    stds = np.std(clipped, axis=0)
+    standardized = (clipped - means) / np.maximum(EPS, stds)
+    return standardized, means, stds
+
+
+def _get_params(trial: FrozenTrial) -> dict[str, Any]:
+    """Params incl. relative params shared via chunked system attrs (RUNNING trials)."""
+    if trial.state.is_finished():
+        return trial.params
+    chunks = []
+    i = 0
+    while chunk := trial.system_attrs.get(f"{_RELATIVE_PARAMS_KEY}:{i}"):
+        chunks.append(chunk)
+        i += 1
+    if not chunks:
+        return trial.params
+    try:
+        params = json.loads("".join(chunks))
+    except json.JSONDecodeError:
+        return trial.params
+    params.update(trial.params)
+    return params
+
+
+def _is_constrained_optimization(trials: list[FrozenTrial]) -> bool:
+    return any("constraints" in t.system_attrs for t in trials)
+
+
+def _get_constraint_vals_and_feasibility(
+    study: "Study", trials: list[FrozenTrial]
+) -> tuple[np.ndarray, np.ndarray]:
+    constraint_vals = np.array(
+        [list(t.constraints.values()) for t in trials], dtype=np.float64
+    )
+    assert constraint_vals.ndim == 2, "each trial must have the same number of constraints"
+    constraint_vals = gp.warn_and_convert_inf(constraint_vals)
+    is_feasible = np.all(constraint_vals <= 0, axis=1)
+    return constraint_vals, is_feasible
+
+
+class GPSampler(BaseSampler):
+    """Gaussian-process sampler (see module docstring)."""
+
+    def __init__(
+        self,
+        *,
+        seed: int | None = None,
+        independent_sampler: BaseSampler | None = None,
+        n_startup_trials: int = 10,
+        deterministic_objective: bool = False,
+        constraints_func: Callable[[FrozenTrial], Sequence[float]] | None = None,
+        warn_independent_sampling: bool = True,
+    ) -> None:
+        self._rng = LazyRandomState(seed)
+        self._independent_sampler = independent_sampler or RandomSampler(seed=seed)
+        self._intersection_search_space = IntersectionSearchSpace()
+        self._n_startup_trials = n_startup_trials
+        self._log_prior = prior.default_log_prior
+        self._minimum_noise: float = prior.DEFAULT_MINIMUM_NOISE_VAR
+        self._gprs_cache_list: list[gp.GPRegressor] | None = None
+        self._constraints_gprs_cache_list: list[gp.GPRegressor] | None = None
+        self._deterministic = deterministic_objective
+        self._constraints_func = constraints_func
+        self._warn_independent_sampling = warn_independent_sampling
+
+        self._n_preliminary_samples = 2048
+        self._n_local_search = 10
+        self._tol = 1e-4
+        self._n_qmc_samples_qei = 128
+        self._n_qmc_samples_ehvi = 128
+
+    def reseed_rng(self) -> None:
+        self._rng.rng.seed()
+        self._independent_sampler.reseed_rng()
+
+    def infer_relative_search_space(
+        self, study: "Study", trial: FrozenTrial
+    ) -> dict[str, BaseDistribution]:
+        search_space = {}
+        for name, distribution in self._intersection_search_space.calculate(study).items():
+            if distribution.single():
+                continue
+            search_space[name] = distribution
+        return search_space
+
+    def _optimize_acqf(
+        self, acqf: acqf_module.BaseAcquisitionFunc, best_params: np.ndarray | None
+    ) -> np.ndarray:
+        assert best_params is None or best_params.ndim == 2
+        normalized_params, _ = optim_mixed.optimize_acqf_mixed(
+            acqf,
+            warmstart_normalized_params_array=best_params,
+            n_preliminary_samples=self._n_preliminary_samples,
+            n_local_search=self._n_local_search,
+            tol=self._tol,
+            rng=self._rng.rng,
+        )
+        return normalized_params
+
+    def _get_constraints_acqf_args(
+        self,
+        constraint_vals: np.ndarray,
+        internal_search_space: gp_search_space.SearchSpace,
+        normalized_params: np.ndarray,
+    ) -> tuple[list[gp.GPRegressor], list[float]]:
+        standardized, means, stds = _standardize_values(-constraint_vals)
+        if (
+            self._constraints_gprs_cache_list is not None
+            and len(self._constraints_gprs_cache_list[0].inverse_squared_lengthscales)
+            != internal_search_space.dim
+        ):
+            self._constraints_gprs_cache_list = None
+        is_categorical = internal_search_space.is_categorical
+        thresholds = (-means / np.maximum(EPS, stds)).tolist()
+        gprs = []
+        for i, vals in enumerate(standardized.T):
+            cache = (
+                self._constraints_gprs_cache_list[i]
+                if self._constraints_gprs_cache_list is not None
+                else None
+            )
+            gprs.append(
+                gp.fit_kernel_params(
+                    X=normalized_params,
+                    Y=vals,
+                    is_categorical=is_categorical,
+                    log_prior=self._log_prior,
+                    minimum_noise=self._minimum_noise,
+                    gpr_cache=cache,
+                    deterministic_objective=self._deterministic,
+                )
+            )
+        self._constraints_gprs_cache_list = gprs
+        return gprs, thresholds
+
+    def _get_best_params_for_multi_objective(
+        self, normalized_params: np.ndarray, standardized_score_vals: np.ndarray
+    ) -> np.ndarray:
+        pareto_params = normalized_params[
+            _is_pareto_front(-standardized_score_vals, assume_unique_lexsorted=False)
+        ]
+        size = min(self._n_local_search // 2, len(pareto_params))
+        chosen = self._rng.rng.choice(len(pareto_params), size=size, replace=False)
+        return pareto_params[chosen]
+
+    def sample_relative(
+        self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
+    ) -> dict[str, Any]:
+        if search_space == {}:
+            return {}
+        states = (TrialState.COMPLETE, TrialState.RUNNING)
+        trials = study._get_trials(deepcopy=False, states=states, use_cache=False)
+        completed_trials = [t for t in trials if t.state == TrialState.COMPLETE]
+        running_trials = [
+            t
+            for t in trials
+            if t.state == TrialState.RUNNING
+            and t._trial_id != trial._trial_id
+            and search_space.keys() <= _get_params(t).keys()
+        ]
+        if len(completed_trials) < self._n_startup_trials:
+            return {}
+
+        with torch.device("cpu"):
+            params = self._sample_relative_impl(
+                study, completed_trials, running_trials, search_space
+            )
+
+        if params != {}:
+            params_str = json.dumps(params)
+            for i in range(0, len(params_str), _SYSTEM_ATTR_MAX_LENGTH):
+                study._storage.set_trial_system_attr(
+                    trial._trial_id,
+                    f"{_RELATIVE_PARAMS_KEY}:{i // _SYSTEM_ATTR_MAX_LENGTH}",
+                    params_str[i : i + _SYSTEM_ATTR_MAX_LENGTH],
+                )
+        return params
+
+    def _sample_relative_impl(
+        self,
+        study: "Study",
+        completed_trials: list[FrozenTrial],
+        running_trials: list[FrozenTrial],
+        search_space: dict[str, BaseDistribution],
+    ) -> dict[str, Any]:
+        internal_search_space = gp_search_space.SearchSpace(search_space)
+        normalized_params = internal_search_space.get_normalized_params(completed_trials)
+        X_running = (
+            internal_search_space.get_normalized_params(
+                running_trials, [_get_params(t) for t in running_trials]
+            )
+            if len(running_trials) > 0
+            else None
+        )
+        signs = np.array(
+            [-1.0 if d == StudyDirection.MINIMIZE else 1.0 for d in study.directions]
+        )
+        standardized_score_vals, _, _ = _standardize_values(
+            signs * np.array([t.values for t in completed_trials])
+        )
+
+        if (
+            self._gprs_cache_list is not None
+            and len(self._gprs_cache_list[0].inverse_squared_lengthscales)
+            != internal_search_space.dim
+        ):
+            self._gprs_cache_list = None
+
+        n_objectives = standardized_score_vals.shape[-1]
+        is_categorical = internal_search_space.is_categorical
+        gprs_list = []
+        for i in range(n_objectives):
+            cache = self._gprs_cache_list[i] if self._gprs_cache_list is not None else None
+            gprs_list.append(
+                gp.fit_kernel_params(
+                    X=normalized_params,
+                    Y=standardized_score_vals[:, i],
+                    is_categorical=is_categorical,
+                    log_prior=self._log_prior,
+                    minimum_noise=self._minimum_noise,
+                    gpr_cache=cache,
+                    deterministic_objective=self._deterministic,
+                )
+            )
+        self._gprs_cache_list = gprs_list
+
+        best_params: np.ndarray | None
+        acqf: acqf_module.BaseAcquisitionFunc
+        qmc_seed = int(self._rng.rng.randint(_MAX_QMC_SEED_VALUE))
+
+        if not _is_constrained_optimization(completed_trials):
+            if n_objectives == 1:
+                threshold = float(standardized_score_vals[:, 0].max())
+                if X_running is None:
+                    acqf = acqf_module.LogEI(
+                        gpr=gprs_list[0],
+                        search_space=internal_search_space,
+                        threshold=threshold,
+                    )
+                else:
+                    acqf = acqf_module.qLogEI(
+                        gpr=gprs_list[0],
+                        search_space=internal_search_space,
+                        threshold=threshold,
+                        n_qmc_samples=self._n_qmc_samples_qei,
+                        qmc_seed=qmc_seed,
+                        normalized_params_of_running_trials=X_running,
+                    )
+                best_params = normalized_params[
+                    np.argmax(standardized_score_vals), np.newaxis
+                ]
+            else:
+                if X_running is None:
+                    acqf = acqf_module.LogEHVI(
+                        gpr_list=gprs_list,
+                        search_space=internal_search_space,
+                        Y_train=torch.from_numpy(standardized_score_vals),
+                        n_qmc_samples=self._n_qmc_samples_ehvi,
+                        qmc_seed=qmc_seed,
+                    )
+                else:
+                    acqf = acqf_module.qLogEHVI(
+                        gpr_list=gprs_list,
+                        search_space=internal_search_space,
+                        Y_train=torch.from_numpy(standardized_score_vals),
+                        n_qmc_samples=self._n_qmc_samples_ehvi,
+                        qmc_seed=qmc_seed,
+                        normalized_params_of_running_trials=X_running,
+                    )
+                best_params = self._get_best_params_for_multi_objective(
+                    normalized_params, standardized_score_vals
+                )
+        else:
+            constraint_vals, is_feasible = _get_constraint_vals_and_feasibility(
+                study, completed_trials
+            )
+            constr_gprs, constr_thresholds = self._get_constraints_acqf_args(
+                constraint_vals, internal_search_space, normalized_params
+            )
+            if n_objectives == 1:
+                y_with_neginf = np.where(is_feasible, standardized_score_vals[:, 0], -np.inf)
+                i_opt = int(np.argmax(y_with_neginf))
+                best_feasible_y = float(y_with_neginf[i_opt])
+                if X_running is None:
+                    acqf = acqf_module.LogCEI(
+                        gpr=gprs_list[0],
+                        search_space=internal_search_space,
+                        threshold=best_feasible_y,
+                        constraints_gpr_list=constr_gprs,
+                        constraints_threshold_list=constr_thresholds,
+                    )
+                else:
+                    acqf = acqf_module.qLogCEI(
+                        gpr=gprs_list[0],
+                        search_space=internal_search_space,
+                        threshold=best_feasible_y,
+                        n_qmc_samples=self._n_qmc_samples_qei,
+                        qmc_seed=qmc_seed,
+                        constraints_gpr_list=constr_gprs,
+                        constraints_threshold_list=constr_thresholds,
+                        normalized_params_of_running_trials=X_running,
+                    )
+                best_params = (
+                    None
+                    if np.isneginf(best_feasible_y)
+                    else normalized_params[i_opt, np.newaxis]
+                )
+            else:
+                is_all_infeasible = not bool(np.any(is_feasible))
+                acqf = acqf_module.LogCEHVI(
+                    gpr_list=gprs_list,
+                    search_space=internal_search_space,
+                    Y_feasible=(
+                        torch.from_numpy(standardized_score_vals[is_feasible])
+                        if not is_all_infeasible
+                        else None
+                    ),
+                    n_qmc_samples=self._n_qmc_samples_ehvi,
+                    qmc_seed=qmc_seed,
+                    constraints_gpr_list=constr_gprs,
+                    constraints_threshold_list=constr_thresholds,
+                    normalized_params_of_running_trials=X_running,
+                )
+                best_params = (
+                    self._get_best_params_for_multi_objective(
+                        normalized_params[is_feasible],
+                        standardized_score_vals[is_feasible],
+                    )
+                    if not is_all_infeasible
+                    else None
+                )
+
+        normalized_param = self._optimize_acqf(acqf, best_params)
+        return internal_search_space.get_unnormalized_param(normalized_param)
+
+    def sample_independent(
+        self,
+        study: "Study",
+        trial: FrozenTrial,
+        param_name: str,
+        param_distribution: BaseDistribution,
+    ) -> Any:
+        if self._warn_independent_sampling:
+            completed = study._get_trials(
+                deepcopy=False, states=(TrialState.COMPLETE,), use_cache=True
+            )
+            if len(completed) >= self._n_startup_trials:
+                _logger.warning(
+                    _INDEPENDENT_SAMPLING_WARNING_TEMPLATE.format(
+                        param_name=param_name,
+                        trial_number=trial.number,
+                        sampler_name=self.__class__.__name__,
+                        fallback_name=self._independent_sampler.__class__.__name__,
+                        reason="dynamic search space is not supported by GPSampler",
+                    )
+                )
+        return self._independent_sampler.sample_independent(
+            study, trial, param_name, param_distribution
+        )
+
+    def before_trial(self, study: "Study", trial: FrozenTrial) -> None:
+        self._independent_sampler.before_trial(study, trial)
+
+    def after_trial(
+        self,
+        study: "Study",
+        trial: FrozenTrial,
+        state: TrialState,
+        values: Sequence[float] | None,
+    ) -> None:
+        if self._constraints_func is not None:
+            _process_constraints_after_trial(self._constraints_func, study, trial, state)
+        self._independent_sampler.after_trial(study, trial, state, values)
