@@ -1,0 +1,284 @@
+"""PED-ANOVA importance (the default evaluator).
+
+Quantile-filter the top-γ' trials, fit per-parameter Scott-bandwidth Parzen
+estimators on an integer grid for the top set and the region set, and score each
+parameter by the Pearson divergence between the two grid densities (conditional
+regimes handled by partitioning on the parameter's distribution).
+
+Parity: reference ``optuna/importance/_ped_anova/`` (evaluator.py:51 —
+_QuantileFilter :26, γ'²-scaled regime sum; scott_parzen_estimator.py — Scott's
+rule bandwidth with IQR guard, grid discretization, prior kernel).
+"""
+from __future__ import annotations
+
+import math
+from collections import defaultdict
+from typing import TYPE_CHECKING, Callable
+
+import numpy as np
+
+from optuna_amd.distributions import (
+    BaseDistribution,
+    CategoricalDistribution,
+    FloatDistribution,
+    IntDistribution,
+)
+from optuna_amd.importance._base import (
+    BaseImportanceEvaluator,
+    _get_filtered_trials,
+    _sort_dict_by_importance,
+)
+from optuna_amd.samplers._tpe import _truncnorm_np as _tn
+from optuna_amd.study._study_direction import StudyDirection
+from optuna_amd.trial import FrozenTrial, TrialState
+
+
+if TYPE_CHECKING:
+    from optuna_amd.study import Study
+
+import warnings
+
+
+class _QuantileFilter:
+    def __init__(
+        self,
+        quantile: float,
+        is_lower_better: bool,
+        target: Callable[[FrozenTrial], float] | None,
+    ) -> None:
+        assert 0 < quantile <= 1, "quantile must be in (0, 1]."
+        self._quantile = quantile
+        self._is_lower_better = is_lower_better
+        self._target = target
+
+    def filter(self, trials: list[FrozenTrial]) -> list[FrozenTrial]:
+        sign = 1.0 if self._is_lower_better else -1.0
+        loss_values = sign * np.asarray(
+            [t.value if self._target is None else self._target(t) for t in trials]
+        )
+        cutoff_index = int(math.ceil(self._quantile * loss_values.size)) - 1
+        cutoff_val = float(np.partition(loss_values, cutoff_index)[cutoff_index])
+        keep = loss_values <= cutoff_val
+        return [t for t, k in zip(trials, keep) if k]
+
+
+class _GridDensity:
+    """Scott-bandwidth Parzen density on an integer grid (one parameter)."""
+
+    def __init__(self, counts: np.ndarray, n_categories: int | None, prior_weight: float) -> None:
+        self._n = len(counts)
+        observations = np.flatnonzero(counts).astype(np.float64)
+        obs_weights = counts[np.flatnonzero(counts)].astype(np.float64)
+        self._categorical = n_categories is not None
+
+        if self._categorical:
+            n_choices = n_categories
+            if len(observations) == 0:
+                self._cat_pdf = np.full(n_choices, 1.0 / n_choices)
+                return
+            n_kernels = len(observations) + 1
+            w = np.full((n_kernels, n_choices), prior_weight / n_kernels)
+            w[np.arange(len(observations)), observations.astype(int)] += obs_weights
+            w /= w.sum(axis=1, keepdims=True)
+            mix = np.append(obs_weights, [prior_weight])
+            mix = mix / mix.sum()
+            self._cat_pdf = mix @ w
+            return
+
+        low, high = 0.0, float(self._n - 1)
+        if len(observations) == 0:
+            mus = np.array([(low + high) / 2])
+            sigmas = np.array([high - low + 1.0])
+            mix = np.array([1.0])
+        else:
+            # Scott's rule with the IQR guard, on the weighted sample.
+            w_cum = np.cumsum(obs_weights)
+            w_sum = w_cum[-1]
+            mean_est = (observations @ obs_weights) / w_sum
+            sigma_est = np.sqrt(
+                ((observations - mean_est) ** 2 @ obs_weights) / max(1, w_sum - 1)
+            )
+            q1 = int(np.searchsorted(w_cum, w_sum // 4, side="left"))
+            q3 = int(np.searchsorted(w_cum, w_sum * 3 // 4, side="right"))
+            iqr = observations[min(observations.size - 1, q3)] - observations[q1]
+            sigma_est = 1.059 * min(iqr / 1.34, sigma_est) * w_sum**-0.2
+            sigma_min = 0.5 / 1.64
+            mus = np.append(observations, [(low + high) / 2])
+            sigmas = np.append(
+                np.full_like(observations, max(sigma_est, sigma_min)), [high - low + 1.0]
+            )
+            mix = np.append(obs_weights, [prior_weight])
+            mix = mix / mix.sum()
+        self._mus = mus
+        self._sigmas = sigmas
+        self._mix = mix
+        self._low = low - 0.5
+        self._high = high + 0.5
+
+    def pdf(self, grid: np.ndarray) -> np.ndarray:
+        if self._categorical:
+            return self._cat_pdf[grid.astype(int)]
+        # Integrated truncated-normal mass over each unit cell.
+        left = (grid - 0.5)[:, None]
+        right = (grid + 0.5)[:, None]
+        m, s = self._mus[None, :], self._sigmas[None, :]
+        cell = _tn._log_gauss_mass((left - m) / s, (right - m) / s)
+        total = _tn._log_gauss_mass(
+            (self._low - self._mus) / self._sigmas, (self._high - self._mus) / self._sigmas
+        )
+        with np.errstate(divide="ignore"):
+            return np.exp(cell - total[None, :]) @ self._mix
+
+
+def _counts_on_grid(
+    param_name: str, dist: BaseDistribution, trials: list[FrozenTrial], n_steps: int
+) -> tuple[np.ndarray, int | None]:
+    if isinstance(dist, CategoricalDistribution):
+        indices = [int(dist.to_internal_repr(t.params[param_name])) for t in trials]
+        return np.bincount(indices, minlength=len(dist.choices)), len(dist.choices)
+    assert isinstance(dist, (FloatDistribution, IntDistribution))
+    if isinstance(dist, IntDistribution) and dist.log:
+        log2_size = int(np.ceil(np.log(dist.high - dist.low + 1) / np.log(2))) + 1
+        n_steps = min(log2_size, n_steps)
+    elif dist.step is not None:
+        assert not dist.log
+        n_steps = min(round((dist.high - dist.low) / dist.step) + 1, n_steps)
+    low, high = (
+        (math.log(dist.low), math.log(dist.high)) if dist.log else (dist.low, dist.high)
+    )
+    values = np.asarray([t.params[param_name] for t in trials], dtype=np.float64)
+    if dist.log:
+        values = np.log(values)
+    step_size = (high - low) / (n_steps - 1)
+    indices = np.clip(np.ceil((values - low) / step_size - 0.5).astype(int), 0, n_steps - 1)
+    return np.bincount(indices, minlength=n_steps), None
+
+
+class PedAnovaImportanceEvaluator(BaseImportanceEvaluator):
+    """PED-ANOVA (see module docstring)."""
+
+    def __init__(
+        self,
+        *,
+        target_quantile: float = 0.1,
+        region_quantile: float = 1.0,
+        evaluate_on_local: bool = True,
+    ) -> None:
+        assert 0.0 < target_quantile < region_quantile <= 1.0, (
+            "condition 0.0 < `target_quantile` < `region_quantile` <= 1.0 must be satisfied"
+        )
+        if region_quantile != 1.0 and not evaluate_on_local:
+            warnings.warn("If `evaluate_on_local` is False, `region_quantile` has no effect.")
+        self._target_quantile = target_quantile
+        self._region_quantile = region_quantile
+        self._evaluate_on_local = evaluate_on_local
+        self._n_steps = 50
+        self._prior_weight = 1.0
+        self._min_n_trials_in_regime = 2
+
+    def _top_quantile_trials(
+        self,
+        study: "Study",
+        trials: list[FrozenTrial],
+        quantile: float,
+        target: Callable[[FrozenTrial], float] | None,
+    ) -> list[FrozenTrial]:
+        if quantile == 1.0:
+            return trials
+        if study._is_multi_objective() and target is None:
+            from optuna_amd.samplers._tpe.sampler import (
+                _split_complete_trials_multi_objective,
+            )
+
+            n_below = math.ceil(quantile * len(trials))
+            top, _ = _split_complete_trials_multi_objective(trials, study, n_below)
+            return top
+        is_lower_better = study.directions[0] == StudyDirection.MINIMIZE
+        if target is not None:
+            warnings.warn(
+                f"{self.__class__.__name__} computes the importances of params to achieve "
+                "low `target` values. If this is not what you want, please modify target, "
+                "e.g., by multiplying the output by -1."
+            )
+            is_lower_better = True
+        return _QuantileFilter(quantile, is_lower_better, target).filter(trials)
+
+    def _pearson_divergence(
+        self,
+        param_name: str,
+        dist: BaseDistribution,
+        target_trials: list[FrozenTrial],
+        region_trials: list[FrozenTrial],
+    ) -> float:
+        counts_top, n_cat = _counts_on_grid(param_name, dist, target_trials, self._n_steps)
+        grid = np.arange(len(counts_top), dtype=np.float64)
+        pdf_top = _GridDensity(counts_top, n_cat, self._prior_weight).pdf(grid) + 1e-12
+        if self._evaluate_on_local:
+            counts_local, n_cat2 = _counts_on_grid(
+                param_name, dist, region_trials, self._n_steps
+            )
+            pdf_local = (
+                _GridDensity(counts_local, n_cat2, self._prior_weight).pdf(grid) + 1e-12
+            )
+        else:
+            pdf_local = np.full(len(grid), 1.0 / len(grid))
+        return float(pdf_local @ ((pdf_top / pdf_local - 1) ** 2))
+
+    def evaluate(
+        self,
+        study: "Study",
+        params: list[str] | None = None,
+        *,
+        target: Callable[[FrozenTrial], float] | None = None,
+    ) -> dict[str, float]:
+        if params is None:
+            # All parameters appearing in completed trials, incl. conditional ones.
+            seen: dict[str, None] = {}
+            for t in study.get_trials(deepcopy=False, states=(TrialState.COMPLETE,)):
+                for name in t.params:
+                    seen.setdefault(name)
+            params = list(seen)
+        trials = _get_filtered_trials(study, target)
+        if len(trials) <= 1:
+            warnings.warn(
+                "The number of trials is too small to compute importances. "
+                "Parameter importances will be equal."
+            )
+            return {k: 0.0 for k in params}
+
+        target_trials = self._top_quantile_trials(study, trials, self._target_quantile, target)
+        region_trials = self._top_quantile_trials(study, trials, self._region_quantile, target)
+        if len(target_trials) == len(region_trials):
+            warnings.warn(
+                "Target and region quantiles select the same set of trials. "
+                "Parameter importances will be equal."
+            )
+        if len(target_trials) == 0:
+            return {k: 0.0 for k in params}
+
+        target_ids = {t._trial_id for t in target_trials}
+        quantile = len(target_trials) / len(region_trials)
+
+        importances = {k: 0.0 for k in params}
+        for param_name in params:
+            # Conditional spaces: partition by the distribution regime.
+            regimes: dict[BaseDistribution | None, list[FrozenTrial]] = defaultdict(list)
+            for t in region_trials:
+                regimes[t.distributions.get(param_name)].append(t)
+            regimes = {
+                k: v for k, v in regimes.items() if len(v) >= self._min_n_trials_in_regime
+            }
+            for dist, regime_region in regimes.items():
+                regime_target = [t for t in regime_region if t._trial_id in target_ids]
+                alpha = len(regime_target) / len(target_trials)
+                beta = len(regime_region) / len(region_trials)
+                if dist is not None and not dist.single() and len(regime_target):
+                    importances[param_name] += (
+                        alpha**2
+                        / beta
+                        * self._pearson_divergence(
+                            param_name, dist, regime_target, regime_region
+                        )
+                    )
+        importances = {k: v * quantile**2 for k, v in importances.items()}
+        return _sort_dict_by_importance(importances)
