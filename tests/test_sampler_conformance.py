@@ -1,0 +1,71 @@
+"""Run the sampler conformance suites over every shipped sampler."""
+from __future__ import annotations
+
+from typing import Callable
+
+import pytest
+
+import optuna_amd
+from optuna_amd.samplers import BaseSampler
+from optuna_amd.testing.pytest_samplers import (
+    BasicSamplerTestCase,
+    MultiObjectiveSamplerTestCase,
+)
+
+
+class TestRandomSampler(BasicSamplerTestCase, MultiObjectiveSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.RandomSampler(seed=seed)
+
+
+class TestTPESampler(BasicSamplerTestCase, MultiObjectiveSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.TPESampler(seed=seed, n_startup_trials=3)
+
+
+class TestTPEMultivariate(BasicSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.TPESampler(
+            seed=seed, n_startup_trials=3, multivariate=True, group=True
+        )
+
+
+class TestCmaEsSampler(BasicSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.CmaEsSampler(
+            seed=seed, n_startup_trials=2, warn_independent_sampling=False
+        )
+
+
+class TestGPSampler(BasicSamplerTestCase):
+    n_trials = 6
+
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.GPSampler(
+            seed=seed, n_startup_trials=3, warn_independent_sampling=False
+        )
+
+
+class TestNSGAIISampler(BasicSamplerTestCase, MultiObjectiveSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.NSGAIISampler(seed=seed, population_size=4)
+
+
+class TestNSGAIIISampler(MultiObjectiveSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.NSGAIIISampler(seed=seed, population_size=4)
+
+
+class TestQMCSampler(BasicSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        return lambda seed: optuna_amd.samplers.QMCSampler(
+            seed=seed, scramble=True, warn_independent_sampling=False
+        )
