@@ -347,7 +347,8 @@ class InMemoryStorage(BaseStorage):
             self._check_study_id(study_id)
             if state is None:
                 return len(self._studies[study_id].trials)
-            return len([t for t in self._studies[study_id].trials if t.state in state])
+            state_rows = self._studies[study_id].state_rows
+            return sum(len(state_rows[s]) for s in TrialState if s in state)
 
     # ---- helpers --------------------------------------------------------------------
 
