@@ -23,6 +23,8 @@
 #include <cstdint>
 #include <stdexcept>
 #include <vector>
+#include <algorithm>
+#include <cstring>
 
 #include "truncnorm_device.h"
 
@@ -203,6 +205,11 @@ struct Workspace {
     double* buf = nullptr;
     size_t capacity = 0;
     hipStream_t stream = nullptr;
+    // Pinned host staging: pageable hipMemcpyAsync silently degrades to a slow
+    // synchronous staged copy; all H2D goes through this buffer instead.
+    char* pinned = nullptr;
+    size_t pinned_capacity = 0;
+    size_t pinned_cursor = 0;
 
     double* ensure(size_t n_doubles) {
         if (n_doubles > capacity) {
@@ -211,6 +218,29 @@ struct Workspace {
             HIP_CHECK(hipMalloc(&buf, capacity * sizeof(double)));
         }
         return buf;
+    }
+    void ensure_pinned(size_t n_bytes) {
+        if (n_bytes > pinned_capacity) {
+            if (pinned) (void)hipHostFree(pinned);
+            pinned_capacity = n_bytes + n_bytes / 2;
+            HIP_CHECK(hipHostMalloc(&pinned, pinned_capacity));
+        }
+    }
+    void begin_uploads() { pinned_cursor = 0; }
+    // memcpy into a fresh pinned slice, then a true-async DMA on the stream.
+    void h2d(void* dst, const void* src, size_t bytes, hipStream_t st) {
+        if (bytes == 0) return;
+        size_t aligned = (bytes + 255) & ~size_t(255);
+        if (pinned_cursor + aligned > pinned_capacity) {
+            // Grow: must drain in-flight DMAs that read the old buffer first.
+            HIP_CHECK(hipStreamSynchronize(st));
+            ensure_pinned(pinned_cursor + aligned);
+            pinned_cursor = 0;
+        }
+        char* slot = pinned + pinned_cursor;
+        pinned_cursor += aligned;
+        memcpy(slot, src, bytes);
+        HIP_CHECK(hipMemcpyAsync(dst, slot, bytes, hipMemcpyHostToDevice, st));
     }
     hipStream_t get_stream() {
         if (!stream) HIP_CHECK(hipStreamCreate(&stream));
@@ -253,10 +283,10 @@ static py::array_t<double> elementwise3(const arr_f64& a, const arr_f64& b,
     hipStream_t st = g_ws.get_stream();
     double* d = g_ws.ensure(4 * (size_t)n);
     double *da = d, *db = d + n, *dc = d + 2 * n, *dout = d + 3 * n;
-    HIP_CHECK(hipMemcpyAsync(da, a.data(), n * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(db, b.data(), n * 8, hipMemcpyHostToDevice, st));
-    if (which == 1)
-        HIP_CHECK(hipMemcpyAsync(dc, c.data(), n * 8, hipMemcpyHostToDevice, st));
+    g_ws.begin_uploads();
+    g_ws.h2d(da, a.data(), n * 8, st);
+    g_ws.h2d(db, b.data(), n * 8, st);
+    if (which == 1) g_ws.h2d(dc, c.data(), n * 8, st);
     const int block = 256;
     const int grid = (int)((n + block - 1) / block);
     if (which == 0)
@@ -291,11 +321,12 @@ py::array_t<double> truncnorm_logpdf(const arr_f64& x, const arr_f64& a,
     double* d = g_ws.ensure(6 * (size_t)n);
     double *dx = d, *da = d + n, *db = d + 2 * n, *dl = d + 3 * n,
            *ds = d + 4 * n, *dout = d + 5 * n;
-    HIP_CHECK(hipMemcpyAsync(dx, x.data(), n * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(da, a.data(), n * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(db, b.data(), n * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(dl, loc.data(), n * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(ds, scale.data(), n * 8, hipMemcpyHostToDevice, st));
+    g_ws.begin_uploads();
+    g_ws.h2d(dx, x.data(), n * 8, st);
+    g_ws.h2d(da, a.data(), n * 8, st);
+    g_ws.h2d(db, b.data(), n * 8, st);
+    g_ws.h2d(dl, loc.data(), n * 8, st);
+    g_ws.h2d(ds, scale.data(), n * 8, st);
     const int block = 256;
     const int grid = (int)((n + block - 1) / block);
     hipLaunchKernelGGL(k_logpdf, dim3(grid), dim3(block), 0, st, dx, da, db, dl,
@@ -346,15 +377,15 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     double* d_out = d_x + (size_t)S * D;
     int64_t* d_sorted = reinterpret_cast<int64_t*>(d_out + S);
 
+    g_ws.begin_uploads();
     if (N > 0) {
-        HIP_CHECK(hipMemcpyAsync(d_obs, obs.data(), n_obs * 8, hipMemcpyHostToDevice, st));
-        HIP_CHECK(hipMemcpyAsync(d_sorted, sorted_pos.data(), n_obs * 8,
-                                 hipMemcpyHostToDevice, st));
+        g_ws.h2d(d_obs, obs.data(), n_obs * 8, st);
+        g_ws.h2d(d_sorted, sorted_pos.data(), n_obs * 8, st);
     }
-    HIP_CHECK(hipMemcpyAsync(d_logw, logw.data(), K * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(d_alow, alow.data(), D * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(d_ahigh, ahigh.data(), D * 8, hipMemcpyHostToDevice, st));
-    HIP_CHECK(hipMemcpyAsync(d_x, x.data(), (size_t)S * D * 8, hipMemcpyHostToDevice, st));
+    g_ws.h2d(d_logw, logw.data(), K * 8, st);
+    g_ws.h2d(d_alow, alow.data(), D * 8, st);
+    g_ws.h2d(d_ahigh, ahigh.data(), D * 8, st);
+    g_ws.h2d(d_x, x.data(), (size_t)S * D * 8, st);
 
     {
         const int block = 256;
@@ -385,6 +416,237 @@ int device_count() {
     return n;
 }
 
+// ---------------------------------------------------------------------------
+// Device-resident TPE history (one per (study, search-space) on the Python side).
+//
+// The parameter table lives in HBM and grows append-only, mirroring the host
+// history (_history.py); a suggest uploads only the per-dim global sorted order
+// (i32), the subset position map, mixture weights and candidates — the big fp64
+// observation matrix never crosses PCIe again after its append.
+//
+// score() pipeline (all on one stream):
+//   1. k_compact: per dim, order-preserving compaction of the global sorted
+//      order down to the "above" subset (block-wide scan over tiles).
+//   2. k_parzen_fit_table: K1 fit reading mus from the resident table.
+//   3. k_mix_logpdf: K2 scoring.
+// ---------------------------------------------------------------------------
+
+__global__ void k_compact(const int32_t* __restrict__ sorted_rows,  // (Nv, D)
+                          const int32_t* __restrict__ pos,          // (n_rows,)
+                          int64_t Nv, int64_t D,
+                          int32_t* __restrict__ sub_rows,  // (D, Na)
+                          int32_t* __restrict__ sub_k) {   // (D, Na)
+    const int64_t d = blockIdx.x;
+    __shared__ int32_t flags[256];
+    __shared__ int32_t scan[256];
+    __shared__ int32_t base_offset;
+    if (threadIdx.x == 0) base_offset = 0;
+    __syncthreads();
+
+    for (int64_t tile = 0; tile < Nv; tile += blockDim.x) {
+        const int64_t r = tile + threadIdx.x;
+        int32_t row = -1, k = -1, flag = 0;
+        if (r < Nv) {
+            row = sorted_rows[r * D + d];
+            k = pos[row];
+            flag = (k >= 0) ? 1 : 0;
+        }
+        flags[threadIdx.x] = flag;
+        __syncthreads();
+        // Hillis-Steele inclusive scan over the tile.
+        scan[threadIdx.x] = flag;
+        __syncthreads();
+        for (int offset = 1; offset < blockDim.x; offset <<= 1) {
+            int32_t v = scan[threadIdx.x];
+            if (threadIdx.x >= (unsigned)offset) v += scan[threadIdx.x - offset];
+            __syncthreads();
+            scan[threadIdx.x] = v;
+            __syncthreads();
+        }
+        if (flag) {
+            const int32_t out_idx = base_offset + scan[threadIdx.x] - 1;
+            sub_rows[d * /*Na cap*/ Nv + out_idx] = row;
+            sub_k[d * Nv + out_idx] = k;
+        }
+        __syncthreads();
+        if (threadIdx.x == 0) base_offset += scan[blockDim.x - 1];
+        __syncthreads();
+    }
+}
+
+__global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_rows, D)
+                                   const int32_t* __restrict__ sub_rows,  // (D, stride)
+                                   const int32_t* __restrict__ sub_k,
+                                   int64_t stride,  // row stride of sub_* (= Nv)
+                                   const double* __restrict__ alow,
+                                   const double* __restrict__ ahigh, int64_t Na,
+                                   int64_t D, int consider_endpoints,
+                                   int magic_clip, double* __restrict__ c1,
+                                   double* __restrict__ c2,
+                                   double* __restrict__ c3) {
+    const int64_t d = blockIdx.y;
+    const double low = alow[d];
+    const double high = ahigh[d];
+    const double range = high - low;
+    double minsigma = 1e-12;
+    if (magic_clip) {
+        minsigma = range / fmin(100.0, (double)(Na + 2));
+    }
+    const int32_t* rows_d = sub_rows + d * stride;
+    const int32_t* ks_d = sub_k + d * stride;
+
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r <= Na;
+         r += (int64_t)gridDim.x * blockDim.x) {
+        double mu, sigma;
+        int64_t k;
+        if (r == Na) {
+            mu = 0.5 * (low + high);
+            sigma = range;
+            k = Na;
+        } else {
+            mu = params[(int64_t)rows_d[r] * D + d];
+            const double v_prev =
+                (r == 0) ? low : params[(int64_t)rows_d[r - 1] * D + d];
+            const double v_next =
+                (r == Na - 1) ? high : params[(int64_t)rows_d[r + 1] * D + d];
+            sigma = fmax(mu - v_prev, v_next - mu);
+            if (!consider_endpoints && Na >= 2) {
+                if (r == 0) {
+                    sigma = params[(int64_t)rows_d[1] * D + d] - mu;
+                } else if (r == Na - 1) {
+                    sigma = mu - params[(int64_t)rows_d[Na - 2] * D + d];
+                }
+            }
+            sigma = fmin(fmax(sigma, minsigma), range);
+            k = ks_d[r];
+        }
+        const double inv_var = 1.0 / (sigma * sigma);
+        const double mass =
+            tn::log_gauss_mass((low - mu) / sigma, (high - mu) / sigma);
+        c1[k * D + d] = -0.5 * inv_var;
+        c2[k * D + d] = mu * inv_var;
+        c3[k * D + d] = -0.5 * mu * mu * inv_var - log(sigma) -
+                        0.9189385332046727418 - mass;
+    }
+}
+
+using arr_i32 = py::array_t<int32_t, py::array::c_style | py::array::forcecast>;
+
+class TpeDeviceHistory {
+  public:
+    explicit TpeDeviceHistory(int64_t D) : D_(D) {}
+
+    ~TpeDeviceHistory() {
+        if (params_) (void)hipFree(params_);
+    }
+
+    int64_t n_rows() const { return n_; }
+
+    void append(const arr_f64& block) {
+        if (block.ndim() != 2 || block.shape(1) != D_)
+            throw std::runtime_error("append: block must be (n_new, D)");
+        const int64_t n_new = block.shape(0);
+        if (n_new == 0) return;
+        hipStream_t st = g_ws.get_stream();
+        if (n_ + n_new > capacity_) {
+            int64_t new_cap = std::max<int64_t>(1024, (n_ + n_new) * 2);
+            double* new_params = nullptr;
+            HIP_CHECK(hipMalloc(&new_params, (size_t)new_cap * D_ * sizeof(double)));
+            if (params_) {
+                HIP_CHECK(hipMemcpyAsync(new_params, params_,
+                                         (size_t)n_ * D_ * sizeof(double),
+                                         hipMemcpyDeviceToDevice, st));
+                HIP_CHECK(hipStreamSynchronize(st));
+                HIP_CHECK(hipFree(params_));
+            }
+            params_ = new_params;
+            capacity_ = new_cap;
+        }
+        g_ws.begin_uploads();
+        g_ws.h2d(params_ + (size_t)n_ * D_, block.data(),
+                 (size_t)n_new * D_ * sizeof(double), st);
+        HIP_CHECK(hipStreamSynchronize(st));
+        n_ += n_new;
+    }
+
+    py::array_t<double> score(const arr_i32& sorted_rows,  // (Nv, D)
+                              const arr_i32& pos,          // (n_rows,)
+                              int64_t n_above,
+                              const arr_f64& logw,  // (n_above + 1,)
+                              const arr_f64& alow, const arr_f64& ahigh,
+                              const arr_f64& x,  // (S, D) KDE domain
+                              bool consider_endpoints, bool magic_clip) {
+        const int64_t Nv = sorted_rows.ndim() == 2 ? sorted_rows.shape(0) : 0;
+        const int64_t S = x.shape(0);
+        const int64_t Na = n_above;
+        const int64_t K = Na + 1;
+        if ((int64_t)pos.size() != n_ || x.shape(1) != D_ ||
+            (Nv > 0 && sorted_rows.shape(1) != D_) || (int64_t)logw.size() != K)
+            throw std::runtime_error("score: shape mismatch");
+
+        hipStream_t st = g_ws.get_stream();
+        const size_t n_c = (size_t)K * D_;
+        // f64 slots: c1|c2|c3|logw|alow|ahigh|x|out  + i32 slots packed after.
+        size_t f64_total = 3 * n_c + K + 2 * D_ + (size_t)S * D_ + S;
+        size_t i32_doubles =
+            ((size_t)Nv * D_ /*sorted*/ + n_ /*pos*/ + 2 * (size_t)Nv * D_ /*sub*/) / 2 + 8;
+        double* base = g_ws.ensure(f64_total + i32_doubles + 16);
+        double* d_c1 = base;
+        double* d_c2 = d_c1 + n_c;
+        double* d_c3 = d_c2 + n_c;
+        double* d_logw = d_c3 + n_c;
+        double* d_alow = d_logw + K;
+        double* d_ahigh = d_alow + D_;
+        double* d_x = d_ahigh + D_;
+        double* d_out = d_x + (size_t)S * D_;
+        int32_t* d_sorted = reinterpret_cast<int32_t*>(d_out + S);
+        int32_t* d_pos = d_sorted + (size_t)Nv * D_;
+        int32_t* d_sub_rows = d_pos + n_;
+        int32_t* d_sub_k = d_sub_rows + (size_t)Nv * D_;
+
+        g_ws.begin_uploads();
+        if (Nv > 0)
+            g_ws.h2d(d_sorted, sorted_rows.data(), (size_t)Nv * D_ * 4, st);
+        if (n_ > 0) g_ws.h2d(d_pos, pos.data(), (size_t)n_ * 4, st);
+        g_ws.h2d(d_logw, logw.data(), K * 8, st);
+        g_ws.h2d(d_alow, alow.data(), D_ * 8, st);
+        g_ws.h2d(d_ahigh, ahigh.data(), D_ * 8, st);
+        g_ws.h2d(d_x, x.data(), (size_t)S * D_ * 8, st);
+
+        if (Na > 0) {
+            hipLaunchKernelGGL(k_compact, dim3((unsigned)D_), dim3(256), 0, st,
+                               d_sorted, d_pos, Nv, D_, d_sub_rows, d_sub_k);
+        }
+        {
+            const int block = 256;
+            const int gx = (int)((K + block - 1) / block);
+            hipLaunchKernelGGL(k_parzen_fit_table, dim3(gx, (unsigned)D_),
+                               dim3(block), 0, st, params_, d_sub_rows, d_sub_k,
+                               Nv, d_alow, d_ahigh, Na, D_,
+                               consider_endpoints ? 1 : 0, magic_clip ? 1 : 0,
+                               d_c1, d_c2, d_c3);
+        }
+        {
+            const int block = 256;
+            const size_t shmem = (2 * (size_t)D_ + 2 * block) * sizeof(double);
+            hipLaunchKernelGGL(k_mix_logpdf, dim3((unsigned)S), dim3(block), shmem,
+                               st, d_x, d_c1, d_c2, d_c3, d_logw, K, D_, d_out);
+        }
+        py::array_t<double> out(S);
+        HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
+                                 hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        HIP_CHECK(hipGetLastError());
+        return out;
+    }
+
+  private:
+    int64_t D_;
+    int64_t n_ = 0;
+    int64_t capacity_ = 0;
+    double* params_ = nullptr;
+};
+
 PYBIND11_MODULE(_hipcore, m) {
     m.doc() = "optuna_amd MI355X (gfx950) HIP kernels: TPE parzen fit + mixture "
               "log-pdf, truncnorm device library";
@@ -396,4 +658,12 @@ PYBIND11_MODULE(_hipcore, m) {
     m.def("kde_logpdf", &kde_logpdf, py::arg("obs"), py::arg("sorted_pos"),
           py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("x"),
           py::arg("consider_endpoints") = false, py::arg("magic_clip") = true);
+    py::class_<TpeDeviceHistory>(m, "TpeDeviceHistory")
+        .def(py::init<int64_t>(), py::arg("dims"))
+        .def_property_readonly("n_rows", &TpeDeviceHistory::n_rows)
+        .def("append", &TpeDeviceHistory::append, py::arg("block"))
+        .def("score", &TpeDeviceHistory::score, py::arg("sorted_rows"),
+             py::arg("pos"), py::arg("n_above"), py::arg("logw"), py::arg("alow"),
+             py::arg("ahigh"), py::arg("x"), py::arg("consider_endpoints") = false,
+             py::arg("magic_clip") = true);
 }

@@ -1,11 +1,18 @@
 """Device dispatch for the TPE hot path (K1 fit + K2 mixture log-pdf on gfx950).
 
-Used by ``TPESampler._sample`` for the "above" KDE when the history is large:
-the (K≈N)×D Parzen fit and the S×K×D log-pdf run in ``optuna_amd._hip._hipcore``
-kernels; the host never materializes mus/sigmas for the big estimator at all.
+Two device modes:
+
+* **Resident history** (the fast path): each (study, search-space) keeps a
+  ``TpeDeviceHistory`` whose parameter table lives in HBM and grows append-only
+  in lockstep with the host mirror (``_history.py``). A suggest uploads only the
+  per-dim sorted order (i32), a subset position map, the mixture weights and the
+  24 candidates — the fp64 observation matrix never crosses PCIe again. Subset
+  compaction, the K1 fit and the K2 scoring all run on device.
+* **Stateless scoring** (`kde_logpdf`): used by golden tests and the
+  constant-liar path (whose observation set includes rows not in the table).
 
 Eligibility: all dimensions continuous (optionally log-scaled). Discrete and
-categorical dimensions currently keep the (already vectorized) host path.
+categorical dimensions keep the (already vectorized) host path.
 """
 from __future__ import annotations
 
@@ -22,6 +29,9 @@ from optuna_amd.distributions import (
 )
 
 
+if TYPE_CHECKING:
+    from optuna_amd.samplers._tpe._history import _SpaceCache
+
 # Below this kernel count the host fit is cheaper than a kernel launch round trip.
 DEVICE_MIN_KERNELS = 512
 
@@ -30,10 +40,8 @@ def space_is_device_eligible(space: dict[str, BaseDistribution]) -> bool:
     if not space:
         return False
     for dist in space.values():
-        if not isinstance(dist, (FloatDistribution, IntDistribution)):
+        if not isinstance(dist, FloatDistribution):
             return False
-        if isinstance(dist, IntDistribution):
-            return False  # int domains are discrete cells
         if dist.step is not None:
             return False
     return True
@@ -41,6 +49,95 @@ def space_is_device_eligible(space: dict[str, BaseDistribution]) -> bool:
 
 def device_ready(n_kernels: int) -> bool:
     return n_kernels >= DEVICE_MIN_KERNELS and _hip.is_available()
+
+
+def _space_domains(space: dict[str, BaseDistribution]) -> tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """(is_log, alow, ahigh) per dim, KDE domain."""
+    is_log = np.array([bool(d.log) for d in space.values()])
+    alow = np.array(
+        [math.log(d.low) if d.log else d.low for d in space.values()], dtype=np.float64
+    )
+    ahigh = np.array(
+        [math.log(d.high) if d.log else d.high for d in space.values()], dtype=np.float64
+    )
+    return is_log, alow, ahigh
+
+
+class _SpaceDeviceMirror:
+    """HBM-resident copy of one space's parameter table (KDE domain)."""
+
+    def __init__(self, space: dict[str, BaseDistribution]) -> None:
+        core = _hip.get()
+        assert core is not None
+        self._is_log, self._alow, self._ahigh = _space_domains(space)
+        self._hist = core.TpeDeviceHistory(len(space))
+        self._n_appended = 0
+
+    def sync(self, cache: "_SpaceCache") -> None:
+        n_total = len(cache.valid)
+        if n_total > self._n_appended:
+            block = np.array(cache.params[self._n_appended :], dtype=np.float64)
+            if self._is_log.any():
+                with np.errstate(invalid="ignore", divide="ignore"):
+                    block[:, self._is_log] = np.log(block[:, self._is_log])
+            self._hist.append(np.ascontiguousarray(block))
+            self._n_appended = n_total
+
+    def score(
+        self,
+        cache: "_SpaceCache",
+        sel: np.ndarray,
+        weights: np.ndarray,
+        samples: dict[str, np.ndarray],
+        consider_endpoints: bool,
+        consider_magic_clip: bool,
+    ) -> np.ndarray:
+        self.sync(cache)
+        n_total = len(cache.valid)
+        pos = np.full(n_total, -1, dtype=np.int32)
+        pos[sel] = np.arange(len(sel), dtype=np.int32)
+        sorted_rows = (
+            np.stack(cache.sorted_rows, axis=1).astype(np.int32)
+            if cache.sorted_rows and len(cache.sorted_rows[0])
+            else np.empty((0, len(cache.names)), dtype=np.int32)
+        )
+        x = np.column_stack(
+            [np.asarray(samples[n], dtype=np.float64) for n in cache.names]
+        )
+        if self._is_log.any():
+            x[:, self._is_log] = np.log(x[:, self._is_log])
+        with np.errstate(divide="ignore"):
+            logw = np.log(weights)
+        return self._hist.score(
+            sorted_rows,
+            pos,
+            int(len(sel)),
+            logw,
+            self._alow,
+            self._ahigh,
+            np.ascontiguousarray(x),
+            consider_endpoints,
+            consider_magic_clip,
+        )
+
+
+def score_above_resident(
+    cache: "_SpaceCache",
+    sel: np.ndarray,
+    weights: np.ndarray,
+    samples: dict[str, np.ndarray],
+    consider_endpoints: bool,
+    consider_magic_clip: bool,
+) -> np.ndarray:
+    """log g(x) for candidates via the device-resident table (creates the mirror
+    on first use; attached to the host space cache so lifetimes match)."""
+    mirror = getattr(cache, "_device_mirror", None)
+    if mirror is None:
+        mirror = _SpaceDeviceMirror(cache.space)
+        cache._device_mirror = mirror  # type: ignore[attr-defined]
+    return mirror.score(
+        cache, sel, weights, samples, consider_endpoints, consider_magic_clip
+    )
 
 
 def kde_logpdf(
@@ -52,7 +149,7 @@ def kde_logpdf(
     consider_endpoints: bool,
     consider_magic_clip: bool,
 ) -> np.ndarray:
-    """log mixture pdf of `samples` under the KDE fit to `observations` (device)."""
+    """Stateless device scoring (observation matrix uploaded per call)."""
     core = _hip.get()
     assert core is not None
     names = list(space.keys())

@@ -255,6 +255,22 @@ class _TpeHistory:
 
     # ---- observations ---------------------------------------------------------------
 
+    def space_cache(self, space: dict[str, BaseDistribution]) -> _SpaceCache:
+        key = tuple((n, d) for n, d in space.items())
+        cache = self._spaces.get(key)
+        if cache is None:
+            cache = _SpaceCache(space)
+            cache.append(self._trials)
+            self._spaces[key] = cache
+        return cache
+
+    def valid_rows(
+        self, space: dict[str, BaseDistribution], row_indices: np.ndarray
+    ) -> np.ndarray:
+        """Subset of row_indices whose trials define every parameter of the space."""
+        cache = self.space_cache(space)
+        return row_indices[cache.valid[row_indices]]
+
     def observations(
         self, space: dict[str, BaseDistribution], row_indices: np.ndarray
     ) -> tuple[dict[str, np.ndarray], dict[str, np.ndarray]]:
@@ -264,12 +280,7 @@ class _TpeHistory:
         order, plus per-param argsort arrays derived from the incrementally-sorted
         index (no per-call sort).
         """
-        key = tuple((n, d) for n, d in space.items())
-        cache = self._spaces.get(key)
-        if cache is None:
-            cache = _SpaceCache(space)
-            cache.append(self._trials)
-            self._spaces[key] = cache
+        cache = self.space_cache(space)
         sel = row_indices[cache.valid[row_indices]]
         mat = cache.params[sel]
         obs = {name: mat[:, c] for c, name in enumerate(cache.names)}

@@ -296,7 +296,21 @@ class TPESampler(BaseSampler):
         below_rows, above_rows = history.split(study, self._gamma(n))
 
         obs_below, orders_below = history.observations(search_space, below_rows)
-        obs_above, orders_above = history.observations(search_space, above_rows)
+
+        from optuna_amd.samplers._tpe import _device
+
+        above_sel = history.valid_rows(search_space, above_rows)
+        use_device = (
+            not self._constant_liar
+            and self._parzen_estimator_cls is _ParzenEstimator
+            and _device.space_is_device_eligible(search_space)
+            and _device.device_ready(len(above_sel) + 1)
+        )
+
+        obs_above: dict[str, np.ndarray] = {}
+        orders_above: dict[str, np.ndarray] | None = None
+        if not use_device:
+            obs_above, orders_above = history.observations(search_space, above_rows)
 
         if self._constant_liar:
             # Treat other workers' RUNNING trials as part of the "above" set, using
@@ -329,23 +343,15 @@ class TPESampler(BaseSampler):
         )
         samples_below = mpe_below.sample(self._rng.rng, self._n_ei_candidates)
 
-        # Device path (K1+K2): fit + score the big "above" KDE on the GPU; the
-        # small "below" estimator stays on host (it also drives candidate
-        # sampling). Falls back to host for discrete/categorical spaces.
-        from optuna_amd.samplers._tpe import _device
-
-        n_above = len(next(iter(obs_above.values()))) if obs_above else 0
-        use_device = (
-            self._parzen_estimator_cls is _ParzenEstimator
-            and _device.space_is_device_eligible(search_space)
-            and _device.device_ready(n_above + 1)
-        )
+        # Device path (K1+K2): the big "above" KDE is fit and scored against the
+        # HBM-resident parameter table; the small "below" estimator stays on host
+        # (it also drives candidate sampling). Host path covers discrete /
+        # categorical spaces and the constant-liar case.
         if use_device:
-            weights_above = self._above_weights(n_above)
-            log_g = _device.kde_logpdf(
-                search_space,
-                obs_above,
-                orders_above,
+            weights_above = self._above_weights(len(above_sel))
+            log_g = _device.score_above_resident(
+                history.space_cache(search_space),
+                above_sel,
                 weights_above,
                 samples_below,
                 self._parzen_estimator_parameters.consider_endpoints,

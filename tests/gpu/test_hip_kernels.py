@@ -102,16 +102,16 @@ def test_kde_logpdf_matches_host(core, n_obs, d, log_dims) -> None:
 
 def test_tpe_sampler_uses_device_path(core, monkeypatch) -> None:
     """End-to-end: at ≥512-kernel history the sampler must call the HIP kernels."""
-    from optuna_amd.samplers._tpe import _device
+    from optuna_amd.samplers._tpe import _device as device_mod
 
     calls = {"n": 0}
-    orig = _device.kde_logpdf
+    orig = device_mod.score_above_resident
 
     def spy(*args, **kwargs):
         calls["n"] += 1
         return orig(*args, **kwargs)
 
-    monkeypatch.setattr(_device, "kde_logpdf", spy)
+    monkeypatch.setattr(device_mod, "score_above_resident", spy)
 
     optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
     study = optuna_amd.create_study(
@@ -179,3 +179,67 @@ def test_device_and_host_sampling_agree_statistically(core) -> None:
     host = run(True)
     device = run(False)
     np.testing.assert_allclose(host, device, rtol=1e-7)
+
+
+def test_resident_history_matches_host(core) -> None:
+    """TpeDeviceHistory.score (compaction + K1 + K2 on the resident table) must
+    reproduce the host estimator's log-pdf over random subsets and appends."""
+    from optuna_amd.samplers._tpe import _device
+    from optuna_amd.samplers._tpe._history import _SpaceCache
+    from optuna_amd.testing.trials import _create_frozen_trial
+
+    rng = np.random.RandomState(5)
+    d = 12
+    space = {}
+    for i in range(d):
+        space[f"x{i}"] = (
+            FloatDistribution(1e-2, 1e2, log=True) if i % 3 == 0 else FloatDistribution(-5, 5)
+        )
+
+    def make_trials(n, offset):
+        out = []
+        for j in range(n):
+            params = {}
+            for i, (name, dist) in enumerate(space.items()):
+                if dist.log:
+                    params[name] = float(np.exp(rng.uniform(np.log(1e-2), np.log(1e2))))
+                else:
+                    params[name] = float(rng.uniform(-5, 5))
+            out.append(
+                _create_frozen_trial(
+                    number=offset + j, values=(float(rng.rand()),), params=params,
+                    distributions=dict(space),
+                )
+            )
+        return out
+
+    cache = _SpaceCache(space)
+    cache.append(make_trials(900, 0))
+
+    for round_i in range(3):
+        n_total = len(cache.valid)
+        sel = np.sort(rng.choice(n_total, size=n_total - 30, replace=False))
+        weights_raw = default_weights(len(sel))
+        weights = np.append(weights_raw, [1.0])
+        weights = weights / weights.sum()
+
+        samples = {
+            name: (
+                np.exp(rng.uniform(np.log(1e-2), np.log(1e2), 24))
+                if dist.log
+                else rng.uniform(-5, 5, 24)
+            )
+            for name, dist in space.items()
+        }
+
+        ours = _device.score_above_resident(cache, sel, weights, samples, False, True)
+
+        obs = {name: cache.params[sel][:, c] for c, name in enumerate(cache.names)}
+        params_obj = _ParzenEstimatorParameters(
+            True, 1.0, True, False, lambda n: weights_raw[:n], True
+        )
+        mpe = _ParzenEstimator(obs, space, params_obj)
+        ref = mpe.log_pdf(samples)
+        np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
+
+        cache.append(make_trials(50, 900 + round_i * 50))
