@@ -35,7 +35,17 @@ def main() -> None:
     parser.add_argument("--warmup", type=int, default=8)
     parser.add_argument("--history", type=int, default=10000)
     parser.add_argument("--dims", type=int, default=20)
+    parser.add_argument(
+        "--suite",
+        choices=("tpe", "random", "gp", "cmaes", "nsgaii"),
+        default="tpe",
+        help="BASELINE.json config to run (default: the headline TPE config)",
+    )
     args = parser.parse_args()
+
+    if args.suite != "tpe":
+        _run_alt_suite(args)
+        return
 
     warnings.simplefilter("ignore")
     import optuna_amd
@@ -159,6 +169,111 @@ def main() -> None:
 
     if dist is not None:
         dist.destroy_process_group()
+
+
+def _run_alt_suite(args: argparse.Namespace) -> None:
+    """The non-headline BASELINE configs (single process; informational)."""
+    import optuna_amd
+
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(0)
+
+    if args.suite == "random":
+        # Config 1: RandomSampler on 2-dim Rosenbrock, InMemoryStorage (plumbing).
+        study = optuna_amd.create_study(sampler=optuna_amd.samplers.RandomSampler(seed=0))
+
+        def one_step() -> None:
+            t = study.ask()
+            x = t.suggest_float("x", -5, 5)
+            y = t.suggest_float("y", -5, 5)
+            study.tell(t, 100 * (y - x**2) ** 2 + (1 - x) ** 2)
+
+        config = {"model": "RandomSampler 2-dim Rosenbrock", "parallelism": "single"}
+    elif args.suite == "gp":
+        # Config 3: GPSampler, 20-dim, batched log-EI (history size via --history).
+        n_obs = min(args.history, 1000)
+        sampler = optuna_amd.samplers.GPSampler(seed=0, n_startup_trials=10)
+        study = optuna_amd.create_study(sampler=sampler)
+        names = [f"x{i}" for i in range(args.dims)]
+        dists = {
+            n: optuna_amd.distributions.FloatDistribution(-5.0, 5.0) for n in names
+        }
+        _populate(study, names, dists, n_obs)
+
+        def one_step() -> None:
+            t = study.ask()
+            x = np.array([t.suggest_float(n, -5, 5) for n in names])
+            study.tell(t, float(np.sum((x - 1) ** 2)))
+
+        config = {
+            "model": "GPSampler log-EI",
+            "observations": n_obs,
+            "dims": args.dims,
+            "parallelism": "single",
+        }
+    elif args.suite == "cmaes":
+        # Config 5 (single-process variant): CMA-ES, 100-dim.
+        dims = max(args.dims, 100)
+        sampler = optuna_amd.samplers.CmaEsSampler(seed=0, n_startup_trials=1)
+        study = optuna_amd.create_study(sampler=sampler)
+
+        def one_step() -> None:
+            t = study.ask()
+            x = np.array([t.suggest_float(f"x{i}", -5, 5) for i in range(dims)])
+            study.tell(t, float(np.sum(x**2)))
+
+        config = {"model": "CmaEsSampler", "dims": dims, "parallelism": "single"}
+    else:  # nsgaii
+        # Config 4 (single-process variant): NSGA-II, 3-objective DTLZ2-like.
+        sampler = optuna_amd.samplers.NSGAIISampler(seed=0, population_size=50)
+        study = optuna_amd.create_study(
+            directions=["minimize"] * 3, sampler=sampler
+        )
+
+        def one_step() -> None:
+            t = study.ask()
+            x = np.array([t.suggest_float(f"x{i}", 0, 1) for i in range(7)])
+            g = float(np.sum((x[2:] - 0.5) ** 2))
+            import math as _math
+
+            f1 = (1 + g) * _math.cos(x[0] * _math.pi / 2) * _math.cos(x[1] * _math.pi / 2)
+            f2 = (1 + g) * _math.cos(x[0] * _math.pi / 2) * _math.sin(x[1] * _math.pi / 2)
+            f3 = (1 + g) * _math.sin(x[0] * _math.pi / 2)
+            study.tell(t, (f1, f2, f3))
+
+        config = {
+            "model": "NSGAIISampler DTLZ2 3-objective",
+            "population_size": 50,
+            "parallelism": "single",
+        }
+
+    for _ in range(args.warmup):
+        one_step()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    elapsed = time.perf_counter() - t0
+
+    print(
+        json.dumps(
+            {
+                "metric": f"{args.suite} sampler suggest()/sec",
+                "value": args.steps / elapsed,
+                "unit": "suggest/s",
+                "n_gpus": 1,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": elapsed / args.steps * 1000.0,
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "fp64",
+                "data": "synthetic",
+                "config": config,
+            }
+        ),
+        flush=True,
+    )
 
 
 def _populate(study, names, dists_def, n_history: int) -> None:
