@@ -280,13 +280,21 @@ class TPESampler(BaseSampler):
     ) -> dict[str, Any]:
         from optuna_amd.samplers._tpe._history import _TpeHistory
 
-        finished = study._get_trials(
-            deepcopy=False, states=(TrialState.COMPLETE, TrialState.PRUNED), use_cache=True
-        )
         history = self._histories.get(study._study_id)
         if history is None:
             history = self._histories[study._study_id] = _TpeHistory()
-        history.update(finished, len(study.directions))
+        # Finished trials only accumulate; an O(1) count check skips the full
+        # storage read on suggests that arrive between tells.
+        n_finished = study._storage.get_n_trials(
+            study._study_id, (TrialState.COMPLETE, TrialState.PRUNED)
+        )
+        if n_finished != len(history):
+            finished = study._get_trials(
+                deepcopy=False,
+                states=(TrialState.COMPLETE, TrialState.PRUNED),
+                use_cache=True,
+            )
+            history.update(finished, len(study.directions))
 
         if self._gamma is None:
             self._gamma = (

@@ -52,6 +52,7 @@ class IntersectionSearchSpace:
         self._search_space: dict[str, BaseDistribution] | None = None
         self._study_id: int | None = None
         self._include_pruned = include_pruned
+        self._n_finished_seen: int = -1
 
     def calculate(self, study: "Study") -> dict[str, BaseDistribution]:
         if self._study_id is None:
@@ -59,6 +60,18 @@ class IntersectionSearchSpace:
         else:
             if self._study_id != study._study_id:
                 raise ValueError("`IntersectionSearchSpace` cannot handle multiple studies.")
+
+        # The intersection depends on finished trials only; an O(1) count check
+        # skips the storage read while nothing new finished.
+        finished_states = (
+            (TrialState.COMPLETE, TrialState.PRUNED)
+            if self._include_pruned
+            else (TrialState.COMPLETE,)
+        )
+        n_finished = study._storage.get_n_trials(study._study_id, finished_states)
+        if n_finished == self._n_finished_seen and self._search_space is not None:
+            return dict(sorted(self._search_space.items(), key=lambda x: x[0]))
+        self._n_finished_seen = n_finished
 
         states_of_interest = [TrialState.COMPLETE, TrialState.WAITING, TrialState.RUNNING]
         if self._include_pruned:
