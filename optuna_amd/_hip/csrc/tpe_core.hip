@@ -196,6 +196,150 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
     }
 }
 
+// Two-phase variant for large K. One block per (sample, K-chunk) pair: with
+// S≈24 EI candidates a one-block-per-sample launch leaves >90% of the 256 CUs
+// idle; S × n_chunks blocks fill the chip. blockIdx.x walks samples fastest so
+// the XCD round-robin distributes every chunk's coefficient rows into all 8
+// L2s once and the co-resident sample-blocks reuse them.
+__global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
+                                     const double* __restrict__ c1,
+                                     const double* __restrict__ c2,
+                                     const double* __restrict__ c3,
+                                     const double* __restrict__ logw, int64_t K,
+                                     int64_t D, int64_t chunk,
+                                     double* __restrict__ part_m,   // (n_chunks, S)
+                                     double* __restrict__ part_s) { // (n_chunks, S)
+    extern __shared__ double lds[];
+    double* xs = lds;
+    double* x2 = lds + D;
+    double* red_m = lds + 2 * D;
+    double* red_s = red_m + blockDim.x;
+
+    const int64_t S = gridDim.x;
+    const int64_t s = blockIdx.x;
+    const int64_t c = blockIdx.y;
+    const int64_t k_lo = c * chunk;
+    const int64_t k_hi = k_lo + chunk < K ? k_lo + chunk : K;
+    for (int64_t d = threadIdx.x; d < D; d += blockDim.x) {
+        const double v = x[s * D + d];
+        xs[d] = v;
+        x2[d] = v * v;
+    }
+    __syncthreads();
+
+    double m = -INFINITY, acc = 0.0;
+    for (int64_t k = k_lo + threadIdx.x; k < k_hi; k += blockDim.x) {
+        double t = logw[k];
+        const double* c1k = c1 + k * D;
+        const double* c2k = c2 + k * D;
+        const double* c3k = c3 + k * D;
+        for (int64_t d = 0; d < D; ++d) {
+            t += x2[d] * c1k[d] + xs[d] * c2k[d] + c3k[d];
+        }
+        if (t > m) {
+            acc = acc * exp(m - t) + 1.0;
+            m = t;
+        } else {
+            acc += exp(t - m);
+        }
+    }
+    red_m[threadIdx.x] = m;
+    red_s[threadIdx.x] = acc;
+    __syncthreads();
+    for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if (threadIdx.x < stride) {
+            double m2 = red_m[threadIdx.x + stride];
+            double s2 = red_s[threadIdx.x + stride];
+            double m1 = red_m[threadIdx.x];
+            double s1 = red_s[threadIdx.x];
+            if (m2 > m1) {
+                s1 = s1 * exp(m1 - m2) + s2;
+                m1 = m2;
+            } else if (m1 != -INFINITY) {
+                s1 = s1 + s2 * exp(m2 - m1);
+            }
+            red_m[threadIdx.x] = m1;
+            red_s[threadIdx.x] = s1;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        part_m[c * S + s] = red_m[0];
+        part_s[c * S + s] = red_s[0];
+    }
+}
+
+__global__ void k_mix_logpdf_merge(const double* __restrict__ part_m,
+                                   const double* __restrict__ part_s,
+                                   int64_t n_chunks, int64_t S,
+                                   double* __restrict__ out) {
+    __shared__ double red_m[64];
+    __shared__ double red_s[64];
+    const int64_t s = blockIdx.x;
+    double m = -INFINITY, acc = 0.0;
+    for (int64_t c = threadIdx.x; c < n_chunks; c += blockDim.x) {
+        const double m2 = part_m[c * S + s];
+        const double s2 = part_s[c * S + s];
+        if (m2 > m) {
+            acc = acc * exp(m - m2) + s2;
+            m = m2;
+        } else if (m != -INFINITY) {
+            acc += s2 * exp(m2 - m);
+        }
+    }
+    red_m[threadIdx.x] = m;
+    red_s[threadIdx.x] = acc;
+    __syncthreads();
+    for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if (threadIdx.x < stride) {
+            double m2 = red_m[threadIdx.x + stride];
+            double s2 = red_s[threadIdx.x + stride];
+            double m1 = red_m[threadIdx.x];
+            double s1 = red_s[threadIdx.x];
+            if (m2 > m1) {
+                s1 = s1 * exp(m1 - m2) + s2;
+                m1 = m2;
+            } else if (m1 != -INFINITY) {
+                s1 = s1 + s2 * exp(m2 - m1);
+            }
+            red_m[threadIdx.x] = m1;
+            red_s[threadIdx.x] = s1;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out[s] = (red_m[0] == -INFINITY) ? -INFINITY : red_m[0] + log(red_s[0]);
+    }
+}
+
+// Chunk rows so S × n_chunks blocks ≈ 2 per CU; below one chunk the original
+// single-phase kernel is cheaper (no scratch round-trip).
+static constexpr int64_t MIX_CHUNK = 512;
+
+static inline int64_t mix_n_chunks(int64_t K) { return (K + MIX_CHUNK - 1) / MIX_CHUNK; }
+
+// d_scratch must hold 2 * mix_n_chunks(K) * S doubles (may alias nothing else).
+static void launch_mix_logpdf(hipStream_t st, const double* d_x, const double* d_c1,
+                              const double* d_c2, const double* d_c3,
+                              const double* d_logw, int64_t K, int64_t D, int64_t S,
+                              double* d_out, double* d_scratch) {
+    const int block = 256;
+    const size_t shmem = (2 * (size_t)D + 2 * block) * sizeof(double);
+    const int64_t n_chunks = mix_n_chunks(K);
+    if (n_chunks <= 1) {
+        hipLaunchKernelGGL(k_mix_logpdf, dim3((unsigned)S), dim3(block), shmem, st,
+                           d_x, d_c1, d_c2, d_c3, d_logw, K, D, d_out);
+        return;
+    }
+    double* d_part_m = d_scratch;
+    double* d_part_s = d_scratch + (size_t)n_chunks * S;
+    hipLaunchKernelGGL(k_mix_logpdf_partial, dim3((unsigned)S, (unsigned)n_chunks),
+                       dim3(block), shmem, st, d_x, d_c1, d_c2, d_c3, d_logw, K, D,
+                       MIX_CHUNK, d_part_m, d_part_s);
+    hipLaunchKernelGGL(k_mix_logpdf_merge, dim3((unsigned)S), dim3(64), 0, st,
+                       d_part_m, d_part_s, n_chunks, S, d_out);
+}
+
 // ---------------------------------------------------------------------------
 // Host-side workspace: grown lazily, reused across calls (no per-suggest
 // hipMalloc). One workspace per process; calls are serialized by the GIL.
@@ -363,8 +507,10 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
 
     const size_t n_obs = (size_t)N * D;
     const size_t n_c = (size_t)K * D;
-    // layout: obs | c1 | c2 | c3 | logw | alow | ahigh | x | out | sorted(i64 as f64 slots)
-    size_t total = n_obs + 3 * n_c + K + 2 * D + (size_t)S * D + S + n_obs + 16;
+    // layout: obs | c1 | c2 | c3 | logw | alow | ahigh | x | out | lse scratch
+    //         | sorted(i64 as f64 slots)
+    const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
+    size_t total = n_obs + 3 * n_c + K + 2 * D + (size_t)S * D + S + n_scratch + n_obs + 16;
     double* base = g_ws.ensure(total);
     double* d_obs = base;
     double* d_c1 = d_obs + n_obs;
@@ -375,7 +521,8 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     double* d_ahigh = d_alow + D;
     double* d_x = d_ahigh + D;
     double* d_out = d_x + (size_t)S * D;
-    int64_t* d_sorted = reinterpret_cast<int64_t*>(d_out + S);
+    double* d_scratch = d_out + S;
+    int64_t* d_sorted = reinterpret_cast<int64_t*>(d_scratch + n_scratch);
 
     g_ws.begin_uploads();
     if (N > 0) {
@@ -395,12 +542,7 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
                            consider_endpoints ? 1 : 0, magic_clip ? 1 : 0, d_c1,
                            d_c2, d_c3);
     }
-    {
-        const int block = 256;
-        const size_t shmem = (2 * (size_t)D + 2 * block) * sizeof(double);
-        hipLaunchKernelGGL(k_mix_logpdf, dim3((unsigned)S), dim3(block), shmem, st,
-                           d_x, d_c1, d_c2, d_c3, d_logw, K, D, d_out);
-    }
+    launch_mix_logpdf(st, d_x, d_c1, d_c2, d_c3, d_logw, K, D, S, d_out, d_scratch);
     HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
                              hipMemcpyDeviceToHost, st));
     HIP_CHECK(hipStreamSynchronize(st));
@@ -586,8 +728,9 @@ class TpeDeviceHistory {
 
         hipStream_t st = g_ws.get_stream();
         const size_t n_c = (size_t)K * D_;
-        // f64 slots: c1|c2|c3|logw|alow|ahigh|x|out  + i32 slots packed after.
-        size_t f64_total = 3 * n_c + K + 2 * D_ + (size_t)S * D_ + S;
+        // f64 slots: c1|c2|c3|logw|alow|ahigh|x|out|lse scratch + i32 after.
+        const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
+        size_t f64_total = 3 * n_c + K + 2 * D_ + (size_t)S * D_ + S + n_scratch;
         size_t i32_doubles =
             ((size_t)Nv * D_ /*sorted*/ + n_ /*pos*/ + 2 * (size_t)Nv * D_ /*sub*/) / 2 + 8;
         double* base = g_ws.ensure(f64_total + i32_doubles + 16);
@@ -599,7 +742,8 @@ class TpeDeviceHistory {
         double* d_ahigh = d_alow + D_;
         double* d_x = d_ahigh + D_;
         double* d_out = d_x + (size_t)S * D_;
-        int32_t* d_sorted = reinterpret_cast<int32_t*>(d_out + S);
+        double* d_scratch = d_out + S;
+        int32_t* d_sorted = reinterpret_cast<int32_t*>(d_scratch + n_scratch);
         int32_t* d_pos = d_sorted + (size_t)Nv * D_;
         int32_t* d_sub_rows = d_pos + n_;
         int32_t* d_sub_k = d_sub_rows + (size_t)Nv * D_;
@@ -626,12 +770,8 @@ class TpeDeviceHistory {
                                consider_endpoints ? 1 : 0, magic_clip ? 1 : 0,
                                d_c1, d_c2, d_c3);
         }
-        {
-            const int block = 256;
-            const size_t shmem = (2 * (size_t)D_ + 2 * block) * sizeof(double);
-            hipLaunchKernelGGL(k_mix_logpdf, dim3((unsigned)S), dim3(block), shmem,
-                               st, d_x, d_c1, d_c2, d_c3, d_logw, K, D_, d_out);
-        }
+        launch_mix_logpdf(st, d_x, d_c1, d_c2, d_c3, d_logw, K, D_, S, d_out,
+                          d_scratch);
         py::array_t<double> out(S);
         HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
                                  hipMemcpyDeviceToHost, st));

@@ -52,8 +52,20 @@ class _SpaceCache:
         self.dists = [space[n] for n in self.names]
         self.params = np.empty((0, len(self.names)), dtype=np.float64)
         self.valid = np.empty(0, dtype=bool)
-        self.sorted_rows = [np.empty(0, dtype=np.int64) for _ in self.names]
-        self.sorted_vals = [np.empty(0, dtype=np.float64) for _ in self.names]
+        # Per-dim sorted index kept in capacity-doubled buffers: the per-tell
+        # insert is a searchsorted + tail shift, avoiding np.insert's full
+        # reallocation per dimension per suggest.
+        self._n_sorted = 0
+        self._vals_buf = [np.empty(64, dtype=np.float64) for _ in self.names]
+        self._rows_buf = [np.empty(64, dtype=np.int64) for _ in self.names]
+
+    @property
+    def sorted_vals(self) -> list[np.ndarray]:
+        return [b[: self._n_sorted] for b in self._vals_buf]
+
+    @property
+    def sorted_rows(self) -> list[np.ndarray]:
+        return [b[: self._n_sorted] for b in self._rows_buf]
 
     def append(self, trials: Sequence[FrozenTrial]) -> None:
         n_new = len(trials)
@@ -76,16 +88,40 @@ class _SpaceCache:
         self.valid = np.concatenate([self.valid, valid], axis=0)
 
         new_rows = base + np.nonzero(valid)[0]
-        if len(new_rows) == 0:
+        m = len(new_rows)
+        if m == 0:
             return
+        n = self._n_sorted
+        if n + m > len(self._vals_buf[0]):
+            cap = max(2 * len(self._vals_buf[0]), n + m)
+            for c in range(len(self.names)):
+                vb = np.empty(cap, dtype=np.float64)
+                rb = np.empty(cap, dtype=np.int64)
+                vb[:n] = self._vals_buf[c][:n]
+                rb[:n] = self._rows_buf[c][:n]
+                self._vals_buf[c] = vb
+                self._rows_buf[c] = rb
         for c in range(len(self.names)):
+            vb = self._vals_buf[c]
+            rb = self._rows_buf[c]
             vals = self.params[new_rows, c]
-            order = np.argsort(vals, kind="stable")
-            vals_sorted = vals[order]
-            rows_sorted = new_rows[order]
-            pos = np.searchsorted(self.sorted_vals[c], vals_sorted)
-            self.sorted_vals[c] = np.insert(self.sorted_vals[c], pos, vals_sorted)
-            self.sorted_rows[c] = np.insert(self.sorted_rows[c], pos, rows_sorted)
+            if m == 1:
+                # side="right": equal values keep insertion (= row) order.
+                i = int(np.searchsorted(vb[:n], vals[0], side="right"))
+                vb[i + 1 : n + 1] = vb[i:n].copy()
+                rb[i + 1 : n + 1] = rb[i:n].copy()
+                vb[i] = vals[0]
+                rb[i] = new_rows[0]
+            else:
+                order = np.argsort(vals, kind="stable")
+                vals_sorted = vals[order]
+                rows_sorted = new_rows[order]
+                pos = np.searchsorted(vb[:n], vals_sorted, side="right")
+                merged_v = np.insert(vb[:n], pos, vals_sorted)
+                merged_r = np.insert(rb[:n], pos, rows_sorted)
+                vb[: n + m] = merged_v
+                rb[: n + m] = merged_r
+        self._n_sorted = n + m
 
 
 class _TpeHistory:
@@ -102,6 +138,15 @@ class _TpeHistory:
         self._seen: set[int] = set()
         self._trials: list[FrozenTrial] = []  # row-aligned frozen trials
         self._spaces: dict[tuple, _SpaceCache] = {}
+        # Rows appended in trial-number order? (True until proven otherwise; lets
+        # split() order the big "above" set with a boolean mask instead of argsort.)
+        self._rows_number_ascending = True
+        # Incrementally-sorted feasible-complete rows by objective value (single-
+        # objective minimize fast path for split(); ties keep row order, matching
+        # a stable argsort).
+        self._comp_vals = np.empty(0, dtype=np.float64)
+        self._comp_rows = np.empty(0, dtype=np.int64)
+        self._comp_cache_valid = True
 
     def __len__(self) -> int:
         return len(self._numbers)
@@ -110,10 +155,34 @@ class _TpeHistory:
     def trials(self) -> list[FrozenTrial]:
         return self._trials
 
-    def update(self, finished: Sequence[FrozenTrial], n_objectives: int) -> None:
-        if len(finished) == len(self._seen):
-            return  # finished trials are immutable and only accumulate
-        new = [t for t in finished if t._trial_id not in self._seen]
+    def update(
+        self, finished: Sequence[FrozenTrial], n_objectives: int, delta: bool = False
+    ) -> None:
+        """Append trials not yet mirrored.
+
+        With ``delta=True``, ``finished`` is a partial list (e.g. from
+        ``InMemoryStorage.get_finished_trials_since``) and is deduped directly.
+        Otherwise it is the complete finished list: the scan runs backwards and
+        stops once the known number of new trials is found — new trials usually
+        sit at the tail, so the common case touches O(new) entries, not
+        O(history).
+        """
+        if delta:
+            new = [t for t in finished if t._trial_id not in self._seen]
+        else:
+            n_missing = len(finished) - len(self._seen)
+            if n_missing <= 0:
+                return  # seen ⊆ finished and same cardinality ⇒ nothing new
+            if n_missing == len(finished):
+                new = list(finished)
+            else:
+                new = []
+                for t in reversed(finished):
+                    if t._trial_id not in self._seen:
+                        new.append(t)
+                        if len(new) == n_missing:
+                            break
+                new.reverse()
         if not new:
             return
         n_new = len(new)
@@ -137,6 +206,11 @@ class _TpeHistory:
                 has_iv[r] = True
             self._seen.add(t._trial_id)
 
+        if self._rows_number_ascending:
+            prev_last = self._numbers[-1] if self._numbers.size else -1
+            if numbers[0] <= prev_last or (n_new > 1 and np.any(np.diff(numbers) <= 0)):
+                self._rows_number_ascending = False
+
         self._numbers = np.concatenate([self._numbers, numbers])
         self._states = np.concatenate([self._states, states])
         if self._values is None:
@@ -150,6 +224,28 @@ class _TpeHistory:
         self._trials.extend(new)
         for cache in self._spaces.values():
             cache.append(new)
+
+        if self._comp_cache_valid:
+            if n_objectives != 1:
+                self._comp_cache_valid = False
+            else:
+                base = len(self._numbers) - n_new
+                fc = np.nonzero(
+                    (states == int(TrialState.COMPLETE)) & (violations <= 0)
+                )[0]
+                if fc.size:
+                    vals = values[fc, 0]
+                    if np.any(np.isnan(vals)):
+                        self._comp_cache_valid = False
+                    else:
+                        order = np.argsort(vals, kind="stable")
+                        vs = vals[order]
+                        rs = (base + fc)[order]
+                        # side="right": equal values keep insertion (= row) order,
+                        # matching a stable argsort over the full column.
+                        pos = np.searchsorted(self._comp_vals, vs, side="right")
+                        self._comp_vals = np.insert(self._comp_vals, pos, vs)
+                        self._comp_rows = np.insert(self._comp_rows, pos, rs)
 
     # ---- split ---------------------------------------------------------------------
 
@@ -171,9 +267,20 @@ class _TpeHistory:
         above_parts: list[np.ndarray] = []
 
         # 1. complete trials
-        comp_rows = rows[is_complete]
-        k = min(n_below, len(comp_rows))
-        if len(directions) <= 1:
+        if (
+            len(directions) == 1
+            and directions[0] != StudyDirection.MAXIMIZE
+            and self._comp_cache_valid
+        ):
+            # Minimize fast path: feasible completes are kept value-sorted
+            # incrementally, so the O(N log N) argsort per suggest disappears.
+            comp_rows = self._comp_rows
+            k = min(n_below, len(comp_rows))
+            below_parts.append(comp_rows[:k])
+            above_parts.append(comp_rows[k:])
+        elif len(directions) <= 1:
+            comp_rows = rows[is_complete]
+            k = min(n_below, len(comp_rows))
             assert self._values is not None
             vals = self._values[comp_rows, 0]
             sign = -1.0 if directions[0] == StudyDirection.MAXIMIZE else 1.0
@@ -181,6 +288,8 @@ class _TpeHistory:
             below_parts.append(comp_rows[order[:k]])
             above_parts.append(comp_rows[order[k:]])
         else:
+            comp_rows = rows[is_complete]
+            k = min(n_below, len(comp_rows))
             below_c, above_c = self._split_complete_mo(comp_rows, directions, k)
             below_parts.append(below_c)
             above_parts.append(above_c)
@@ -214,10 +323,20 @@ class _TpeHistory:
             above_parts.append(inf_rows[order[k:]])
 
         below = np.concatenate(below_parts) if below_parts else np.empty(0, dtype=np.int64)
-        above = np.concatenate(above_parts) if above_parts else np.empty(0, dtype=np.int64)
-        # Estimator observations are ordered by trial number.
+        # Estimator observations are ordered by trial number. `below` is small
+        # (the gamma quantile); `above` is everything else, so when rows were
+        # appended number-ascending its number order is just the row-order
+        # complement of `below` — a boolean mask instead of an O(N log N) sort.
         below = below[np.argsort(self._numbers[below], kind="stable")]
-        above = above[np.argsort(self._numbers[above], kind="stable")]
+        if self._rows_number_ascending:
+            mask = np.ones(n, dtype=bool)
+            mask[below] = False
+            above = rows[mask]
+        else:
+            above = (
+                np.concatenate(above_parts) if above_parts else np.empty(0, dtype=np.int64)
+            )
+            above = above[np.argsort(self._numbers[above], kind="stable")]
         return below, above
 
     def _split_complete_mo(
@@ -286,6 +405,14 @@ class _TpeHistory:
         obs = {name: mat[:, c] for c, name in enumerate(cache.names)}
 
         n_total = len(cache.valid)
+        if len(sel) * 16 < n_total:
+            # Tiny subset (e.g. the "below" gamma quantile): sorting it directly
+            # is cheaper than filtering the full presorted index per dimension.
+            orders_small = {
+                name: np.argsort(mat[:, c], kind="stable")
+                for c, name in enumerate(cache.names)
+            }
+            return obs, orders_small
         in_sel = np.zeros(n_total, dtype=bool)
         in_sel[sel] = True
         pos = np.empty(n_total, dtype=np.int64)

@@ -204,8 +204,8 @@ class TPESampler(BaseSampler):
         if search_space == {}:
             return {}
         states = (TrialState.COMPLETE, TrialState.PRUNED)
-        trials = study._get_trials(deepcopy=False, states=states, use_cache=True)
-        if len(trials) < self._n_startup_trials:
+        n_finished = study._storage.get_n_trials(study._study_id, states)
+        if n_finished < self._n_startup_trials:
             return {}
         return self._sample(study, trial, search_space)
 
@@ -289,12 +289,20 @@ class TPESampler(BaseSampler):
             study._study_id, (TrialState.COMPLETE, TrialState.PRUNED)
         )
         if n_finished != len(history):
-            finished = study._get_trials(
-                deepcopy=False,
-                states=(TrialState.COMPLETE, TrialState.PRUNED),
-                use_cache=True,
-            )
-            history.update(finished, len(study.directions))
+            delta_read = getattr(study._storage, "get_finished_trials_since", None)
+            if delta_read is not None:
+                history.update(
+                    delta_read(study._study_id, len(history)),
+                    len(study.directions),
+                    delta=True,
+                )
+            else:
+                finished = study._get_trials(
+                    deepcopy=False,
+                    states=(TrialState.COMPLETE, TrialState.PRUNED),
+                    use_cache=True,
+                )
+                history.update(finished, len(study.directions))
 
         if self._gamma is None:
             self._gamma = (

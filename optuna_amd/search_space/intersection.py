@@ -53,6 +53,7 @@ class IntersectionSearchSpace:
         self._study_id: int | None = None
         self._include_pruned = include_pruned
         self._n_finished_seen: int = -1
+        self._log_idx: int = 0
 
     def calculate(self, study: "Study") -> dict[str, BaseDistribution]:
         if self._study_id is None:
@@ -72,6 +73,18 @@ class IntersectionSearchSpace:
         if n_finished == self._n_finished_seen and self._search_space is not None:
             return dict(sorted(self._search_space.items(), key=lambda x: x[0]))
         self._n_finished_seen = n_finished
+
+        # Delta path: _calculate only looks at finished trials, and finished
+        # trials are immutable — read just the ones logged since the last call.
+        delta_read = getattr(study._storage, "get_finished_trials_since", None)
+        if delta_read is not None:
+            new_finished = delta_read(study._study_id, self._log_idx)
+            self._log_idx += len(new_finished)
+            self._search_space = _calculate(
+                new_finished, self._include_pruned, self._search_space
+            )
+            search_space = self._search_space or {}
+            return dict(sorted(search_space.items(), key=lambda x: x[0]))
 
         states_of_interest = [TrialState.COMPLETE, TrialState.WAITING, TrialState.RUNNING]
         if self._include_pruned:

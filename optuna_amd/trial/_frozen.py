@@ -28,6 +28,28 @@ from optuna_amd.trial._state import TrialState
 
 _logger = _logging.get_logger(__name__)
 
+# Value types that deepcopy may share (immutable or treated as such everywhere).
+_ATOMIC_TYPES = (int, float, str, bool, type(None), datetime.datetime, TrialState)
+
+
+def _fast_deepcopy_value(v: Any, memo: dict) -> Any:
+    """Deepcopy with fast paths for FrozenTrial's typical attribute shapes."""
+    if isinstance(v, _ATOMIC_TYPES):
+        return v
+    tv = type(v)
+    if tv is dict:
+        return {k: _fast_deepcopy_value(x, memo) for k, x in v.items()}
+    if tv is list:
+        return [_fast_deepcopy_value(x, memo) for x in v]
+    if tv is tuple:
+        return tuple(_fast_deepcopy_value(x, memo) for x in v)
+    if tv in (FloatDistribution, IntDistribution):
+        # All fields are scalars; reconstruct without __init__ revalidation.
+        d = object.__new__(tv)
+        d.__dict__.update(v.__dict__)
+        return d
+    return copy.deepcopy(v, memo)
+
 
 class FrozenTrial(BaseTrial):
     """A finished (or snapshotted) trial.
@@ -75,6 +97,20 @@ class FrozenTrial(BaseTrial):
         if not isinstance(other, FrozenTrial):
             return NotImplemented
         return other.__dict__ == self.__dict__
+
+    def __deepcopy__(self, memo: dict) -> "FrozenTrial":
+        # Storages deepcopy trials on every read/tell to isolate their internal
+        # state from callers; the generic copy.deepcopy walk dominates tell()
+        # latency. Attribute values are overwhelmingly scalars, flat dicts of
+        # scalars, and parameter distributions with scalar fields — copy those
+        # directly and fall back to copy.deepcopy only for anything unusual
+        # (e.g. categorical choices holding user objects).
+        cls = self.__class__
+        new = cls.__new__(cls)
+        memo[id(self)] = new
+        for key, v in self.__dict__.items():
+            new.__dict__[key] = _fast_deepcopy_value(v, memo)
+        return new
 
     def __lt__(self, other: Any) -> bool:
         if not isinstance(other, FrozenTrial):
