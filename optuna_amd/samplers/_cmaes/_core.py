@@ -310,6 +310,195 @@ class SepCMA(CMA):
         return False
 
 
+class CMAwM(CMA):
+    """CMA-ES with Margin for mixed-integer spaces (Hamano et al., GECCO 2022).
+
+    Discrete coordinates (``steps[i] > 0``) are optimized in the continuous
+    relaxation; ``ask`` returns both the discretized point (for evaluation) and
+    the raw sample (for ``tell``). After every generation update, the margin
+    correction lower-bounds the probability that a sample leaves the mean's
+    current discrete cell by ``margin`` (default ``1/(n_dim·λ)``), adjusting the
+    mean and a per-coordinate sampling expansion factor ``A``.
+
+    Interface parity: reference ``optuna/samplers/_cmaes.py`` :432-456 (ask
+    returns ``(x_for_eval, x_for_tell)``; tell consumes the raw samples), with
+    ``steps`` given in the 0-1-transformed space (:568-579).
+    """
+
+    def __init__(
+        self,
+        mean: np.ndarray,
+        sigma: float,
+        bounds: np.ndarray,
+        steps: np.ndarray,
+        n_max_resampling: int = 100,
+        seed: int | None = None,
+        population_size: int | None = None,
+        cov: np.ndarray | None = None,
+        margin: float | None = None,
+    ) -> None:
+        super().__init__(
+            mean,
+            sigma,
+            bounds=np.asarray(bounds, dtype=np.float64),
+            n_max_resampling=n_max_resampling,
+            seed=seed,
+            population_size=population_size,
+            cov=cov,
+        )
+        steps = np.asarray(steps, dtype=np.float64)
+        if len(steps) != self._n_dim:
+            raise ValueError("steps must have one entry per dimension")
+        self._steps = steps
+        self._disc_idx = np.nonzero(steps > 0)[0]
+        self._cont_idx = np.nonzero(steps == 0)[0]
+        # Candidate grids and midpoint thresholds per discrete dim.
+        self._z_space: list[np.ndarray] = []
+        self._z_lims: list[np.ndarray] = []
+        for i in self._disc_idx:
+            lo, hi = self._bounds[i]
+            n_cand = max(int(round((hi - lo) / steps[i])) + 1, 2)
+            cand = lo + steps[i] * np.arange(n_cand)
+            self._z_space.append(cand)
+            self._z_lims.append((cand[:-1] + cand[1:]) / 2.0)
+        self._A = np.ones(self._n_dim)
+        self._margin = (
+            margin if margin is not None else 1.0 / (self._n_dim * self._popsize)
+        )
+
+    # ---- sampling -------------------------------------------------------------------
+
+    def _sample_one(self) -> np.ndarray:
+        B, D = self._eigen()
+        z = self._rng.randn(self._n_dim)
+        return self._mean + self._sigma * self._A * (B @ (D * z))
+
+    def _in_bounds(self, x: np.ndarray) -> bool:
+        # Discrete dims are snapped onto the grid, so only continuous dims
+        # constrain resampling.
+        c = self._cont_idx
+        if len(c) == 0:
+            return True
+        return bool(
+            np.all(x[c] >= self._bounds[c, 0]) and np.all(x[c] <= self._bounds[c, 1])
+        )
+
+    def _encode(self, x: np.ndarray) -> np.ndarray:
+        out = x.copy()
+        c = self._cont_idx
+        out[c] = np.clip(out[c], self._bounds[c, 0], self._bounds[c, 1])
+        for k, i in enumerate(self._disc_idx):
+            cand = self._z_space[k]
+            out[i] = cand[np.argmin(np.abs(cand - x[i]))]
+        return out
+
+    def ask(self) -> tuple[np.ndarray, np.ndarray]:  # type: ignore[override]
+        for _ in range(self._n_max_resampling):
+            x = self._sample_one()
+            if self._in_bounds(x):
+                return self._encode(x), x
+        x = self._sample_one()
+        x[self._cont_idx] = np.clip(
+            x[self._cont_idx],
+            self._bounds[self._cont_idx, 0],
+            self._bounds[self._cont_idx, 1],
+        )
+        return self._encode(x), x
+
+    # ---- update ---------------------------------------------------------------------
+
+    def tell(self, solutions: list[tuple[np.ndarray, float]]) -> None:
+        if len(solutions) != self._popsize:
+            raise ValueError("Must tell popsize-length solutions.")
+        self._g += 1
+        solutions.sort(key=lambda s: s[1])
+
+        B, D = self._eigen()
+        self._B, self._D = None, None
+
+        xs = np.array([s[0] for s in solutions])  # raw samples (λ, n)
+        # Underlying Gaussian variates: x = m + σ A∘(BDz) ⇒ y = (x−m)/(σA).
+        ys = (xs - self._mean) / (self._sigma * self._A)
+
+        w = self._weights
+        y_w = w[: self._mu] @ ys[: self._mu]
+        # Mean moves in the sampled (A-scaled) space.
+        self._mean = self._mean + self._cm * self._sigma * self._A * y_w
+
+        C_inv_half = B @ np.diag(1.0 / D) @ B.T
+        self._p_sigma = (1 - self._c_sigma) * self._p_sigma + math.sqrt(
+            self._c_sigma * (2 - self._c_sigma) * self._mu_eff
+        ) * (C_inv_half @ y_w)
+        norm_p_sigma = float(np.linalg.norm(self._p_sigma))
+        self._sigma *= math.exp(
+            (self._c_sigma / self._d_sigma) * (norm_p_sigma / self._chi_n - 1)
+        )
+        self._sigma = min(self._sigma, _SIGMA_MAX)
+
+        h_sigma_cond = norm_p_sigma / math.sqrt(
+            1 - (1 - self._c_sigma) ** (2 * (self._g + 1))
+        )
+        h_sigma = 1.0 if h_sigma_cond < (1.4 + 2 / (self._n_dim + 1)) * self._chi_n else 0.0
+        self._pc = (1 - self._cc) * self._pc + h_sigma * math.sqrt(
+            self._cc * (2 - self._cc) * self._mu_eff
+        ) * y_w
+
+        w_circ = np.where(
+            w >= 0,
+            w,
+            w * self._n_dim / (np.linalg.norm(ys @ C_inv_half.T, axis=1) ** 2 + _EPS),
+        )
+        delta_h = (1 - h_sigma) * self._cc * (2 - self._cc)
+        rank_one = np.outer(self._pc, self._pc)
+        rank_mu = (ys.T * w_circ) @ ys
+        self._C = (
+            (1 + self._c1 * delta_h - self._c1 - self._cmu * w.sum()) * self._C
+            + self._c1 * rank_one
+            + self._cmu * rank_mu
+        )
+
+        self._margin_correction()
+
+    def _margin_correction(self) -> None:
+        from scipy.special import ndtr, ndtri
+
+        alpha = self._margin
+        diag_C = np.diag(self._C)
+        for k, i in enumerate(self._disc_idx):
+            lims = self._z_lims[k]
+            m = float(self._mean[i])
+            s_dev = self._sigma * self._A[i] * math.sqrt(max(diag_C[i], _EPS**2))
+            pos = int(np.searchsorted(lims, m))
+            n_cand = len(lims) + 1
+            if pos == 0:
+                # Mean sits in the lowest cell: keep P(x > lims[0]) ≥ α.
+                lo_lim = lims[0]
+                if 1.0 - ndtr((lo_lim - m) / s_dev) < alpha:
+                    self._mean[i] = lo_lim - s_dev * ndtri(1.0 - alpha)
+            elif pos == n_cand - 1:
+                # Highest cell: keep P(x < lims[-1]) ≥ α.
+                up_lim = lims[-1]
+                if ndtr((up_lim - m) / s_dev) < alpha:
+                    self._mean[i] = up_lim + s_dev * ndtri(1.0 - alpha)
+            else:
+                lo_lim = lims[pos - 1]
+                up_lim = lims[pos]
+                p_low = float(ndtr((lo_lim - m) / s_dev))
+                p_up = 1.0 - float(ndtr((up_lim - m) / s_dev))
+                if p_low >= alpha / 2 and p_up >= alpha / 2:
+                    continue
+                p_low_t = max(p_low, alpha / 2)
+                p_up_t = max(p_up, alpha / 2)
+                q_low = float(ndtri(p_low_t))
+                q_up = float(ndtri(1.0 - p_up_t))
+                if not q_up > q_low:
+                    continue  # degenerate (extremely wide cell); leave untouched
+                s_new = (up_lim - lo_lim) / (q_up - q_low)
+                self._mean[i] = lo_lim - s_new * q_low
+                self._A[i] = s_new / (
+                    self._sigma * math.sqrt(max(diag_C[i], _EPS**2))
+                )
+
 def get_warm_start_mgd(
     source_solutions: list[tuple[np.ndarray, float]],
     gamma: float = 0.1,
