@@ -115,6 +115,20 @@ class _ParzenEstimator:
         self._n_kernels = n_obs + 1
 
         # ---- build per-dimension model, packing numerical dims into SoA ------------
+        dists = [search_space[n] for n in self._param_names]
+        if dists and all(
+            not isinstance(d, CategoricalDistribution) for d in dists
+        ):
+            # All-numerical space: one vectorized fit over the (n, D) matrix
+            # instead of D small per-dim numpy calls (the per-call overhead
+            # dominates for the "below" estimator's ~25 observations).
+            self._numerical = self._numerical_kernels_batched(
+                observations, dists, parameters, n_obs
+            )
+            self._categoricals = []
+            self._precompute_logpdf_coefficients()
+            return
+
         num_idx: list[int] = []
         kinds: list[int] = []
         lows: list[float] = []
@@ -264,6 +278,89 @@ class _ParzenEstimator:
         w[np.arange(n_obs), idx] += 1.0
         row_sums = w.sum(axis=1, keepdims=True)
         return w / np.where(row_sums == 0, 1.0, row_sums)
+
+    def _numerical_kernels_batched(
+        self,
+        observations: dict[str, np.ndarray],
+        dists: list[BaseDistribution],
+        parameters: _ParzenEstimatorParameters,
+        n_obs: int,
+    ) -> _NumericalDims:
+        """Vectorized equivalent of the per-dim `_numerical_kernels` loop."""
+        D = len(dists)
+        steps = np.array(
+            [float(d.step) if d.step is not None else 0.0 for d in dists]
+        )
+        is_log = np.array([bool(d.log) for d in dists])
+        lows = np.array([float(d.low) for d in dists])
+        highs = np.array([float(d.high) for d in dists])
+        a_lows = np.where(steps != 0.0, lows - steps / 2, lows)
+        a_highs = np.where(steps != 0.0, highs + steps / 2, highs)
+        if is_log.any():
+            a_lows[is_log] = np.log(a_lows[is_log])
+            a_highs[is_log] = np.log(a_highs[is_log])
+        kinds = np.where(
+            is_log,
+            np.where(steps != 0.0, KIND_LOG_DISC, KIND_LOG),
+            np.where(steps != 0.0, KIND_DISC, KIND_CONT),
+        ).astype(np.int8)
+
+        ranges = a_highs - a_lows
+        if n_obs == 0:
+            mus_full = (0.5 * (a_lows + a_highs))[None, :]
+            sig_full = ranges[None, :]
+        else:
+            mus_mat = np.column_stack(
+                [np.asarray(observations[n], dtype=np.float64) for n in self._param_names]
+            )
+            if is_log.any():
+                mus_mat[:, is_log] = np.log(mus_mat[:, is_log])
+            order_cols = [
+                self._sorted_orders.get(name) for name in self._param_names
+            ]
+            if all(o is not None for o in order_cols):
+                order_mat = np.column_stack(order_cols)
+            else:
+                order_mat = np.column_stack(
+                    [
+                        o if o is not None else np.argsort(mus_mat[:, c])
+                        for c, o in enumerate(order_cols)
+                    ]
+                )
+            padded = np.empty((n_obs + 2, D), dtype=np.float64)
+            padded[0] = a_lows
+            padded[-1] = a_highs
+            padded[1:-1] = np.take_along_axis(mus_mat, order_mat, axis=0)
+            gaps_left = padded[1:-1] - padded[:-2]
+            gaps_right = padded[2:] - padded[1:-1]
+            sorted_sigmas = np.maximum(gaps_left, gaps_right)
+            if not parameters.consider_endpoints and n_obs >= 2:
+                sorted_sigmas[0] = padded[2] - padded[1]
+                sorted_sigmas[-1] = padded[-2] - padded[-3]
+            inverse = np.empty_like(order_mat)
+            np.put_along_axis(
+                inverse, order_mat, np.arange(n_obs, dtype=order_mat.dtype)[:, None], axis=0
+            )
+            sigmas_mat = np.take_along_axis(sorted_sigmas, inverse, axis=0)
+            if parameters.consider_magic_clip:
+                minsig = ranges / min(100.0, 1.0 + (n_obs + 1))
+            else:
+                minsig = np.full(D, EPS)
+            sigmas_mat = np.clip(sigmas_mat, minsig, ranges)
+            mus_full = np.vstack([mus_mat, 0.5 * (a_lows + a_highs)])
+            sig_full = np.vstack([sigmas_mat, ranges])
+
+        return _NumericalDims(
+            dim_indices=np.arange(D, dtype=np.int64),
+            kinds=kinds,
+            lows=lows,
+            highs=highs,
+            steps=steps,
+            adapted_lows=a_lows,
+            adapted_highs=a_highs,
+            mus=mus_full,
+            sigmas=sig_full,
+        )
 
     def _numerical_kernels(
         self,
