@@ -19,6 +19,7 @@
 #include <hip/hip_runtime.h>
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
 
 #include <cstdint>
 #include <stdexcept>
@@ -578,7 +579,7 @@ int device_count() {
 // (a 20-block launch leaves >90% of the chip idle and was the top kernel in the
 // round-4 profile at 71 µs; the tiled scan reads the same bytes chip-wide).
 
-__global__ void k_compact_count(const int32_t* __restrict__ sorted_rows,  // (Nv, D)
+__global__ void k_compact_count(const int32_t* __restrict__ sorted_rows,  // (D, Nv) col-major
                                 const int32_t* __restrict__ pos,          // (n_rows,)
                                 int64_t Nv, int64_t D,
                                 int32_t* __restrict__ counts) {  // (D, n_tiles)
@@ -586,7 +587,7 @@ __global__ void k_compact_count(const int32_t* __restrict__ sorted_rows,  // (Nv
     const int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     __shared__ int32_t red[256];
     int32_t flag = 0;
-    if (r < Nv) flag = (pos[sorted_rows[r * D + d]] >= 0) ? 1 : 0;
+    if (r < Nv) flag = (pos[sorted_rows[d * Nv + r]] >= 0) ? 1 : 0;
     red[threadIdx.x] = flag;
     __syncthreads();
     for (int s = blockDim.x / 2; s > 0; s >>= 1) {
@@ -622,7 +623,7 @@ __global__ void k_compact_write(const int32_t* __restrict__ sorted_rows,
     __shared__ int32_t scan[256];
     int32_t row = -1, k = -1, flag = 0;
     if (r < Nv) {
-        row = sorted_rows[r * D + d];
+        row = sorted_rows[d * Nv + r];
         k = pos[row];
         flag = (k >= 0) ? 1 : 0;
     }
@@ -738,20 +739,25 @@ class TpeDeviceHistory {
         n_ += n_new;
     }
 
-    py::array_t<double> score(const arr_i32& sorted_rows,  // (Nv, D)
-                              const arr_i32& pos,          // (n_rows,)
+    py::array_t<double> score(const std::vector<arr_i32>& sorted_cols,  // D × (Nv,)
+                              const arr_i32& pos,  // (n_rows,)
                               int64_t n_above,
                               const arr_f64& logw,  // (n_above + 1,)
                               const arr_f64& alow, const arr_f64& ahigh,
                               const arr_f64& x,  // (S, D) KDE domain
                               bool consider_endpoints, bool magic_clip) {
-        const int64_t Nv = sorted_rows.ndim() == 2 ? sorted_rows.shape(0) : 0;
+        const int64_t Nv =
+            sorted_cols.empty() ? 0 : (int64_t)sorted_cols[0].size();
         const int64_t S = x.shape(0);
         const int64_t Na = n_above;
         const int64_t K = Na + 1;
         if ((int64_t)pos.size() != n_ || x.shape(1) != D_ ||
-            (Nv > 0 && sorted_rows.shape(1) != D_) || (int64_t)logw.size() != K)
+            (Nv > 0 && (int64_t)sorted_cols.size() != D_) ||
+            (int64_t)logw.size() != K)
             throw std::runtime_error("score: shape mismatch");
+        for (const auto& col : sorted_cols)
+            if ((int64_t)col.size() != Nv)
+                throw std::runtime_error("score: ragged sorted columns");
 
         hipStream_t st = g_ws.get_stream();
         const size_t n_c = (size_t)K * D_;
@@ -781,8 +787,8 @@ class TpeDeviceHistory {
         int32_t* d_counts = d_sub_k + (size_t)Nv * D_;
 
         g_ws.begin_uploads();
-        if (Nv > 0)
-            g_ws.h2d(d_sorted, sorted_rows.data(), (size_t)Nv * D_ * 4, st);
+        for (int64_t d = 0; d < (int64_t)sorted_cols.size() && Nv > 0; ++d)
+            g_ws.h2d(d_sorted + d * Nv, sorted_cols[d].data(), (size_t)Nv * 4, st);
         if (n_ > 0) g_ws.h2d(d_pos, pos.data(), (size_t)n_ * 4, st);
         g_ws.h2d(d_logw, logw.data(), K * 8, st);
         g_ws.h2d(d_alow, alow.data(), D_ * 8, st);

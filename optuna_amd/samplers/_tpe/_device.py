@@ -96,11 +96,9 @@ class _SpaceDeviceMirror:
         n_total = len(cache.valid)
         pos = np.full(n_total, -1, dtype=np.int32)
         pos[sel] = np.arange(len(sel), dtype=np.int32)
-        sorted_rows = (
-            np.stack(cache.sorted_rows, axis=1).astype(np.int32)
-            if cache.sorted_rows and len(cache.sorted_rows[0])
-            else np.empty((0, len(cache.names)), dtype=np.int32)
-        )
+        # Per-dim contiguous int32 prefixes go to the device as-is — no
+        # (Nv, D) restack per suggest (was the top host cost on the GPU box).
+        sorted_cols = list(cache.sorted_rows)
         x = np.column_stack(
             [np.asarray(samples[n], dtype=np.float64) for n in cache.names]
         )
@@ -109,7 +107,7 @@ class _SpaceDeviceMirror:
         with np.errstate(divide="ignore"):
             logw = np.log(weights)
         return self._hist.score(
-            sorted_rows,
+            sorted_cols,
             pos,
             int(len(sel)),
             logw,

@@ -55,9 +55,11 @@ class _SpaceCache:
         # Per-dim sorted index kept in capacity-doubled buffers: the per-tell
         # insert is a searchsorted + tail shift, avoiding np.insert's full
         # reallocation per dimension per suggest.
+        # Rows as int32: the device compaction kernels consume these columns
+        # directly (contiguous per-dim prefixes, no per-suggest restack/cast).
         self._n_sorted = 0
         self._vals_buf = [np.empty(64, dtype=np.float64) for _ in self.names]
-        self._rows_buf = [np.empty(64, dtype=np.int64) for _ in self.names]
+        self._rows_buf = [np.empty(64, dtype=np.int32) for _ in self.names]
 
     @property
     def sorted_vals(self) -> list[np.ndarray]:
@@ -96,7 +98,7 @@ class _SpaceCache:
             cap = max(2 * len(self._vals_buf[0]), n + m)
             for c in range(len(self.names)):
                 vb = np.empty(cap, dtype=np.float64)
-                rb = np.empty(cap, dtype=np.int64)
+                rb = np.empty(cap, dtype=np.int32)
                 vb[:n] = self._vals_buf[c][:n]
                 rb[:n] = self._rows_buf[c][:n]
                 self._vals_buf[c] = vb

@@ -104,3 +104,48 @@ def test_warn_and_convert_inf() -> None:
     with pytest.warns(UserWarning):
         out = gp.warn_and_convert_inf(vals)
     assert out.max() == 2.0 and out.min() == 1.0
+
+
+def test_numpy_loss_grad_matches_torch_autograd() -> None:
+    """The closed-form fit loss/grad must equal the torch-autograd path."""
+    import torch
+
+    from optuna_amd._gp import gp as gp_mod
+    from optuna_amd._gp import prior
+
+    rng = np.random.RandomState(0)
+    n, d = 40, 5
+    X = rng.rand(n, d)
+    Y = rng.randn(n)
+    is_cat = np.zeros(d, dtype=bool)
+    gpr = gp_mod.GPRegressor(
+        is_categorical=torch.from_numpy(is_cat),
+        X_train=torch.from_numpy(X),
+        y_train=torch.from_numpy(Y),
+        inverse_squared_lengthscales=torch.ones(d, dtype=torch.float64),
+        kernel_scale=torch.tensor(1.0, dtype=torch.float64),
+        noise_var=torch.tensor(1.0, dtype=torch.float64),
+    )
+    minimum_noise = 1e-6
+    sqd = gpr._squared_X_diff.numpy()
+
+    for det in (False, True):
+        for trial_i in range(5):
+            raw = rng.randn(d + 2) * 0.7
+            loss_np, grad_np = gpr._loss_and_grad_numpy(raw, sqd, Y, minimum_noise, det)
+
+            raw_t = torch.from_numpy(raw).requires_grad_(True)
+            with torch.enable_grad():
+                gpr.inverse_squared_lengthscales = torch.exp(raw_t[:d])
+                gpr.kernel_scale = torch.exp(raw_t[d])
+                gpr.noise_var = (
+                    torch.tensor(minimum_noise, dtype=torch.float64)
+                    if det
+                    else torch.exp(raw_t[d + 1]) + minimum_noise
+                )
+                loss_t = -gpr.marginal_log_likelihood() - prior.default_log_prior(gpr)
+                loss_t.backward()
+            np.testing.assert_allclose(loss_np, loss_t.item(), rtol=1e-10)
+            np.testing.assert_allclose(
+                grad_np, raw_t.grad.detach().numpy(), rtol=1e-8, atol=1e-10
+            )

@@ -243,3 +243,45 @@ def test_resident_history_matches_host(core) -> None:
         np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
 
         cache.append(make_trials(50, 900 + round_i * 50))
+
+
+def test_gp_device_fit_matches_cpu() -> None:
+    """The MI355X closed-form MLL fit must agree with the CPU numpy fit."""
+    import torch
+
+    from optuna_amd._gp import gp as gp_mod
+    from optuna_amd._gp import prior
+
+    rng = np.random.RandomState(2)
+    n, d = 600, 8  # above _DEVICE_FIT_MIN_OBS → device branch
+    X = rng.rand(n, d)
+    Y = np.sum((X - 0.3) ** 2, axis=1) + 0.05 * rng.randn(n)
+    Y = (Y - Y.mean()) / Y.std()
+    is_cat = np.zeros(d, dtype=bool)
+
+    assert torch.cuda.is_available()
+    gpr_dev = gp_mod.fit_kernel_params(
+        X, Y, is_cat, prior.default_log_prior, 1e-6, False
+    )
+
+    # Force the CPU branch by lowering the threshold temporarily.
+    old = gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS
+    gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = 10**9
+    try:
+        gpr_cpu = gp_mod.fit_kernel_params(
+            X, Y, is_cat, prior.default_log_prior, 1e-6, False
+        )
+    finally:
+        gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = old
+
+    np.testing.assert_allclose(
+        gpr_dev.inverse_squared_lengthscales.numpy(),
+        gpr_cpu.inverse_squared_lengthscales.numpy(),
+        rtol=1e-4,
+    )
+    np.testing.assert_allclose(
+        gpr_dev.kernel_scale.item(), gpr_cpu.kernel_scale.item(), rtol=1e-4
+    )
+    np.testing.assert_allclose(
+        gpr_dev.noise_var.item(), gpr_cpu.noise_var.item(), rtol=1e-3
+    )
