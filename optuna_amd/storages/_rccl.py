@@ -347,6 +347,25 @@ class RcclStorage(BaseStorage):
                 trials = copy.deepcopy(trials)
             return trials
 
+    def get_finished_trials_since(self, study_id: int, start: int) -> list[FrozenTrial]:
+        """O(delta) read of COMPLETE/PRUNED trials in finish order (see
+        InMemoryStorage.get_finished_trials_since for the contract)."""
+        with self._thread_lock:
+            self._sync()
+            return self._replay.finished_trials_since(study_id, start)
+
+    def get_n_trials(
+        self, study_id: int, state: tuple[TrialState, ...] | TrialState | None = None
+    ) -> int:
+        if isinstance(state, tuple) and set(state) == {
+            TrialState.COMPLETE,
+            TrialState.PRUNED,
+        }:
+            with self._thread_lock:
+                self._sync()
+                return len(self._replay._study_id_to_finished.get(study_id, []))
+        return super().get_n_trials(study_id, state)
+
 
 # Descriptive alias: the class name advertises the MI355X deployment (one rank per
 # GPU over RCCL/xGMI); the mechanism is the distributed op-log table.

@@ -344,6 +344,27 @@ class JournalStorage(BaseStorage):
                 trials = copy.deepcopy(trials)
             return trials
 
+    def get_finished_trials_since(self, study_id: int, start: int) -> list[FrozenTrial]:
+        """O(delta) read of COMPLETE/PRUNED trials in finish order (see
+        InMemoryStorage.get_finished_trials_since for the contract)."""
+        with self._thread_lock:
+            self._sync()
+            return self._replay_result.finished_trials_since(study_id, start)
+
+    def get_n_trials(
+        self, study_id: int, state: tuple[TrialState, ...] | TrialState | None = None
+    ) -> int:
+        if isinstance(state, tuple) and set(state) == {
+            TrialState.COMPLETE,
+            TrialState.PRUNED,
+        }:
+            with self._thread_lock:
+                self._sync()
+                return len(
+                    self._replay_result._study_id_to_finished.get(study_id, [])
+                )
+        return super().get_n_trials(study_id, state)
+
 
 class _ReplayState:
     """In-memory state reconstructed from the op log (one per storage object)."""
@@ -354,6 +375,9 @@ class _ReplayState:
         self._studies: dict[int, FrozenStudy] = {}
         self._trials: dict[int, FrozenTrial] = {}
         self._study_id_to_trial_ids: dict[int, list[int]] = {}
+        # COMPLETE/PRUNED trials per study in finish order; finished trials are
+        # immutable, so storages expose O(delta) history reads from this log.
+        self._study_id_to_finished: dict[int, list[FrozenTrial]] = {}
         self._trial_id_to_study_id: dict[int, int] = {}
         self._next_study_id = 0
         self._worker_id_to_owned_trial_id: dict[str, int] = {}
@@ -367,6 +391,20 @@ class _ReplayState:
     def __setstate__(self, state: dict[str, Any]) -> None:
         self.__dict__.update(state)
         self.last_created_trial_id = -1
+        if "_study_id_to_finished" not in self.__dict__:
+            # Snapshot from an older build: rebuild the finished log (trial-id
+            # order; consumers dedupe by id, so order only affects row order).
+            self._study_id_to_finished = {}
+            for study_id, trial_ids in self._study_id_to_trial_ids.items():
+                self._study_id_to_finished[study_id] = [
+                    self._trials[tid]
+                    for tid in trial_ids
+                    if self._trials[tid].state
+                    in (TrialState.COMPLETE, TrialState.PRUNED)
+                ]
+
+    def finished_trials_since(self, study_id: int, start: int) -> list[FrozenTrial]:
+        return list(self._study_id_to_finished.get(study_id, [])[start:])
 
     # ---- queries --------------------------------------------------------------------
 
@@ -523,6 +561,10 @@ class _ReplayState:
         )
         self._study_id_to_trial_ids[study_id].append(trial_id)
         self._trial_id_to_study_id[trial_id] = study_id
+        if self._trials[trial_id].state in (TrialState.COMPLETE, TrialState.PRUNED):
+            self._study_id_to_finished.setdefault(study_id, []).append(
+                self._trials[trial_id]
+            )
         if self._mine(log):
             self.last_created_trial_id = trial_id
             if self._trials[trial_id].state == TrialState.RUNNING:
@@ -573,6 +615,10 @@ class _ReplayState:
         if log["values"] is not None:
             trial.values = log["values"]
         self._trials[trial_id] = trial
+        if state in (TrialState.COMPLETE, TrialState.PRUNED):
+            self._study_id_to_finished.setdefault(
+                self._trial_id_to_study_id[trial_id], []
+            ).append(trial)
 
     def _on_trial_intermediate(self, log: dict[str, Any]) -> None:
         trial_id = log["trial_id"]

@@ -42,3 +42,42 @@ class TestGrpcSqliteStorage(StorageTestCase):
     def storage(self):  # type: ignore[override]
         with StorageSupplier("grpc_sqlite") as s:
             yield s
+
+
+def test_finished_trials_since_inmemory_and_journal(tmp_path) -> None:
+    import optuna_amd
+    from optuna_amd.study import StudyDirection
+    from optuna_amd.trial import TrialState
+
+    """Delta-read contract: slice of the finish-order log, immutable entries."""
+    from optuna_amd.storages import InMemoryStorage, JournalStorage
+    from optuna_amd.storages.journal import JournalFileBackend
+
+    for make in (
+        lambda: InMemoryStorage(),
+        lambda: JournalStorage(JournalFileBackend(str(tmp_path / "log.jsonl"))),
+    ):
+        storage = make()
+        sid = storage.create_new_study([StudyDirection.MINIMIZE], study_name=None)
+        # 3 running trials; finish them out of creation order.
+        tids = [storage.create_new_trial(sid) for _ in range(3)]
+        for tid in (tids[1], tids[0]):
+            storage.set_trial_state_values(tid, TrialState.RUNNING)
+            storage.set_trial_state_values(tid, TrialState.COMPLETE, [1.0])
+        log0 = storage.get_finished_trials_since(sid, 0)
+        assert [t._trial_id for t in log0] == [tids[1], tids[0]]
+        assert storage.get_n_trials(sid, (TrialState.COMPLETE, TrialState.PRUNED)) == 2
+        # Delta after cursor 2 is empty until another trial finishes.
+        assert storage.get_finished_trials_since(sid, 2) == []
+        storage.set_trial_state_values(tids[2], TrialState.RUNNING)
+        storage.set_trial_state_values(tids[2], TrialState.FAIL)
+        assert storage.get_finished_trials_since(sid, 2) == []  # FAIL not logged
+        # Template-created finished trial lands in the log too.
+        from optuna_amd.trial import FrozenTrial
+
+        t = optuna_amd.create_trial(
+            params={}, distributions={}, value=3.0
+        )
+        storage.create_new_trial(sid, template_trial=t)
+        delta = storage.get_finished_trials_since(sid, 2)
+        assert len(delta) == 1 and delta[0].value == 3.0
