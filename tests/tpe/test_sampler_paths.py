@@ -1,0 +1,50 @@
+"""Cross-path determinism checks for the TPE fast paths."""
+from __future__ import annotations
+
+import numpy as np
+
+from optuna_amd.distributions import FloatDistribution
+
+
+def test_same_suggestions_across_storages_and_cache_paths(tmp_path) -> None:
+    """The delta-log, value-order cache, and mask-complement fast paths must not
+    change sampling decisions: same seed -> identical suggestions on
+    InMemoryStorage, JournalStorage, and with the caches force-disabled."""
+    import optuna_amd
+    from optuna_amd.storages import JournalStorage
+    from optuna_amd.storages.journal import JournalFileBackend
+
+    def run(storage, disable_caches: bool) -> list[float]:
+        sampler = optuna_amd.samplers.TPESampler(seed=9, n_startup_trials=5)
+        study = optuna_amd.create_study(storage=storage, sampler=sampler)
+        rng = np.random.RandomState(1)
+        dists = {f"x{i}": FloatDistribution(-5.0, 5.0) for i in range(6)}
+        study.add_trials(
+            [
+                optuna_amd.create_trial(
+                    params={f"x{i}": float(rng.uniform(-5, 5)) for i in range(6)},
+                    distributions=dists,
+                    value=float(rng.rand()),
+                )
+                for _ in range(60)
+            ]
+        )
+        out: list[float] = []
+        for _ in range(5):
+            t = study.ask()
+            xs = [t.suggest_float(f"x{i}", -5, 5) for i in range(6)]
+            if disable_caches:
+                hist = sampler._histories[study._study_id]
+                hist._comp_cache_valid = False
+                hist._rows_number_ascending = False
+            out.extend(xs)
+            study.tell(t, float(sum(x * x for x in xs)))
+        return out
+
+    base = run(None, False)
+    journal = run(
+        JournalStorage(JournalFileBackend(str(tmp_path / "j.jsonl"))), False
+    )
+    no_caches = run(None, True)
+    np.testing.assert_allclose(base, journal)
+    np.testing.assert_allclose(base, no_caches)
