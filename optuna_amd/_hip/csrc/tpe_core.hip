@@ -551,6 +551,126 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     return out;
 }
 
+// ---------------------------------------------------------------------------
+// K6: non-domination rank (NSGA-II / MO-TPE). The O(N²M) dominance test is the
+// heavy part: pack it into a bitmatrix (bit j of row i ⇔ j dominates i), then
+// peel Pareto fronts with two tiny kernels per round against a shrinking
+// "unranked" bitmask. The value matrix (N×M fp64 ≤ a few MB) stays L2-resident
+// across the whole N²M sweep.
+// ---------------------------------------------------------------------------
+
+__global__ void k_dominance_words(const double* __restrict__ vals,  // (N, M)
+                                  int64_t N, int64_t M, int64_t W,
+                                  uint64_t* __restrict__ words) {  // (N, W)
+    const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t i = idx / W;
+    const int64_t w = idx % W;
+    if (i >= N) return;
+    const double* vi = vals + i * M;
+    uint64_t bits = 0;
+    const int64_t j0 = w * 64;
+    const int64_t j1 = j0 + 64 < N ? j0 + 64 : N;
+    for (int64_t j = j0; j < j1; ++j) {
+        if (j == i) continue;
+        const double* vj = vals + j * M;
+        bool leq = true, lt = false;
+        for (int64_t m = 0; m < M; ++m) {
+            const double a = vj[m], b = vi[m];
+            if (a > b) { leq = false; break; }
+            if (a < b) lt = true;
+        }
+        if (leq && lt) bits |= (uint64_t)1 << (j - j0);
+    }
+    words[i * W + w] = bits;
+}
+
+__global__ void k_front_detect(const uint64_t* __restrict__ words, int64_t N,
+                               int64_t W, const uint64_t* __restrict__ mask,
+                               int32_t* __restrict__ front) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= N) return;
+    if (!((mask[i / 64] >> (i % 64)) & 1)) { front[i] = 0; return; }
+    const uint64_t* row = words + i * W;
+    uint64_t any = 0;
+    for (int64_t w = 0; w < W; ++w) any |= row[w] & mask[w];
+    front[i] = any == 0 ? 1 : 0;
+}
+
+__global__ void k_front_apply(const int32_t* __restrict__ front, int64_t N,
+                              int32_t rank, int64_t* __restrict__ ranks,
+                              uint64_t* __restrict__ mask,
+                              int32_t* __restrict__ n_front) {
+    const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= N || !front[i]) return;
+    ranks[i] = rank;
+    atomicAnd(
+        reinterpret_cast<unsigned long long*>(mask + i / 64),
+        ~((unsigned long long)1 << (i % 64)));
+    atomicAdd(n_front, 1);
+}
+
+// Rank the (pre-filtered, all-finite) rows; early-stop once n_below rows are
+// exactly ranked. Unranked rows return -1 (host lumps them into a bottom
+// rank), mirroring _calculate_nondomination_rank in study/_multi_objective.py.
+py::array_t<int64_t> nondomination_rank(const arr_f64& vals, int64_t n_below) {
+    if (vals.ndim() != 2) throw std::runtime_error("vals must be 2-D");
+    const int64_t N = vals.shape(0);
+    const int64_t M = vals.shape(1);
+    const int64_t W = (N + 63) / 64;
+    py::array_t<int64_t> out(N);
+    if (N == 0) return out;
+    hipStream_t st = g_ws.get_stream();
+
+    const size_t f64_vals = (size_t)N * M;
+    // u64 words (N*W) + mask (W) + ranks (N i64) + front (N i32) + counter.
+    const size_t u64_slots = (size_t)N * W + W + N + (N + 1) / 2 + 2;
+    double* base = g_ws.ensure(f64_vals + u64_slots + 16);
+    double* d_vals = base;
+    uint64_t* d_words = reinterpret_cast<uint64_t*>(d_vals + f64_vals);
+    uint64_t* d_mask = d_words + (size_t)N * W;
+    int64_t* d_ranks = reinterpret_cast<int64_t*>(d_mask + W);
+    int32_t* d_front = reinterpret_cast<int32_t*>(d_ranks + N);
+    int32_t* d_nfront = d_front + N;
+
+    g_ws.begin_uploads();
+    g_ws.h2d(d_vals, vals.data(), f64_vals * 8, st);
+    std::vector<uint64_t> mask_init(W, 0);
+    for (int64_t i = 0; i < N; ++i) mask_init[i / 64] |= (uint64_t)1 << (i % 64);
+    g_ws.h2d(d_mask, mask_init.data(), W * 8, st);
+    std::vector<int64_t> ranks_init(N, -1);
+    g_ws.h2d(d_ranks, ranks_init.data(), N * 8, st);
+
+    {
+        const int64_t total = N * W;
+        const int block = 256;
+        hipLaunchKernelGGL(k_dominance_words,
+                           dim3((unsigned)((total + block - 1) / block)),
+                           dim3(block), 0, st, d_vals, N, M, W, d_words);
+    }
+    const int block = 256;
+    const unsigned gridN = (unsigned)((N + block - 1) / block);
+    int64_t n_assigned = 0;
+    int32_t rank = -1;
+    int32_t h_nfront = 0;
+    while (n_assigned < n_below) {
+        ++rank;
+        HIP_CHECK(hipMemsetAsync(d_nfront, 0, 4, st));
+        hipLaunchKernelGGL(k_front_detect, dim3(gridN), dim3(block), 0, st,
+                           d_words, N, W, d_mask, d_front);
+        hipLaunchKernelGGL(k_front_apply, dim3(gridN), dim3(block), 0, st,
+                           d_front, N, rank, d_ranks, d_mask, d_nfront);
+        HIP_CHECK(hipMemcpyAsync(&h_nfront, d_nfront, 4, hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        if (h_nfront == 0) break;
+        n_assigned += h_nfront;
+    }
+    HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_ranks, (size_t)N * 8,
+                             hipMemcpyDeviceToHost, st));
+    HIP_CHECK(hipStreamSynchronize(st));
+    HIP_CHECK(hipGetLastError());
+    return out;
+}
+
 bool available() { return device_available(); }
 
 int device_count() {
@@ -838,6 +958,8 @@ PYBIND11_MODULE(_hipcore, m) {
     m.def("log_gauss_mass", &log_gauss_mass);
     m.def("truncnorm_ppf", &truncnorm_ppf);
     m.def("truncnorm_logpdf", &truncnorm_logpdf);
+    m.def("nondomination_rank", &nondomination_rank, py::arg("vals"),
+          py::arg("n_below"));
     m.def("kde_logpdf", &kde_logpdf, py::arg("obs"), py::arg("sorted_pos"),
           py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("x"),
           py::arg("consider_endpoints") = false, py::arg("magic_clip") = true);

@@ -285,3 +285,33 @@ def test_gp_device_fit_matches_cpu() -> None:
     np.testing.assert_allclose(
         gpr_dev.noise_var.item(), gpr_cpu.noise_var.item(), rtol=1e-3
     )
+
+
+@pytest.mark.parametrize("m,n", [(2, 5000), (3, 6000), (4, 4096)])
+def test_nondomination_rank_device_matches_host(core, m, n) -> None:
+    """K6 bitmatrix + device peel must reproduce the host front-peeling ranks."""
+    from optuna_amd.study import _multi_objective as mo
+
+    rng = np.random.RandomState(m)
+    vals = rng.randn(n, m)
+    # Inject duplicates and ties to exercise the non-strict dominance edge.
+    vals[100:140] = vals[0]
+    vals[200:220, 0] = vals[5, 0]
+
+    dev = mo._nondomination_rank_device(vals, n)
+    assert dev is not None
+
+    old = mo._DEVICE_RANK_MIN_ROWS
+    mo._DEVICE_RANK_MIN_ROWS = 10**9
+    try:
+        host, _ = mo._calculate_nondomination_rank(vals)
+    finally:
+        mo._DEVICE_RANK_MIN_ROWS = old
+    np.testing.assert_array_equal(dev, host)
+
+    # Early-stop contract: top-k ranks exact, the rest lumped at -1.
+    k = 500
+    dev_k = mo._nondomination_rank_device(vals, k)
+    ranked = dev_k >= 0
+    assert ranked.sum() >= k
+    np.testing.assert_array_equal(dev_k[ranked], host[ranked])
