@@ -97,9 +97,21 @@ def _mean_max_log_utility(log_utils: "torch.Tensor") -> "torch.Tensor":
 
 
 class BaseAcquisitionFunc(ABC):
+    # Device the underlying GP tensors live on (None = CPU). The numpy entry
+    # points below are the only host/device boundary: candidates go up, scalars
+    # and gradients come back.
+    _device: "torch.device | None" = None
+
     def __init__(self, length_scales: np.ndarray, search_space: "SearchSpace") -> None:
         self.length_scales = length_scales
         self.search_space = search_space
+
+    def set_device(self, device: "torch.device | None") -> None:
+        self._device = device
+
+    def _up(self, x: np.ndarray) -> "torch.Tensor":
+        t = torch.from_numpy(x)
+        return t.to(self._device) if self._device is not None else t
 
     @abstractmethod
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
@@ -107,14 +119,24 @@ class BaseAcquisitionFunc(ABC):
 
     def eval_acqf_no_grad(self, x: np.ndarray) -> np.ndarray:
         with torch.no_grad():
-            return self.eval_acqf(torch.from_numpy(x)).detach().numpy()
+            return self.eval_acqf(self._up(x)).detach().cpu().numpy()
 
     def eval_acqf_with_grad(self, x: np.ndarray) -> tuple[float, np.ndarray]:
         assert x.ndim == 1
-        x_tensor = torch.from_numpy(x).requires_grad_(True)
+        x_tensor = self._up(x).requires_grad_(True)
         val = self.eval_acqf(x_tensor)
         val.backward()
-        return val.item(), x_tensor.grad.detach().numpy()  # type: ignore[union-attr]
+        return val.item(), x_tensor.grad.detach().cpu().numpy()  # type: ignore[union-attr]
+
+    def eval_acqf_batched_with_grad(self, x: np.ndarray) -> tuple[np.ndarray, np.ndarray]:
+        """(fvals, grads) for a (B, D) batch — one graph, one transfer each way."""
+        x_tensor = self._up(x).requires_grad_(True)
+        fvals = self.eval_acqf(x_tensor)
+        fvals.sum().backward()
+        return (
+            np.atleast_1d(fvals.detach().cpu().numpy()),
+            x_tensor.grad.detach().cpu().numpy(),  # type: ignore[union-attr]
+        )
 
 
 class LogEI(BaseAcquisitionFunc):
@@ -132,7 +154,7 @@ class LogEI(BaseAcquisitionFunc):
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         if np.isneginf(self._threshold):
-            return torch.zeros(x.shape[:-1], dtype=torch.float64)
+            return torch.zeros(x.shape[:-1], dtype=torch.float64, device=x.device)
         mean, var = self._gpr.posterior(x)
         return logei(mean=mean, var=var + self._stabilizing_noise, f0=self._threshold)
 

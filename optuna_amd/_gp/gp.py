@@ -130,6 +130,30 @@ class GPRegressor:
     def length_scales(self) -> np.ndarray:
         return 1.0 / np.sqrt(self.inverse_squared_lengthscales.detach().cpu().numpy())
 
+    def to(self, device: "torch.device") -> "GPRegressor":
+        """Move every tensor (incl. the cached Cholesky) to `device` in place.
+
+        Used by the sampler to run acquisition evaluations on the MI355X once
+        the history is large enough that posterior GEMMs dominate.
+        """
+        for name in (
+            "_is_categorical",
+            "_X_train",
+            "_y_train",
+            "_X_all",
+            "_y_all",
+            "_squared_X_diff",
+            "_cov_Y_Y_chol",
+            "_cov_Y_Y_inv_Y",
+            "inverse_squared_lengthscales",
+            "kernel_scale",
+            "noise_var",
+        ):
+            t = getattr(self, name)
+            if t is not None:
+                setattr(self, name, t.to(device))
+        return self
+
     def kernel(
         self, X1: "torch.Tensor | None" = None, X2: "torch.Tensor | None" = None
     ) -> "torch.Tensor":

@@ -46,14 +46,8 @@ def _gradient_ascent_batched(
     ) -> tuple[np.ndarray, np.ndarray]:
         next_params = np.array(fixed_params)
         next_params[:, continuous_indices] = scaled_x * lengthscales
-        x_tensor = torch.from_numpy(next_params).requires_grad_(True)
-        neg_fvals = -acqf.eval_acqf(x_tensor)
-        neg_fvals.sum().backward()
-        grads = x_tensor.grad.detach().numpy()  # type: ignore[union-attr]
-        return (
-            np.atleast_1d(neg_fvals.detach().numpy()),
-            grads[:, continuous_indices] * lengthscales,
-        )
+        fvals, grads = acqf.eval_acqf_batched_with_grad(next_params)
+        return (-fvals, -grads[:, continuous_indices] * lengthscales)
 
     with limit_threads_in_optimization():
         scaled_opt, neg_fvals_opt, n_iterations = batched_lbfgsb.batched_lbfgsb(

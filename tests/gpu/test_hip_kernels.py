@@ -315,3 +315,42 @@ def test_nondomination_rank_device_matches_host(core, m, n) -> None:
     ranked = dev_k >= 0
     assert ranked.sum() >= k
     np.testing.assert_array_equal(dev_k[ranked], host[ranked])
+
+
+def test_gp_sampler_device_acqf_end_to_end() -> None:
+    """GPSampler with a big history: GP + acqf evaluate on the MI355X and the
+    suggestions still optimize the objective."""
+    import torch
+    import warnings
+
+    import optuna_amd
+    from optuna_amd._gp import gp as gp_mod
+
+    assert torch.cuda.is_available()
+    warnings.simplefilter("ignore")
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(7)
+    names = [f"x{i}" for i in range(6)]
+    dists = {n: FloatDistribution(-3.0, 3.0) for n in names}
+    sampler = optuna_amd.samplers.GPSampler(seed=0, n_startup_trials=5)
+    study = optuna_amd.create_study(sampler=sampler)
+    study.add_trials(
+        [
+            optuna_amd.create_trial(
+                params={n: float(rng.uniform(-3, 3)) for n in names},
+                distributions=dists,
+                value=float(
+                    sum((rng.uniform(-3, 3) - 0.5) ** 2 for _ in names)
+                ),
+            )
+            for _ in range(600)
+        ]
+    )
+
+    def objective(trial):
+        return sum((trial.suggest_float(n, -3, 3) - 0.5) ** 2 for n in names)
+
+    study.optimize(objective, n_trials=2)
+    # The cached GP must be device-resident after the device-path sample.
+    cached = sampler._gprs_cache_list[0]
+    assert cached._X_train.device.type == "cuda"
