@@ -44,14 +44,19 @@ class RetryFailedTrialCallback:
     def __call__(self, study: "Study", trial: FrozenTrial) -> None:
         from optuna_amd.trial import create_trial
 
-        system_attrs = dict(trial.system_attrs)
-        retry_history = list(system_attrs.get("failed_trial_numbers", []))
-        retry_history.append(trial.number)
-        if self._max_retry is not None and len(retry_history) > self._max_retry:
-            return
-        system_attrs["failed_trial_numbers"] = retry_history
-        system_attrs["retry_of_trial_number"] = retry_history[0]
-        system_attrs["fixed_params"] = trial.params
+        # Attr names match the reference exactly (storages/_callbacks.py:72-77):
+        # "failed_trial" keeps the ORIGINAL failed number (existing attrs win in
+        # the merge), "retry_history" accumulates every retried number.
+        system_attrs: dict = {
+            "failed_trial": trial.number,
+            "retry_history": [],
+            **trial.system_attrs,
+        }
+        system_attrs["retry_history"] = list(system_attrs["retry_history"])
+        system_attrs["retry_history"].append(trial.number)
+        if self._max_retry is not None:
+            if self._max_retry < len(system_attrs["retry_history"]):
+                return
 
         study.add_trial(
             create_trial(
@@ -68,8 +73,10 @@ class RetryFailedTrialCallback:
 
     @staticmethod
     def retried_trial_number(trial: FrozenTrial) -> int | None:
-        return trial.system_attrs.get("retry_of_trial_number")
+        """Number of the first failed trial in the retry series (or None)."""
+        return trial.system_attrs.get("failed_trial", None)
 
     @staticmethod
     def retry_history(trial: FrozenTrial) -> list[int]:
-        return list(trial.system_attrs.get("failed_trial_numbers", []))
+        """Retried trial numbers in series order, oldest first."""
+        return trial.system_attrs.get("retry_history", [])

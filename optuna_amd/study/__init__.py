@@ -1,3 +1,4 @@
+from optuna_amd._callbacks import MaxTrialsCallback
 from optuna_amd.study._study_direction import StudyDirection
 from optuna_amd.study._study_summary import StudySummary
 from optuna_amd.study.study import (
@@ -13,6 +14,7 @@ from optuna_amd.study.study import (
 
 
 __all__ = [
+    "MaxTrialsCallback",
     "ObjectiveFuncType",
     "Study",
     "StudyDirection",
