@@ -374,6 +374,30 @@ class TPESampler(BaseSampler):
                 self._parzen_estimator_parameters.consider_magic_clip,
             )
             acq_func_vals = mpe_below.log_pdf(samples_below) - log_g
+        elif (
+            self._constant_liar
+            and self._parzen_estimator_cls is _ParzenEstimator
+            and obs_above
+            and _device.space_is_device_eligible(search_space)
+            and _device.device_ready(len(next(iter(obs_above.values()))) + 1)
+        ):
+            # Constant-liar path (multi-worker): liar rows from RUNNING trials
+            # extend the "above" set, so the resident table's sorted orders
+            # don't apply — score through the stateless device KDE instead
+            # (per-call observation upload; still ~20x the host estimator at
+            # 10k history, which is what every rank runs during weak scaling).
+            n_combined = len(next(iter(obs_above.values())))
+            weights_above = self._above_weights(n_combined)
+            log_g = _device.kde_logpdf(
+                search_space,
+                obs_above,
+                orders_above,
+                weights_above,
+                samples_below,
+                self._parzen_estimator_parameters.consider_endpoints,
+                self._parzen_estimator_parameters.consider_magic_clip,
+            )
+            acq_func_vals = mpe_below.log_pdf(samples_below) - log_g
         else:
             mpe_above = self._build_mpe(
                 study, search_space, obs_above, handle_below=False, orders=orders_above

@@ -354,3 +354,49 @@ def test_gp_sampler_device_acqf_end_to_end() -> None:
     # The cached GP must be device-resident after the device-path sample.
     cached = sampler._gprs_cache_list[0]
     assert cached._X_train.device.type == "cuda"
+
+
+def test_constant_liar_uses_stateless_device_path(core, monkeypatch) -> None:
+    """Multi-worker (constant_liar) TPE must score on device too — that's the
+    path every rank runs in the driver's weak-scaling bench."""
+    import warnings
+
+    from optuna_amd.samplers._tpe import _device as device_mod
+
+    calls = {"kde": 0}
+    orig = device_mod.kde_logpdf
+
+    def spy(*args, **kwargs):
+        calls["kde"] += 1
+        return orig(*args, **kwargs)
+
+    monkeypatch.setattr(device_mod, "kde_logpdf", spy)
+    warnings.simplefilter("ignore")
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(0)
+    dists = {f"x{i}": FloatDistribution(-5.0, 5.0) for i in range(8)}
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.TPESampler(
+            seed=1, n_startup_trials=5, constant_liar=True
+        )
+    )
+    study.add_trials(
+        [
+            optuna_amd.create_trial(
+                params={f"x{i}": float(rng.uniform(-5, 5)) for i in range(8)},
+                distributions=dists,
+                value=float(rng.rand()),
+            )
+            for _ in range(700)
+        ]
+    )
+    # A RUNNING trial from a "peer" so the liar path actually engages.
+    peer = study.ask()
+    for i in range(8):
+        peer.suggest_float(f"x{i}", -5, 5)
+
+    def objective(trial):
+        return sum(trial.suggest_float(f"x{i}", -5, 5) ** 2 for i in range(8))
+
+    study.optimize(objective, n_trials=3)
+    assert calls["kde"] >= 3
