@@ -61,24 +61,31 @@ class _TcpStoreLog:
 
     def append(self, records: list[dict[str, Any]]) -> None:
         n = len(records)
-        # Reserve a contiguous index range, then publish the records.
+        # Reserve a contiguous index range, then publish the records — one
+        # round trip for the whole batch when the store supports multi_set.
         end = self._store.add(_SEQ_KEY, n)
         start = end - n
-        for i, rec in enumerate(records):
-            self._store.set(
-                _REC_KEY.format(idx=start + i),
-                json.dumps(rec, separators=(",", ":")),
-            )
+        keys = [_REC_KEY.format(idx=start + i) for i in range(n)]
+        payloads = [json.dumps(rec, separators=(",", ":")) for rec in records]
+        if hasattr(self._store, "multi_set"):
+            self._store.multi_set(keys, payloads)
+        else:
+            for key, payload in zip(keys, payloads):
+                self._store.set(key, payload)
 
     def read_from(self, start: int) -> list[dict[str, Any]]:
         end = self._store.add(_SEQ_KEY, 0)
-        out = []
-        for idx in range(start, end):
-            # set() may lag the counter by a moment on another worker; get() blocks
-            # until the key appears (bounded by the store timeout).
-            payload = self._store.get(_REC_KEY.format(idx=idx))
-            out.append(json.loads(payload))
-        return out
+        if end <= start:
+            return []
+        keys = [_REC_KEY.format(idx=idx) for idx in range(start, end)]
+        # set() may lag the counter by a moment on another worker; get() blocks
+        # until the key appears (bounded by the store timeout). multi_get has
+        # the same wait-for-key semantics, one round trip for the batch.
+        if hasattr(self._store, "multi_get"):
+            payloads = self._store.multi_get(keys)
+        else:
+            payloads = [self._store.get(k) for k in keys]
+        return [json.loads(p) for p in payloads]
 
 
 class RcclStorage(BaseStorage):
@@ -89,6 +96,15 @@ class RcclStorage(BaseStorage):
         self._worker_id_prefix = (worker_label or str(uuid.uuid4())) + "-" + str(uuid.uuid4())[:8] + "-"
         self._thread_lock = threading.Lock()
         self._replay = _ReplayState(self._worker_id_prefix)
+        # Param writes on this worker's own RUNNING trial are buffered here and
+        # flushed as one log batch with the finishing state record: locally they
+        # are previewed into the replay state immediately, and nobody else needs
+        # them before the trial finishes (constant-liar shares params via system
+        # attrs, which are never buffered). Cuts the per-suggest sequencer round
+        # trips from ~20 to 0.
+        self._pending: list[dict[str, Any]] = []
+        self._sync_ttl = float(os.environ.get("OPTUNA_AMD_RCCL_SYNC_TTL", "0.002"))
+        self._last_sync_at = 0.0
         with self._thread_lock:
             self._sync()
 
@@ -123,13 +139,36 @@ class RcclStorage(BaseStorage):
     # ---- log plumbing ---------------------------------------------------------------
 
     def _append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
-        self._log.append(
-            [{"op_code": int(op), "worker_id": self._replay.worker_id, **fields}]
-        )
+        rec = {"op_code": int(op), "worker_id": self._replay.worker_id, **fields}
+        # Pending records must precede this one in the global order.
+        batch = self._pending + [rec]
+        self._pending = []
+        self._log.append(batch)
+        self._last_sync_at = 0.0  # own record pending: next sync must not skip
 
-    def _sync(self) -> None:
+    def _defer_append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
+        rec = {"op_code": int(op), "worker_id": self._replay.worker_id, **fields}
+        # Preview locally (no counter advance): the flushed record re-applies
+        # idempotently when it comes back through the log.
+        self._replay.apply_logs([rec], advance=False)
+        self._pending.append(rec)
+
+    def _sync(self, force: bool = False) -> None:
+        """Pull unseen records into the local replay state.
+
+        Reads tolerate slightly stale views of OTHER workers' records (standard
+        optimistic-concurrency storage semantics; our own writes are applied on
+        their forced post-append sync or previewed when buffered), so read-side
+        syncs are rate-limited to one sequencer round trip per TTL window.
+        """
+        import time as _time
+
+        now = _time.monotonic()
+        if not force and now - self._last_sync_at < self._sync_ttl:
+            return
         records = self._log.read_from(self._replay.log_number_read)
         self._replay.apply_logs(records)
+        self._last_sync_at = _time.monotonic()
 
     # ---- studies --------------------------------------------------------------------
 
@@ -206,7 +245,8 @@ class RcclStorage(BaseStorage):
 
     # ---- trials ---------------------------------------------------------------------
 
-    def create_new_trial(self, study_id: int, template_trial: FrozenTrial | None = None) -> int:
+    @staticmethod
+    def _create_trial_record(study_id: int, template_trial: FrozenTrial | None) -> dict[str, Any]:
         from datetime import timezone
 
         log: dict[str, Any] = {"study_id": study_id, "datetime_start": _utcnow_iso()}
@@ -238,10 +278,41 @@ class RcclStorage(BaseStorage):
             log["user_attrs"] = template_trial.user_attrs
             log["system_attrs"] = template_trial.system_attrs
             log["intermediate_values"] = template_trial.intermediate_values
+        return log
+
+    def create_new_trial(self, study_id: int, template_trial: FrozenTrial | None = None) -> int:
+        log = self._create_trial_record(study_id, template_trial)
         with self._thread_lock:
             self._append(JournalOperation.CREATE_TRIAL, log)
             self._sync()
             return self._replay.last_created_trial_id
+
+    def bulk_create_trials(
+        self, study_id: int, template_trials: Sequence[FrozenTrial]
+    ) -> list[int]:
+        """Create many trials in one sequencer batch (bulk study population).
+
+        Non-standard extension discovered via getattr (Study.add_trials); a
+        10k-trial populate costs ~1 round trip instead of 10k sequenced appends.
+        """
+        if not template_trials:
+            return []
+        records = [
+            {
+                "op_code": int(JournalOperation.CREATE_TRIAL),
+                "worker_id": self._replay.worker_id,
+                **self._create_trial_record(study_id, t),
+            }
+            for t in template_trials
+        ]
+        with self._thread_lock:
+            before = len(self._replay.my_created_trial_ids)
+            batch = self._pending + records
+            self._pending = []
+            self._log.append(batch)
+            self._last_sync_at = 0.0
+            self._sync()
+            return list(self._replay.my_created_trial_ids[before:])
 
     def set_trial_param(
         self,
@@ -250,16 +321,23 @@ class RcclStorage(BaseStorage):
         param_value_internal: float,
         distribution: BaseDistribution,
     ) -> None:
+        fields = {
+            "trial_id": trial_id,
+            "param_name": param_name,
+            "param_value_internal": param_value_internal,
+            "distribution": distribution_to_json(distribution),
+        }
         with self._thread_lock:
-            self._append(
-                JournalOperation.SET_TRIAL_PARAM,
-                {
-                    "trial_id": trial_id,
-                    "param_name": param_name,
-                    "param_value_internal": param_value_internal,
-                    "distribution": distribution_to_json(distribution),
-                },
-            )
+            existing = self._replay._trials.get(trial_id)
+            if (
+                existing is not None
+                and existing.state == TrialState.RUNNING
+                and trial_id == self._replay.owned_trial_id
+            ):
+                # Our own running trial: buffer — flushed with the finish record.
+                self._defer_append(JournalOperation.SET_TRIAL_PARAM, fields)
+                return
+            self._append(JournalOperation.SET_TRIAL_PARAM, fields)
             self._sync()
 
     def get_trial_id_from_study_id_trial_number(self, study_id: int, trial_number: int) -> int:

@@ -382,6 +382,9 @@ class _ReplayState:
         self._next_study_id = 0
         self._worker_id_to_owned_trial_id: dict[str, int] = {}
         self.last_created_trial_id = -1
+        # Trial ids created by THIS worker, in creation order (bulk-create
+        # bookkeeping for RcclStorage.bulk_create_trials).
+        self.my_created_trial_ids: list[int] = []
 
     def __getstate__(self) -> dict[str, Any]:
         state = self.__dict__.copy()
@@ -391,6 +394,7 @@ class _ReplayState:
     def __setstate__(self, state: dict[str, Any]) -> None:
         self.__dict__.update(state)
         self.last_created_trial_id = -1
+        self.__dict__.setdefault("my_created_trial_ids", [])
         if "_study_id_to_finished" not in self.__dict__:
             # Snapshot from an older build: rebuild the finished log (trial-id
             # order; consumers dedupe by id, so order only affects row order).
@@ -441,7 +445,10 @@ class _ReplayState:
 
     # ---- replay ---------------------------------------------------------------------
 
-    def apply_logs(self, logs: Iterable[dict[str, Any]]) -> None:
+    def apply_logs(self, logs: Iterable[dict[str, Any]], advance: bool = True) -> None:
+        """Apply records; advance=False previews records not yet in the log
+        (RcclStorage write batching) — the log counter stays put so the same
+        records re-apply idempotently once they arrive through the log."""
         handlers = {
             int(JournalOperation.CREATE_STUDY): self._on_create_study,
             int(JournalOperation.DELETE_STUDY): self._on_delete_study,
@@ -455,7 +462,8 @@ class _ReplayState:
             int(JournalOperation.SET_TRIAL_SYSTEM_ATTR): self._on_trial_system_attr,
         }
         for log in logs:
-            self.log_number_read += 1
+            if advance:
+                self.log_number_read += 1
             op = log["op_code"]
             handler = handlers.get(op)
             if handler is None:
@@ -567,6 +575,7 @@ class _ReplayState:
             )
         if self._mine(log):
             self.last_created_trial_id = trial_id
+            self.my_created_trial_ids.append(trial_id)
             if self._trials[trial_id].state == TrialState.RUNNING:
                 self._worker_id_to_owned_trial_id[self.worker_id] = trial_id
 

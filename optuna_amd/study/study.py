@@ -314,6 +314,13 @@ class Study:
         self._storage.create_new_trial(self._study_id, template_trial=trial)
 
     def add_trials(self, trials: Iterable[FrozenTrial]) -> None:
+        trials = list(trials)
+        bulk = getattr(self._storage, "bulk_create_trials", None)
+        if bulk is not None and len(trials) > 1:
+            for trial in trials:
+                trial._validate()
+            bulk(self._study_id, trials)
+            return
         for trial in trials:
             self.add_trial(trial)
 
