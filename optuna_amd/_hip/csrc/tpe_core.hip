@@ -371,7 +371,25 @@ struct Workspace {
             HIP_CHECK(hipHostMalloc(&pinned, pinned_capacity));
         }
     }
-    void begin_uploads() { pinned_cursor = 0; }
+    // Event guard: an upload batch that ends without a stream sync (append's
+    // fast path) records this event; the next begin_uploads waits on it before
+    // reusing the staging buffer.
+    hipEvent_t uploads_done = nullptr;
+    bool uploads_pending = false;
+
+    void end_uploads_async(hipStream_t st) {
+        if (!uploads_done)
+            HIP_CHECK(hipEventCreateWithFlags(&uploads_done, hipEventDisableTiming));
+        HIP_CHECK(hipEventRecord(uploads_done, st));
+        uploads_pending = true;
+    }
+    void begin_uploads() {
+        if (uploads_pending) {
+            HIP_CHECK(hipEventSynchronize(uploads_done));
+            uploads_pending = false;
+        }
+        pinned_cursor = 0;
+    }
     // memcpy into a fresh pinned slice, then a true-async DMA on the stream.
     void h2d(void* dst, const void* src, size_t bytes, hipStream_t st) {
         if (bytes == 0) return;
@@ -855,7 +873,9 @@ class TpeDeviceHistory {
         g_ws.begin_uploads();
         g_ws.h2d(params_ + (size_t)n_ * D_, block.data(),
                  (size_t)n_new * D_ * sizeof(double), st);
-        HIP_CHECK(hipStreamSynchronize(st));
+        // No stream sync: the table write is stream-ordered before any later
+        // kernel on this stream; staging reuse is guarded by the upload event.
+        g_ws.end_uploads_async(st);
         n_ += n_new;
     }
 

@@ -50,8 +50,11 @@ class _SpaceCache:
         self.space = dict(space)
         self.names = list(space.keys())
         self.dists = [space[n] for n in self.names]
-        self.params = np.empty((0, len(self.names)), dtype=np.float64)
-        self.valid = np.empty(0, dtype=bool)
+        # Capacity-doubled row store: `params`/`valid` are views of the filled
+        # prefix, so a per-tell append is a row write, not an O(N) reallocation.
+        self._n_rows = 0
+        self._params_buf = np.empty((64, len(self.names)), dtype=np.float64)
+        self._valid_buf = np.empty(64, dtype=bool)
         # Per-dim sorted index kept in capacity-doubled buffers: the per-tell
         # insert is a searchsorted + tail shift, avoiding np.insert's full
         # reallocation per dimension per suggest.
@@ -60,6 +63,14 @@ class _SpaceCache:
         self._n_sorted = 0
         self._vals_buf = [np.empty(64, dtype=np.float64) for _ in self.names]
         self._rows_buf = [np.empty(64, dtype=np.int32) for _ in self.names]
+
+    @property
+    def params(self) -> np.ndarray:
+        return self._params_buf[: self._n_rows]
+
+    @property
+    def valid(self) -> np.ndarray:
+        return self._valid_buf[: self._n_rows]
 
     @property
     def sorted_vals(self) -> list[np.ndarray]:
@@ -73,9 +84,17 @@ class _SpaceCache:
         n_new = len(trials)
         if n_new == 0:
             return
-        base = len(self.valid)
-        block = np.empty((n_new, len(self.names)), dtype=np.float64)
-        valid = np.empty(n_new, dtype=bool)
+        base = self._n_rows
+        if base + n_new > len(self._valid_buf):
+            cap = max(2 * len(self._valid_buf), base + n_new)
+            pb = np.empty((cap, len(self.names)), dtype=np.float64)
+            vb_ = np.empty(cap, dtype=bool)
+            pb[:base] = self._params_buf[:base]
+            vb_[:base] = self._valid_buf[:base]
+            self._params_buf = pb
+            self._valid_buf = vb_
+        block = self._params_buf[base : base + n_new]
+        valid = self._valid_buf[base : base + n_new]
         for r, t in enumerate(trials):
             params = t.params
             ok = True
@@ -86,8 +105,7 @@ class _SpaceCache:
                     ok = False
                     break
             valid[r] = ok
-        self.params = np.concatenate([self.params, block], axis=0)
-        self.valid = np.concatenate([self.valid, valid], axis=0)
+        self._n_rows = base + n_new
 
         new_rows = base + np.nonzero(valid)[0]
         m = len(new_rows)
@@ -130,13 +148,18 @@ class _TpeHistory:
     """Append-only mirror of one study's finished trials (see module docstring)."""
 
     def __init__(self) -> None:
-        self._numbers = np.empty(0, dtype=np.int64)
-        self._states = np.empty(0, dtype=np.int8)
-        self._values: np.ndarray | None = None  # (N, M)
-        self._violations = np.empty(0, dtype=np.float64)
-        self._pruned_step = np.empty(0, dtype=np.float64)
-        self._pruned_value = np.empty(0, dtype=np.float64)  # unsigned; sign at split
-        self._has_intermediate = np.empty(0, dtype=bool)
+        # Master SoA columns live in capacity-doubled buffers; the public
+        # `_numbers` etc. are views of the filled prefix (appends are row
+        # writes, not O(N) reallocations per tell).
+        self._m_n = 0
+        self._m_cap = 64
+        self._numbers_buf = np.empty(64, dtype=np.int64)
+        self._states_buf = np.empty(64, dtype=np.int8)
+        self._values_buf: np.ndarray | None = None  # (cap, M)
+        self._violations_buf = np.empty(64, dtype=np.float64)
+        self._pruned_step_buf = np.empty(64, dtype=np.float64)
+        self._pruned_value_buf = np.empty(64, dtype=np.float64)
+        self._has_intermediate_buf = np.empty(64, dtype=bool)
         self._seen: set[int] = set()
         self._trials: list[FrozenTrial] = []  # row-aligned frozen trials
         self._spaces: dict[tuple, _SpaceCache] = {}
@@ -151,7 +174,35 @@ class _TpeHistory:
         self._comp_cache_valid = True
 
     def __len__(self) -> int:
-        return len(self._numbers)
+        return self._m_n
+
+    @property
+    def _numbers(self) -> np.ndarray:
+        return self._numbers_buf[: self._m_n]
+
+    @property
+    def _states(self) -> np.ndarray:
+        return self._states_buf[: self._m_n]
+
+    @property
+    def _values(self) -> np.ndarray | None:
+        return None if self._values_buf is None else self._values_buf[: self._m_n]
+
+    @property
+    def _violations(self) -> np.ndarray:
+        return self._violations_buf[: self._m_n]
+
+    @property
+    def _pruned_step(self) -> np.ndarray:
+        return self._pruned_step_buf[: self._m_n]
+
+    @property
+    def _pruned_value(self) -> np.ndarray:
+        return self._pruned_value_buf[: self._m_n]
+
+    @property
+    def _has_intermediate(self) -> np.ndarray:
+        return self._has_intermediate_buf[: self._m_n]
 
     @property
     def trials(self) -> list[FrozenTrial]:
@@ -209,20 +260,37 @@ class _TpeHistory:
             self._seen.add(t._trial_id)
 
         if self._rows_number_ascending:
-            prev_last = self._numbers[-1] if self._numbers.size else -1
+            prev_last = self._numbers_buf[self._m_n - 1] if self._m_n else -1
             if numbers[0] <= prev_last or (n_new > 1 and np.any(np.diff(numbers) <= 0)):
                 self._rows_number_ascending = False
 
-        self._numbers = np.concatenate([self._numbers, numbers])
-        self._states = np.concatenate([self._states, states])
-        if self._values is None:
-            self._values = values
-        else:
-            self._values = np.concatenate([self._values, values], axis=0)
-        self._violations = np.concatenate([self._violations, violations])
-        self._pruned_step = np.concatenate([self._pruned_step, p_step])
-        self._pruned_value = np.concatenate([self._pruned_value, p_value])
-        self._has_intermediate = np.concatenate([self._has_intermediate, has_iv])
+        base_m = self._m_n
+        if self._values_buf is None:
+            self._values_buf = np.empty((self._m_cap, n_objectives), dtype=np.float64)
+        if base_m + n_new > self._m_cap:
+            cap = max(2 * self._m_cap, base_m + n_new)
+            for attr in (
+                "_numbers_buf",
+                "_states_buf",
+                "_violations_buf",
+                "_pruned_step_buf",
+                "_pruned_value_buf",
+                "_has_intermediate_buf",
+                "_values_buf",
+            ):
+                old = getattr(self, attr)
+                grown = np.empty((cap,) + old.shape[1:], dtype=old.dtype)
+                grown[:base_m] = old[:base_m]
+                setattr(self, attr, grown)
+            self._m_cap = cap
+        self._numbers_buf[base_m : base_m + n_new] = numbers
+        self._states_buf[base_m : base_m + n_new] = states
+        self._values_buf[base_m : base_m + n_new] = values
+        self._violations_buf[base_m : base_m + n_new] = violations
+        self._pruned_step_buf[base_m : base_m + n_new] = p_step
+        self._pruned_value_buf[base_m : base_m + n_new] = p_value
+        self._has_intermediate_buf[base_m : base_m + n_new] = has_iv
+        self._m_n = base_m + n_new
         self._trials.extend(new)
         for cache in self._spaces.values():
             cache.append(new)
@@ -231,7 +299,7 @@ class _TpeHistory:
             if n_objectives != 1:
                 self._comp_cache_valid = False
             else:
-                base = len(self._numbers) - n_new
+                base = self._m_n - n_new
                 fc = np.nonzero(
                     (states == int(TrialState.COMPLETE)) & (violations <= 0)
                 )[0]
