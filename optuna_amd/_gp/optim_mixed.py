@@ -35,8 +35,6 @@ def _gradient_ascent_batched(
     tol: float,
 ) -> tuple[np.ndarray, np.ndarray, np.ndarray]:
     """L-BFGS-B on z = x/ℓ (lengthscale preconditioning equalizes curvature)."""
-    import torch
-
     assert initial_params_batched.ndim == 2
     if len(continuous_indices) == 0:
         return initial_params_batched, initial_fvals, np.zeros(len(initial_fvals), dtype=bool)

@@ -15,9 +15,7 @@ with the reference's ask/tell protocol: raw samples stored per trial under the
 """
 from __future__ import annotations
 
-import math
 import pickle
-import warnings
 from typing import TYPE_CHECKING, Any, Sequence
 
 import numpy as np

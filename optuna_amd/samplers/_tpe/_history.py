@@ -21,7 +21,6 @@ what makes the append-only mirror sound.
 """
 from __future__ import annotations
 
-import math
 from typing import TYPE_CHECKING, Sequence
 
 import numpy as np

@@ -8,7 +8,6 @@ populations (host numpy below typical pop sizes).
 """
 from __future__ import annotations
 
-import warnings
 from typing import TYPE_CHECKING, Any, Callable, Sequence
 
 from optuna_amd.distributions import BaseDistribution

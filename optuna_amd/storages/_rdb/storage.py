@@ -10,7 +10,6 @@ schemas is not supported in this build and raises with a clear message).
 from __future__ import annotations
 
 import json
-import threading
 import time
 from contextlib import contextmanager
 from datetime import datetime, timedelta, timezone

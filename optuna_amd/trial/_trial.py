@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import copy
 import datetime
-import math
 import warnings
 from typing import TYPE_CHECKING, Any, Sequence
 
