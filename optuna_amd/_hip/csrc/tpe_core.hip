@@ -120,11 +120,12 @@ __global__ void k_parzen_fit(const double* __restrict__ obs,
         const double mass =
             tn::log_gauss_mass((low - mu) / sigma, (high - mu) / sigma);
         const int64_t k = (r == N) ? N : sorted_pos[r * D + d];
-        c1[k * D + d] = -0.5 * inv_var;
-        c2[k * D + d] = mu * inv_var;
-        c3[k * D + d] = -0.5 * mu * mu * inv_var - log(sigma) -
+        // k-major (D, K) layout: lane index runs along k, so the scoring
+        // kernel's loads (and these writes) are coalesced.
+        c1[d * K + k] = -0.5 * inv_var;
+        c2[d * K + k] = mu * inv_var;
+        c3[d * K + k] = -0.5 * mu * mu * inv_var - log(sigma) -
                         0.9189385332046727418 - mass;
-        (void)K;
     }
 }
 
@@ -158,11 +159,9 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
     double m = -INFINITY, acc = 0.0;
     for (int64_t k = threadIdx.x; k < K; k += blockDim.x) {
         double t = logw[k];
-        const double* c1k = c1 + k * D;
-        const double* c2k = c2 + k * D;
-        const double* c3k = c3 + k * D;
         for (int64_t d = 0; d < D; ++d) {
-            t += x2[d] * c1k[d] + xs[d] * c2k[d] + c3k[d];
+            // k-major (D, K): lanes of a wavefront read consecutive k.
+            t += x2[d] * c1[d * K + k] + xs[d] * c2[d * K + k] + c3[d * K + k];
         }
         if (t > m) {
             acc = acc * exp(m - t) + 1.0;
@@ -231,11 +230,9 @@ __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
     double m = -INFINITY, acc = 0.0;
     for (int64_t k = k_lo + threadIdx.x; k < k_hi; k += blockDim.x) {
         double t = logw[k];
-        const double* c1k = c1 + k * D;
-        const double* c2k = c2 + k * D;
-        const double* c3k = c3 + k * D;
         for (int64_t d = 0; d < D; ++d) {
-            t += x2[d] * c1k[d] + xs[d] * c2k[d] + c3k[d];
+            // k-major (D, K): lanes of a wavefront read consecutive k.
+            t += x2[d] * c1[d * K + k] + xs[d] * c2[d * K + k] + c3[d * K + k];
         }
         if (t > m) {
             acc = acc * exp(m - t) + 1.0;
@@ -831,9 +828,9 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
         const double inv_var = 1.0 / (sigma * sigma);
         const double mass =
             tn::log_gauss_mass((low - mu) / sigma, (high - mu) / sigma);
-        c1[k * D + d] = -0.5 * inv_var;
-        c2[k * D + d] = mu * inv_var;
-        c3[k * D + d] = -0.5 * mu * mu * inv_var - log(sigma) -
+        c1[d * (Na + 1) + k] = -0.5 * inv_var;
+        c2[d * (Na + 1) + k] = mu * inv_var;
+        c3[d * (Na + 1) + k] = -0.5 * mu * mu * inv_var - log(sigma) -
                         0.9189385332046727418 - mass;
     }
 }
