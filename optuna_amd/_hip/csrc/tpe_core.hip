@@ -779,7 +779,66 @@ __global__ void k_compact_write(const int32_t* __restrict__ sorted_rows,
     }
 }
 
+// Row ids < n_table index the resident table; ids >= n_table index the small
+// per-suggest "extras" buffer (constant-liar rows from RUNNING trials).
+__device__ inline double row_val(const double* __restrict__ params,
+                                 const double* __restrict__ extras,
+                                 int64_t n_table, int64_t row, int64_t D,
+                                 int64_t d) {
+    return row < n_table ? params[row * D + d] : extras[(row - n_table) * D + d];
+}
+
+// Merge the per-dim sorted liar rows into the compacted subset on device.
+// extras_sorted (D, L): per-dim ascending values; extras_sorted_idx (D, L):
+// original extra index. Host-stable tie rule: extras sort AFTER equal finished
+// values (they sit at the end of the combined observation array), and among
+// themselves in original order.
+__global__ void k_merge_extras(const int32_t* __restrict__ sub_rows,  // (D, stride)
+                               const int32_t* __restrict__ sub_k,
+                               int64_t stride, int64_t Na,
+                               const double* __restrict__ params,
+                               int64_t n_table, int64_t D,
+                               const double* __restrict__ extras_sorted,
+                               const int32_t* __restrict__ extras_sorted_idx,
+                               int64_t L,
+                               int32_t* __restrict__ out_rows,  // (D, out_stride)
+                               int32_t* __restrict__ out_k,
+                               int64_t out_stride) {
+    const int64_t d = blockIdx.y;
+    const double* ev = extras_sorted + d * L;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < Na + L;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        if (i < Na) {
+            const double v = params[(int64_t)sub_rows[d * stride + i] * D + d];
+            // #extras strictly below v (ties go after finished).
+            int64_t lo = 0, hi = L;
+            while (lo < hi) {
+                const int64_t mid = (lo + hi) / 2;
+                if (ev[mid] < v) lo = mid + 1; else hi = mid;
+            }
+            out_rows[d * out_stride + i + lo] = sub_rows[d * stride + i];
+            out_k[d * out_stride + i + lo] = sub_k[d * stride + i];
+        } else {
+            const int64_t j = i - Na;  // j-th smallest extra in this dim
+            const double v = ev[j];
+            // #finished with value <= v (upper bound).
+            int64_t lo = 0, hi = Na;
+            while (lo < hi) {
+                const int64_t mid = (lo + hi) / 2;
+                const double fv =
+                    params[(int64_t)sub_rows[d * stride + mid] * D + d];
+                if (fv <= v) lo = mid + 1; else hi = mid;
+            }
+            const int32_t orig = extras_sorted_idx[d * L + j];
+            out_rows[d * out_stride + lo + j] = (int32_t)n_table + orig;
+            out_k[d * out_stride + lo + j] = (int32_t)Na + orig;
+        }
+    }
+}
+
 __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_rows, D)
+                                   const double* __restrict__ extras,  // (L, D) or null
+                                   int64_t n_table,
                                    const int32_t* __restrict__ sub_rows,  // (D, stride)
                                    const int32_t* __restrict__ sub_k,
                                    int64_t stride,  // row stride of sub_* (= Nv)
@@ -809,17 +868,21 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
             sigma = range;
             k = Na;
         } else {
-            mu = params[(int64_t)rows_d[r] * D + d];
+            mu = row_val(params, extras, n_table, rows_d[r], D, d);
             const double v_prev =
-                (r == 0) ? low : params[(int64_t)rows_d[r - 1] * D + d];
+                (r == 0) ? low
+                         : row_val(params, extras, n_table, rows_d[r - 1], D, d);
             const double v_next =
-                (r == Na - 1) ? high : params[(int64_t)rows_d[r + 1] * D + d];
+                (r == Na - 1)
+                    ? high
+                    : row_val(params, extras, n_table, rows_d[r + 1], D, d);
             sigma = fmax(mu - v_prev, v_next - mu);
             if (!consider_endpoints && Na >= 2) {
                 if (r == 0) {
-                    sigma = params[(int64_t)rows_d[1] * D + d] - mu;
+                    sigma = row_val(params, extras, n_table, rows_d[1], D, d) - mu;
                 } else if (r == Na - 1) {
-                    sigma = mu - params[(int64_t)rows_d[Na - 2] * D + d];
+                    sigma =
+                        mu - row_val(params, extras, n_table, rows_d[Na - 2], D, d);
                 }
             }
             sigma = fmin(fmax(sigma, minsigma), range);
@@ -879,19 +942,29 @@ class TpeDeviceHistory {
     py::array_t<double> score(const std::vector<arr_i32>& sorted_cols,  // D × (Nv,)
                               const arr_i32& pos,  // (n_rows,)
                               int64_t n_above,
-                              const arr_f64& logw,  // (n_above + 1,)
+                              const arr_f64& logw,  // (n_above + L + 1,)
                               const arr_f64& alow, const arr_f64& ahigh,
                               const arr_f64& x,  // (S, D) KDE domain
-                              bool consider_endpoints, bool magic_clip) {
+                              bool consider_endpoints, bool magic_clip,
+                              const arr_f64& extras_raw,         // (L, D) or empty
+                              const arr_f64& extras_sorted,      // (D, L)
+                              const arr_i32& extras_sorted_idx)  // (D, L)
+    {
         const int64_t Nv =
             sorted_cols.empty() ? 0 : (int64_t)sorted_cols[0].size();
         const int64_t S = x.shape(0);
         const int64_t Na = n_above;
-        const int64_t K = Na + 1;
+        const int64_t L = extras_raw.ndim() == 2 ? extras_raw.shape(0) : 0;
+        const int64_t Nk = Na + L;  // mixture kernels excluding the prior
+        const int64_t K = Nk + 1;
         if ((int64_t)pos.size() != n_ || x.shape(1) != D_ ||
             (Nv > 0 && (int64_t)sorted_cols.size() != D_) ||
             (int64_t)logw.size() != K)
             throw std::runtime_error("score: shape mismatch");
+        if (L > 0 && ((int64_t)extras_sorted.size() != L * D_ ||
+                      (int64_t)extras_sorted_idx.size() != L * D_ ||
+                      extras_raw.shape(1) != D_))
+            throw std::runtime_error("score: extras shape mismatch");
         for (const auto& col : sorted_cols)
             if ((int64_t)col.size() != Nv)
                 throw std::runtime_error("score: ragged sorted columns");
@@ -899,11 +972,16 @@ class TpeDeviceHistory {
         hipStream_t st = g_ws.get_stream();
         const size_t n_c = (size_t)K * D_;
         const int64_t n_tiles = (Nv + 255) / 256;
-        // f64 slots: c1|c2|c3|logw|alow|ahigh|x|out|lse scratch + i32 after.
+        const int64_t mstride = Nv + L;  // merged subset stride
+        // f64 slots: c1|c2|c3|logw|alow|ahigh|x|out|lse scratch|extras + i32.
         const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
-        size_t f64_total = 3 * n_c + K + 2 * D_ + (size_t)S * D_ + S + n_scratch;
+        const size_t n_extras_f64 = 2 * (size_t)L * D_;  // raw + per-dim sorted
+        size_t f64_total =
+            3 * n_c + K + 2 * D_ + (size_t)S * D_ + S + n_scratch + n_extras_f64;
         size_t i32_doubles = ((size_t)Nv * D_ /*sorted*/ + n_ /*pos*/ +
                               2 * (size_t)Nv * D_ /*sub*/ +
+                              2 * (size_t)mstride * D_ /*merged*/ +
+                              (size_t)L * D_ /*extra idx*/ +
                               (size_t)n_tiles * D_ /*tile counts*/) /
                                  2 +
                              8;
@@ -917,11 +995,17 @@ class TpeDeviceHistory {
         double* d_x = d_ahigh + D_;
         double* d_out = d_x + (size_t)S * D_;
         double* d_scratch = d_out + S;
-        int32_t* d_sorted = reinterpret_cast<int32_t*>(d_scratch + n_scratch);
+        double* d_extras_raw = d_scratch + n_scratch;
+        double* d_extras_sorted = d_extras_raw + (size_t)L * D_;
+        int32_t* d_sorted =
+            reinterpret_cast<int32_t*>(d_extras_sorted + (size_t)L * D_);
         int32_t* d_pos = d_sorted + (size_t)Nv * D_;
         int32_t* d_sub_rows = d_pos + n_;
         int32_t* d_sub_k = d_sub_rows + (size_t)Nv * D_;
-        int32_t* d_counts = d_sub_k + (size_t)Nv * D_;
+        int32_t* d_merged_rows = d_sub_k + (size_t)Nv * D_;
+        int32_t* d_merged_k = d_merged_rows + (size_t)mstride * D_;
+        int32_t* d_extra_idx = d_merged_k + (size_t)mstride * D_;
+        int32_t* d_counts = d_extra_idx + (size_t)L * D_;
 
         g_ws.begin_uploads();
         for (int64_t d = 0; d < (int64_t)sorted_cols.size() && Nv > 0; ++d)
@@ -931,6 +1015,11 @@ class TpeDeviceHistory {
         g_ws.h2d(d_alow, alow.data(), D_ * 8, st);
         g_ws.h2d(d_ahigh, ahigh.data(), D_ * 8, st);
         g_ws.h2d(d_x, x.data(), (size_t)S * D_ * 8, st);
+        if (L > 0) {
+            g_ws.h2d(d_extras_raw, extras_raw.data(), (size_t)L * D_ * 8, st);
+            g_ws.h2d(d_extras_sorted, extras_sorted.data(), (size_t)L * D_ * 8, st);
+            g_ws.h2d(d_extra_idx, extras_sorted_idx.data(), (size_t)L * D_ * 4, st);
+        }
 
         if (Na > 0) {
             const dim3 grid((unsigned)n_tiles, (unsigned)D_);
@@ -941,12 +1030,27 @@ class TpeDeviceHistory {
             hipLaunchKernelGGL(k_compact_write, grid, dim3(256), 0, st, d_sorted,
                                d_pos, Nv, D_, d_counts, d_sub_rows, d_sub_k);
         }
+        const int32_t* fit_rows = d_sub_rows;
+        const int32_t* fit_k = d_sub_k;
+        int64_t fit_stride = Nv;
+        if (L > 0) {
+            const int block = 256;
+            const int gx = (int)((Na + L + block - 1) / block);
+            hipLaunchKernelGGL(k_merge_extras, dim3(gx, (unsigned)D_), dim3(block),
+                               0, st, d_sub_rows, d_sub_k, Nv, Na, params_, n_,
+                               D_, d_extras_sorted, d_extra_idx, L,
+                               d_merged_rows, d_merged_k, mstride);
+            fit_rows = d_merged_rows;
+            fit_k = d_merged_k;
+            fit_stride = mstride;
+        }
         {
             const int block = 256;
             const int gx = (int)((K + block - 1) / block);
             hipLaunchKernelGGL(k_parzen_fit_table, dim3(gx, (unsigned)D_),
-                               dim3(block), 0, st, params_, d_sub_rows, d_sub_k,
-                               Nv, d_alow, d_ahigh, Na, D_,
+                               dim3(block), 0, st, params_,
+                               L > 0 ? d_extras_raw : nullptr, n_, fit_rows,
+                               fit_k, fit_stride, d_alow, d_ahigh, Nk, D_,
                                consider_endpoints ? 1 : 0, magic_clip ? 1 : 0,
                                d_c1, d_c2, d_c3);
         }
@@ -987,5 +1091,8 @@ PYBIND11_MODULE(_hipcore, m) {
         .def("score", &TpeDeviceHistory::score, py::arg("sorted_rows"),
              py::arg("pos"), py::arg("n_above"), py::arg("logw"), py::arg("alow"),
              py::arg("ahigh"), py::arg("x"), py::arg("consider_endpoints") = false,
-             py::arg("magic_clip") = true);
+             py::arg("magic_clip") = true,
+             py::arg("extras_raw") = arr_f64(),
+             py::arg("extras_sorted") = arr_f64(),
+             py::arg("extras_sorted_idx") = arr_i32());
 }

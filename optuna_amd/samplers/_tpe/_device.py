@@ -91,7 +91,15 @@ class _SpaceDeviceMirror:
         samples: dict[str, np.ndarray],
         consider_endpoints: bool,
         consider_magic_clip: bool,
+        extras: np.ndarray | None = None,
     ) -> np.ndarray:
+        """Score candidates against the resident table.
+
+        ``extras`` ((L, D), internal repr) are constant-liar rows from other
+        workers' RUNNING trials: they are appended to the mixture after the
+        ``sel`` kernels (matching the host estimator's observation order) and
+        merged into the per-dim sorted subsets on device.
+        """
         self.sync(cache)
         n_total = len(cache.valid)
         pos = np.full(n_total, -1, dtype=np.int32)
@@ -106,6 +114,21 @@ class _SpaceDeviceMirror:
             x[:, self._is_log] = np.log(x[:, self._is_log])
         with np.errstate(divide="ignore"):
             logw = np.log(weights)
+        kwargs = {}
+        if extras is not None and len(extras):
+            E = np.array(extras, dtype=np.float64)
+            if self._is_log.any():
+                E[:, self._is_log] = np.log(E[:, self._is_log])
+            order = np.argsort(E, axis=0, kind="stable")  # (L, D) per column
+            kwargs = dict(
+                extras_raw=np.ascontiguousarray(E),
+                extras_sorted=np.ascontiguousarray(
+                    np.take_along_axis(E, order, axis=0).T  # (D, L)
+                ),
+                extras_sorted_idx=np.ascontiguousarray(
+                    order.T.astype(np.int32)  # (D, L)
+                ),
+            )
         return self._hist.score(
             sorted_cols,
             pos,
@@ -116,6 +139,7 @@ class _SpaceDeviceMirror:
             np.ascontiguousarray(x),
             consider_endpoints,
             consider_magic_clip,
+            **kwargs,
         )
 
 
@@ -126,6 +150,7 @@ def score_above_resident(
     samples: dict[str, np.ndarray],
     consider_endpoints: bool,
     consider_magic_clip: bool,
+    extras: np.ndarray | None = None,
 ) -> np.ndarray:
     """log g(x) for candidates via the device-resident table (creates the mirror
     on first use; attached to the host space cache so lifetimes match)."""
@@ -134,7 +159,8 @@ def score_above_resident(
         mirror = _SpaceDeviceMirror(cache.space)
         cache._device_mirror = mirror  # type: ignore[attr-defined]
     return mirror.score(
-        cache, sel, weights, samples, consider_endpoints, consider_magic_clip
+        cache, sel, weights, samples, consider_endpoints, consider_magic_clip,
+        extras=extras,
     )
 
 
