@@ -120,6 +120,13 @@ class TPESampler(BaseSampler):
         self._parzen_estimator_cls = _ParzenEstimator
         # Per-study incremental history mirrors (see _history.py).
         self._histories: dict[int, Any] = {}
+        # n_jobs>1 runs suggests from worker threads; the history mirror is
+        # stateful (capacity buffers, incrementally sorted indices), so its
+        # read-modify-write cycle must be serialized. The reference's samplers
+        # are effectively GIL-serialized because they rebuild state per call.
+        import threading
+
+        self._history_lock = threading.RLock()
 
         if group:
             if multivariate is False:
@@ -276,6 +283,12 @@ class TPESampler(BaseSampler):
         return {k: np.asarray(v) for k, v in values.items()}
 
     def _sample(
+        self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
+    ) -> dict[str, Any]:
+        with self._history_lock:
+            return self._sample_locked(study, trial, search_space)
+
+    def _sample_locked(
         self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
     ) -> dict[str, Any]:
         from optuna_amd.samplers._tpe._history import _TpeHistory

@@ -48,3 +48,28 @@ def test_same_suggestions_across_storages_and_cache_paths(tmp_path) -> None:
     no_caches = run(None, True)
     np.testing.assert_allclose(base, journal)
     np.testing.assert_allclose(base, no_caches)
+
+
+def test_tpe_n_jobs_threaded_consistency() -> None:
+    """Concurrent suggests (n_jobs) must not corrupt the history mirror."""
+    import optuna_amd
+
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.TPESampler(seed=3, n_startup_trials=5)
+    )
+
+    def objective(trial):
+        return sum(trial.suggest_float(f"x{i}", -5, 5) ** 2 for i in range(4))
+
+    study.optimize(objective, n_trials=120, n_jobs=4)
+    assert len(study.trials) == 120
+    hist = study.sampler._histories[study._study_id]
+    # Mirror is consistent with storage.
+    assert len(hist) == len(study.get_trials(deepcopy=False))
+    import numpy as np
+
+    cache = next(iter(hist._spaces.values()), None)
+    if cache is not None:
+        for c in range(len(cache.names)):
+            sv = cache.sorted_vals[c]
+            assert np.all(np.diff(sv) >= 0)
