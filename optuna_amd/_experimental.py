@@ -75,3 +75,18 @@ def experimental_class(version: str, name: str | None = None) -> Callable[[CT], 
         return cls
 
     return decorator
+
+
+def warn_experimental_argument(option_name: str) -> None:
+    """Emit an ExperimentalWarning for an experimental keyword argument
+    (parity: reference optuna/_experimental.py warn_experimental_argument)."""
+    import warnings
+
+    from optuna_amd.exceptions import ExperimentalWarning
+
+    warnings.warn(
+        f"Argument ``{option_name}`` is an experimental feature."
+        " The interface can change in the future.",
+        ExperimentalWarning,
+        stacklevel=2,
+    )

@@ -85,6 +85,19 @@ class CmaEsSampler(BaseSampler):
         else:
             self._attr_prefix = "cma:"
 
+        from optuna_amd._experimental import warn_experimental_argument
+
+        if consider_pruned_trials:
+            warn_experimental_argument("consider_pruned_trials")
+        if use_separable_cma:
+            warn_experimental_argument("use_separable_cma")
+        if source_trials is not None:
+            warn_experimental_argument("source_trials")
+        if with_margin:
+            warn_experimental_argument("with_margin")
+        if lr_adapt:
+            warn_experimental_argument("lr_adapt")
+
         if source_trials is not None and (x0 is not None or sigma0 is not None):
             raise ValueError(
                 "It is prohibited to pass `source_trials` argument when x0 or sigma0 is "

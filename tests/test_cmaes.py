@@ -272,3 +272,16 @@ def test_sampler_with_margin_state_roundtrip() -> None:
     clone = pickle.loads(pickle.dumps(opt))
     np.testing.assert_array_equal(clone._A, opt._A)
     np.testing.assert_array_equal(clone.mean, opt.mean)
+
+
+@pytest.mark.parametrize("kwargs", [
+    {"with_margin": True},
+    {"lr_adapt": True},
+    {"use_separable_cma": True},
+    {"consider_pruned_trials": True},
+])
+def test_experimental_argument_warnings(kwargs) -> None:
+    from optuna_amd.exceptions import ExperimentalWarning
+
+    with pytest.warns(ExperimentalWarning):
+        optuna_amd.samplers.CmaEsSampler(**kwargs)
