@@ -119,6 +119,10 @@ class GPSampler(BaseSampler):
         self._minimum_noise: float = prior.DEFAULT_MINIMUM_NOISE_VAR
         self._gprs_cache_list: list[gp.GPRegressor] | None = None
         self._constraints_gprs_cache_list: list[gp.GPRegressor] | None = None
+        if deterministic_objective:
+            from optuna_amd._experimental import warn_experimental_argument
+
+            warn_experimental_argument("deterministic_objective")
         self._deterministic = deterministic_objective
         self._constraints_func = constraints_func
         self._warn_independent_sampling = warn_independent_sampling

@@ -110,6 +110,10 @@ class TPESampler(BaseSampler):
         self._rng = LazyRandomState(seed)
         self._random_sampler = RandomSampler(seed=seed)
         self._multivariate = multivariate
+        if group:
+            from optuna_amd._experimental import warn_experimental_argument
+
+            warn_experimental_argument("group")
         self._group = group
         self._group_decomposed_search_space: _GroupDecomposedSearchSpace | None = None
         self._search_space_group: _SearchSpaceGroup | None = None

@@ -82,6 +82,10 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
         skip_table_creation: bool = False,
     ) -> None:
         engine_kwargs = engine_kwargs or {}
+        if heartbeat_interval is not None:
+            from optuna_amd._experimental import warn_experimental_argument
+
+            warn_experimental_argument("heartbeat_interval")
         if heartbeat_interval is not None and heartbeat_interval <= 0:
             raise ValueError("The value of `heartbeat_interval` should be a positive integer.")
         if grace_period is not None and grace_period <= 0:
