@@ -61,14 +61,18 @@ def main() -> None:
     has_gpu = torch.cuda.is_available()
     if has_gpu:
         local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
 
     dist = None
     if world_size > 1:
         import torch.distributed as torch_dist
 
         dist = torch_dist
-        backend = "nccl" if has_gpu else "gloo"
+        # OPTUNA_AMD_BENCH_BACKEND=gloo lets N ranks share one GPU for testing
+        # (RCCL needs a GPU per rank).
+        backend = os.environ.get("OPTUNA_AMD_BENCH_BACKEND") or (
+            "nccl" if has_gpu else "gloo"
+        )
         dist.init_process_group(backend=backend)
 
     D = args.dims
