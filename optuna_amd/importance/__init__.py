@@ -41,6 +41,10 @@ def get_param_importances(
     Defaults to :class:`PedAnovaImportanceEvaluator`; the result is sorted
     descending and, with ``normalize=True``, sums to 1.
     """
+    if not normalize:
+        from optuna_amd._experimental import warn_experimental_argument
+
+        warn_experimental_argument("normalize")
     if evaluator is None:
         evaluator = PedAnovaImportanceEvaluator()
     if not isinstance(evaluator, BaseImportanceEvaluator):

@@ -70,6 +70,14 @@ class NSGAIISampler(BaseGASampler):
         self._rng = LazyRandomState(seed)
         self._constraints_func = constraints_func
         self._search_space = IntersectionSearchSpace()
+        from optuna_amd._experimental import warn_experimental_argument
+
+        if after_trial_strategy is not None:
+            warn_experimental_argument("after_trial_strategy")
+        if child_generation_strategy is not None:
+            warn_experimental_argument("child_generation_strategy")
+        if elite_population_selection_strategy is not None:
+            warn_experimental_argument("elite_population_selection_strategy")
         self._elite_population_selection_strategy = (
             elite_population_selection_strategy
             or NSGAIIElitePopulationSelectionStrategy(
