@@ -432,6 +432,35 @@ class RcclStorage(BaseStorage):
             self._sync()
             return self._replay.finished_trials_since(study_id, start)
 
+    def to_journal(self, backend: Any) -> None:
+        """Checkpoint the shared op log into a journal backend.
+
+        The records use the journal op format verbatim, so the resulting file
+        replays into the identical study state under ``JournalStorage`` (and
+        interoperates with the reference's journal format).
+        """
+        with self._thread_lock:
+            self._sync(force=True)
+            records = self._log.read_from(0)
+        backend.append_logs(records)
+
+    def load_journal(self, backend: Any) -> None:
+        """Resume: seed an EMPTY shared log from a journal backend's records.
+
+        Call on one rank before any study operation; other ranks replay the
+        seeded log on their next sync. RUNNING trials from the previous job
+        keep their old worker ids (no rank of this job owns them), matching
+        journal-restart semantics.
+        """
+        with self._thread_lock:
+            if self._log.read_from(0):
+                raise RuntimeError("load_journal requires an empty shared log.")
+            records = backend.read_logs(0)
+            if records:
+                self._log.append(records)
+            self._last_sync_at = 0.0
+            self._sync(force=True)
+
     def get_n_trials(
         self, study_id: int, state: tuple[TrialState, ...] | TrialState | None = None
     ) -> int:

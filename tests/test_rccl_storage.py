@@ -162,3 +162,39 @@ def test_waiting_claim_cas_across_processes() -> None:
         assert isinstance(won, bool), f"rank {rank} failed:\n{won}"
         wins.append(won)
     assert sum(wins) == 1
+
+
+def test_rccl_journal_checkpoint_roundtrip(tmp_path) -> None:
+    """RcclStorage <-> journal file: checkpoint and resume share the log format."""
+    import datetime
+
+    from torch.distributed import TCPStore
+
+    import optuna_amd
+    from optuna_amd.storages import JournalStorage
+    from optuna_amd.storages._rccl import RcclStorage
+    from optuna_amd.storages.journal import JournalFileBackend
+
+    store = TCPStore("127.0.0.1", 29645, 1, is_master=True,
+                     timeout=datetime.timedelta(seconds=30))
+    storage = RcclStorage(store)
+    study = optuna_amd.create_study(storage=storage, study_name="ckpt")
+    study.optimize(lambda t: t.suggest_float("x", -1, 1) ** 2, n_trials=8)
+
+    path = str(tmp_path / "ckpt.jsonl")
+    storage.to_journal(JournalFileBackend(path))
+
+    js = JournalStorage(JournalFileBackend(path))
+    resumed = optuna_amd.load_study(study_name="ckpt", storage=js)
+    assert len(resumed.trials) == 8
+    assert resumed.best_value == study.best_value
+
+    # Resume the journal into a fresh shared log.
+    store2 = TCPStore("127.0.0.1", 29646, 1, is_master=True,
+                      timeout=datetime.timedelta(seconds=30))
+    storage2 = RcclStorage(store2)
+    storage2.load_journal(JournalFileBackend(path))
+    study2 = optuna_amd.load_study(study_name="ckpt", storage=storage2)
+    assert len(study2.trials) == 8
+    study2.optimize(lambda t: t.suggest_float("x", -1, 1) ** 2, n_trials=4)
+    assert len(study2.trials) == 12
