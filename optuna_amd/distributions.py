@@ -73,9 +73,23 @@ def _adjust_discrete_high(low: float, high: float, step: float) -> float:
     d_high = decimal.Decimal(str(high))
     d_low = decimal.Decimal(str(low))
     d_step = decimal.Decimal(str(step))
-    d_r = (d_high - d_low) % d_step
-    if d_r != 0:
-        adjusted = float((d_high - d_r).quantize(decimal.Decimal(str(step))))
+    d_r = d_high - d_low
+    if d_r % d_step != 0:
+        # Largest grid point low + k*step ≤ high, in exact decimal arithmetic
+        # (reference distributions.py:661-675; a quantize to the step's decimal
+        # places would corrupt non-decimal lows). float() rounds the exact grid
+        # point, so iterate to a fixed point — otherwise a JSON round-trip of
+        # the stored (rounded) high can renormalize to a smaller grid point.
+        k = d_r // d_step
+        adjusted = float(k * d_step + d_low)
+        while (
+            k > 0
+            and (decimal.Decimal(str(adjusted)) - d_low) % d_step != 0
+        ):
+            k -= 1
+            adjusted = float(k * d_step + d_low)
+        if k == 0:
+            adjusted = float(low)
         warnings.warn(
             f"The distribution is specified by [{low}, {high}] and step={step}, but the range "
             f"is not divisible by `step`. It will be replaced by [{low}, {adjusted}]."

@@ -1,0 +1,100 @@
+"""Property-based checks (hypothesis) for the numeric substrate."""
+from __future__ import annotations
+
+import numpy as np
+from hypothesis import given, settings, strategies as st
+
+from optuna_amd._transform import _SearchSpaceTransform
+from optuna_amd.distributions import (
+    CategoricalDistribution,
+    FloatDistribution,
+    IntDistribution,
+    check_distribution_compatibility,
+    distribution_to_json,
+    json_to_distribution,
+)
+
+
+finite = st.floats(-1e6, 1e6, allow_nan=False, allow_infinity=False)
+
+
+@st.composite
+def float_dists(draw):
+    low = draw(st.floats(-1e5, 1e5, allow_nan=False))
+    width = draw(st.floats(1e-3, 1e5, allow_nan=False))
+    log = draw(st.booleans())
+    if log:
+        low = abs(low) + 1e-3
+    step = None if log or draw(st.booleans()) else draw(st.floats(1e-3, width))
+    return FloatDistribution(low, low + width, log=log, step=step)
+
+
+@st.composite
+def int_dists(draw):
+    low = draw(st.integers(-10000, 10000))
+    width = draw(st.integers(1, 10000))
+    log = draw(st.booleans())
+    if log:
+        low = abs(low) + 1
+    step = 1 if log else draw(st.integers(1, max(1, width)))
+    return IntDistribution(low, low + width, log=log, step=step)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.one_of(float_dists(), int_dists()))
+def test_distribution_json_roundtrip(dist) -> None:
+    clone = json_to_distribution(distribution_to_json(dist))
+    assert clone == dist
+    check_distribution_compatibility(dist, clone)
+
+
+@settings(max_examples=60, deadline=None)
+@given(float_dists(), st.floats(0, 1, allow_nan=False))
+def test_float_internal_repr_roundtrip(dist, q) -> None:
+    v = dist.low + (dist.high - dist.low) * q
+    if dist.step is not None:
+        v = dist.low + round((v - dist.low) / dist.step) * dist.step
+        v = min(v, dist.high)
+    internal = dist.to_internal_repr(v)
+    assert dist._contains(internal)
+    ext = dist.to_external_repr(internal)
+    np.testing.assert_allclose(ext, v, rtol=1e-12, atol=1e-12)
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    st.lists(st.one_of(float_dists(), int_dists()), min_size=1, max_size=6),
+    st.randoms(use_true_random=False),
+)
+def test_transform_untransform_inverse(dists, rnd) -> None:
+    space = {f"p{i}": d for i, d in enumerate(dists)}
+    trans = _SearchSpaceTransform(space)
+    params = {}
+    for name, d in space.items():
+        if isinstance(d, IntDistribution):
+            n_steps = (d.high - d.low) // (d.step or 1)
+            params[name] = d.low + (d.step or 1) * rnd.randint(0, max(n_steps, 0))
+        else:
+            if d.step is not None:
+                n_steps = int((d.high - d.low) / d.step)
+                params[name] = min(d.low + d.step * rnd.randint(0, max(n_steps, 0)), d.high)
+            else:
+                params[name] = rnd.uniform(d.low, d.high)
+    x = trans.transform(params)
+    back = trans.untransform(x)
+    for name, d in space.items():
+        if isinstance(d, IntDistribution):
+            assert back[name] == params[name]
+        else:
+            np.testing.assert_allclose(back[name], params[name], rtol=1e-9, atol=1e-9)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.sampled_from(["a", "b", "c", "d"]), min_size=1, max_size=4, unique=True))
+def test_categorical_repr(choices) -> None:
+    d = CategoricalDistribution(tuple(choices))
+    for i, c in enumerate(choices):
+        assert d.to_internal_repr(c) == i
+        assert d.to_external_repr(i) == c
+    clone = json_to_distribution(distribution_to_json(d))
+    assert clone == d
