@@ -59,7 +59,9 @@ class _TcpStoreLog:
     def __init__(self, store: Any) -> None:
         self._store = store
 
-    def append(self, records: list[dict[str, Any]]) -> None:
+    def append(self, records: list[dict[str, Any]]) -> int:
+        """Publish a batch; returns the end index of the reserved range so the
+        caller's read-back sync can skip the counter query."""
         n = len(records)
         # Reserve a contiguous index range, then publish the records — one
         # round trip for the whole batch when the store supports multi_set.
@@ -72,6 +74,17 @@ class _TcpStoreLog:
         else:
             for key, payload in zip(keys, payloads):
                 self._store.set(key, payload)
+        return end
+
+    def read_range(self, start: int, end: int) -> list[dict[str, Any]]:
+        if end <= start:
+            return []
+        keys = [_REC_KEY.format(idx=idx) for idx in range(start, end)]
+        if hasattr(self._store, "multi_get"):
+            payloads = self._store.multi_get(keys)
+        else:
+            payloads = [self._store.get(k) for k in keys]
+        return [json.loads(p) for p in payloads]
 
     def read_from(self, start: int) -> list[dict[str, Any]]:
         end = self._store.add(_SEQ_KEY, 0)
@@ -143,8 +156,12 @@ class RcclStorage(BaseStorage):
         # Pending records must precede this one in the global order.
         batch = self._pending + [rec]
         self._pending = []
-        self._log.append(batch)
-        self._last_sync_at = 0.0  # own record pending: next sync must not skip
+        end = self._log.append(batch)
+        # Apply everything up to our own record without another counter query;
+        # records appended concurrently after `end` arrive on a later sync.
+        self._replay.apply_logs(
+            self._log.read_range(self._replay.log_number_read, end)
+        )
 
     def _defer_append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
         rec = {"op_code": int(op), "worker_id": self._replay.worker_id, **fields}
