@@ -76,7 +76,8 @@ __global__ void k_logpdf(const double* x, const double* a, const double* b,
 __global__ void k_parzen_fit(const double* __restrict__ obs,
                              const int64_t* __restrict__ sorted_pos,
                              const double* __restrict__ alow,
-                             const double* __restrict__ ahigh, int64_t N,
+                             const double* __restrict__ ahigh,
+                             const double* __restrict__ steps, int64_t N,
                              int64_t D, int consider_endpoints, int magic_clip,
                              double* __restrict__ c1, double* __restrict__ c2,
                              double* __restrict__ c3) {
@@ -115,17 +116,25 @@ __global__ void k_parzen_fit(const double* __restrict__ obs,
             }
             sigma = fmin(fmax(sigma, minsigma), range);
         }
-        // Quadratic expansion with truncation-mass normalization folded in.
-        const double inv_var = 1.0 / (sigma * sigma);
+        // Truncation-mass normalization; for continuous dims fold everything
+        // into the quadratic expansion, for discrete (step>0) dims store
+        // (mu, sigma, -logZ) — the scoring kernel integrates the step cell.
         const double mass =
             tn::log_gauss_mass((low - mu) / sigma, (high - mu) / sigma);
         const int64_t k = (r == N) ? N : sorted_pos[r * D + d];
         // k-major (D, K) layout: lane index runs along k, so the scoring
         // kernel's loads (and these writes) are coalesced.
-        c1[d * K + k] = -0.5 * inv_var;
-        c2[d * K + k] = mu * inv_var;
-        c3[d * K + k] = -0.5 * mu * mu * inv_var - log(sigma) -
-                        0.9189385332046727418 - mass;
+        if (steps[d] > 0.0) {
+            c1[d * K + k] = mu;
+            c2[d * K + k] = sigma;
+            c3[d * K + k] = -mass;
+        } else {
+            const double inv_var = 1.0 / (sigma * sigma);
+            c1[d * K + k] = -0.5 * inv_var;
+            c2[d * K + k] = mu * inv_var;
+            c3[d * K + k] = -0.5 * mu * mu * inv_var - log(sigma) -
+                            0.9189385332046727418 - mass;
+        }
     }
 }
 
@@ -135,16 +144,41 @@ __global__ void k_parzen_fit(const double* __restrict__ obs,
 // logsumexp, merged through LDS at the end.
 // ---------------------------------------------------------------------------
 
+// Per-(kernel, dim) mixture term: continuous dims use the precomputed
+// quadratic form; discrete dims integrate the step cell [xl, xr] against the
+// stored (mu, sigma) with the cached -logZ in c3.
+__device__ inline double mix_term(const double* __restrict__ c1,
+                                  const double* __restrict__ c2,
+                                  const double* __restrict__ c3,
+                                  const double* __restrict__ steps,
+                                  const double* __restrict__ xs,
+                                  const double* __restrict__ x2,
+                                  const double* __restrict__ xl,
+                                  const double* __restrict__ xr, int64_t K,
+                                  int64_t k, int64_t d) {
+    if (steps[d] > 0.0) {
+        const double mu = c1[d * K + k];
+        const double sig = c2[d * K + k];
+        return tn::log_gauss_mass((xl[d] - mu) / sig, (xr[d] - mu) / sig) +
+               c3[d * K + k];
+    }
+    return x2[d] * c1[d * K + k] + xs[d] * c2[d * K + k] + c3[d * K + k];
+}
+
 __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
+                             const double* __restrict__ xedges,  // (S, 2D) lo|hi
                              const double* __restrict__ c1,
                              const double* __restrict__ c2,
                              const double* __restrict__ c3,
-                             const double* __restrict__ logw, int64_t K,
+                             const double* __restrict__ logw,
+                             const double* __restrict__ steps, int64_t K,
                              int64_t D, double* __restrict__ out) {
-    extern __shared__ double lds[];  // D xs + D x2 + 2*blockDim reduction
+    extern __shared__ double lds[];  // 4 D-vectors + 2*blockDim reduction
     double* xs = lds;
     double* x2 = lds + D;
-    double* red_m = lds + 2 * D;
+    double* xl = lds + 2 * D;
+    double* xr = lds + 3 * D;
+    double* red_m = lds + 4 * D;
     double* red_s = red_m + blockDim.x;
 
     const int64_t s = blockIdx.x;
@@ -152,6 +186,8 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
         const double v = x[s * D + d];
         xs[d] = v;
         x2[d] = v * v;
+        xl[d] = xedges[s * 2 * D + d];
+        xr[d] = xedges[s * 2 * D + D + d];
     }
     __syncthreads();
 
@@ -161,7 +197,7 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
         double t = logw[k];
         for (int64_t d = 0; d < D; ++d) {
             // k-major (D, K): lanes of a wavefront read consecutive k.
-            t += x2[d] * c1[d * K + k] + xs[d] * c2[d * K + k] + c3[d * K + k];
+            t += mix_term(c1, c2, c3, steps, xs, x2, xl, xr, K, k, d);
         }
         if (t > m) {
             acc = acc * exp(m - t) + 1.0;
@@ -202,17 +238,21 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
 // the XCD round-robin distributes every chunk's coefficient rows into all 8
 // L2s once and the co-resident sample-blocks reuse them.
 __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
+                                     const double* __restrict__ xedges,  // (S, 2D)
                                      const double* __restrict__ c1,
                                      const double* __restrict__ c2,
                                      const double* __restrict__ c3,
-                                     const double* __restrict__ logw, int64_t K,
+                                     const double* __restrict__ logw,
+                                     const double* __restrict__ steps, int64_t K,
                                      int64_t D, int64_t chunk,
                                      double* __restrict__ part_m,   // (n_chunks, S)
                                      double* __restrict__ part_s) { // (n_chunks, S)
     extern __shared__ double lds[];
     double* xs = lds;
     double* x2 = lds + D;
-    double* red_m = lds + 2 * D;
+    double* xl = lds + 2 * D;
+    double* xr = lds + 3 * D;
+    double* red_m = lds + 4 * D;
     double* red_s = red_m + blockDim.x;
 
     const int64_t S = gridDim.x;
@@ -224,6 +264,8 @@ __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
         const double v = x[s * D + d];
         xs[d] = v;
         x2[d] = v * v;
+        xl[d] = xedges[s * 2 * D + d];
+        xr[d] = xedges[s * 2 * D + D + d];
     }
     __syncthreads();
 
@@ -232,7 +274,7 @@ __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
         double t = logw[k];
         for (int64_t d = 0; d < D; ++d) {
             // k-major (D, K): lanes of a wavefront read consecutive k.
-            t += x2[d] * c1[d * K + k] + xs[d] * c2[d * K + k] + c3[d * K + k];
+            t += mix_term(c1, c2, c3, steps, xs, x2, xl, xr, K, k, d);
         }
         if (t > m) {
             acc = acc * exp(m - t) + 1.0;
@@ -317,23 +359,26 @@ static constexpr int64_t MIX_CHUNK = 512;
 static inline int64_t mix_n_chunks(int64_t K) { return (K + MIX_CHUNK - 1) / MIX_CHUNK; }
 
 // d_scratch must hold 2 * mix_n_chunks(K) * S doubles (may alias nothing else).
-static void launch_mix_logpdf(hipStream_t st, const double* d_x, const double* d_c1,
+static void launch_mix_logpdf(hipStream_t st, const double* d_x,
+                              const double* d_xedges, const double* d_c1,
                               const double* d_c2, const double* d_c3,
-                              const double* d_logw, int64_t K, int64_t D, int64_t S,
-                              double* d_out, double* d_scratch) {
+                              const double* d_logw, const double* d_steps,
+                              int64_t K, int64_t D, int64_t S, double* d_out,
+                              double* d_scratch) {
     const int block = 256;
-    const size_t shmem = (2 * (size_t)D + 2 * block) * sizeof(double);
+    const size_t shmem = (4 * (size_t)D + 2 * block) * sizeof(double);
     const int64_t n_chunks = mix_n_chunks(K);
     if (n_chunks <= 1) {
         hipLaunchKernelGGL(k_mix_logpdf, dim3((unsigned)S), dim3(block), shmem, st,
-                           d_x, d_c1, d_c2, d_c3, d_logw, K, D, d_out);
+                           d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps, K, D,
+                           d_out);
         return;
     }
     double* d_part_m = d_scratch;
     double* d_part_s = d_scratch + (size_t)n_chunks * S;
     hipLaunchKernelGGL(k_mix_logpdf_partial, dim3((unsigned)S, (unsigned)n_chunks),
-                       dim3(block), shmem, st, d_x, d_c1, d_c2, d_c3, d_logw, K, D,
-                       MIX_CHUNK, d_part_m, d_part_s);
+                       dim3(block), shmem, st, d_x, d_xedges, d_c1, d_c2, d_c3,
+                       d_logw, d_steps, K, D, MIX_CHUNK, d_part_m, d_part_s);
     hipLaunchKernelGGL(k_mix_logpdf_merge, dim3((unsigned)S), dim3(64), 0, st,
                        d_part_m, d_part_s, n_chunks, S, d_out);
 }
@@ -506,7 +551,8 @@ py::array_t<double> truncnorm_logpdf(const arr_f64& x, const arr_f64& a,
 // returns (S,) log mixture pdf.
 py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
                                const arr_f64& logw, const arr_f64& alow,
-                               const arr_f64& ahigh, const arr_f64& x,
+                               const arr_f64& ahigh, const arr_f64& steps,
+                               const arr_f64& x, const arr_f64& xedges,
                                bool consider_endpoints, bool magic_clip) {
     if (obs.ndim() != 2 || x.ndim() != 2) throw std::runtime_error("obs/x must be 2-D");
     const int64_t N = obs.shape(0);
@@ -515,6 +561,7 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     const int64_t K = N + 1;
     if (x.shape(1) != D || (int64_t)logw.size() != K ||
         (int64_t)alow.size() != D || (int64_t)ahigh.size() != D ||
+        (int64_t)steps.size() != D || (int64_t)xedges.size() != 2 * S * D ||
         sorted_pos.shape(0) != N || (N > 0 && sorted_pos.shape(1) != D))
         throw std::runtime_error("shape mismatch in kde_logpdf");
 
@@ -523,10 +570,11 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
 
     const size_t n_obs = (size_t)N * D;
     const size_t n_c = (size_t)K * D;
-    // layout: obs | c1 | c2 | c3 | logw | alow | ahigh | x | out | lse scratch
-    //         | sorted(i64 as f64 slots)
+    // layout: obs | c1 | c2 | c3 | logw | alow | ahigh | steps | x | xedges
+    //         | out | lse scratch | sorted(i64 as f64 slots)
     const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
-    size_t total = n_obs + 3 * n_c + K + 2 * D + (size_t)S * D + S + n_scratch + n_obs + 16;
+    size_t total = n_obs + 3 * n_c + K + 3 * D + 3 * (size_t)S * D + S + n_scratch +
+                   n_obs + 16;
     double* base = g_ws.ensure(total);
     double* d_obs = base;
     double* d_c1 = d_obs + n_obs;
@@ -535,8 +583,10 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     double* d_logw = d_c3 + n_c;
     double* d_alow = d_logw + K;
     double* d_ahigh = d_alow + D;
-    double* d_x = d_ahigh + D;
-    double* d_out = d_x + (size_t)S * D;
+    double* d_steps = d_ahigh + D;
+    double* d_x = d_steps + D;
+    double* d_xedges = d_x + (size_t)S * D;
+    double* d_out = d_xedges + 2 * (size_t)S * D;
     double* d_scratch = d_out + S;
     int64_t* d_sorted = reinterpret_cast<int64_t*>(d_scratch + n_scratch);
 
@@ -548,17 +598,20 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     g_ws.h2d(d_logw, logw.data(), K * 8, st);
     g_ws.h2d(d_alow, alow.data(), D * 8, st);
     g_ws.h2d(d_ahigh, ahigh.data(), D * 8, st);
+    g_ws.h2d(d_steps, steps.data(), D * 8, st);
     g_ws.h2d(d_x, x.data(), (size_t)S * D * 8, st);
+    g_ws.h2d(d_xedges, xedges.data(), 2 * (size_t)S * D * 8, st);
 
     {
         const int block = 256;
         const int gx = (int)((K + block - 1) / block);
         hipLaunchKernelGGL(k_parzen_fit, dim3(gx, (unsigned)D), dim3(block), 0, st,
-                           d_obs, d_sorted, d_alow, d_ahigh, N, D,
+                           d_obs, d_sorted, d_alow, d_ahigh, d_steps, N, D,
                            consider_endpoints ? 1 : 0, magic_clip ? 1 : 0, d_c1,
                            d_c2, d_c3);
     }
-    launch_mix_logpdf(st, d_x, d_c1, d_c2, d_c3, d_logw, K, D, S, d_out, d_scratch);
+    launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps, K, D, S,
+                      d_out, d_scratch);
     HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
                              hipMemcpyDeviceToHost, st));
     HIP_CHECK(hipStreamSynchronize(st));
@@ -843,7 +896,8 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
                                    const int32_t* __restrict__ sub_k,
                                    int64_t stride,  // row stride of sub_* (= Nv)
                                    const double* __restrict__ alow,
-                                   const double* __restrict__ ahigh, int64_t Na,
+                                   const double* __restrict__ ahigh,
+                                   const double* __restrict__ steps, int64_t Na,
                                    int64_t D, int consider_endpoints,
                                    int magic_clip, double* __restrict__ c1,
                                    double* __restrict__ c2,
@@ -888,13 +942,19 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
             sigma = fmin(fmax(sigma, minsigma), range);
             k = ks_d[r];
         }
-        const double inv_var = 1.0 / (sigma * sigma);
         const double mass =
             tn::log_gauss_mass((low - mu) / sigma, (high - mu) / sigma);
-        c1[d * (Na + 1) + k] = -0.5 * inv_var;
-        c2[d * (Na + 1) + k] = mu * inv_var;
-        c3[d * (Na + 1) + k] = -0.5 * mu * mu * inv_var - log(sigma) -
-                        0.9189385332046727418 - mass;
+        if (steps[d] > 0.0) {
+            c1[d * (Na + 1) + k] = mu;
+            c2[d * (Na + 1) + k] = sigma;
+            c3[d * (Na + 1) + k] = -mass;
+        } else {
+            const double inv_var = 1.0 / (sigma * sigma);
+            c1[d * (Na + 1) + k] = -0.5 * inv_var;
+            c2[d * (Na + 1) + k] = mu * inv_var;
+            c3[d * (Na + 1) + k] = -0.5 * mu * mu * inv_var - log(sigma) -
+                            0.9189385332046727418 - mass;
+        }
     }
 }
 
@@ -944,7 +1004,9 @@ class TpeDeviceHistory {
                               int64_t n_above,
                               const arr_f64& logw,  // (n_above + L + 1,)
                               const arr_f64& alow, const arr_f64& ahigh,
-                              const arr_f64& x,  // (S, D) KDE domain
+                              const arr_f64& steps,   // (D,) 0 = continuous
+                              const arr_f64& x,       // (S, D) KDE domain
+                              const arr_f64& xedges,  // (S, 2D) cell lo|hi
                               bool consider_endpoints, bool magic_clip,
                               const arr_f64& extras_raw,         // (L, D) or empty
                               const arr_f64& extras_sorted,      // (D, L)
@@ -973,11 +1035,11 @@ class TpeDeviceHistory {
         const size_t n_c = (size_t)K * D_;
         const int64_t n_tiles = (Nv + 255) / 256;
         const int64_t mstride = Nv + L;  // merged subset stride
-        // f64 slots: c1|c2|c3|logw|alow|ahigh|x|out|lse scratch|extras + i32.
+        // f64 slots: c1|c2|c3|logw|alow|ahigh|steps|x|xedges|out|lse|extras + i32.
         const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
         const size_t n_extras_f64 = 2 * (size_t)L * D_;  // raw + per-dim sorted
-        size_t f64_total =
-            3 * n_c + K + 2 * D_ + (size_t)S * D_ + S + n_scratch + n_extras_f64;
+        size_t f64_total = 3 * n_c + K + 3 * D_ + 3 * (size_t)S * D_ + S +
+                           n_scratch + n_extras_f64;
         size_t i32_doubles = ((size_t)Nv * D_ /*sorted*/ + n_ /*pos*/ +
                               2 * (size_t)Nv * D_ /*sub*/ +
                               2 * (size_t)mstride * D_ /*merged*/ +
@@ -992,8 +1054,10 @@ class TpeDeviceHistory {
         double* d_logw = d_c3 + n_c;
         double* d_alow = d_logw + K;
         double* d_ahigh = d_alow + D_;
-        double* d_x = d_ahigh + D_;
-        double* d_out = d_x + (size_t)S * D_;
+        double* d_steps = d_ahigh + D_;
+        double* d_x = d_steps + D_;
+        double* d_xedges = d_x + (size_t)S * D_;
+        double* d_out = d_xedges + 2 * (size_t)S * D_;
         double* d_scratch = d_out + S;
         double* d_extras_raw = d_scratch + n_scratch;
         double* d_extras_sorted = d_extras_raw + (size_t)L * D_;
@@ -1014,7 +1078,9 @@ class TpeDeviceHistory {
         g_ws.h2d(d_logw, logw.data(), K * 8, st);
         g_ws.h2d(d_alow, alow.data(), D_ * 8, st);
         g_ws.h2d(d_ahigh, ahigh.data(), D_ * 8, st);
+        g_ws.h2d(d_steps, steps.data(), D_ * 8, st);
         g_ws.h2d(d_x, x.data(), (size_t)S * D_ * 8, st);
+        g_ws.h2d(d_xedges, xedges.data(), 2 * (size_t)S * D_ * 8, st);
         if (L > 0) {
             g_ws.h2d(d_extras_raw, extras_raw.data(), (size_t)L * D_ * 8, st);
             g_ws.h2d(d_extras_sorted, extras_sorted.data(), (size_t)L * D_ * 8, st);
@@ -1050,12 +1116,12 @@ class TpeDeviceHistory {
             hipLaunchKernelGGL(k_parzen_fit_table, dim3(gx, (unsigned)D_),
                                dim3(block), 0, st, params_,
                                L > 0 ? d_extras_raw : nullptr, n_, fit_rows,
-                               fit_k, fit_stride, d_alow, d_ahigh, Nk, D_,
-                               consider_endpoints ? 1 : 0, magic_clip ? 1 : 0,
-                               d_c1, d_c2, d_c3);
+                               fit_k, fit_stride, d_alow, d_ahigh, d_steps, Nk,
+                               D_, consider_endpoints ? 1 : 0,
+                               magic_clip ? 1 : 0, d_c1, d_c2, d_c3);
         }
-        launch_mix_logpdf(st, d_x, d_c1, d_c2, d_c3, d_logw, K, D_, S, d_out,
-                          d_scratch);
+        launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps, K,
+                          D_, S, d_out, d_scratch);
         py::array_t<double> out(S);
         HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
                                  hipMemcpyDeviceToHost, st));
@@ -1082,7 +1148,8 @@ PYBIND11_MODULE(_hipcore, m) {
     m.def("nondomination_rank", &nondomination_rank, py::arg("vals"),
           py::arg("n_below"));
     m.def("kde_logpdf", &kde_logpdf, py::arg("obs"), py::arg("sorted_pos"),
-          py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("x"),
+          py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("steps"),
+          py::arg("x"), py::arg("xedges"),
           py::arg("consider_endpoints") = false, py::arg("magic_clip") = true);
     py::class_<TpeDeviceHistory>(m, "TpeDeviceHistory")
         .def(py::init<int64_t>(), py::arg("dims"))
@@ -1090,7 +1157,8 @@ PYBIND11_MODULE(_hipcore, m) {
         .def("append", &TpeDeviceHistory::append, py::arg("block"))
         .def("score", &TpeDeviceHistory::score, py::arg("sorted_rows"),
              py::arg("pos"), py::arg("n_above"), py::arg("logw"), py::arg("alow"),
-             py::arg("ahigh"), py::arg("x"), py::arg("consider_endpoints") = false,
+             py::arg("ahigh"), py::arg("steps"), py::arg("x"), py::arg("xedges"),
+             py::arg("consider_endpoints") = false,
              py::arg("magic_clip") = true,
              py::arg("extras_raw") = arr_f64(),
              py::arg("extras_sorted") = arr_f64(),

@@ -11,8 +11,9 @@ Two device modes:
 * **Stateless scoring** (`kde_logpdf`): used by golden tests and the
   constant-liar path (whose observation set includes rows not in the table).
 
-Eligibility: all dimensions continuous (optionally log-scaled). Discrete and
-categorical dimensions keep the (already vectorized) host path.
+Eligibility: any numerical dimensions — continuous, log, int, or
+step-discretized (the scoring kernel integrates the step cell). Categorical
+dimensions keep the (already vectorized) host path.
 """
 from __future__ import annotations
 
@@ -39,28 +40,67 @@ DEVICE_MIN_KERNELS = 512
 def space_is_device_eligible(space: dict[str, BaseDistribution]) -> bool:
     if not space:
         return False
-    for dist in space.values():
-        if not isinstance(dist, FloatDistribution):
-            return False
-        if dist.step is not None:
-            return False
-    return True
+    return all(
+        isinstance(dist, (FloatDistribution, IntDistribution))
+        for dist in space.values()
+    )
 
 
 def device_ready(n_kernels: int) -> bool:
     return n_kernels >= DEVICE_MIN_KERNELS and _hip.is_available()
 
 
-def _space_domains(space: dict[str, BaseDistribution]) -> tuple[np.ndarray, np.ndarray, np.ndarray]:
-    """(is_log, alow, ahigh) per dim, KDE domain."""
+def _space_domains(
+    space: dict[str, BaseDistribution],
+) -> tuple[np.ndarray, np.ndarray, np.ndarray, np.ndarray]:
+    """(is_log, alow, ahigh, steps) per dim, KDE domain.
+
+    Step-discretized dims widen the domain by half a step on each side before
+    the log transform — identical to the host estimator's adapted bounds
+    (parzen.py `_numerical_kernels_batched`).
+    """
     is_log = np.array([bool(d.log) for d in space.values()])
-    alow = np.array(
-        [math.log(d.low) if d.log else d.low for d in space.values()], dtype=np.float64
+    steps = np.array(
+        [float(d.step) if d.step is not None else 0.0 for d in space.values()],
+        dtype=np.float64,
     )
-    ahigh = np.array(
-        [math.log(d.high) if d.log else d.high for d in space.values()], dtype=np.float64
-    )
-    return is_log, alow, ahigh
+    alow = np.empty(len(steps))
+    ahigh = np.empty(len(steps))
+    for c, d in enumerate(space.values()):
+        lo = float(d.low) - steps[c] / 2 if steps[c] else float(d.low)
+        hi = float(d.high) + steps[c] / 2 if steps[c] else float(d.high)
+        if d.log:
+            lo, hi = math.log(lo), math.log(hi)
+        alow[c] = lo
+        ahigh[c] = hi
+    return is_log, alow, ahigh, steps
+
+
+def _cell_edges(
+    x_raw: np.ndarray, is_log: np.ndarray, steps: np.ndarray
+) -> np.ndarray:
+    """(S, 2D) per-candidate step-cell edges [lo | hi] in KDE domain.
+
+    Continuous dims get zeros (the kernel never reads them); discrete dims get
+    x ∓ step/2, logged for log-discrete dims — exactly the host integration
+    cell (parzen.py `_log_pdf_array`).
+    """
+    S, D = x_raw.shape
+    edges = np.zeros((S, 2 * D), dtype=np.float64)
+    disc = steps > 0
+    if disc.any():
+        half = steps[disc] / 2
+        lo = x_raw[:, disc] - half
+        hi = x_raw[:, disc] + half
+        log_disc = is_log[disc]
+        if log_disc.any():
+            with np.errstate(divide="ignore", invalid="ignore"):
+                lo[:, log_disc] = np.log(lo[:, log_disc])
+                hi[:, log_disc] = np.log(hi[:, log_disc])
+        idx = np.nonzero(disc)[0]
+        edges[:, idx] = lo
+        edges[:, D + idx] = hi
+    return edges
 
 
 class _SpaceDeviceMirror:
@@ -69,7 +109,7 @@ class _SpaceDeviceMirror:
     def __init__(self, space: dict[str, BaseDistribution]) -> None:
         core = _hip.get()
         assert core is not None
-        self._is_log, self._alow, self._ahigh = _space_domains(space)
+        self._is_log, self._alow, self._ahigh, self._steps = _space_domains(space)
         self._hist = core.TpeDeviceHistory(len(space))
         self._n_appended = 0
 
@@ -107,9 +147,11 @@ class _SpaceDeviceMirror:
         # Per-dim contiguous int32 prefixes go to the device as-is — no
         # (Nv, D) restack per suggest (was the top host cost on the GPU box).
         sorted_cols = list(cache.sorted_rows)
-        x = np.column_stack(
+        x_raw = np.column_stack(
             [np.asarray(samples[n], dtype=np.float64) for n in cache.names]
         )
+        xedges = _cell_edges(x_raw, self._is_log, self._steps)
+        x = x_raw.copy()
         if self._is_log.any():
             x[:, self._is_log] = np.log(x[:, self._is_log])
         with np.errstate(divide="ignore"):
@@ -136,7 +178,9 @@ class _SpaceDeviceMirror:
             logw,
             self._alow,
             self._ahigh,
+            self._steps,
             np.ascontiguousarray(x),
+            np.ascontiguousarray(xedges),
             consider_endpoints,
             consider_magic_clip,
             **kwargs,
@@ -180,28 +224,24 @@ def kde_logpdf(
     D = len(names)
     N = len(observations[names[0]])
 
+    is_log, alow, ahigh, steps = _space_domains(space)
     obs = np.empty((N, D), dtype=np.float64)
-    x = np.column_stack([np.asarray(samples[n], dtype=np.float64) for n in names])
-    alow = np.empty(D)
-    ahigh = np.empty(D)
+    x_raw = np.column_stack(
+        [np.asarray(samples[n], dtype=np.float64) for n in names]
+    )
     sorted_pos = np.empty((N, D), dtype=np.int64)
     for c, name in enumerate(names):
-        dist = space[name]
-        assert isinstance(dist, FloatDistribution) and dist.step is None
         col = np.asarray(observations[name], dtype=np.float64)
-        if dist.log:
-            obs[:, c] = np.log(col)
-            x[:, c] = np.log(x[:, c])
-            alow[c] = math.log(dist.low)
-            ahigh[c] = math.log(dist.high)
-        else:
-            obs[:, c] = col
-            alow[c] = dist.low
-            ahigh[c] = dist.high
+        obs[:, c] = np.log(col) if is_log[c] else col
         if orders is not None:
             sorted_pos[:, c] = orders[name]
         else:
             sorted_pos[:, c] = np.argsort(col, kind="stable")
+
+    xedges = _cell_edges(x_raw, is_log, steps)
+    x = x_raw.copy()
+    if is_log.any():
+        x[:, is_log] = np.log(x[:, is_log])
 
     with np.errstate(divide="ignore"):
         logw = np.log(weights)
@@ -211,7 +251,9 @@ def kde_logpdf(
         logw,
         alow,
         ahigh,
+        steps,
         x,
+        np.ascontiguousarray(xedges),
         consider_endpoints,
         consider_magic_clip,
     )
