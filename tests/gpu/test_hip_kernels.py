@@ -472,3 +472,96 @@ def test_resident_with_extras_matches_host(core) -> None:
         mpe = _ParzenEstimator(obs, space, params_obj)
         ref = mpe.log_pdf(samples)
         np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
+
+
+def test_kde_logpdf_discrete_dims_matches_host(core) -> None:
+    """Device scoring with int / step / log-int dims (cell-integral path) must
+    match the host estimator."""
+    from optuna_amd.distributions import IntDistribution
+    from optuna_amd.samplers._tpe import _device
+
+    rng = np.random.RandomState(23)
+    n_obs = 900
+    space = {
+        "a": FloatDistribution(-5.0, 5.0),
+        "b": IntDistribution(1, 40),
+        "c": FloatDistribution(0.0, 10.0, step=0.5),
+        "d": IntDistribution(1, 256, log=True),
+        "e": FloatDistribution(1e-3, 1e3, log=True),
+    }
+    observations = {
+        "a": rng.uniform(-5, 5, n_obs),
+        "b": rng.randint(1, 41, n_obs).astype(float),
+        "c": np.round(rng.uniform(0, 20, n_obs)) * 0.5,
+        "d": np.exp(rng.uniform(0, np.log(256), n_obs)).round().clip(1, 256),
+        "e": np.exp(rng.uniform(np.log(1e-3), np.log(1e3), n_obs)),
+    }
+    samples = {
+        "a": rng.uniform(-5, 5, 24),
+        "b": rng.randint(1, 41, 24).astype(float),
+        "c": np.round(rng.uniform(0, 20, 24)) * 0.5,
+        "d": np.exp(rng.uniform(0, np.log(256), 24)).round().clip(1, 256),
+        "e": np.exp(rng.uniform(np.log(1e-3), np.log(1e3), 24)),
+    }
+    w = default_weights(n_obs)
+    weights = np.append(w, [1.0])
+    weights = weights / weights.sum()
+
+    from optuna_amd.samplers._tpe import _device as dev
+
+    assert dev.space_is_device_eligible(space)
+    ours = dev.kde_logpdf(space, observations, None, weights, samples, False, True)
+
+    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, True)
+    mpe = _ParzenEstimator(observations, space, params)
+    ref = mpe.log_pdf(samples)
+    np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
+
+
+def test_resident_discrete_matches_host(core) -> None:
+    """Resident-table scoring over a mixed int/float space matches host."""
+    from optuna_amd.distributions import IntDistribution
+    from optuna_amd.samplers._tpe import _device
+    from optuna_amd.samplers._tpe._history import _SpaceCache
+    from optuna_amd.testing.trials import _create_frozen_trial
+
+    rng = np.random.RandomState(29)
+    space = {
+        "a": FloatDistribution(-5.0, 5.0),
+        "b": IntDistribution(0, 100),
+        "c": FloatDistribution(1e-2, 1e2, log=True),
+    }
+    trials = [
+        _create_frozen_trial(
+            number=j,
+            values=(float(rng.rand()),),
+            params={
+                "a": float(rng.uniform(-5, 5)),
+                "b": int(rng.randint(0, 101)),
+                "c": float(np.exp(rng.uniform(np.log(1e-2), np.log(1e2)))),
+            },
+            distributions=dict(space),
+        )
+        for j in range(800)
+    ]
+    cache = _SpaceCache(space)
+    cache.append(trials)
+    n_total = len(cache.valid)
+    sel = np.sort(rng.choice(n_total, size=n_total - 25, replace=False))
+    weights_raw = default_weights(len(sel))
+    weights = np.append(weights_raw, [1.0])
+    weights = weights / weights.sum()
+    samples = {
+        "a": rng.uniform(-5, 5, 24),
+        "b": rng.randint(0, 101, 24).astype(float),
+        "c": np.exp(rng.uniform(np.log(1e-2), np.log(1e2), 24)),
+    }
+    ours = _device.score_above_resident(cache, sel, weights, samples, False, True)
+
+    obs = {name: cache.params[sel][:, c] for c, name in enumerate(cache.names)}
+    params_obj = _ParzenEstimatorParameters(
+        True, 1.0, True, False, lambda n: weights_raw[:n], True
+    )
+    mpe = _ParzenEstimator(obs, space, params_obj)
+    ref = mpe.log_pdf(samples)
+    np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
