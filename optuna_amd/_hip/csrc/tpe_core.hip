@@ -151,11 +151,23 @@ __device__ inline double mix_term(const double* __restrict__ c1,
                                   const double* __restrict__ c2,
                                   const double* __restrict__ c3,
                                   const double* __restrict__ steps,
+                                  const double* __restrict__ n_choices,
+                                  double cat_base,  // prior_weight / K
                                   const double* __restrict__ xs,
                                   const double* __restrict__ x2,
                                   const double* __restrict__ xl,
                                   const double* __restrict__ xr, int64_t K,
                                   int64_t k, int64_t d) {
+    const double C = n_choices[d];
+    if (C > 0.0) {
+        // Categorical: prior-smoothed one-hot weights in closed form —
+        // row k of the (K, C) weight matrix is base everywhere plus 1 at the
+        // kernel's own category, row-normalized; the prior row is uniform.
+        const double v = c1[d * K + k];  // kernel's category (-1 ⇒ prior row)
+        if (v < 0.0) return -log(C);
+        const double hit = (xs[d] == v) ? 1.0 : 0.0;
+        return log(cat_base + hit) - log(C * cat_base + 1.0);
+    }
     if (steps[d] > 0.0) {
         const double mu = c1[d * K + k];
         const double sig = c2[d * K + k];
@@ -171,7 +183,9 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
                              const double* __restrict__ c2,
                              const double* __restrict__ c3,
                              const double* __restrict__ logw,
-                             const double* __restrict__ steps, int64_t K,
+                             const double* __restrict__ steps,
+                             const double* __restrict__ n_choices,
+                             double cat_base, int64_t K,
                              int64_t D, double* __restrict__ out) {
     extern __shared__ double lds[];  // 4 D-vectors + 2*blockDim reduction
     double* xs = lds;
@@ -197,7 +211,7 @@ __global__ void k_mix_logpdf(const double* __restrict__ x,  // (S, D)
         double t = logw[k];
         for (int64_t d = 0; d < D; ++d) {
             // k-major (D, K): lanes of a wavefront read consecutive k.
-            t += mix_term(c1, c2, c3, steps, xs, x2, xl, xr, K, k, d);
+            t += mix_term(c1, c2, c3, steps, n_choices, cat_base, xs, x2, xl, xr, K, k, d);
         }
         if (t > m) {
             acc = acc * exp(m - t) + 1.0;
@@ -243,7 +257,9 @@ __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
                                      const double* __restrict__ c2,
                                      const double* __restrict__ c3,
                                      const double* __restrict__ logw,
-                                     const double* __restrict__ steps, int64_t K,
+                                     const double* __restrict__ steps,
+                                     const double* __restrict__ n_choices,
+                                     double cat_base, int64_t K,
                                      int64_t D, int64_t chunk,
                                      double* __restrict__ part_m,   // (n_chunks, S)
                                      double* __restrict__ part_s) { // (n_chunks, S)
@@ -274,7 +290,7 @@ __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
         double t = logw[k];
         for (int64_t d = 0; d < D; ++d) {
             // k-major (D, K): lanes of a wavefront read consecutive k.
-            t += mix_term(c1, c2, c3, steps, xs, x2, xl, xr, K, k, d);
+            t += mix_term(c1, c2, c3, steps, n_choices, cat_base, xs, x2, xl, xr, K, k, d);
         }
         if (t > m) {
             acc = acc * exp(m - t) + 1.0;
@@ -363,6 +379,7 @@ static void launch_mix_logpdf(hipStream_t st, const double* d_x,
                               const double* d_xedges, const double* d_c1,
                               const double* d_c2, const double* d_c3,
                               const double* d_logw, const double* d_steps,
+                              const double* d_nchoices, double cat_base,
                               int64_t K, int64_t D, int64_t S, double* d_out,
                               double* d_scratch) {
     const int block = 256;
@@ -370,15 +387,16 @@ static void launch_mix_logpdf(hipStream_t st, const double* d_x,
     const int64_t n_chunks = mix_n_chunks(K);
     if (n_chunks <= 1) {
         hipLaunchKernelGGL(k_mix_logpdf, dim3((unsigned)S), dim3(block), shmem, st,
-                           d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps, K, D,
-                           d_out);
+                           d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps,
+                           d_nchoices, cat_base, K, D, d_out);
         return;
     }
     double* d_part_m = d_scratch;
     double* d_part_s = d_scratch + (size_t)n_chunks * S;
     hipLaunchKernelGGL(k_mix_logpdf_partial, dim3((unsigned)S, (unsigned)n_chunks),
                        dim3(block), shmem, st, d_x, d_xedges, d_c1, d_c2, d_c3,
-                       d_logw, d_steps, K, D, MIX_CHUNK, d_part_m, d_part_s);
+                       d_logw, d_steps, d_nchoices, cat_base, K, D, MIX_CHUNK,
+                       d_part_m, d_part_s);
     hipLaunchKernelGGL(k_mix_logpdf_merge, dim3((unsigned)S), dim3(64), 0, st,
                        d_part_m, d_part_s, n_chunks, S, d_out);
 }

@@ -323,7 +323,9 @@ class _ParzenEstimator:
             else:
                 order_mat = np.column_stack(
                     [
-                        o if o is not None else np.argsort(mus_mat[:, c])
+                        o
+                        if o is not None
+                        else np.argsort(mus_mat[:, c], kind="stable")
                         for c, o in enumerate(order_cols)
                     ]
                 )
@@ -376,7 +378,9 @@ class _ParzenEstimator:
             return np.array([0.5 * (low + high)]), np.array([high - low])
 
         if order is None:
-            order = np.argsort(mus)
+            # Stable: tied observations keep index order, so sigma assignment is
+            # deterministic and matches the incrementally-sorted device orders.
+            order = np.argsort(mus, kind="stable")
         padded = np.empty(n + 2, dtype=np.float64)
         padded[0] = low
         padded[1:-1] = mus[order]
