@@ -77,7 +77,8 @@ __global__ void k_parzen_fit(const double* __restrict__ obs,
                              const int64_t* __restrict__ sorted_pos,
                              const double* __restrict__ alow,
                              const double* __restrict__ ahigh,
-                             const double* __restrict__ steps, int64_t N,
+                             const double* __restrict__ steps,
+                             const double* __restrict__ n_choices, int64_t N,
                              int64_t D, int consider_endpoints, int magic_clip,
                              double* __restrict__ c1, double* __restrict__ c2,
                              double* __restrict__ c3) {
@@ -94,6 +95,17 @@ __global__ void k_parzen_fit(const double* __restrict__ obs,
 
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r <= N;
          r += (int64_t)gridDim.x * blockDim.x) {
+        if (n_choices[d] > 0.0) {
+            // Categorical: store the kernel's category (-1 ⇒ prior row); the
+            // scoring kernel evaluates the smoothed one-hot weight directly.
+            const int64_t kc =
+                (r == N) ? N : sorted_pos[r * D + d];
+            c1[d * K + kc] =
+                (r == N) ? -1.0 : obs[sorted_pos[r * D + d] * D + d];
+            c2[d * K + kc] = 0.0;
+            c3[d * K + kc] = 0.0;
+            continue;
+        }
         double mu, sigma;
         if (r == N) {
             // Prior kernel: midpoint, full-range width.
@@ -570,6 +582,7 @@ py::array_t<double> truncnorm_logpdf(const arr_f64& x, const arr_f64& a,
 py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
                                const arr_f64& logw, const arr_f64& alow,
                                const arr_f64& ahigh, const arr_f64& steps,
+                               const arr_f64& n_choices, double prior_weight,
                                const arr_f64& x, const arr_f64& xedges,
                                bool consider_endpoints, bool magic_clip) {
     if (obs.ndim() != 2 || x.ndim() != 2) throw std::runtime_error("obs/x must be 2-D");
@@ -591,7 +604,7 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     // layout: obs | c1 | c2 | c3 | logw | alow | ahigh | steps | x | xedges
     //         | out | lse scratch | sorted(i64 as f64 slots)
     const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
-    size_t total = n_obs + 3 * n_c + K + 3 * D + 3 * (size_t)S * D + S + n_scratch +
+    size_t total = n_obs + 3 * n_c + K + 4 * D + 3 * (size_t)S * D + S + n_scratch +
                    n_obs + 16;
     double* base = g_ws.ensure(total);
     double* d_obs = base;
@@ -602,7 +615,8 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     double* d_alow = d_logw + K;
     double* d_ahigh = d_alow + D;
     double* d_steps = d_ahigh + D;
-    double* d_x = d_steps + D;
+    double* d_nchoices = d_steps + D;
+    double* d_x = d_nchoices + D;
     double* d_xedges = d_x + (size_t)S * D;
     double* d_out = d_xedges + 2 * (size_t)S * D;
     double* d_scratch = d_out + S;
@@ -617,19 +631,21 @@ py::array_t<double> kde_logpdf(const arr_f64& obs, const arr_i64& sorted_pos,
     g_ws.h2d(d_alow, alow.data(), D * 8, st);
     g_ws.h2d(d_ahigh, ahigh.data(), D * 8, st);
     g_ws.h2d(d_steps, steps.data(), D * 8, st);
+    g_ws.h2d(d_nchoices, n_choices.data(), D * 8, st);
     g_ws.h2d(d_x, x.data(), (size_t)S * D * 8, st);
     g_ws.h2d(d_xedges, xedges.data(), 2 * (size_t)S * D * 8, st);
 
+    const double cat_base = prior_weight / (double)K;
     {
         const int block = 256;
         const int gx = (int)((K + block - 1) / block);
         hipLaunchKernelGGL(k_parzen_fit, dim3(gx, (unsigned)D), dim3(block), 0, st,
-                           d_obs, d_sorted, d_alow, d_ahigh, d_steps, N, D,
-                           consider_endpoints ? 1 : 0, magic_clip ? 1 : 0, d_c1,
-                           d_c2, d_c3);
+                           d_obs, d_sorted, d_alow, d_ahigh, d_steps, d_nchoices,
+                           N, D, consider_endpoints ? 1 : 0, magic_clip ? 1 : 0,
+                           d_c1, d_c2, d_c3);
     }
-    launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps, K, D, S,
-                      d_out, d_scratch);
+    launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps,
+                      d_nchoices, cat_base, K, D, S, d_out, d_scratch);
     HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
                              hipMemcpyDeviceToHost, st));
     HIP_CHECK(hipStreamSynchronize(st));
@@ -915,7 +931,9 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
                                    int64_t stride,  // row stride of sub_* (= Nv)
                                    const double* __restrict__ alow,
                                    const double* __restrict__ ahigh,
-                                   const double* __restrict__ steps, int64_t Na,
+                                   const double* __restrict__ steps,
+                                   const double* __restrict__ n_choices,
+                                   int64_t Na,
                                    int64_t D, int consider_endpoints,
                                    int magic_clip, double* __restrict__ c1,
                                    double* __restrict__ c2,
@@ -933,6 +951,15 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
 
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r <= Na;
          r += (int64_t)gridDim.x * blockDim.x) {
+        if (n_choices[d] > 0.0) {
+            const int64_t kc = (r == Na) ? Na : (int64_t)ks_d[r];
+            c1[d * (Na + 1) + kc] =
+                (r == Na) ? -1.0
+                          : row_val(params, extras, n_table, rows_d[r], D, d);
+            c2[d * (Na + 1) + kc] = 0.0;
+            c3[d * (Na + 1) + kc] = 0.0;
+            continue;
+        }
         double mu, sigma;
         int64_t k;
         if (r == Na) {
@@ -1023,6 +1050,8 @@ class TpeDeviceHistory {
                               const arr_f64& logw,  // (n_above + L + 1,)
                               const arr_f64& alow, const arr_f64& ahigh,
                               const arr_f64& steps,   // (D,) 0 = continuous
+                              const arr_f64& n_choices,  // (D,) 0 = numerical
+                              double prior_weight,
                               const arr_f64& x,       // (S, D) KDE domain
                               const arr_f64& xedges,  // (S, 2D) cell lo|hi
                               bool consider_endpoints, bool magic_clip,
@@ -1056,7 +1085,7 @@ class TpeDeviceHistory {
         // f64 slots: c1|c2|c3|logw|alow|ahigh|steps|x|xedges|out|lse|extras + i32.
         const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
         const size_t n_extras_f64 = 2 * (size_t)L * D_;  // raw + per-dim sorted
-        size_t f64_total = 3 * n_c + K + 3 * D_ + 3 * (size_t)S * D_ + S +
+        size_t f64_total = 3 * n_c + K + 4 * D_ + 3 * (size_t)S * D_ + S +
                            n_scratch + n_extras_f64;
         size_t i32_doubles = ((size_t)Nv * D_ /*sorted*/ + n_ /*pos*/ +
                               2 * (size_t)Nv * D_ /*sub*/ +
@@ -1073,7 +1102,8 @@ class TpeDeviceHistory {
         double* d_alow = d_logw + K;
         double* d_ahigh = d_alow + D_;
         double* d_steps = d_ahigh + D_;
-        double* d_x = d_steps + D_;
+        double* d_nchoices = d_steps + D_;
+        double* d_x = d_nchoices + D_;
         double* d_xedges = d_x + (size_t)S * D_;
         double* d_out = d_xedges + 2 * (size_t)S * D_;
         double* d_scratch = d_out + S;
@@ -1097,6 +1127,7 @@ class TpeDeviceHistory {
         g_ws.h2d(d_alow, alow.data(), D_ * 8, st);
         g_ws.h2d(d_ahigh, ahigh.data(), D_ * 8, st);
         g_ws.h2d(d_steps, steps.data(), D_ * 8, st);
+        g_ws.h2d(d_nchoices, n_choices.data(), D_ * 8, st);
         g_ws.h2d(d_x, x.data(), (size_t)S * D_ * 8, st);
         g_ws.h2d(d_xedges, xedges.data(), 2 * (size_t)S * D_ * 8, st);
         if (L > 0) {
@@ -1134,12 +1165,13 @@ class TpeDeviceHistory {
             hipLaunchKernelGGL(k_parzen_fit_table, dim3(gx, (unsigned)D_),
                                dim3(block), 0, st, params_,
                                L > 0 ? d_extras_raw : nullptr, n_, fit_rows,
-                               fit_k, fit_stride, d_alow, d_ahigh, d_steps, Nk,
-                               D_, consider_endpoints ? 1 : 0,
+                               fit_k, fit_stride, d_alow, d_ahigh, d_steps,
+                               d_nchoices, Nk, D_, consider_endpoints ? 1 : 0,
                                magic_clip ? 1 : 0, d_c1, d_c2, d_c3);
         }
-        launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps, K,
-                          D_, S, d_out, d_scratch);
+        const double cat_base = prior_weight / (double)K;
+        launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps,
+                          d_nchoices, cat_base, K, D_, S, d_out, d_scratch);
         py::array_t<double> out(S);
         HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, S * 8,
                                  hipMemcpyDeviceToHost, st));
@@ -1167,6 +1199,7 @@ PYBIND11_MODULE(_hipcore, m) {
           py::arg("n_below"));
     m.def("kde_logpdf", &kde_logpdf, py::arg("obs"), py::arg("sorted_pos"),
           py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("steps"),
+          py::arg("n_choices"), py::arg("prior_weight"),
           py::arg("x"), py::arg("xedges"),
           py::arg("consider_endpoints") = false, py::arg("magic_clip") = true);
     py::class_<TpeDeviceHistory>(m, "TpeDeviceHistory")
@@ -1175,7 +1208,8 @@ PYBIND11_MODULE(_hipcore, m) {
         .def("append", &TpeDeviceHistory::append, py::arg("block"))
         .def("score", &TpeDeviceHistory::score, py::arg("sorted_rows"),
              py::arg("pos"), py::arg("n_above"), py::arg("logw"), py::arg("alow"),
-             py::arg("ahigh"), py::arg("steps"), py::arg("x"), py::arg("xedges"),
+             py::arg("ahigh"), py::arg("steps"), py::arg("n_choices"),
+             py::arg("prior_weight"), py::arg("x"), py::arg("xedges"),
              py::arg("consider_endpoints") = false,
              py::arg("magic_clip") = true,
              py::arg("extras_raw") = arr_f64(),

@@ -25,6 +25,7 @@ import numpy as np
 from optuna_amd import _hip
 from optuna_amd.distributions import (
     BaseDistribution,
+    CategoricalDistribution,
     FloatDistribution,
     IntDistribution,
 )
@@ -38,12 +39,9 @@ DEVICE_MIN_KERNELS = 512
 
 
 def space_is_device_eligible(space: dict[str, BaseDistribution]) -> bool:
-    if not space:
-        return False
-    return all(
-        isinstance(dist, (FloatDistribution, IntDistribution))
-        for dist in space.values()
-    )
+    # Float / int / discrete dims use the (possibly cell-integrated) truncnorm
+    # kernels; categorical dims use the closed-form smoothed one-hot weights.
+    return bool(space)
 
 
 def device_ready(n_kernels: int) -> bool:
@@ -53,27 +51,40 @@ def device_ready(n_kernels: int) -> bool:
 def _space_domains(
     space: dict[str, BaseDistribution],
 ) -> tuple[np.ndarray, np.ndarray, np.ndarray, np.ndarray]:
-    """(is_log, alow, ahigh, steps) per dim, KDE domain.
+    """(is_log, alow, ahigh, steps, n_choices) per dim, KDE domain.
 
     Step-discretized dims widen the domain by half a step on each side before
     the log transform — identical to the host estimator's adapted bounds
     (parzen.py `_numerical_kernels_batched`).
     """
-    is_log = np.array([bool(d.log) for d in space.values()])
+    dists = list(space.values())
+    is_cat = np.array([isinstance(d, CategoricalDistribution) for d in dists])
+    is_log = np.array(
+        [False if c else bool(d.log) for c, d in zip(is_cat, dists)]
+    )
     steps = np.array(
-        [float(d.step) if d.step is not None else 0.0 for d in space.values()],
+        [
+            0.0 if c else (float(d.step) if d.step is not None else 0.0)
+            for c, d in zip(is_cat, dists)
+        ],
         dtype=np.float64,
     )
-    alow = np.empty(len(steps))
-    ahigh = np.empty(len(steps))
-    for c, d in enumerate(space.values()):
+    n_choices = np.array(
+        [float(len(d.choices)) if c else 0.0 for c, d in zip(is_cat, dists)],
+        dtype=np.float64,
+    )
+    alow = np.zeros(len(dists))
+    ahigh = np.ones(len(dists))
+    for c, d in enumerate(dists):
+        if is_cat[c]:
+            continue  # unused by the categorical branch
         lo = float(d.low) - steps[c] / 2 if steps[c] else float(d.low)
         hi = float(d.high) + steps[c] / 2 if steps[c] else float(d.high)
         if d.log:
             lo, hi = math.log(lo), math.log(hi)
         alow[c] = lo
         ahigh[c] = hi
-    return is_log, alow, ahigh, steps
+    return is_log, alow, ahigh, steps, n_choices
 
 
 def _cell_edges(
@@ -109,7 +120,13 @@ class _SpaceDeviceMirror:
     def __init__(self, space: dict[str, BaseDistribution]) -> None:
         core = _hip.get()
         assert core is not None
-        self._is_log, self._alow, self._ahigh, self._steps = _space_domains(space)
+        (
+            self._is_log,
+            self._alow,
+            self._ahigh,
+            self._steps,
+            self._n_choices,
+        ) = _space_domains(space)
         self._hist = core.TpeDeviceHistory(len(space))
         self._n_appended = 0
 
@@ -131,6 +148,7 @@ class _SpaceDeviceMirror:
         samples: dict[str, np.ndarray],
         consider_endpoints: bool,
         consider_magic_clip: bool,
+        prior_weight: float = 1.0,
         extras: np.ndarray | None = None,
     ) -> np.ndarray:
         """Score candidates against the resident table.
@@ -179,6 +197,8 @@ class _SpaceDeviceMirror:
             self._alow,
             self._ahigh,
             self._steps,
+            self._n_choices,
+            float(prior_weight),
             np.ascontiguousarray(x),
             np.ascontiguousarray(xedges),
             consider_endpoints,
@@ -194,6 +214,7 @@ def score_above_resident(
     samples: dict[str, np.ndarray],
     consider_endpoints: bool,
     consider_magic_clip: bool,
+    prior_weight: float = 1.0,
     extras: np.ndarray | None = None,
 ) -> np.ndarray:
     """log g(x) for candidates via the device-resident table (creates the mirror
@@ -204,7 +225,7 @@ def score_above_resident(
         cache._device_mirror = mirror  # type: ignore[attr-defined]
     return mirror.score(
         cache, sel, weights, samples, consider_endpoints, consider_magic_clip,
-        extras=extras,
+        prior_weight=prior_weight, extras=extras,
     )
 
 
@@ -216,6 +237,7 @@ def kde_logpdf(
     samples: dict[str, np.ndarray],
     consider_endpoints: bool,
     consider_magic_clip: bool,
+    prior_weight: float = 1.0,
 ) -> np.ndarray:
     """Stateless device scoring (observation matrix uploaded per call)."""
     core = _hip.get()
@@ -224,7 +246,7 @@ def kde_logpdf(
     D = len(names)
     N = len(observations[names[0]])
 
-    is_log, alow, ahigh, steps = _space_domains(space)
+    is_log, alow, ahigh, steps, n_choices = _space_domains(space)
     obs = np.empty((N, D), dtype=np.float64)
     x_raw = np.column_stack(
         [np.asarray(samples[n], dtype=np.float64) for n in names]
@@ -252,6 +274,8 @@ def kde_logpdf(
         alow,
         ahigh,
         steps,
+        n_choices,
+        float(prior_weight),
         x,
         np.ascontiguousarray(xedges),
         consider_endpoints,

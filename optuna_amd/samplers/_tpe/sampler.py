@@ -398,6 +398,7 @@ class TPESampler(BaseSampler):
                 samples_below,
                 self._parzen_estimator_parameters.consider_endpoints,
                 self._parzen_estimator_parameters.consider_magic_clip,
+                prior_weight=self._parzen_estimator_parameters.prior_weight,
                 extras=liar_extras,
             )
             acq_func_vals = mpe_below.log_pdf(samples_below) - log_g

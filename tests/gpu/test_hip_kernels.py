@@ -565,3 +565,94 @@ def test_resident_discrete_matches_host(core) -> None:
     mpe = _ParzenEstimator(obs, space, params_obj)
     ref = mpe.log_pdf(samples)
     np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
+
+
+def test_kde_logpdf_categorical_dims_matches_host(core) -> None:
+    """Closed-form categorical weights on device must match the host's
+    prior-smoothed one-hot weight matrix."""
+    from optuna_amd.distributions import CategoricalDistribution, IntDistribution
+    from optuna_amd.samplers._tpe import _device as dev
+
+    rng = np.random.RandomState(31)
+    n_obs = 700
+    space = {
+        "a": FloatDistribution(-5.0, 5.0),
+        "b": CategoricalDistribution(("x", "y", "z")),
+        "c": IntDistribution(0, 20),
+        "d": CategoricalDistribution(tuple(range(7))),
+    }
+    observations = {
+        "a": rng.uniform(-5, 5, n_obs),
+        "b": rng.randint(0, 3, n_obs).astype(float),   # internal repr: index
+        "c": rng.randint(0, 21, n_obs).astype(float),
+        "d": rng.randint(0, 7, n_obs).astype(float),
+    }
+    samples = {
+        "a": rng.uniform(-5, 5, 24),
+        "b": rng.randint(0, 3, 24).astype(float),
+        "c": rng.randint(0, 21, 24).astype(float),
+        "d": rng.randint(0, 7, 24).astype(float),
+    }
+    w = default_weights(n_obs)
+    weights = np.append(w, [1.0])
+    weights = weights / weights.sum()
+
+    assert dev.space_is_device_eligible(space)
+    ours = dev.kde_logpdf(space, observations, None, weights, samples, False, True)
+
+    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, True)
+    mpe = _ParzenEstimator(observations, space, params)
+    ref = mpe.log_pdf(samples)
+    np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
+
+
+def test_tpe_end_to_end_categorical_device(core, monkeypatch) -> None:
+    """suggest_categorical at large history runs through the device path."""
+    import warnings
+
+    from optuna_amd.samplers._tpe import _device as device_mod
+
+    calls = {"n": 0}
+    orig = device_mod.score_above_resident
+
+    def spy(*args, **kwargs):
+        calls["n"] += 1
+        return orig(*args, **kwargs)
+
+    monkeypatch.setattr(device_mod, "score_above_resident", spy)
+    warnings.simplefilter("ignore")
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(2)
+    from optuna_amd.distributions import CategoricalDistribution, IntDistribution
+
+    dists = {
+        "x": FloatDistribution(-5.0, 5.0),
+        "k": CategoricalDistribution(("a", "b", "c")),
+        "i": IntDistribution(1, 64, log=True),
+    }
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.TPESampler(seed=4, n_startup_trials=5)
+    )
+    study.add_trials(
+        [
+            optuna_amd.create_trial(
+                params={
+                    "x": float(rng.uniform(-5, 5)),
+                    "k": ["a", "b", "c"][rng.randint(3)],
+                    "i": int(np.exp(rng.uniform(0, np.log(64)))),
+                },
+                distributions=dict(dists),
+                value=float(rng.rand()),
+            )
+            for _ in range(700)
+        ]
+    )
+
+    def objective(trial):
+        x = trial.suggest_float("x", -5, 5)
+        k = trial.suggest_categorical("k", ("a", "b", "c"))
+        i = trial.suggest_int("i", 1, 64, log=True)
+        return x * x + {"a": 0.0, "b": 1.0, "c": 2.0}[k] + abs(i - 8)
+
+    study.optimize(objective, n_trials=3)
+    assert calls["n"] >= 3
