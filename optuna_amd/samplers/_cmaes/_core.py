@@ -130,14 +130,34 @@ class CMA:
 
     # ---- sampling -------------------------------------------------------------------
 
+    # K8: below this dimension the host eigh wins (an n=100 symmetric
+    # eigendecomposition is ~0.5 ms on CPU, amortized over a whole generation,
+    # while a device round trip costs more in launch latency than it saves);
+    # above it the O(n³) work dominates and rocSOLVER syevd takes over.
+    _DEVICE_EIGH_MIN_DIM = 256
+
     def _eigen(self) -> tuple[np.ndarray, np.ndarray]:
         if self._B is None or self._D is None:
             self._C = (self._C + self._C.T) / 2  # enforce symmetry
-            d2, B = np.linalg.eigh(self._C)
+            d2, B = self._eigh(self._C)
             D = np.sqrt(np.maximum(d2, _EPS**2))
             self._C = B @ np.diag(D**2) @ B.T
             self._B, self._D = B, D
         return self._B, self._D
+
+    def _eigh(self, C: np.ndarray) -> tuple[np.ndarray, np.ndarray]:
+        if self._n_dim >= self._DEVICE_EIGH_MIN_DIM:
+            try:
+                import torch
+
+                if torch.cuda.is_available():
+                    d2, B = torch.linalg.eigh(
+                        torch.from_numpy(C).to("cuda")
+                    )
+                    return d2.cpu().numpy(), B.cpu().numpy()
+            except ImportError:
+                pass
+        return np.linalg.eigh(C)
 
     def _sample_one(self) -> np.ndarray:
         B, D = self._eigen()

@@ -656,3 +656,22 @@ def test_tpe_end_to_end_categorical_device(core, monkeypatch) -> None:
 
     study.optimize(objective, n_trials=3)
     assert calls["n"] >= 3
+
+
+def test_cma_device_eigh_matches_host(core) -> None:
+    """K8: rocSOLVER eigendecomposition above the dim threshold reconstructs C."""
+    import torch
+
+    from optuna_amd.samplers._cmaes._core import CMA
+
+    assert torch.cuda.is_available()
+    n = 300
+    rng = np.random.RandomState(5)
+    A = rng.randn(n, n)
+    C = A @ A.T / n + np.eye(n)
+    opt = CMA(mean=np.zeros(n), sigma=1.0, seed=0)
+    assert n >= opt._DEVICE_EIGH_MIN_DIM
+    d2, B = opt._eigh(C)
+    np.testing.assert_allclose(B @ np.diag(d2) @ B.T, C, rtol=1e-8, atol=1e-8)
+    d2h, _ = np.linalg.eigh(C)
+    np.testing.assert_allclose(np.sort(d2), np.sort(d2h), rtol=1e-8, atol=1e-8)
