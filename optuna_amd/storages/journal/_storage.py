@@ -378,6 +378,10 @@ class _ReplayState:
         # COMPLETE/PRUNED trials per study in finish order; finished trials are
         # immutable, so storages expose O(delta) history reads from this log.
         self._study_id_to_finished: dict[int, list[FrozenTrial]] = {}
+        # Live-state index: RUNNING/WAITING trial ids per study, so the
+        # per-suggest constant-liar fetch and the per-ask WAITING scan cost
+        # O(matching) instead of O(all trials).
+        self._study_id_to_live: dict[int, dict[TrialState, set[int]]] = {}
         self._trial_id_to_study_id: dict[int, int] = {}
         self._next_study_id = 0
         self._worker_id_to_owned_trial_id: dict[str, int] = {}
@@ -395,6 +399,15 @@ class _ReplayState:
         self.__dict__.update(state)
         self.last_created_trial_id = -1
         self.__dict__.setdefault("my_created_trial_ids", [])
+        if "_study_id_to_live" not in self.__dict__:
+            self._study_id_to_live = {}
+            for sid, trial_ids in self._study_id_to_trial_ids.items():
+                live: dict[TrialState, set[int]] = {}
+                for tid in trial_ids:
+                    st0 = self._trials[tid].state
+                    if st0 in (TrialState.RUNNING, TrialState.WAITING):
+                        live.setdefault(st0, set()).add(tid)
+                self._study_id_to_live[sid] = live
         if "_study_id_to_finished" not in self.__dict__:
             # Snapshot from an older build: rebuild the finished log (trial-id
             # order; consumers dedupe by id, so order only affects row order).
@@ -436,6 +449,20 @@ class _ReplayState:
     def all_trials(self, study_id: int, states: Container[TrialState] | None) -> list[FrozenTrial]:
         if study_id not in self._studies:
             raise KeyError(NOT_FOUND_MSG)
+        try:
+            state_set = set(states) if states is not None else None
+        except TypeError:
+            state_set = None
+        if state_set is not None and state_set <= {
+            TrialState.RUNNING,
+            TrialState.WAITING,
+        }:
+            live = self._study_id_to_live.get(study_id, {})
+            ids: list[int] = []
+            for st0 in state_set:
+                ids.extend(live.get(st0, ()))
+            # trial ids are created in number order within a study
+            return [self._trials[tid] for tid in sorted(ids)]
         out = []
         for trial_id in self._study_id_to_trial_ids[study_id]:
             t = self._trials[trial_id]
@@ -569,6 +596,11 @@ class _ReplayState:
         )
         self._study_id_to_trial_ids[study_id].append(trial_id)
         self._trial_id_to_study_id[trial_id] = study_id
+        st0 = self._trials[trial_id].state
+        if st0 in (TrialState.RUNNING, TrialState.WAITING):
+            self._study_id_to_live.setdefault(study_id, {}).setdefault(
+                st0, set()
+            ).add(trial_id)
         if self._trials[trial_id].state in (TrialState.COMPLETE, TrialState.PRUNED):
             self._study_id_to_finished.setdefault(study_id, []).append(
                 self._trials[trial_id]
@@ -620,10 +652,17 @@ class _ReplayState:
                 self._worker_id_to_owned_trial_id[self.worker_id] = trial_id
         if state.is_finished():
             trial.datetime_complete = _iso_to_local_naive(log["datetime_complete"])
+        prev_state = self._trials[trial_id].state
         trial.state = state
         if log["values"] is not None:
             trial.values = log["values"]
         self._trials[trial_id] = trial
+        sid = self._trial_id_to_study_id[trial_id]
+        live = self._study_id_to_live.setdefault(sid, {})
+        if prev_state in (TrialState.RUNNING, TrialState.WAITING):
+            live.get(prev_state, set()).discard(trial_id)
+        if state in (TrialState.RUNNING, TrialState.WAITING):
+            live.setdefault(state, set()).add(trial_id)
         if state in (TrialState.COMPLETE, TrialState.PRUNED):
             self._study_id_to_finished.setdefault(
                 self._trial_id_to_study_id[trial_id], []
