@@ -9,6 +9,7 @@ from optuna_amd.samplers._tpe.sampler import TPESampler
 
 
 __all__ = [
+    "BaseGASampler",
     "BaseSampler",
     "BruteForceSampler",
     "CmaEsSampler",
@@ -56,4 +57,8 @@ def __getattr__(name: str):  # lazy heavy/optional samplers
         from optuna_amd.samplers._nsgaiii import NSGAIIISampler
 
         return NSGAIIISampler
+    if name == "BaseGASampler":
+        from optuna_amd.samplers._ga._base import BaseGASampler
+
+        return BaseGASampler
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")

@@ -81,6 +81,43 @@ def __getattr__(name: str):  # lazy to avoid import cycles / optional deps
         from optuna_amd.storages.journal import BaseJournalBackend
 
         return BaseJournalBackend
+    if name in ("JournalFileSymlinkLock", "JournalFileOpenLock"):
+        from optuna_amd.storages.journal import _file
+
+        return getattr(_file, name)
+    # Deprecated aliases kept for drop-in parity with the reference
+    # (storages/__init__.py): old names forward to the new classes.
+    if name == "JournalFileStorage" or name == "BaseJournalLogStorage":
+        import warnings
+
+        from optuna_amd.storages.journal import BaseJournalBackend, JournalFileBackend
+
+        warnings.warn(
+            f"{name} is deprecated; use the *Backend classes instead.",
+            FutureWarning,
+        )
+        return JournalFileBackend if name == "JournalFileStorage" else BaseJournalBackend
+    if name == "JournalRedisStorage":
+        import warnings
+
+        from optuna_amd.storages.journal._redis import JournalRedisBackend
+
+        warnings.warn(
+            "JournalRedisStorage is deprecated; use JournalRedisBackend.",
+            FutureWarning,
+        )
+        return JournalRedisBackend
+    if name == "RetryHeartbeatStaleTrialCallback":
+        import warnings
+
+        from optuna_amd._callbacks import RetryFailedTrialCallback
+
+        warnings.warn(
+            "RetryHeartbeatStaleTrialCallback is deprecated; use "
+            "RetryFailedTrialCallback.",
+            FutureWarning,
+        )
+        return RetryFailedTrialCallback
     raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
 
 
@@ -96,5 +133,11 @@ __all__ = [
     "RetryFailedTrialCallback",
     "fail_stale_trials",
     "BaseHeartbeat",
+    "JournalFileSymlinkLock",
+    "JournalFileOpenLock",
+    "JournalFileStorage",
+    "JournalRedisStorage",
+    "BaseJournalLogStorage",
+    "RetryHeartbeatStaleTrialCallback",
     "get_storage",
 ]
