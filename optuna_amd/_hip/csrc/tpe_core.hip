@@ -297,18 +297,42 @@ __global__ void k_mix_logpdf_partial(const double* __restrict__ x,  // (S, D)
     }
     __syncthreads();
 
+    // All-continuous spaces (the common case, incl. the headline bench) take
+    // a branch-free two-FMA inner loop; any discrete/categorical dim switches
+    // to the general per-dim dispatch.
+    bool any_special = false;
+    for (int64_t d = 0; d < D; ++d)
+        any_special |= (steps[d] > 0.0) || (n_choices[d] > 0.0);
+
     double m = -INFINITY, acc = 0.0;
-    for (int64_t k = k_lo + threadIdx.x; k < k_hi; k += blockDim.x) {
-        double t = logw[k];
-        for (int64_t d = 0; d < D; ++d) {
-            // k-major (D, K): lanes of a wavefront read consecutive k.
-            t += mix_term(c1, c2, c3, steps, n_choices, cat_base, xs, x2, xl, xr, K, k, d);
+    if (!any_special) {
+        for (int64_t k = k_lo + threadIdx.x; k < k_hi; k += blockDim.x) {
+            double t = logw[k];
+            for (int64_t d = 0; d < D; ++d) {
+                // k-major (D, K): lanes of a wavefront read consecutive k.
+                t += x2[d] * c1[d * K + k] + xs[d] * c2[d * K + k] +
+                     c3[d * K + k];
+            }
+            if (t > m) {
+                acc = acc * exp(m - t) + 1.0;
+                m = t;
+            } else {
+                acc += exp(t - m);
+            }
         }
-        if (t > m) {
-            acc = acc * exp(m - t) + 1.0;
-            m = t;
-        } else {
-            acc += exp(t - m);
+    } else {
+        for (int64_t k = k_lo + threadIdx.x; k < k_hi; k += blockDim.x) {
+            double t = logw[k];
+            for (int64_t d = 0; d < D; ++d) {
+                t += mix_term(c1, c2, c3, steps, n_choices, cat_base, xs, x2,
+                              xl, xr, K, k, d);
+            }
+            if (t > m) {
+                acc = acc * exp(m - t) + 1.0;
+                m = t;
+            } else {
+                acc += exp(t - m);
+            }
         }
     }
     red_m[threadIdx.x] = m;
