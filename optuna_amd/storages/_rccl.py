@@ -70,7 +70,10 @@ class _TcpStoreLog:
         keys = [_REC_KEY.format(idx=start + i) for i in range(n)]
         payloads = [json.dumps(rec, separators=(",", ":")) for rec in records]
         if hasattr(self._store, "multi_set"):
-            self._store.multi_set(keys, payloads)
+            # Chunked: a 10k-record bulk populate would otherwise be one
+            # multi-MB RPC.
+            for i in range(0, n, 2048):
+                self._store.multi_set(keys[i : i + 2048], payloads[i : i + 2048])
         else:
             for key, payload in zip(keys, payloads):
                 self._store.set(key, payload)
@@ -81,7 +84,9 @@ class _TcpStoreLog:
             return []
         keys = [_REC_KEY.format(idx=idx) for idx in range(start, end)]
         if hasattr(self._store, "multi_get"):
-            payloads = self._store.multi_get(keys)
+            payloads = []
+            for i in range(0, len(keys), 2048):
+                payloads.extend(self._store.multi_get(keys[i : i + 2048]))
         else:
             payloads = [self._store.get(k) for k in keys]
         return [json.loads(p) for p in payloads]
