@@ -92,18 +92,10 @@ class _TcpStoreLog:
         return [json.loads(p) for p in payloads]
 
     def read_from(self, start: int) -> list[dict[str, Any]]:
-        end = self._store.add(_SEQ_KEY, 0)
-        if end <= start:
-            return []
-        keys = [_REC_KEY.format(idx=idx) for idx in range(start, end)]
         # set() may lag the counter by a moment on another worker; get() blocks
         # until the key appears (bounded by the store timeout). multi_get has
-        # the same wait-for-key semantics, one round trip for the batch.
-        if hasattr(self._store, "multi_get"):
-            payloads = self._store.multi_get(keys)
-        else:
-            payloads = [self._store.get(k) for k in keys]
-        return [json.loads(p) for p in payloads]
+        # the same wait-for-key semantics.
+        return self.read_range(start, self._store.add(_SEQ_KEY, 0))
 
 
 class RcclStorage(BaseStorage):

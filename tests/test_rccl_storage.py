@@ -175,7 +175,7 @@ def test_rccl_journal_checkpoint_roundtrip(tmp_path) -> None:
     from optuna_amd.storages._rccl import RcclStorage
     from optuna_amd.storages.journal import JournalFileBackend
 
-    store = TCPStore("127.0.0.1", 29645, 1, is_master=True,
+    store = TCPStore("127.0.0.1", _free_port(), 1, is_master=True,
                      timeout=datetime.timedelta(seconds=30))
     storage = RcclStorage(store)
     study = optuna_amd.create_study(storage=storage, study_name="ckpt")
@@ -190,7 +190,7 @@ def test_rccl_journal_checkpoint_roundtrip(tmp_path) -> None:
     assert resumed.best_value == study.best_value
 
     # Resume the journal into a fresh shared log.
-    store2 = TCPStore("127.0.0.1", 29646, 1, is_master=True,
+    store2 = TCPStore("127.0.0.1", _free_port(), 1, is_master=True,
                       timeout=datetime.timedelta(seconds=30))
     storage2 = RcclStorage(store2)
     storage2.load_journal(JournalFileBackend(path))
