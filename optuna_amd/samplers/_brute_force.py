@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import decimal
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from functools import lru_cache
 from numbers import Real
 from typing import TYPE_CHECKING, Any, Sequence

@@ -9,7 +9,7 @@ from __future__ import annotations
 from typing import TYPE_CHECKING
 
 from optuna_amd.distributions import BaseDistribution
-from optuna_amd.trial import FrozenTrial, TrialState
+from optuna_amd.trial import TrialState
 
 
 if TYPE_CHECKING:

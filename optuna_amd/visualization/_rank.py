@@ -14,8 +14,6 @@ from optuna_amd.visualization._utils import (
     _check_plot_args,
     _filter_nonfinite,
     _get_completed_trials,
-    _is_log_scale,
-    _is_numerical,
 )
 
 
