@@ -31,8 +31,8 @@ def _log(msg: str) -> None:
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=64)
-    parser.add_argument("--warmup", type=int, default=8)
+    parser.add_argument("--steps", type=int, default=256)
+    parser.add_argument("--warmup", type=int, default=16)
     parser.add_argument("--history", type=int, default=10000)
     parser.add_argument("--dims", type=int, default=20)
     parser.add_argument(
