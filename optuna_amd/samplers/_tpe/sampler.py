@@ -386,9 +386,10 @@ class TPESampler(BaseSampler):
         samples_below = mpe_below.sample(self._rng.rng, self._n_ei_candidates)
 
         # Device path (K1+K2): the big "above" KDE is fit and scored against the
-        # HBM-resident parameter table; the small "below" estimator stays on host
-        # (it also drives candidate sampling). Host path covers discrete /
-        # categorical spaces.
+        # HBM-resident parameter table for every distribution type (continuous,
+        # discrete cells, categorical); the small "below" estimator stays on
+        # host (it also drives candidate sampling). The host path remains for
+        # small histories and custom estimator classes.
         if use_device:
             weights_above = self._above_weights(len(above_sel) + n_liar)
             log_g = _device.score_above_resident(
