@@ -148,10 +148,44 @@ class RcclStorage(BaseStorage):
 
     # ---- log plumbing ---------------------------------------------------------------
 
+    def _coalesce_pending(self) -> list[dict[str, Any]]:
+        """Compress runs of buffered SET_TRIAL_PARAM records for one trial into
+        a single SET_TRIAL_PARAMS_BATCH record (one JSON parse per trial on
+        every replaying rank instead of ~n_params)."""
+        out: list[dict[str, Any]] = []
+        for rec in self._pending:
+            if (
+                rec["op_code"] == int(JournalOperation.SET_TRIAL_PARAM)
+                and out
+                and out[-1]["op_code"] == int(JournalOperation.SET_TRIAL_PARAMS_BATCH)
+                and out[-1]["trial_id"] == rec["trial_id"]
+            ):
+                out[-1]["params"][rec["param_name"]] = {
+                    "value": rec["param_value_internal"],
+                    "distribution": rec["distribution"],
+                }
+            elif rec["op_code"] == int(JournalOperation.SET_TRIAL_PARAM):
+                out.append(
+                    {
+                        "op_code": int(JournalOperation.SET_TRIAL_PARAMS_BATCH),
+                        "worker_id": rec["worker_id"],
+                        "trial_id": rec["trial_id"],
+                        "params": {
+                            rec["param_name"]: {
+                                "value": rec["param_value_internal"],
+                                "distribution": rec["distribution"],
+                            }
+                        },
+                    }
+                )
+            else:
+                out.append(rec)
+        return out
+
     def _append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
         rec = {"op_code": int(op), "worker_id": self._replay.worker_id, **fields}
         # Pending records must precede this one in the global order.
-        batch = self._pending + [rec]
+        batch = self._coalesce_pending() + [rec]
         self._pending = []
         end = self._log.append(batch)
         # Apply everything up to our own record without another counter query;
@@ -456,7 +490,23 @@ class RcclStorage(BaseStorage):
         with self._thread_lock:
             self._sync(force=True)
             records = self._log.read_from(0)
-        backend.append_logs(records)
+        expanded: list[dict[str, Any]] = []
+        for rec in records:
+            if rec["op_code"] == int(JournalOperation.SET_TRIAL_PARAMS_BATCH):
+                for name, fields in rec["params"].items():
+                    expanded.append(
+                        {
+                            "op_code": int(JournalOperation.SET_TRIAL_PARAM),
+                            "worker_id": rec["worker_id"],
+                            "trial_id": rec["trial_id"],
+                            "param_name": name,
+                            "param_value_internal": fields["value"],
+                            "distribution": fields["distribution"],
+                        }
+                    )
+            else:
+                expanded.append(rec)
+        backend.append_logs(expanded)
 
     def load_journal(self, backend: Any) -> None:
         """Resume: seed an EMPTY shared log from a journal backend's records.

@@ -55,6 +55,11 @@ class JournalOperation(enum.IntEnum):
     SET_TRIAL_USER_ATTR = 8
     SET_TRIAL_SYSTEM_ATTR = 9
     DISCARD_TRIALS = 10  # written by other implementations; replay skips it
+    # Private RcclStorage extension (never written to journal files; to_journal
+    # expands it into standard SET_TRIAL_PARAM records): all of one trial's
+    # buffered param writes as a single record, so N-rank replay parses one
+    # JSON object instead of ~20 per finished trial.
+    SET_TRIAL_PARAMS_BATCH = 100
 
 
 def _utcnow_iso() -> str:
@@ -487,6 +492,7 @@ class _ReplayState:
             int(JournalOperation.SET_TRIAL_INTERMEDIATE_VALUE): self._on_trial_intermediate,
             int(JournalOperation.SET_TRIAL_USER_ATTR): self._on_trial_user_attr,
             int(JournalOperation.SET_TRIAL_SYSTEM_ATTR): self._on_trial_system_attr,
+            int(JournalOperation.SET_TRIAL_PARAMS_BATCH): self._on_trial_params_batch,
         }
         for log in logs:
             if advance:
@@ -637,6 +643,19 @@ class _ReplayState:
         }
         trial.distributions = {**trial.distributions, param_name: distribution}
         self._trials[trial_id] = trial
+
+    def _on_trial_params_batch(self, log: dict[str, Any]) -> None:
+        base = {"op_code": int(JournalOperation.SET_TRIAL_PARAM),
+                "worker_id": log["worker_id"], "trial_id": log["trial_id"]}
+        for name, fields in log["params"].items():
+            self._on_trial_param(
+                {
+                    **base,
+                    "param_name": name,
+                    "param_value_internal": fields["value"],
+                    "distribution": fields["distribution"],
+                }
+            )
 
     def _on_trial_state_values(self, log: dict[str, Any]) -> None:
         trial_id = log["trial_id"]
