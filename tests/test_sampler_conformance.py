@@ -41,6 +41,42 @@ class TestCmaEsSampler(BasicSamplerTestCase):
         )
 
 
+class TestCmaEsSamplerWithMargin(BasicSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        import warnings
+
+        def make(seed: int) -> BaseSampler:
+            with warnings.catch_warnings():
+                warnings.simplefilter("ignore")
+                return optuna_amd.samplers.CmaEsSampler(
+                    seed=seed,
+                    n_startup_trials=2,
+                    with_margin=True,
+                    warn_independent_sampling=False,
+                )
+
+        return make
+
+
+class TestCmaEsSamplerSeparable(BasicSamplerTestCase):
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        import warnings
+
+        def make(seed: int) -> BaseSampler:
+            with warnings.catch_warnings():
+                warnings.simplefilter("ignore")
+                return optuna_amd.samplers.CmaEsSampler(
+                    seed=seed,
+                    n_startup_trials=2,
+                    use_separable_cma=True,
+                    warn_independent_sampling=False,
+                )
+
+        return make
+
+
 class TestGPSampler(BasicSamplerTestCase):
     n_trials = 6
 
