@@ -37,7 +37,7 @@ def main() -> None:
     parser.add_argument("--dims", type=int, default=20)
     parser.add_argument(
         "--suite",
-        choices=("tpe", "random", "gp", "cmaes", "nsgaii"),
+        choices=("tpe", "random", "gp", "cmaes", "nsgaii", "motpe"),
         default="tpe",
         help="BASELINE.json config to run (default: the headline TPE config)",
     )
@@ -227,6 +227,39 @@ def _run_alt_suite(args: argparse.Namespace) -> None:
             study.tell(t, float(np.sum(x**2)))
 
         config = {"model": "CmaEsSampler", "dims": dims, "parallelism": "single"}
+    elif args.suite == "motpe":
+        # Multi-objective TPE at a large history: the non-domination split runs
+        # through the K6 dominance-bitmatrix kernel on a GPU box (>=4096 rows).
+        n_hist = max(args.history, 6000)
+        sampler = optuna_amd.samplers.TPESampler(seed=0, n_startup_trials=10)
+        study = optuna_amd.create_study(directions=["minimize", "minimize"], sampler=sampler)
+        names = [f"x{i}" for i in range(10)]
+        dists = {n: optuna_amd.distributions.FloatDistribution(0.0, 1.0) for n in names}
+        pm = rng.uniform(0, 1, size=(n_hist, 10))
+        trials = [
+            optuna_amd.create_trial(
+                params={n: float(pm[r, i]) for i, n in enumerate(names)},
+                distributions=dists,
+                values=[float(pm[r, 0]), float(1.0 - pm[r, 0] + 0.1 * pm[r, 1])],
+            )
+            for r in range(n_hist)
+        ]
+        study.add_trials(trials)
+
+        def one_step() -> None:
+            t = study.ask()
+            x = np.array([t.suggest_float(n, 0, 1) for n in names])
+            f1 = float(x[0])
+            g = 1.0 + 9.0 * float(np.mean(x[1:]))
+            f2 = g * (1.0 - (f1 / g) ** 0.5)
+            study.tell(t, (f1, f2))
+
+        config = {
+            "model": "MO-TPE (2-objective ZDT1-like), K6 non-domination split",
+            "history_trials": n_hist,
+            "dims": 10,
+            "parallelism": "single",
+        }
     else:  # nsgaii
         # Config 4 (single-process variant): NSGA-II, 3-objective DTLZ2-like.
         sampler = optuna_amd.samplers.NSGAIISampler(seed=0, population_size=50)
