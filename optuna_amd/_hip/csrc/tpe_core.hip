@@ -825,15 +825,15 @@ int device_count() {
 // (a 20-block launch leaves >90% of the chip idle and was the top kernel in the
 // round-4 profile at 71 µs; the tiled scan reads the same bytes chip-wide).
 
-__global__ void k_compact_count(const int32_t* __restrict__ sorted_rows,  // (D, Nv) col-major
+__global__ void k_compact_count(const int32_t* __restrict__ sorted_rows,  // (D, s_stride)
                                 const int32_t* __restrict__ pos,          // (n_rows,)
-                                int64_t Nv, int64_t D,
+                                int64_t Nv, int64_t s_stride, int64_t D,
                                 int32_t* __restrict__ counts) {  // (D, n_tiles)
     const int64_t d = blockIdx.y;
     const int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     __shared__ int32_t red[256];
     int32_t flag = 0;
-    if (r < Nv) flag = (pos[sorted_rows[d * Nv + r]] >= 0) ? 1 : 0;
+    if (r < Nv) flag = (pos[sorted_rows[d * s_stride + r]] >= 0) ? 1 : 0;
     red[threadIdx.x] = flag;
     __syncthreads();
     for (int s = blockDim.x / 2; s > 0; s >>= 1) {
@@ -860,7 +860,7 @@ __global__ void k_compact_scan(int32_t* __restrict__ counts,  // (D, n_tiles)
 
 __global__ void k_compact_write(const int32_t* __restrict__ sorted_rows,
                                 const int32_t* __restrict__ pos, int64_t Nv,
-                                int64_t D,
+                                int64_t s_stride, int64_t D,
                                 const int32_t* __restrict__ offsets,  // (D, n_tiles)
                                 int32_t* __restrict__ sub_rows,  // (D, Nv stride)
                                 int32_t* __restrict__ sub_k) {
@@ -869,7 +869,7 @@ __global__ void k_compact_write(const int32_t* __restrict__ sorted_rows,
     __shared__ int32_t scan[256];
     int32_t row = -1, k = -1, flag = 0;
     if (r < Nv) {
-        row = sorted_rows[d * Nv + r];
+        row = sorted_rows[d * s_stride + r];
         k = pos[row];
         flag = (k >= 0) ? 1 : 0;
     }
@@ -1027,6 +1027,36 @@ __global__ void k_parzen_fit_table(const double* __restrict__ params,  // (n_row
     }
 }
 
+// Incremental insert into the device-resident per-dim sorted index: one block
+// per dim; rows are inserted sequentially (positions are final indices, i.e.
+// already offset by earlier inserts of the same batch). The tail shift is a
+// block-parallel backward memmove in descending chunks: within a chunk every
+// thread reads before any thread writes; a later (lower) chunk only writes
+// cells the earlier chunk has already read.
+__global__ void k_sorted_insert(int32_t* __restrict__ cols,  // (D, cap)
+                                int64_t cap, int64_t n_sorted,
+                                const int32_t* __restrict__ pos,   // (n_new, D)
+                                const int32_t* __restrict__ rows,  // (n_new, D)
+                                int64_t n_new, int64_t D) {
+    const int64_t d = blockIdx.x;
+    int32_t* col = cols + d * cap;
+    int64_t n = n_sorted;
+    for (int64_t i = 0; i < n_new; ++i) {
+        const int64_t p = pos[i * D + d];
+        for (int64_t hi = n; hi > p; hi -= blockDim.x) {
+            const int64_t j = hi - 1 - threadIdx.x;
+            int32_t v = 0;
+            if (j >= p) v = col[j];
+            __syncthreads();
+            if (j >= p) col[j + 1] = v;
+            __syncthreads();
+        }
+        if (threadIdx.x == 0) col[p] = rows[i * D + d];
+        __syncthreads();
+        ++n;
+    }
+}
+
 using arr_i32 = py::array_t<int32_t, py::array::c_style | py::array::forcecast>;
 
 class TpeDeviceHistory {
@@ -1035,6 +1065,7 @@ class TpeDeviceHistory {
 
     ~TpeDeviceHistory() {
         if (params_) (void)hipFree(params_);
+        if (sorted_) (void)hipFree(sorted_);
     }
 
     int64_t n_rows() const { return n_; }
@@ -1083,15 +1114,18 @@ class TpeDeviceHistory {
                               const arr_f64& extras_sorted,      // (D, L)
                               const arr_i32& extras_sorted_idx)  // (D, L)
     {
+        const bool resident_sorted = sorted_cols.empty() && sorted_valid_;
         const int64_t Nv =
-            sorted_cols.empty() ? 0 : (int64_t)sorted_cols[0].size();
+            resident_sorted ? n_sorted_
+                            : (sorted_cols.empty() ? 0
+                                                   : (int64_t)sorted_cols[0].size());
         const int64_t S = x.shape(0);
         const int64_t Na = n_above;
         const int64_t L = extras_raw.ndim() == 2 ? extras_raw.shape(0) : 0;
         const int64_t Nk = Na + L;  // mixture kernels excluding the prior
         const int64_t K = Nk + 1;
         if ((int64_t)pos.size() != n_ || x.shape(1) != D_ ||
-            (Nv > 0 && (int64_t)sorted_cols.size() != D_) ||
+            (!resident_sorted && Nv > 0 && (int64_t)sorted_cols.size() != D_) ||
             (int64_t)logw.size() != K)
             throw std::runtime_error("score: shape mismatch");
         if (L > 0 && ((int64_t)extras_sorted.size() != L * D_ ||
@@ -1144,8 +1178,10 @@ class TpeDeviceHistory {
         int32_t* d_counts = d_extra_idx + (size_t)L * D_;
 
         g_ws.begin_uploads();
-        for (int64_t d = 0; d < (int64_t)sorted_cols.size() && Nv > 0; ++d)
-            g_ws.h2d(d_sorted + d * Nv, sorted_cols[d].data(), (size_t)Nv * 4, st);
+        if (!resident_sorted)
+            for (int64_t d = 0; d < (int64_t)sorted_cols.size() && Nv > 0; ++d)
+                g_ws.h2d(d_sorted + d * Nv, sorted_cols[d].data(), (size_t)Nv * 4,
+                         st);
         if (n_ > 0) g_ws.h2d(d_pos, pos.data(), (size_t)n_ * 4, st);
         g_ws.h2d(d_logw, logw.data(), K * 8, st);
         g_ws.h2d(d_alow, alow.data(), D_ * 8, st);
@@ -1161,13 +1197,16 @@ class TpeDeviceHistory {
         }
 
         if (Na > 0) {
+            const int32_t* s_src = resident_sorted ? sorted_ : d_sorted;
+            const int64_t s_stride = resident_sorted ? sorted_cap_ : Nv;
             const dim3 grid((unsigned)n_tiles, (unsigned)D_);
-            hipLaunchKernelGGL(k_compact_count, grid, dim3(256), 0, st, d_sorted,
-                               d_pos, Nv, D_, d_counts);
+            hipLaunchKernelGGL(k_compact_count, grid, dim3(256), 0, st, s_src,
+                               d_pos, Nv, s_stride, D_, d_counts);
             hipLaunchKernelGGL(k_compact_scan, dim3((unsigned)D_), dim3(64), 0, st,
                                d_counts, n_tiles);
-            hipLaunchKernelGGL(k_compact_write, grid, dim3(256), 0, st, d_sorted,
-                               d_pos, Nv, D_, d_counts, d_sub_rows, d_sub_k);
+            hipLaunchKernelGGL(k_compact_write, grid, dim3(256), 0, st, s_src,
+                               d_pos, Nv, s_stride, D_, d_counts, d_sub_rows,
+                               d_sub_k);
         }
         const int32_t* fit_rows = d_sub_rows;
         const int32_t* fit_k = d_sub_k;
@@ -1204,11 +1243,79 @@ class TpeDeviceHistory {
         return out;
     }
 
+    // ---- device-resident per-dim sorted index ------------------------------
+
+    void upload_sorted(const std::vector<arr_i32>& cols) {
+        const int64_t Nv = cols.empty() ? 0 : (int64_t)cols[0].size();
+        if ((int64_t)cols.size() != D_ && Nv > 0)
+            throw std::runtime_error("upload_sorted: wrong number of columns");
+        hipStream_t st = g_ws.get_stream();
+        ensure_sorted_capacity(Nv, st);
+        g_ws.begin_uploads();
+        for (int64_t d = 0; d < (int64_t)cols.size() && Nv > 0; ++d)
+            g_ws.h2d(sorted_ + d * sorted_cap_, cols[d].data(), (size_t)Nv * 4, st);
+        g_ws.end_uploads_async(st);
+        n_sorted_ = Nv;
+        sorted_valid_ = true;
+    }
+
+    void insert_sorted(const arr_i32& pos, const arr_i32& rows) {
+        // pos (n_new, D): FINAL indices (already offset by earlier rows of the
+        // same batch); rows (n_new, D): per-dim row ids in per-dim value order.
+        if (!sorted_valid_) throw std::runtime_error("insert_sorted before upload");
+        const int64_t n_new = pos.ndim() == 2 ? pos.shape(0) : 0;
+        if (n_new == 0) return;
+        if (pos.shape(1) != D_ || rows.shape(0) != n_new || rows.shape(1) != D_)
+            throw std::runtime_error("insert_sorted: shape mismatch");
+        hipStream_t st = g_ws.get_stream();
+        ensure_sorted_capacity(n_sorted_ + n_new, st);
+        // small staging via the workspace device buffer tail is unsafe (ensure
+        // may realloc); use a dedicated tiny upload through pinned staging.
+        const size_t bytes = (size_t)n_new * D_ * 4;
+        double* scratch = g_ws.ensure((2 * bytes) / 8 + 8);
+        int32_t* d_pos = reinterpret_cast<int32_t*>(scratch);
+        int32_t* d_rows = d_pos + (size_t)n_new * D_;
+        g_ws.begin_uploads();
+        g_ws.h2d(d_pos, pos.data(), bytes, st);
+        g_ws.h2d(d_rows, rows.data(), bytes, st);
+        hipLaunchKernelGGL(k_sorted_insert, dim3((unsigned)D_), dim3(256), 0, st,
+                           sorted_, sorted_cap_, n_sorted_, d_pos, d_rows, n_new,
+                           D_);
+        // The insert kernel reads workspace memory that a later ensure() may
+        // reallocate; drain before returning (≈ the kernel's own few µs).
+        HIP_CHECK(hipStreamSynchronize(st));
+        HIP_CHECK(hipGetLastError());
+        n_sorted_ += n_new;
+    }
+
+    int64_t n_sorted() const { return sorted_valid_ ? n_sorted_ : -1; }
+
   private:
+    void ensure_sorted_capacity(int64_t need, hipStream_t st) {
+        if (need <= sorted_cap_) return;
+        const int64_t new_cap = std::max<int64_t>(1024, need * 2);
+        int32_t* grown = nullptr;
+        HIP_CHECK(hipMalloc(&grown, (size_t)new_cap * D_ * 4));
+        if (sorted_ && n_sorted_ > 0) {
+            HIP_CHECK(hipMemcpy2DAsync(grown, (size_t)new_cap * 4, sorted_,
+                                       (size_t)sorted_cap_ * 4,
+                                       (size_t)n_sorted_ * 4, (size_t)D_,
+                                       hipMemcpyDeviceToDevice, st));
+            HIP_CHECK(hipStreamSynchronize(st));
+        }
+        if (sorted_) (void)hipFree(sorted_);
+        sorted_ = grown;
+        sorted_cap_ = new_cap;
+    }
+
     int64_t D_;
     int64_t n_ = 0;
     int64_t capacity_ = 0;
     double* params_ = nullptr;
+    int32_t* sorted_ = nullptr;
+    int64_t n_sorted_ = 0;
+    int64_t sorted_cap_ = 0;
+    bool sorted_valid_ = false;
 };
 
 PYBIND11_MODULE(_hipcore, m) {
@@ -1230,6 +1337,10 @@ PYBIND11_MODULE(_hipcore, m) {
         .def(py::init<int64_t>(), py::arg("dims"))
         .def_property_readonly("n_rows", &TpeDeviceHistory::n_rows)
         .def("append", &TpeDeviceHistory::append, py::arg("block"))
+        .def("upload_sorted", &TpeDeviceHistory::upload_sorted, py::arg("cols"))
+        .def("insert_sorted", &TpeDeviceHistory::insert_sorted, py::arg("pos"),
+             py::arg("rows"))
+        .def_property_readonly("n_sorted", &TpeDeviceHistory::n_sorted)
         .def("score", &TpeDeviceHistory::score, py::arg("sorted_rows"),
              py::arg("pos"), py::arg("n_above"), py::arg("logw"), py::arg("alow"),
              py::arg("ahigh"), py::arg("steps"), py::arg("n_choices"),

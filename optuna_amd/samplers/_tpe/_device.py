@@ -129,6 +129,10 @@ class _SpaceDeviceMirror:
         ) = _space_domains(space)
         self._hist = core.TpeDeviceHistory(len(space))
         self._n_appended = 0
+        # Device-resident sorted index bookkeeping: entries of the host cache's
+        # insert log already replayed, and the resident index length.
+        self._sorted_log_cursor = 0
+        self._n_sorted_synced = -1
 
     def sync(self, cache: "_SpaceCache") -> None:
         n_total = len(cache.valid)
@@ -139,6 +143,27 @@ class _SpaceDeviceMirror:
                     block[:, self._is_log] = np.log(block[:, self._is_log])
             self._hist.append(np.ascontiguousarray(block))
             self._n_appended = n_total
+        self._sync_sorted(cache)
+
+    def _sync_sorted(self, cache: "_SpaceCache") -> None:
+        log = cache._insert_log
+        if self._n_sorted_synced == cache._n_sorted:
+            return
+        pending = log[self._sorted_log_cursor :]
+        # Bulk (re-)upload when starting fresh or when a large batch arrived
+        # (the sequential device insert is O(batch · N)); per-tell single-row
+        # inserts replay incrementally.
+        if (
+            self._n_sorted_synced < 0
+            or len(pending) > 16
+            or any(len(rows) > 64 for _, rows in pending)
+        ):
+            self._hist.upload_sorted(list(cache.sorted_rows))
+        else:
+            for pos, rows in pending:
+                self._hist.insert_sorted(pos, rows)
+        self._sorted_log_cursor = len(log)
+        self._n_sorted_synced = cache._n_sorted
 
     def score(
         self,
@@ -162,9 +187,13 @@ class _SpaceDeviceMirror:
         n_total = len(cache.valid)
         pos = np.full(n_total, -1, dtype=np.int32)
         pos[sel] = np.arange(len(sel), dtype=np.int32)
-        # Per-dim contiguous int32 prefixes go to the device as-is — no
-        # (Nv, D) restack per suggest (was the top host cost on the GPU box).
-        sorted_cols = list(cache.sorted_rows)
+        # The per-dim sorted index is device-resident (kept current by
+        # _sync_sorted); pass no columns unless the resident copy somehow
+        # lagged, in which case fall back to a one-off upload.
+        if self._n_sorted_synced == cache._n_sorted:
+            sorted_cols: list = []
+        else:
+            sorted_cols = list(cache.sorted_rows)
         x_raw = np.column_stack(
             [np.asarray(samples[n], dtype=np.float64) for n in cache.names]
         )

@@ -62,6 +62,11 @@ class _SpaceCache:
         self._n_sorted = 0
         self._vals_buf = [np.empty(64, dtype=np.float64) for _ in self.names]
         self._rows_buf = [np.empty(64, dtype=np.int32) for _ in self.names]
+        # Insert log for the device-resident sorted index: per append, the
+        # FINAL insert position and row id per dim (consumed by the GPU mirror
+        # so each suggest replays O(new) inserts instead of re-uploading the
+        # whole (Nv, D) order).
+        self._insert_log: list[tuple[np.ndarray, np.ndarray]] = []
 
     @property
     def params(self) -> np.ndarray:
@@ -120,6 +125,8 @@ class _SpaceCache:
                 rb[:n] = self._rows_buf[c][:n]
                 self._vals_buf[c] = vb
                 self._rows_buf[c] = rb
+        log_pos = np.empty((m, len(self.names)), dtype=np.int32)
+        log_rows = np.empty((m, len(self.names)), dtype=np.int32)
         for c in range(len(self.names)):
             vb = self._vals_buf[c]
             rb = self._rows_buf[c]
@@ -131,6 +138,8 @@ class _SpaceCache:
                 rb[i + 1 : n + 1] = rb[i:n].copy()
                 vb[i] = vals[0]
                 rb[i] = new_rows[0]
+                log_pos[0, c] = i
+                log_rows[0, c] = new_rows[0]
             else:
                 order = np.argsort(vals, kind="stable")
                 vals_sorted = vals[order]
@@ -140,6 +149,11 @@ class _SpaceCache:
                 merged_r = np.insert(rb[:n], pos, rows_sorted)
                 vb[: n + m] = merged_v
                 rb[: n + m] = merged_r
+                # Final indices: np.insert places the j-th inserted value at
+                # pos[j] + j in the merged array.
+                log_pos[:, c] = pos + np.arange(m)
+                log_rows[:, c] = rows_sorted
+        self._insert_log.append((log_pos, log_rows))
         self._n_sorted = n + m
 
 
