@@ -135,13 +135,27 @@ class RcclStorage(BaseStorage):
         )
         rank = int(os.environ.get("RANK", "0"))
         world_size = int(os.environ.get("WORLD_SIZE", "1"))
-        store = TCPStore(
-            host,
-            port,
-            world_size,
-            is_master=(rank == 0),
-            timeout=datetime.timedelta(seconds=120),
-        )
+        # Back-to-back launches (the driver's 1/2/4/8 scaling sweep) can leave
+        # the previous master's port briefly unbindable; retry a few times.
+        import time as _time
+
+        last_err: Exception | None = None
+        store = None
+        for _attempt in range(5):
+            try:
+                store = TCPStore(
+                    host,
+                    port,
+                    world_size,
+                    is_master=(rank == 0),
+                    timeout=datetime.timedelta(seconds=120),
+                )
+                break
+            except (RuntimeError, OSError) as e:
+                last_err = e
+                _time.sleep(2.0)
+        if store is None:
+            raise RuntimeError(f"TCPStore init failed on {host}:{port}: {last_err}")
         storage = cls(store, worker_label=f"rank{rank}")
         storage._store = store  # keep the master store object alive
         return storage
