@@ -195,6 +195,13 @@ class Trial(BaseTrial):
                 f"The `value` argument is of type '{type(value).__name__}' but supposed to be a "
                 "float."
             ) from e
+        try:
+            step = int(step)
+        except (TypeError, ValueError) as e:
+            raise TypeError(
+                f"The `step` argument is of type '{type(step).__name__}' but supposed "
+                "to be an int."
+            ) from e
         if step < 0:
             raise ValueError(f"The `step` argument is {step} but cannot be negative.")
         if step in self._cached_frozen_trial.intermediate_values:
