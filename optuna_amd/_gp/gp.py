@@ -166,7 +166,12 @@ class GPRegressor:
             sqd = (X1 - X2 if X1.ndim == 1 else X1.unsqueeze(-2) - X2.unsqueeze(-3)).square_()
             if self._is_categorical.any():
                 sqd[..., self._is_categorical] = (sqd[..., self._is_categorical] > 0.0).double()
-        sqdist = sqd.matmul(self.inverse_squared_lengthscales)
+        if sqd.is_cuda:
+            # rocBLAS's (…, N, D) @ (D,) gemv path measures ~100× slower than a
+            # broadcast reduction at these tall-skinny fp64 shapes.
+            sqdist = (sqd * self.inverse_squared_lengthscales).sum(-1)
+        else:
+            sqdist = sqd.matmul(self.inverse_squared_lengthscales)
         return matern52_of_sqdist(sqdist) * self.kernel_scale
 
     def _cache_matrix(self) -> None:
