@@ -170,3 +170,55 @@ def test_dominates_semantics() -> None:
     assert _dominates(t([0.0, 2.0]), t([1.0, 1.0]), d2)
     assert not _dominates(t([0.0, 1.0]), t([1.0, 2.0]), d2)
     assert not _dominates(t([0.0, 1.0]), t([0.0, 1.0]), d2)
+
+
+def test_optimize_catch_and_failed_state() -> None:
+    study = optuna_amd.create_study()
+
+    def obj(t):
+        t.suggest_float("x", 0, 1)
+        if t.number % 2 == 0:
+            raise ValueError("boom")
+        return 1.0
+
+    study.optimize(obj, n_trials=6, catch=(ValueError,))
+    states = [t.state for t in study.trials]
+    assert states.count(TrialState.FAIL) == 3
+    assert states.count(TrialState.COMPLETE) == 3
+    # without catch, the exception propagates
+    with pytest.raises(ValueError):
+        study.optimize(obj, n_trials=1)
+
+
+def test_optimize_callbacks_see_final_state() -> None:
+    seen = []
+    study = optuna_amd.create_study()
+
+    def cb(st, trial):
+        seen.append((trial.number, trial.state))
+
+    def obj(t):
+        t.suggest_float("x", 0, 1)
+        if t.number == 1:
+            raise optuna_amd.TrialPruned()
+        return 0.0
+
+    study.optimize(obj, n_trials=3, callbacks=[cb])
+    assert [s for _, s in seen] == [
+        TrialState.COMPLETE,
+        TrialState.PRUNED,
+        TrialState.COMPLETE,
+    ]
+
+
+def test_fail_reason_system_attr_recorded() -> None:
+    study = optuna_amd.create_study()
+
+    def obj(t):
+        t.suggest_float("x", 0, 1)
+        raise RuntimeError("xyz failure")
+
+    study.optimize(obj, n_trials=1, catch=(RuntimeError,))
+    t = study.trials[0]
+    assert t.state == TrialState.FAIL
+    assert "xyz failure" in str(t.system_attrs.get("fail_reason", ""))
