@@ -105,6 +105,12 @@ def test_backends_agree_on_random_sequences(tmp_path, seed) -> None:
         timeout=datetime.timedelta(seconds=30),
     )
     results["rccl"] = _apply_sequence(RcclStorage(store), seed)
+    from optuna_amd.storages import RDBStorage
+
+    results["rdb"] = _apply_sequence(
+        RDBStorage(f"sqlite:///{tmp_path}/f{seed}.db"), seed
+    )
 
     assert results["inmemory"] == results["journal"], "inmemory vs journal"
     assert results["inmemory"] == results["rccl"], "inmemory vs rccl"
+    assert results["inmemory"] == results["rdb"], "inmemory vs rdb"
