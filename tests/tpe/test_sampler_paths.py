@@ -45,8 +45,12 @@ def test_same_suggestions_across_storages_and_cache_paths(tmp_path) -> None:
     journal = run(
         JournalStorage(JournalFileBackend(str(tmp_path / "j.jsonl"))), False
     )
+    from optuna_amd.storages import RDBStorage
+
+    rdb = run(RDBStorage(f"sqlite:///{tmp_path}/s.db"), False)
     no_caches = run(None, True)
     np.testing.assert_allclose(base, journal)
+    np.testing.assert_allclose(base, rdb)
     np.testing.assert_allclose(base, no_caches)
 
 
