@@ -65,6 +65,14 @@ def __getattr__(name: str):  # lazy to avoid import cycles / optional deps
         from optuna_amd.storages._grpc.server import run_grpc_proxy_server
 
         return run_grpc_proxy_server
+    if name == "GrpcWireStorageProxy":
+        from optuna_amd.storages._grpc.wire_client import GrpcWireStorageProxy
+
+        return GrpcWireStorageProxy
+    if name == "run_grpc_wire_proxy_server":
+        from optuna_amd.storages._grpc.wire_server import run_grpc_wire_proxy_server
+
+        return run_grpc_wire_proxy_server
     if name == "RetryFailedTrialCallback":
         from optuna_amd._callbacks import RetryFailedTrialCallback
 
@@ -130,6 +138,8 @@ __all__ = [
     "RcclStorage",
     "GrpcStorageProxy",
     "run_grpc_proxy_server",
+    "GrpcWireStorageProxy",
+    "run_grpc_wire_proxy_server",
     "RetryFailedTrialCallback",
     "fail_stale_trials",
     "BaseHeartbeat",
