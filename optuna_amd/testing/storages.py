@@ -23,6 +23,7 @@ STORAGE_MODES: list[str] = [
     "journal",
     "grpc_sqlite",
     "grpc_journal",
+    "grpc_wire",
 ]
 
 SQLITE3_TIMEOUT = 300
@@ -48,6 +49,8 @@ class StorageSupplier:
             if len(self.extra_args) > 0:
                 raise ValueError("InMemoryStorage does not accept any arguments!")
             return optuna_amd.storages.InMemoryStorage()
+        elif self.storage_specifier == "grpc_wire":
+            return self._start_grpc(optuna_amd.storages.InMemoryStorage())
         elif "sqlite" in self.storage_specifier or "journal" in self.storage_specifier:
             if self.storage_specifier in ("journal", "grpc_journal"):
                 self.tempfile = tempfile.NamedTemporaryFile(suffix=".log", delete=False)
@@ -77,10 +80,18 @@ class StorageSupplier:
             raise ValueError(f"Unknown storage specifier {self.storage_specifier}")
 
     def _start_grpc(self, backend: BaseStorage) -> BaseStorage:
+        port = _find_free_port()
+        if self.storage_specifier == "grpc_wire":
+            from optuna_amd.storages._grpc.wire_client import GrpcWireStorageProxy
+            from optuna_amd.storages._grpc.wire_server import make_wire_server
+
+            self.server = make_wire_server(backend, "127.0.0.1", port)
+            self.server.start()
+            self.proxy = GrpcWireStorageProxy(host="127.0.0.1", port=port)
+            return self.proxy
         from optuna_amd.storages._grpc.client import GrpcStorageProxy
         from optuna_amd.storages._grpc.server import make_server
 
-        port = _find_free_port()
         self.server = make_server(backend, "127.0.0.1", port)
         self.server.start()
         self.proxy = GrpcStorageProxy(host="127.0.0.1", port=port)

@@ -44,6 +44,15 @@ class TestGrpcSqliteStorage(StorageTestCase):
             yield s
 
 
+class TestGrpcWireStorage(StorageTestCase):
+    """Wire-protocol (reference api.proto) proxy over an in-memory backend."""
+
+    @pytest.fixture
+    def storage(self):  # type: ignore[override]
+        with StorageSupplier("grpc_wire") as s:
+            yield s
+
+
 def test_finished_trials_since_inmemory_and_journal(tmp_path) -> None:
     import optuna_amd
     from optuna_amd.study import StudyDirection
