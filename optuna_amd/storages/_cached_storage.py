@@ -162,7 +162,9 @@ class _CachedStorage(BaseStorage, BaseHeartbeat):
             if mapped is not None:
                 study_id, number = mapped
                 trial = self._studies[study_id].trials.get(number)
-                if trial is not None:
+                # Only finished trials are immutable; an unfinished cached
+                # snapshot can be stale (reference _cached_storage.py:207-223).
+                if trial is not None and trial.state.is_finished():
                     return copy.deepcopy(trial)
         return self._backend.get_trial(trial_id)
 

@@ -90,3 +90,38 @@ def test_finished_trials_since_inmemory_and_journal(tmp_path) -> None:
         storage.create_new_trial(sid, template_trial=t)
         delta = storage.get_finished_trials_since(sid, 2)
         assert len(delta) == 1 and delta[0].value == 3.0
+
+
+def test_heartbeat_fail_stale_and_retry(tmp_path) -> None:
+    """Heartbeat-enabled RDB: a stale RUNNING trial is failed and re-enqueued by
+    the retry callback (reference storages/_heartbeat.py:156 semantics)."""
+    import time
+
+    import optuna_amd
+    from optuna_amd.storages import RDBStorage, RetryFailedTrialCallback, fail_stale_trials
+    from optuna_amd.trial import TrialState
+
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        storage = RDBStorage(
+            f"sqlite:///{tmp_path}/hb.db",
+            heartbeat_interval=1,
+            grace_period=1,
+            failed_trial_callback=RetryFailedTrialCallback(max_retry=2),
+        )
+    study = optuna_amd.create_study(storage=storage, study_name="hb")
+    t = study.ask()
+    t.suggest_float("x", 0, 1)
+    storage.record_heartbeat(t._trial_id)
+    time.sleep(2.1)  # heartbeat is now stale (grace 1s past 1s interval)
+    fail_stale_trials(study)
+    trials = study.get_trials(deepcopy=False)
+    states = [tr.state for tr in trials]
+    assert TrialState.FAIL in states
+    # the retry callback enqueued a WAITING clone with the same params
+    waiting = [tr for tr in trials if tr.state == TrialState.WAITING]
+    assert len(waiting) == 1
+    assert waiting[0].params == t.params
+    assert RetryFailedTrialCallback.retried_trial_number(waiting[0]) == 0
