@@ -48,9 +48,13 @@ def test_same_suggestions_across_storages_and_cache_paths(tmp_path) -> None:
     from optuna_amd.storages import RDBStorage
 
     rdb = run(RDBStorage(f"sqlite:///{tmp_path}/s.db"), False)
+    from optuna_amd.storages._cached_storage import _CachedStorage
+
+    cached = run(_CachedStorage(RDBStorage(f"sqlite:///{tmp_path}/c.db")), False)
     no_caches = run(None, True)
     np.testing.assert_allclose(base, journal)
     np.testing.assert_allclose(base, rdb)
+    np.testing.assert_allclose(base, cached)
     np.testing.assert_allclose(base, no_caches)
 
 
