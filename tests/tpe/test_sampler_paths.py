@@ -83,3 +83,64 @@ def test_tpe_n_jobs_threaded_consistency() -> None:
         for c in range(len(cache.names)):
             sv = cache.sorted_vals[c]
             assert np.all(np.diff(sv) >= 0)
+
+
+def test_split_with_pruned_and_failed_matches_reference_semantics() -> None:
+    """Mirror split over a history with COMPLETE/PRUNED(+intermediates)/infeasible
+    trials: below has gamma members, partitions are disjoint, pruned ordering
+    follows (-last_step, value) after completes."""
+    import optuna_amd
+    from optuna_amd.samplers._tpe.sampler import default_gamma
+    from optuna_amd.trial import TrialState
+
+    rng = np.random.RandomState(0)
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.TPESampler(seed=1, n_startup_trials=5)
+    )
+    dists = {"x": FloatDistribution(-5.0, 5.0)}
+    trials = []
+    for j in range(300):
+        r = j % 4
+        if r == 0:
+            t = optuna_amd.create_trial(
+                params={"x": float(rng.uniform(-5, 5))}, distributions=dists,
+                value=float(rng.rand()),
+            )
+        elif r == 1:
+            t = optuna_amd.create_trial(
+                params={"x": float(rng.uniform(-5, 5))}, distributions=dists,
+                state=TrialState.PRUNED,
+                intermediate_values={int(rng.randint(1, 5)): float(rng.rand())},
+            )
+        elif r == 2:
+            t = optuna_amd.create_trial(
+                params={"x": float(rng.uniform(-5, 5))}, distributions=dists,
+                value=float(rng.rand()),
+                system_attrs={"constraints": [float(rng.rand())]},  # infeasible
+            )
+        else:
+            t = optuna_amd.create_trial(
+                params={"x": float(rng.uniform(-5, 5))}, distributions=dists,
+                state=TrialState.PRUNED, intermediate_values={},
+            )
+        trials.append(t)
+    study.add_trials(trials)
+
+    # Drive one suggest so the mirror builds, then inspect the split directly.
+    tr = study.ask()
+    tr.suggest_float("x", -5, 5)
+    study.tell(tr, 0.5)
+    hist = study.sampler._histories[study._study_id]
+    n = len(hist)
+    below, above = hist.split(study, default_gamma(n))
+    assert len(below) == default_gamma(n)
+    assert len(set(below) & set(above)) == 0
+    assert len(below) + len(above) == n
+    # below must be filled by feasible completes first (best values)
+    vals = hist._values[below, 0]
+    states = hist._states[below]
+    assert all(s == int(TrialState.COMPLETE) for s in states)
+    comp_vals = hist._values[
+        (hist._states == int(TrialState.COMPLETE)) & ~(hist._violations > 0), 0
+    ]
+    assert np.nanmax(vals) <= np.sort(comp_vals)[len(below) - 1] + 1e-12
