@@ -185,6 +185,11 @@ class _TpeHistory:
         self._comp_vals = np.empty(0, dtype=np.float64)
         self._comp_rows = np.empty(0, dtype=np.int64)
         self._comp_cache_valid = True
+        # Split memo: independent-mode TPE (e.g. multi-objective) runs one
+        # _sample per dimension within a single suggest; the history is
+        # unchanged between them, so the (rows, gamma) split — including the
+        # K6 non-domination rank — is identical and cached per (n, n_below).
+        self._split_memo: tuple[int, int, np.ndarray, np.ndarray] | None = None
 
     def __len__(self) -> int:
         return self._m_n
@@ -341,6 +346,10 @@ class _TpeHistory:
         """
         directions = study.directions
         n = len(self._numbers)
+        n_below_arg = n_below  # n_below is consumed section-by-section below
+        memo = self._split_memo
+        if memo is not None and memo[0] == n and memo[1] == n_below_arg:
+            return memo[2], memo[3]
         rows = np.arange(n)
         is_infeasible = self._violations > 0
         is_complete = (self._states == int(TrialState.COMPLETE)) & ~is_infeasible
@@ -420,6 +429,7 @@ class _TpeHistory:
                 np.concatenate(above_parts) if above_parts else np.empty(0, dtype=np.int64)
             )
             above = above[np.argsort(self._numbers[above], kind="stable")]
+        self._split_memo = (n, n_below_arg, below, above)
         return below, above
 
     def _split_complete_mo(
