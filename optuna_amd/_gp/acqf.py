@@ -28,32 +28,45 @@ else:
 
     torch = _LazyImport("torch")
 
-_SQRT_HALF = math.sqrt(0.5)
-_INV_SQRT_2PI = 1 / math.sqrt(2 * math.pi)
-_SQRT_HALF_PI = math.sqrt(0.5 * math.pi)
-_LOG_SQRT_2PI = math.log(math.sqrt(2 * math.pi))
 _EPS = 1e-12  # zero eps makes gradients NaN
+_HALF_LOG_2PI = 0.5 * math.log(2.0 * math.pi)
+# Below this z the two terms of z*Phi(z)+phi(z) cancel; switch to the erfcx form.
+_TAIL_Z = -1.0
 
 
 def standard_logei(z: "torch.Tensor") -> "torch.Tensor":
-    """log E_{x~N(0,1)}[max(0, x+z)], tail-stable (erfcx in the far left tail)."""
-    out = (
-        (z_half := 0.5 * z) * torch.special.erfc(-_SQRT_HALF * z)
-        + (-z_half * z).exp() * _INV_SQRT_2PI
-    ).log()
-    small = z < -25
-    if (z_small := z[small]).numel():
-        out[small] = (
-            -0.5 * z_small**2
-            - _LOG_SQRT_2PI
-            + (1 + _SQRT_HALF_PI * z_small * torch.special.erfcx(-_SQRT_HALF * z_small)).log()
-        )
-    return out
+    """log E_{x~N(0,1)}[max(0, x+z)] = log(z*Phi(z) + phi(z)), tail-stable.
+
+    Two algebraically-equivalent evaluations selected by ``torch.where``:
+
+    * right half (z >= _TAIL_Z): the textbook sum, whose terms are the same
+      order of magnitude there, logged directly;
+    * left half: factor phi(z) out first.  Since Phi(z) = phi(z) *
+      sqrt(pi/2) * erfcx(-z/sqrt(2)), the whole expectation is
+      phi(z) * (1 + sqrt(pi/2) * z * erfcx(-z/sqrt(2))) and its log is the
+      log-density plus a log1p of a quantity in (-1, 0] — no cancellation
+      however deep the tail.
+
+    Both branches are evaluated on range-clamped copies of ``z`` so the
+    inactive branch can neither overflow nor poison gradients through where().
+    """
+    inv_sqrt2 = math.sqrt(0.5)
+    zr = z.clamp(min=_TAIL_Z - 0.5)
+    phi_r = torch.exp(-0.5 * zr * zr - _HALF_LOG_2PI)
+    right = torch.log(zr * torch.special.ndtr(zr) + phi_r)
+
+    zl = z.clamp(max=_TAIL_Z + 0.5)
+    scaled_mills = math.sqrt(0.5 * math.pi) * zl * torch.special.erfcx(-zl * inv_sqrt2)
+    left = (-0.5 * zl * zl - _HALF_LOG_2PI) + torch.log1p(scaled_mills)
+
+    return torch.where(z >= _TAIL_Z, right, left)
 
 
 def logei(mean: "torch.Tensor", var: "torch.Tensor", f0: float) -> "torch.Tensor":
-    sigma = var.sqrt_()
-    return standard_logei((mean - f0) / sigma) + sigma.log()
+    """log E_{y~N(mean, var)}[max(0, y - f0)] elementwise."""
+    half_log_var = 0.5 * torch.log(var)
+    z = (mean - f0) * torch.exp(-half_log_var)
+    return half_log_var + standard_logei(z)
 
 
 def logehvi(
@@ -78,22 +91,50 @@ def _per_sample_log_hvi(
 
 
 def _get_reference_point(Y: "torch.Tensor") -> np.ndarray:
-    loss_vals = -Y.numpy()
-    ref = np.max(loss_vals, axis=0)
-    return np.nextafter(np.maximum(1.1 * ref, 0.9 * ref), np.inf)
+    # Work in loss (minimization) space: worst observed loss per objective,
+    # pushed 10% further from the origin, then one ulp more so every kept
+    # point is strictly inside the reference box.
+    worst = np.asarray((-Y).amax(dim=0))
+    pushed = np.where(worst >= 0.0, 1.1 * worst, 0.9 * worst)
+    return np.nextafter(pushed, np.inf)
 
 
 def _get_boxes(Y: "torch.Tensor", ref_point: np.ndarray) -> tuple["torch.Tensor", "torch.Tensor"]:
-    loss_vals = -Y.numpy()
-    loss_vals = loss_vals[np.all(loss_vals < ref_point, axis=-1)]
-    pareto = loss_vals[_is_pareto_front(loss_vals, assume_unique_lexsorted=False)]
-    lbs, ubs = get_non_dominated_box_bounds(pareto, ref_point)
-    return torch.from_numpy(-ubs), torch.from_numpy(-lbs)
+    """Non-dominated boxes of -Y under ``ref_point``, returned in max space.
+
+    Negating swaps the roles of the bounds, so the decomposition's (lower,
+    upper) pairs come back as (-upper, -lower).
+    """
+    losses = np.asarray(-Y)
+    kept = losses[(losses < ref_point).all(axis=-1)]
+    front = kept[_is_pareto_front(kept, assume_unique_lexsorted=False)]
+    box_lo, box_hi = get_non_dominated_box_bounds(front, ref_point)
+    return torch.from_numpy(-box_hi), torch.from_numpy(-box_lo)
+
+
+def _fantasize(
+    gpr: GPRegressor,
+    X_running: np.ndarray,
+    n_qmc_samples: int,
+    qmc_seed: int,
+    stabilizing_noise: float,
+) -> ConditionalGPRegressor:
+    """Condition ``gpr`` on QMC fantasy outcomes at the running trials' points."""
+    return ConditionalGPRegressor(
+        gpr=gpr,
+        X_running=torch.from_numpy(X_running),
+        n_qmc_samples=n_qmc_samples,
+        qmc_seed=qmc_seed,
+        stabilizing_noise=stabilizing_noise,
+    )
 
 
 def _mean_max_log_utility(log_utils: "torch.Tensor") -> "torch.Tensor":
-    max_in_q = torch.amax(log_utils, dim=-1)
-    return torch.special.logsumexp(max_in_q, dim=-1) - math.log(max_in_q.shape[-1])
+    # MC estimate in log space: per posterior draw keep the best of the
+    # q-batch, then average the draws.
+    per_draw_best = log_utils.amax(dim=-1)
+    n_draws = per_draw_best.shape[-1]
+    return per_draw_best.logsumexp(dim=-1) - math.log(n_draws)
 
 
 class BaseAcquisitionFunc(ABC):
@@ -165,18 +206,14 @@ class qLogEI(BaseAcquisitionFunc):
         gpr: GPRegressor,
         search_space: "SearchSpace",
         threshold: float,
+        normalized_params_of_running_trials: np.ndarray,
         n_qmc_samples: int,
         qmc_seed: int,
-        normalized_params_of_running_trials: np.ndarray,
         stabilizing_noise: float = 1e-12,
     ) -> None:
         self._threshold = threshold
-        self._cond_gpr = ConditionalGPRegressor(
-            gpr=gpr,
-            X_running=torch.from_numpy(normalized_params_of_running_trials),
-            n_qmc_samples=n_qmc_samples,
-            qmc_seed=qmc_seed,
-            stabilizing_noise=stabilizing_noise,
+        self._cond_gpr = _fantasize(
+            gpr, normalized_params_of_running_trials, n_qmc_samples, qmc_seed, stabilizing_noise
         )
         n_running = len(normalized_params_of_running_trials)
         self._per_sample_shape = (n_qmc_samples, n_running + 1)
@@ -222,20 +259,16 @@ class qLogPI(BaseAcquisitionFunc):
         gpr: GPRegressor,
         search_space: "SearchSpace",
         threshold: float,
+        normalized_params_of_running_trials: np.ndarray,
         n_qmc_samples: int,
         qmc_seed: int,
-        normalized_params_of_running_trials: np.ndarray,
         stabilizing_noise: float = 1e-12,
         tau: float = 1e-2,
     ) -> None:
         self._threshold = threshold
         self._tau = tau
-        self._cond_gpr = ConditionalGPRegressor(
-            gpr=gpr,
-            X_running=torch.from_numpy(normalized_params_of_running_trials),
-            n_qmc_samples=n_qmc_samples,
-            qmc_seed=qmc_seed,
-            stabilizing_noise=stabilizing_noise,
+        self._cond_gpr = _fantasize(
+            gpr, normalized_params_of_running_trials, n_qmc_samples, qmc_seed, stabilizing_noise
         )
         super().__init__(gpr.length_scales, search_space)
 
@@ -303,24 +336,19 @@ class qLogCEI(BaseAcquisitionFunc):
         gpr: GPRegressor,
         search_space: "SearchSpace",
         threshold: float,
+        normalized_params_of_running_trials: np.ndarray,
         n_qmc_samples: int,
         qmc_seed: int,
         constraints_gpr_list: list[GPRegressor],
         constraints_threshold_list: list[float],
-        normalized_params_of_running_trials: np.ndarray,
         stabilizing_noise: float = 1e-12,
     ) -> None:
         assert constraints_gpr_list and len(constraints_gpr_list) == len(
             constraints_threshold_list
         )
         self._acqf = qLogEI(
-            gpr,
-            search_space,
-            threshold,
-            n_qmc_samples,
-            qmc_seed,
-            normalized_params_of_running_trials,
-            stabilizing_noise,
+            gpr, search_space, threshold,
+            normalized_params_of_running_trials, n_qmc_samples, qmc_seed, stabilizing_noise,
         )
         self._constraints_acqf_list = [
             qLogPI(
@@ -351,9 +379,10 @@ class LogEHVI(BaseAcquisitionFunc):
         gpr_list: list[GPRegressor],
         search_space: "SearchSpace",
         Y_train: "torch.Tensor",
+        *,
+        normalized_params_of_running_trials: np.ndarray | None = None,
         n_qmc_samples: int,
         qmc_seed: int,
-        normalized_params_of_running_trials: np.ndarray | None = None,
         stabilizing_noise: float = 1e-12,
     ) -> None:
         self._stabilizing_noise = stabilizing_noise
@@ -391,19 +420,15 @@ class qLogEHVI(BaseAcquisitionFunc):
         gpr_list: list[GPRegressor],
         search_space: "SearchSpace",
         Y_train: "torch.Tensor",
+        normalized_params_of_running_trials: np.ndarray,
         n_qmc_samples: int,
         qmc_seed: int,
-        normalized_params_of_running_trials: np.ndarray,
         stabilizing_noise: float = 1e-12,
     ) -> None:
         self._Y_train = Y_train
         self._cond_gpr_list = [
-            ConditionalGPRegressor(
-                gpr=gpr,
-                X_running=torch.from_numpy(normalized_params_of_running_trials),
-                n_qmc_samples=n_qmc_samples,
-                qmc_seed=qmc_seed + i,
-                stabilizing_noise=stabilizing_noise,
+            _fantasize(
+                gpr, normalized_params_of_running_trials, n_qmc_samples, qmc_seed + i, stabilizing_noise
             )
             for i, gpr in enumerate(gpr_list)
         ]
@@ -448,11 +473,12 @@ class LogCEHVI(BaseAcquisitionFunc):
         gpr_list: list[GPRegressor],
         search_space: "SearchSpace",
         Y_feasible: "torch.Tensor | None",
+        *,
+        normalized_params_of_running_trials: np.ndarray | None = None,
         n_qmc_samples: int,
         qmc_seed: int,
         constraints_gpr_list: list[GPRegressor],
         constraints_threshold_list: list[float],
-        normalized_params_of_running_trials: np.ndarray | None = None,
         stabilizing_noise: float = 1e-12,
     ) -> None:
         assert constraints_gpr_list and len(constraints_gpr_list) == len(

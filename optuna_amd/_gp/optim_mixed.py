@@ -205,16 +205,16 @@ def local_search_mixed_batched(
 def optimize_acqf_mixed(
     acqf: "BaseAcquisitionFunc",
     *,
-    warmstart_normalized_params_array: np.ndarray | None = None,
+    warmstart_points: np.ndarray | None = None,
     n_preliminary_samples: int = 2048,
     n_local_search: int = 10,
     tol: float = 1e-4,
     rng: np.random.RandomState | None = None,
 ) -> tuple[np.ndarray, float]:
     rng = rng or np.random.RandomState()
-    if warmstart_normalized_params_array is None:
-        warmstart_normalized_params_array = np.empty((0, acqf.search_space.dim))
-    assert len(warmstart_normalized_params_array) <= n_local_search - 1
+    if warmstart_points is None:
+        warmstart_points = np.empty((0, acqf.search_space.dim))
+    assert len(warmstart_points) <= n_local_search - 1
 
     sampled_xs = acqf.search_space.sample_normalized_params(n_preliminary_samples, rng=rng)
     f_vals = acqf.eval_acqf_no_grad(sampled_xs)
@@ -226,7 +226,7 @@ def optimize_acqf_mixed(
     probs /= probs.sum()
     n_improving = int(np.count_nonzero(probs > 0.0))
     n_additional = min(
-        n_local_search - len(warmstart_normalized_params_array) - 1, n_improving
+        n_local_search - len(warmstart_points) - 1, n_improving
     )
     if n_additional == n_improving:
         _logger.warning("Study already converged, so the number of local search is reduced.")
@@ -235,7 +235,7 @@ def optimize_acqf_mixed(
         chosen = np.append(
             chosen, rng.choice(len(sampled_xs), size=n_additional, replace=False, p=probs)
         )
-    x_warmstarts = np.vstack([sampled_xs[chosen, :], warmstart_normalized_params_array])
+    x_warmstarts = np.vstack([sampled_xs[chosen, :], warmstart_points])
     best_xs, best_fvals = local_search_mixed_batched(acqf, x_warmstarts, tol=tol)
     best_idx = int(np.argmax(best_fvals))
     return best_xs[best_idx], float(best_fvals[best_idx])

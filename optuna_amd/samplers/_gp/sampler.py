@@ -151,15 +151,15 @@ class GPSampler(BaseSampler):
         self, acqf: acqf_module.BaseAcquisitionFunc, best_params: np.ndarray | None
     ) -> np.ndarray:
         assert best_params is None or best_params.ndim == 2
-        normalized_params, _ = optim_mixed.optimize_acqf_mixed(
+        found, _ = optim_mixed.optimize_acqf_mixed(
             acqf,
-            warmstart_normalized_params_array=best_params,
-            n_preliminary_samples=self._n_preliminary_samples,
-            n_local_search=self._n_local_search,
-            tol=self._tol,
+            warmstart_points=best_params,
             rng=self._rng.rng,
+            tol=self._tol,
+            n_local_search=self._n_local_search,
+            n_preliminary_samples=self._n_preliminary_samples,
         )
-        return normalized_params
+        return found
 
     def _get_constraints_acqf_args(
         self,
@@ -240,6 +240,34 @@ class GPSampler(BaseSampler):
                 )
         return params
 
+    def _fit_objective_gps(
+        self,
+        X: np.ndarray,
+        Y_std: np.ndarray,
+        space: "gp_search_space.SearchSpace",
+    ) -> list["gp.GPRegressor"]:
+        # One GP per objective, warm-started from the previous suggest's fitted
+        # kernel parameters; a dimensionality change invalidates the warm start.
+        warm: list["gp.GPRegressor | None"]
+        warm = list(self._gprs_cache_list or [])
+        if warm and len(warm[0].inverse_squared_lengthscales) != space.dim:
+            warm = []
+        warm += [None] * (Y_std.shape[-1] - len(warm))
+        fitted = [
+            gp.fit_kernel_params(
+                X=X,
+                Y=Y_std[:, i],
+                is_categorical=space.is_categorical,
+                log_prior=self._log_prior,
+                minimum_noise=self._minimum_noise,
+                gpr_cache=prev,
+                deterministic_objective=self._deterministic,
+            )
+            for i, prev in enumerate(warm)
+        ]
+        self._gprs_cache_list = fitted
+        return fitted
+
     def _sample_relative_impl(
         self,
         study: "Study",
@@ -263,30 +291,10 @@ class GPSampler(BaseSampler):
             signs * np.array([t.values for t in completed_trials])
         )
 
-        if (
-            self._gprs_cache_list is not None
-            and len(self._gprs_cache_list[0].inverse_squared_lengthscales)
-            != internal_search_space.dim
-        ):
-            self._gprs_cache_list = None
-
         n_objectives = standardized_score_vals.shape[-1]
-        is_categorical = internal_search_space.is_categorical
-        gprs_list = []
-        for i in range(n_objectives):
-            cache = self._gprs_cache_list[i] if self._gprs_cache_list is not None else None
-            gprs_list.append(
-                gp.fit_kernel_params(
-                    X=normalized_params,
-                    Y=standardized_score_vals[:, i],
-                    is_categorical=is_categorical,
-                    log_prior=self._log_prior,
-                    minimum_noise=self._minimum_noise,
-                    gpr_cache=cache,
-                    deterministic_objective=self._deterministic,
-                )
-            )
-        self._gprs_cache_list = gprs_list
+        gprs_list = self._fit_objective_gps(
+            normalized_params, standardized_score_vals, internal_search_space
+        )
 
         best_params: np.ndarray | None
         acqf: acqf_module.BaseAcquisitionFunc
@@ -388,28 +396,26 @@ class GPSampler(BaseSampler):
                 )
             else:
                 is_all_infeasible = not bool(np.any(is_feasible))
+                if is_all_infeasible:
+                    feasible_scores = None
+                else:
+                    feasible_scores = torch.from_numpy(standardized_score_vals[is_feasible])
                 acqf = acqf_module.LogCEHVI(
                     gpr_list=gprs_list,
                     search_space=internal_search_space,
-                    Y_feasible=(
-                        torch.from_numpy(standardized_score_vals[is_feasible])
-                        if not is_all_infeasible
-                        else None
-                    ),
+                    Y_feasible=feasible_scores,
                     n_qmc_samples=self._n_qmc_samples_ehvi,
                     qmc_seed=qmc_seed,
                     constraints_gpr_list=constr_gprs,
                     constraints_threshold_list=constr_thresholds,
                     normalized_params_of_running_trials=X_running,
                 )
-                best_params = (
-                    self._get_best_params_for_multi_objective(
-                        normalized_params[is_feasible],
-                        standardized_score_vals[is_feasible],
+                if is_all_infeasible:
+                    best_params = None
+                else:
+                    best_params = self._get_best_params_for_multi_objective(
+                        normalized_params[is_feasible], standardized_score_vals[is_feasible]
                     )
-                    if not is_all_infeasible
-                    else None
-                )
 
         normalized_param = self._optimize_acqf(acqf, best_params)
         return internal_search_space.get_unnormalized_param(normalized_param)

@@ -99,29 +99,26 @@ class InMemoryStorage(BaseStorage):
 
     def get_study_id_from_name(self, study_name: str) -> int:
         with self._lock:
-            if study_name not in self._study_name_to_id:
+            sid = self._study_name_to_id.get(study_name)
+            if sid is None:
                 raise KeyError(f"No such study {study_name}.")
-            return self._study_name_to_id[study_name]
+            return sid
 
     def get_study_name_from_id(self, study_id: int) -> str:
         with self._lock:
-            self._check_study_id(study_id)
-            return self._studies[study_id].name
+            return self._study(study_id).name
 
     def get_study_directions(self, study_id: int) -> list[StudyDirection]:
         with self._lock:
-            self._check_study_id(study_id)
-            return self._studies[study_id].directions
+            return self._study(study_id).directions
 
     def get_study_user_attrs(self, study_id: int) -> dict[str, Any]:
         with self._lock:
-            self._check_study_id(study_id)
-            return self._studies[study_id].user_attrs
+            return self._study(study_id).user_attrs
 
     def get_study_system_attrs(self, study_id: int) -> dict[str, Any]:
         with self._lock:
-            self._check_study_id(study_id)
-            return self._studies[study_id].system_attrs
+            return self._study(study_id).system_attrs
 
     def get_all_studies(self) -> list[FrozenStudy]:
         with self._lock:
@@ -159,18 +156,19 @@ class InMemoryStorage(BaseStorage):
 
     @staticmethod
     def _create_running_trial() -> FrozenTrial:
+        # Fresh RUNNING record; id/number are assigned by create_new_trial.
         return FrozenTrial(
-            trial_id=-1,
-            number=-1,
             state=TrialState.RUNNING,
+            datetime_start=datetime.now(),
+            datetime_complete=None,
+            value=None,
+            intermediate_values={},
             params={},
             distributions={},
             user_attrs={},
             system_attrs={},
-            value=None,
-            intermediate_values={},
-            datetime_start=datetime.now(),
-            datetime_complete=None,
+            trial_id=-1,
+            number=-1,
         )
 
     def set_trial_param(
@@ -259,29 +257,25 @@ class InMemoryStorage(BaseStorage):
             return True
 
     def _update_cache(self, trial_id: int, study_id: int) -> None:
-        """Maintain the incremental best-trial cache (reference :296-331)."""
+        """Incremental best-trial cache: compare the newly finished trial only."""
+        info = self._studies[study_id]
+        if len(info.directions) > 1:
+            return  # single-objective cache only
         trial = self._get_trial(trial_id)
         if trial.state != TrialState.COMPLETE:
             return
-        if len(self._studies[study_id].directions) > 1:
-            return
-        best_trial_id = self._studies[study_id].best_trial_id
-        if best_trial_id is None:
-            self._studies[study_id].best_trial_id = trial_id
-            return
-        best_trial = self._get_trial(best_trial_id)
-        assert best_trial.value is not None
         assert trial.value is not None
-        best_value = best_trial.value
-        new_value = trial.value
-        if math.isnan(new_value):
+        if math.isnan(trial.value):
+            return  # NaN never becomes the incumbent
+        incumbent_id = info.best_trial_id
+        if incumbent_id is None:
+            info.best_trial_id = trial_id
             return
-        if self._studies[study_id].directions[0] == StudyDirection.MAXIMIZE:
-            if math.isnan(best_value) or new_value > best_value:
-                self._studies[study_id].best_trial_id = trial_id
-        else:
-            if math.isnan(best_value) or new_value < best_value:
-                self._studies[study_id].best_trial_id = trial_id
+        incumbent_value = self._get_trial(incumbent_id).value
+        assert incumbent_value is not None
+        sign = -1.0 if info.directions[0] == StudyDirection.MINIMIZE else 1.0
+        if math.isnan(incumbent_value) or sign * trial.value > sign * incumbent_value:
+            info.best_trial_id = trial_id
 
     def set_trial_intermediate_value(
         self, trial_id: int, step: int, intermediate_value: float
@@ -313,14 +307,20 @@ class InMemoryStorage(BaseStorage):
         with self._lock:
             return self._get_trial(trial_id)
 
+    def _locate(self, trial_id: int) -> tuple[_StudyInfo, int]:
+        """Resolve a trial id to its study record and row number."""
+        where = self._trial_id_to_study_id_and_number.get(trial_id)
+        if where is None:
+            raise KeyError(f"No trial with trial_id {trial_id} exists.")
+        return self._studies[where[0]], where[1]
+
     def _get_trial(self, trial_id: int) -> FrozenTrial:
-        self._check_trial_id(trial_id)
-        study_id, trial_number = self._trial_id_to_study_id_and_number[trial_id]
-        return self._studies[study_id].trials[trial_number]
+        info, number = self._locate(trial_id)
+        return info.trials[number]
 
     def _set_trial(self, trial_id: int, trial: FrozenTrial) -> None:
-        study_id, trial_number = self._trial_id_to_study_id_and_number[trial_id]
-        self._studies[study_id].trials[trial_number] = trial
+        info, number = self._locate(trial_id)
+        info.trials[number] = trial
 
     def get_all_trials(
         self,
@@ -373,10 +373,15 @@ class InMemoryStorage(BaseStorage):
 
     # ---- helpers --------------------------------------------------------------------
 
-    def _check_study_id(self, study_id: int) -> None:
-        if study_id not in self._studies:
+    def _study(self, study_id: int) -> _StudyInfo:
+        """Checked lookup of the study record."""
+        info = self._studies.get(study_id)
+        if info is None:
             raise KeyError(f"No study with study_id {study_id} exists.")
+        return info
+
+    def _check_study_id(self, study_id: int) -> None:
+        self._study(study_id)
 
     def _check_trial_id(self, trial_id: int) -> None:
-        if trial_id not in self._trial_id_to_study_id_and_number:
-            raise KeyError(f"No trial with trial_id {trial_id} exists.")
+        self._locate(trial_id)
