@@ -143,9 +143,16 @@ class BaseAcquisitionFunc(ABC):
     # and gradients come back.
     _device: "torch.device | None" = None
 
-    def __init__(self, length_scales: np.ndarray, search_space: "SearchSpace") -> None:
+    def __init__(
+        self,
+        length_scales: np.ndarray,
+        search_space: "SearchSpace",
+        device: "torch.device | None" = None,
+    ) -> None:
         self.length_scales = length_scales
         self.search_space = search_space
+        if device is not None and device.type != "cpu":
+            self._device = device
 
     def set_device(self, device: "torch.device | None") -> None:
         self._device = device
@@ -191,7 +198,7 @@ class LogEI(BaseAcquisitionFunc):
         self._gpr = gpr
         self._stabilizing_noise = stabilizing_noise
         self._threshold = threshold
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         if np.isneginf(self._threshold):
@@ -217,11 +224,13 @@ class qLogEI(BaseAcquisitionFunc):
         )
         n_running = len(normalized_params_of_running_trials)
         self._per_sample_shape = (n_qmc_samples, n_running + 1)
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def compute_per_sample_log_utility(self, x: "torch.Tensor") -> "torch.Tensor":
         if np.isneginf(self._threshold):
-            return torch.zeros(x.shape[:-1] + self._per_sample_shape, dtype=torch.float64)
+            return torch.zeros(
+                x.shape[:-1] + self._per_sample_shape, dtype=torch.float64, device=x.device
+            )
         y_post = self._cond_gpr.sample_joint_posterior(x)
         return (y_post - self._threshold).clamp_min_(_EPS).log()
 
@@ -243,9 +252,9 @@ class LogPI(BaseAcquisitionFunc):
         self._threshold = threshold
         if normalized_params_of_running_trials is not None:
             # Kriging Believer: append running points at their posterior mean.
-            X_running = torch.from_numpy(normalized_params_of_running_trials)
+            X_running = torch.from_numpy(normalized_params_of_running_trials).to(gpr.device)
             self._gpr.append_running_data(X_running, gpr.posterior(X_running)[0])
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         mean, var = self._gpr.posterior(x)
@@ -270,7 +279,7 @@ class qLogPI(BaseAcquisitionFunc):
         self._cond_gpr = _fantasize(
             gpr, normalized_params_of_running_trials, n_qmc_samples, qmc_seed, stabilizing_noise
         )
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def compute_per_sample_log_utility(self, x: "torch.Tensor") -> "torch.Tensor":
         y_post = self._cond_gpr.sample_joint_posterior(x)
@@ -284,7 +293,7 @@ class UCB(BaseAcquisitionFunc):
     def __init__(self, gpr: GPRegressor, search_space: "SearchSpace", beta: float) -> None:
         self._gpr = gpr
         self._beta = beta
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         mean, var = self._gpr.posterior(x)
@@ -295,7 +304,7 @@ class LCB(BaseAcquisitionFunc):
     def __init__(self, gpr: GPRegressor, search_space: "SearchSpace", beta: float) -> None:
         self._gpr = gpr
         self._beta = beta
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         mean, var = self._gpr.posterior(x)
@@ -322,7 +331,7 @@ class LogCEI(BaseAcquisitionFunc):
             LogPI(c_gpr, search_space, c_threshold, None, stabilizing_noise)
             for c_gpr, c_threshold in zip(constraints_gpr_list, constraints_threshold_list)
         ]
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         return self._acqf.eval_acqf(x) + sum(
@@ -364,7 +373,7 @@ class qLogCEI(BaseAcquisitionFunc):
                 zip(constraints_gpr_list, constraints_threshold_list)
             )
         ]
-        super().__init__(gpr.length_scales, search_space)
+        super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
         log_feasible_improvement = self._acqf.compute_per_sample_log_utility(x) + sum(
@@ -387,18 +396,20 @@ class LogEHVI(BaseAcquisitionFunc):
     ) -> None:
         self._stabilizing_noise = stabilizing_noise
         self._gpr_list = gpr_list
+        dev = gpr_list[0].device
         if normalized_params_of_running_trials is not None:
-            X_running = torch.from_numpy(normalized_params_of_running_trials)
+            X_running = torch.from_numpy(normalized_params_of_running_trials).to(dev)
             for gpr in self._gpr_list:
                 gpr.append_running_data(X_running, gpr.posterior(X_running)[0])
         self._fixed_samples = sample_from_normal_sobol(
             dim=Y_train.shape[-1], n_samples=n_qmc_samples, seed=qmc_seed
-        )
+        ).to(dev)
         ref_point = _get_reference_point(Y_train)
-        self._box_lower, box_upper = _get_boxes(Y_train, ref_point)
-        self._box_intervals = (box_upper - self._box_lower).clamp_min_(_EPS)
+        box_lower, box_upper = _get_boxes(Y_train, ref_point)
+        self._box_lower = box_lower.to(dev)
+        self._box_intervals = (box_upper - box_lower).clamp_min_(_EPS).to(dev)
         super().__init__(
-            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space
+            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space, dev
         )
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
@@ -432,21 +443,22 @@ class qLogEHVI(BaseAcquisitionFunc):
             )
             for i, gpr in enumerate(gpr_list)
         ]
+        dev = gpr_list[0].device
         ref_point = _get_reference_point(Y_train)
         lower_list, interval_list = [], []
         for fantasy in torch.stack(
-            [cg.get_fantasy_samples() for cg in self._cond_gpr_list], dim=-1
+            [cg.get_fantasy_samples().cpu() for cg in self._cond_gpr_list], dim=-1
         ):
             Y_fantasy = torch.cat([Y_train, fantasy], dim=0)
             lower, upper = _get_boxes(Y_fantasy, ref_point)
             lower_list.append(lower)
             interval_list.append((upper - lower).clamp_min_(_EPS))
-        self._box_lower = torch.nn.utils.rnn.pad_sequence(lower_list, batch_first=True)
+        self._box_lower = torch.nn.utils.rnn.pad_sequence(lower_list, batch_first=True).to(dev)
         self._box_intervals = torch.nn.utils.rnn.pad_sequence(
             interval_list, batch_first=True, padding_value=_EPS
-        )
+        ).to(dev)
         super().__init__(
-            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space
+            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space, dev
         )
 
     def compute_per_sample_log_utility(self, x: "torch.Tensor") -> "torch.Tensor":
@@ -508,7 +520,9 @@ class LogCEHVI(BaseAcquisitionFunc):
             for c_gpr, c_threshold in zip(constraints_gpr_list, constraints_threshold_list)
         ]
         super().__init__(
-            np.mean([gpr.length_scales for gpr in gpr_list], axis=0), search_space
+            np.mean([gpr.length_scales for gpr in gpr_list], axis=0),
+            search_space,
+            gpr_list[0].device,
         )
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":

@@ -66,7 +66,9 @@ def _extend_cholesky(
     """Grow chol(K11) into chol([[K11,K12],[K21,K22]]) without refactorizing."""
     n1 = L11.shape[-1]
     n2 = K22.shape[-1]
-    L = torch.zeros(L11.shape[:-2] + (n1 + n2, n1 + n2), dtype=torch.float64)
+    L = torch.zeros(
+        L11.shape[:-2] + (n1 + n2, n1 + n2), dtype=torch.float64, device=L11.device
+    )
     L21_T = torch.linalg.solve_triangular(L11, K21.transpose(-1, -2), upper=False)
     L21 = L21_T.transpose(-1, -2)
     L[..., :n1, :n1] = L11
@@ -99,7 +101,31 @@ def matern52_of_sqdist(squared_distance: "torch.Tensor") -> "torch.Tensor":
     return _Matern52.apply(squared_distance)  # type: ignore[attr-defined]
 
 
+def _ard_sqdist_gemm(
+    X1: "torch.Tensor", X2: "torch.Tensor", eta: "torch.Tensor"
+) -> "torch.Tensor":
+    """ARD-weighted squared distance via one GEMM: no (B, N, D) materialization.
+
+    sum_d eta_d (x_d - y_d)^2 = |x'|^2 + |y'|^2 - 2 x'.y'  with x' = x*sqrt(eta).
+    Cancellation near coincident points can dip slightly negative — clamped.
+    """
+    root = torch.sqrt(eta)
+    A = X1 * root
+    B = X2 * root
+    sq = (
+        A.square().sum(-1).unsqueeze(-1)
+        + B.square().sum(-1).unsqueeze(-2)
+        - 2.0 * A.matmul(B.transpose(-1, -2))
+    )
+    return sq.clamp_min_(0.0)
+
+
 class GPRegressor:
+    # Build the (N, N, D) per-dim squared-difference tensor only up to this many
+    # observations; above it every training-covariance evaluation goes through
+    # the GEMM identity instead (4 GB at N=5k, D=20 — and O(N^2 D) host time).
+    _MAX_DENSE_SQDIFF_OBS = 2048
+
     def __init__(
         self,
         is_categorical: "torch.Tensor",
@@ -115,13 +141,20 @@ class GPRegressor:
         self._y_train = y_train.unsqueeze(-1)
         self._X_all = X_train
         self._y_all = self._y_train
-        sqd = (X_train.unsqueeze(-2) - X_train.unsqueeze(-3)).square_()
-        if is_categorical.any():
-            # Hamming distance on categorical axes.
-            sqd[..., is_categorical] = (sqd[..., is_categorical] > 0.0).double()
-        self._squared_X_diff = sqd
+        self._squared_X_diff: "torch.Tensor | None" = None
+        n_obs = X_train.shape[0]
+        going_to_device = n_obs >= self._DEVICE_FIT_MIN_OBS and torch.cuda.is_available()
+        if is_categorical.any() or (
+            n_obs <= self._MAX_DENSE_SQDIFF_OBS and not going_to_device
+        ):
+            sqd = (X_train.unsqueeze(-2) - X_train.unsqueeze(-3)).square_()
+            if is_categorical.any():
+                # Hamming distance on categorical axes.
+                sqd[..., is_categorical] = (sqd[..., is_categorical] > 0.0).double()
+            self._squared_X_diff = sqd
         self._cov_Y_Y_chol: "torch.Tensor | None" = None
         self._cov_Y_Y_inv_Y: "torch.Tensor | None" = None
+        self._cov_Y_Y_inv: "torch.Tensor | None" = None
         self.inverse_squared_lengthscales = inverse_squared_lengthscales
         self.kernel_scale = kernel_scale
         self.noise_var = noise_var
@@ -129,6 +162,10 @@ class GPRegressor:
     @property
     def length_scales(self) -> np.ndarray:
         return 1.0 / np.sqrt(self.inverse_squared_lengthscales.detach().cpu().numpy())
+
+    @property
+    def device(self) -> "torch.device":
+        return self._X_train.device
 
     def to(self, device: "torch.device") -> "GPRegressor":
         """Move every tensor (incl. the cached Cholesky) to `device` in place.
@@ -145,6 +182,7 @@ class GPRegressor:
             "_squared_X_diff",
             "_cov_Y_Y_chol",
             "_cov_Y_Y_inv_Y",
+            "_cov_Y_Y_inv",
             "inverse_squared_lengthscales",
             "kernel_scale",
             "noise_var",
@@ -157,21 +195,30 @@ class GPRegressor:
     def kernel(
         self, X1: "torch.Tensor | None" = None, X2: "torch.Tensor | None" = None
     ) -> "torch.Tensor":
+        eta = self.inverse_squared_lengthscales
         if X1 is None:
             assert X2 is None
+            if self._squared_X_diff is None:
+                # Large no-categorical history: GEMM identity, O(N^2 D) flops
+                # through MFMA instead of an (N, N, D) broadcast tensor.
+                sqdist = _ard_sqdist_gemm(self._X_train, self._X_train, eta)
+                return matern52_of_sqdist(sqdist) * self.kernel_scale
             sqd = self._squared_X_diff
         else:
             if X2 is None:
                 X2 = self._X_train
+            if not self._is_categorical.any() and X1.ndim >= 2:
+                sqdist = _ard_sqdist_gemm(X1, X2, eta)
+                return matern52_of_sqdist(sqdist) * self.kernel_scale
             sqd = (X1 - X2 if X1.ndim == 1 else X1.unsqueeze(-2) - X2.unsqueeze(-3)).square_()
             if self._is_categorical.any():
                 sqd[..., self._is_categorical] = (sqd[..., self._is_categorical] > 0.0).double()
         if sqd.is_cuda:
             # rocBLAS's (…, N, D) @ (D,) gemv path measures ~100× slower than a
             # broadcast reduction at these tall-skinny fp64 shapes.
-            sqdist = (sqd * self.inverse_squared_lengthscales).sum(-1)
+            sqdist = (sqd * eta).sum(-1)
         else:
-            sqdist = sqd.matmul(self.inverse_squared_lengthscales)
+            sqdist = sqd.matmul(eta)
         return matern52_of_sqdist(sqdist) * self.kernel_scale
 
     def _cache_matrix(self) -> None:
@@ -184,6 +231,10 @@ class GPRegressor:
         cov_Y_Y.diagonal().add_(self.noise_var)
         self._cov_Y_Y_chol = torch.linalg.cholesky(cov_Y_Y)
         self._cov_Y_Y_inv_Y = _solve_cholesky(self._cov_Y_Y_chol, self._y_train).squeeze(-1)
+        if self._X_train.is_cuda:
+            # Device path: a resident explicit inverse turns the two N×N
+            # triangular solves per acquisition evaluation into one MFMA GEMM.
+            self._cov_Y_Y_inv = torch.cholesky_inverse(self._cov_Y_Y_chol)
 
     def append_running_data(self, X_running: "torch.Tensor", y_running: "torch.Tensor") -> None:
         """Kriging-Believer append: extend the Cholesky with running-trial rows."""
@@ -196,6 +247,8 @@ class GPRegressor:
         k_rr.diagonal().add_(self.noise_var)
         self._cov_Y_Y_chol = _extend_cholesky(L11=self._cov_Y_Y_chol, K21=k_rt, K22=k_rr)
         self._cov_Y_Y_inv_Y = _solve_cholesky(self._cov_Y_Y_chol, self._y_all).squeeze(-1)
+        if self._cov_Y_Y_inv is not None:
+            self._cov_Y_Y_inv = torch.cholesky_inverse(self._cov_Y_Y_chol)
 
     def posterior(
         self, x: "torch.Tensor", joint: bool = False
@@ -205,7 +258,10 @@ class GPRegressor:
         x_ = x.unsqueeze(0) if is_single else x
         cov_fx_fX = self.kernel(x_, self._X_all)
         mean = torch.linalg.vecdot(cov_fx_fX, self._cov_Y_Y_inv_Y)
-        V = _solve_cholesky(self._cov_Y_Y_chol, cov_fx_fX, left=False)
+        if self._cov_Y_Y_inv is not None:
+            V = cov_fx_fX.matmul(self._cov_Y_Y_inv)
+        else:
+            V = _solve_cholesky(self._cov_Y_Y_chol, cov_fx_fX, left=False)
         if joint:
             assert not is_single
             var_ = self.kernel(x_, x_) - V.matmul(cov_fx_fX.transpose(-1, -2))
@@ -294,24 +350,27 @@ class GPRegressor:
         )
         return loss, grad
 
-    def _loss_and_grad_torch_device(
+    def _loss_and_grad_closed_form_torch(
         self,
         raw_params: np.ndarray,
-        sqd_flat: "torch.Tensor",  # (N·N, D) resident on the GPU
-        y: "torch.Tensor",  # (N,) on the GPU
+        X: "torch.Tensor",  # (N, D) — resident wherever the fit runs (HBM on MI355X)
+        y: "torch.Tensor",  # (N,)
         minimum_noise: float,
         deterministic_objective: bool,
     ) -> tuple[float, np.ndarray]:
-        """Same closed-form loss/grad as `_loss_and_grad_numpy`, on the MI355X.
+        """Closed-form negative-MLL-plus-prior and gradient, GEMM-structured.
 
-        One rocSOLVER Cholesky + cholesky_inverse plus a handful of fused
-        elementwise kernels per evaluation; only the 22-element raw-parameter
-        vector and the scalar loss/gradient cross the PCIe bus per L-BFGS
-        iteration. fp64 throughout (CDNA4 fp64 MFMA backs the GEMV).
+        Everything is O(N²) matrices plus one rocSOLVER Cholesky +
+        cholesky_inverse; the (N, N, D) per-dim tensor never exists. The
+        lengthscale gradient uses
+          d/d eta_d = sum_ij W_ij (x_id - x_jd)^2
+                    = ((r + c) · X_d²) − 2 X_dᵀ (W X_d)
+        with r/c the row/col sums of W — three GEMM-class ops total. Only the
+        raw-parameter vector and the scalar loss/gradient cross the PCIe bus
+        per L-BFGS iteration when X lives on the device.
         """
         device = y.device
-        n = y.shape[0]
-        n_params = sqd_flat.shape[1]
+        n_params = X.shape[1]
         with torch.no_grad():
             raw = torch.from_numpy(raw_params).to(device)
             eta = torch.exp(raw[:n_params])
@@ -322,12 +381,10 @@ class GPRegressor:
                 if deterministic_objective
                 else noise_raw + minimum_noise
             )
-            # Tall skinny reductions: rocBLAS gemv/gemm paths measure 0.7–107 ms
-            # at (N², D) fp64; broadcast-multiply + sum measures ~0.1 ms.
-            r2 = (sqd_flat * eta).sum(1)
+            r2 = _ard_sqdist_gemm(X, X, eta)
             u = torch.sqrt(5.0 * r2)
             eu = torch.exp(-u)
-            M = (eu * ((5.0 / 3.0) * r2 + u + 1.0)).reshape(n, n)
+            M = eu * ((5.0 / 3.0) * r2 + u + 1.0)
             C = scale * M
             C.diagonal().add_(noise)
             L = torch.linalg.cholesky(C)
@@ -343,9 +400,10 @@ class GPRegressor:
             loss = -(mll + log_prior_val)
 
             A = torch.outer(alpha, alpha) - Cinv
-            Mp = ((-5.0 / 6.0) * (1.0 + u) * eu).reshape(n, n)
+            Mp = (-5.0 / 6.0) * (1.0 + u) * eu
             W = (0.5 * scale) * (A * Mp)
-            g_eta = (sqd_flat * W.reshape(-1, 1)).sum(0)
+            rc = W.sum(1) + W.sum(0)
+            g_eta = X.square().T.mv(rc) - 2.0 * (X * W.matmul(X)).sum(0)
             g_scale = 0.5 * (A * M).sum()
             g_noise = 0.5 * A.diagonal().sum()
             gp_eta = 0.1 / (eta * eta) - 0.1
@@ -386,25 +444,27 @@ class GPRegressor:
 
         from optuna_amd._gp.prior import default_log_prior
 
-        if (
-            log_prior is default_log_prior
-            and self._X_train.shape[0] >= self._DEVICE_FIT_MIN_OBS
+        use_device = (
+            self._X_train.shape[0] >= self._DEVICE_FIT_MIN_OBS
             and torch.cuda.is_available()
-        ):
-            dev = torch.device("cuda")
-            sqd_flat_dev = (
-                self._squared_X_diff.reshape(-1, n_params).to(dev)
-            )
-            y_dev = self._y_train.squeeze(-1).to(dev)
+            and not self._is_categorical.any()
+        )
+        if log_prior is default_log_prior and (use_device or self._squared_X_diff is None):
+            # Closed-form GEMM-structured loss/grad. On the MI355X the whole
+            # regressor moves to HBM first, so the fit, the cached Cholesky and
+            # every later posterior/acqf evaluation stay device-resident.
+            if use_device:
+                self.to(torch.device("cuda"))
+            X = self._X_train
+            y = self._y_train.squeeze(-1)
 
-            def loss_func_dev(raw_params: np.ndarray) -> tuple[float, np.ndarray]:
-                return self._loss_and_grad_torch_device(
-                    raw_params, sqd_flat_dev, y_dev, minimum_noise,
-                    deterministic_objective,
+            def loss_func_closed(raw_params: np.ndarray) -> tuple[float, np.ndarray]:
+                return self._loss_and_grad_closed_form_torch(
+                    raw_params, X, y, minimum_noise, deterministic_objective
                 )
 
             res = scipy.optimize.minimize(
-                loss_func_dev,
+                loss_func_closed,
                 initial_raw_params,
                 jac=True,
                 method="l-bfgs-b",
@@ -412,11 +472,11 @@ class GPRegressor:
             )
             if not res.success:
                 raise RuntimeError(f"Optimization failed: {res.message}")
-            raw_opt = torch.from_numpy(res.x)
+            raw_opt = torch.from_numpy(res.x).to(X.device)
             self.inverse_squared_lengthscales = torch.exp(raw_opt[:n_params])
             self.kernel_scale = torch.exp(raw_opt[n_params])
             self.noise_var = (
-                torch.tensor(minimum_noise, dtype=torch.float64)
+                torch.tensor(minimum_noise, dtype=torch.float64, device=X.device)
                 if deterministic_objective
                 else minimum_noise + torch.exp(raw_opt[n_params + 1])
             )
@@ -504,10 +564,11 @@ class ConditionalGPRegressor:
         stabilizing_noise: float,
     ) -> None:
         self._gpr = gpr
+        X_running = X_running.to(gpr.device)
         self._X_running = X_running
         fixed_samples = sample_from_normal_sobol(
             dim=X_running.shape[0] + 1, n_samples=n_qmc_samples, seed=qmc_seed
-        )
+        ).to(gpr.device)
         self._fixed_samples_x = fixed_samples[..., -1]
         self._stabilizing_noise = stabilizing_noise
         with torch.no_grad():

@@ -304,28 +304,15 @@ class GPSampler(BaseSampler):
             if n_objectives == 1:
                 threshold = float(standardized_score_vals[:, 0].max())
                 if X_running is None:
-                    # MI355X path: with a large history the posterior GEMMs
-                    # dominate acquisition evaluation — move the fitted GP
-                    # (incl. its cached Cholesky) to the device and evaluate
-                    # the acqf there (BaseAcquisitionFunc transfers candidates).
-                    if (
-                        torch.cuda.is_available()
-                        and len(normalized_params)
-                        >= gp.GPRegressor._DEVICE_FIT_MIN_OBS
-                    ):
-                        gprs_list[0] = gprs_list[0].to(torch.device("cuda"))
-                        acqf = acqf_module.LogEI(
-                            gpr=gprs_list[0],
-                            search_space=internal_search_space,
-                            threshold=threshold,
-                        )
-                        acqf.set_device(torch.device("cuda"))
-                    else:
-                        acqf = acqf_module.LogEI(
-                            gpr=gprs_list[0],
-                            search_space=internal_search_space,
-                            threshold=threshold,
-                        )
+                    # MI355X path: at large histories the fit leaves the GP
+                    # (incl. its cached Cholesky/inverse) resident on the
+                    # device, and the acqf inherits that device — candidates
+                    # go up, scalars/gradients come back.
+                    acqf = acqf_module.LogEI(
+                        gpr=gprs_list[0],
+                        search_space=internal_search_space,
+                        threshold=threshold,
+                    )
                 else:
                     acqf = acqf_module.qLogEI(
                         gpr=gprs_list[0],

@@ -149,3 +149,65 @@ def test_numpy_loss_grad_matches_torch_autograd() -> None:
             np.testing.assert_allclose(
                 grad_np, raw_t.grad.detach().numpy(), rtol=1e-8, atol=1e-10
             )
+
+
+def test_ard_sqdist_gemm_matches_broadcast() -> None:
+    rng = np.random.RandomState(3)
+    X1 = torch.from_numpy(rng.rand(37, 6))
+    X2 = torch.from_numpy(rng.rand(53, 6))
+    eta = torch.from_numpy(np.exp(rng.randn(6)))
+    got = gp._ard_sqdist_gemm(X1, X2, eta)
+    want = ((X1.unsqueeze(-2) - X2.unsqueeze(-3)).square() * eta).sum(-1)
+    torch.testing.assert_close(got, want, rtol=1e-9, atol=1e-10)
+    # coincident points: exactly clamped at 0, never negative
+    same = gp._ard_sqdist_gemm(X1, X1.clone(), eta)
+    assert (same.diagonal() >= 0).all()
+
+
+def test_closed_form_torch_loss_matches_numpy() -> None:
+    rng = np.random.RandomState(7)
+    N, D = 64, 4
+    X = rng.rand(N, D)
+    y = rng.randn(N)
+    gpr = _make_gpr(X, y, fit=False)
+    raw = rng.randn(D + 2) * 0.3
+    sqd = (X[:, None, :] - X[None, :, :]) ** 2
+    loss_np, grad_np = gpr._loss_and_grad_numpy(raw, sqd, y, 1e-6, False)
+    loss_t, grad_t = gpr._loss_and_grad_closed_form_torch(
+        raw, torch.from_numpy(X), torch.from_numpy(y), 1e-6, False
+    )
+    assert loss_t == pytest.approx(loss_np, rel=1e-8)
+    np.testing.assert_allclose(grad_t, grad_np, rtol=1e-6, atol=1e-8)
+
+
+def test_posterior_with_explicit_inverse_matches_solves() -> None:
+    rng = np.random.RandomState(11)
+    X = rng.rand(40, 3)
+    y = rng.randn(40)
+    gpr = _make_gpr(X, y, fit=False)
+    x_eval = torch.from_numpy(rng.rand(9, 3))
+    mean_ref, var_ref = gpr.posterior(x_eval)
+    gpr._cov_Y_Y_inv = torch.cholesky_inverse(gpr._cov_Y_Y_chol)
+    mean_inv, var_inv = gpr.posterior(x_eval)
+    torch.testing.assert_close(mean_inv, mean_ref, rtol=1e-8, atol=1e-10)
+    torch.testing.assert_close(var_inv, var_ref, rtol=1e-6, atol=1e-9)
+
+
+def test_large_history_skips_dense_sqdiff_and_still_fits() -> None:
+    # Above _MAX_DENSE_SQDIFF_OBS (no categorical, no cuda) the regressor must
+    # fit through the GEMM identity without the (N, N, D) tensor.
+    rng = np.random.RandomState(13)
+    N = gp.GPRegressor._MAX_DENSE_SQDIFF_OBS + 8
+    X = rng.rand(N, 2)
+    y = np.sin(3 * X[:, 0]) + rng.randn(N) * 0.1
+    gpr = gp.fit_kernel_params(
+        X=X,
+        Y=y,
+        is_categorical=np.zeros(2, dtype=bool),
+        log_prior=prior.default_log_prior,
+        minimum_noise=prior.DEFAULT_MINIMUM_NOISE_VAR,
+        deterministic_objective=False,
+    )
+    assert gpr._squared_X_diff is None
+    mean, var = gpr.posterior(torch.from_numpy(X[:5]))
+    assert torch.isfinite(mean).all() and (var >= 0).all()

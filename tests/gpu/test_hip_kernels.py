@@ -275,8 +275,8 @@ def test_gp_device_fit_matches_cpu() -> None:
         gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = old
 
     np.testing.assert_allclose(
-        gpr_dev.inverse_squared_lengthscales.numpy(),
-        gpr_cpu.inverse_squared_lengthscales.numpy(),
+        gpr_dev.inverse_squared_lengthscales.cpu().numpy(),
+        gpr_cpu.inverse_squared_lengthscales.cpu().numpy(),
         rtol=1e-4,
     )
     np.testing.assert_allclose(
@@ -675,3 +675,137 @@ def test_cma_device_eigh_matches_host(core) -> None:
     np.testing.assert_allclose(B @ np.diag(d2) @ B.T, C, rtol=1e-8, atol=1e-8)
     d2h, _ = np.linalg.eigh(C)
     np.testing.assert_allclose(np.sort(d2), np.sort(d2h), rtol=1e-8, atol=1e-8)
+
+
+def test_gp_device_acqfs_match_cpu() -> None:
+    """Every acquisition family must produce the same values/argmax whether the
+    GP lives on the MI355X or on the host (golden device-vs-CPU parity)."""
+    import torch
+
+    import optuna_amd
+    from optuna_amd._gp import acqf as acqf_mod
+    from optuna_amd._gp import gp as gp_mod
+    from optuna_amd._gp import prior
+    from optuna_amd._gp import search_space as gp_ss
+    from optuna_amd.distributions import FloatDistribution
+
+    assert torch.cuda.is_available()
+    rng = np.random.RandomState(5)
+    n, d = 700, 5  # above _DEVICE_FIT_MIN_OBS
+    X = rng.rand(n, d)
+    Y = np.sum((X - 0.4) ** 2, axis=1)
+    Y = (Y - Y.mean()) / Y.std()
+    X_running = rng.rand(3, d)
+    space = gp_ss.SearchSpace({f"x{i}": FloatDistribution(0.0, 1.0) for i in range(d)})
+    is_cat = np.zeros(d, dtype=bool)
+    cands = rng.rand(32, d)
+
+    def fit(force_cpu: bool) -> gp_mod.GPRegressor:
+        old = gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS
+        if force_cpu:
+            gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = 10**9
+        try:
+            return gp_mod.fit_kernel_params(
+                X, Y, is_cat, prior.default_log_prior, 1e-6, False
+            )
+        finally:
+            gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = old
+
+    gpr_dev = fit(force_cpu=False)
+    gpr_cpu = fit(force_cpu=True)
+    assert gpr_dev.device.type == "cuda"
+    assert gpr_cpu.device.type == "cpu"
+    thr = float(Y.max())
+
+    pairs = []
+    for make in (
+        lambda g: acqf_mod.LogEI(gpr=g, search_space=space, threshold=thr),
+        lambda g: acqf_mod.qLogEI(
+            gpr=g,
+            search_space=space,
+            threshold=thr,
+            n_qmc_samples=64,
+            qmc_seed=9,
+            normalized_params_of_running_trials=X_running,
+        ),
+        lambda g: acqf_mod.LogPI(g, space, thr, None, 1e-12),
+        lambda g: acqf_mod.UCB(g, space, 2.0),
+    ):
+        a_dev = make(gpr_dev)
+        a_cpu = make(gpr_cpu)
+        v_dev = a_dev.eval_acqf_no_grad(cands)
+        v_cpu = a_cpu.eval_acqf_no_grad(cands)
+        pairs.append((v_dev, v_cpu))
+        np.testing.assert_allclose(v_dev, v_cpu, rtol=5e-3, atol=5e-3)
+        f_dev, g_dev = a_dev.eval_acqf_batched_with_grad(cands[:4].copy())
+        f_cpu, g_cpu = a_cpu.eval_acqf_batched_with_grad(cands[:4].copy())
+        np.testing.assert_allclose(f_dev, f_cpu, rtol=5e-3, atol=5e-3)
+        np.testing.assert_allclose(g_dev, g_cpu, rtol=5e-2, atol=5e-2)
+
+
+def test_gp_device_ehvi_matches_cpu() -> None:
+    """LogEHVI / qLogEHVI device-vs-CPU parity (2-objective)."""
+    import torch
+
+    from optuna_amd._gp import acqf as acqf_mod
+    from optuna_amd._gp import gp as gp_mod
+    from optuna_amd._gp import prior
+    from optuna_amd._gp import search_space as gp_ss
+    from optuna_amd.distributions import FloatDistribution
+
+    assert torch.cuda.is_available()
+    rng = np.random.RandomState(6)
+    n, d = 640, 4
+    X = rng.rand(n, d)
+    Y = np.stack([np.sum(X**2, axis=1), np.sum((X - 1) ** 2, axis=1)], axis=1)
+    Y = (Y - Y.mean(0)) / Y.std(0)
+    space = gp_ss.SearchSpace({f"x{i}": FloatDistribution(0.0, 1.0) for i in range(d)})
+    is_cat = np.zeros(d, dtype=bool)
+    X_running = rng.rand(2, d)
+    cands = rng.rand(16, d)
+
+    def fit_pair(force_cpu: bool):
+        old = gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS
+        if force_cpu:
+            gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = 10**9
+        try:
+            return [
+                gp_mod.fit_kernel_params(
+                    X, Y[:, i], is_cat, prior.default_log_prior, 1e-6, False
+                )
+                for i in range(2)
+            ]
+        finally:
+            gp_mod.GPRegressor._DEVICE_FIT_MIN_OBS = old
+
+    gprs_dev = fit_pair(False)
+    gprs_cpu = fit_pair(True)
+    Y_t = torch.from_numpy(Y)
+
+    for kwargs in (dict(), dict(normalized_params_of_running_trials=None)):
+        a_dev = acqf_mod.LogEHVI(
+            gpr_list=gprs_dev, search_space=space, Y_train=Y_t,
+            n_qmc_samples=64, qmc_seed=3, **kwargs
+        )
+        a_cpu = acqf_mod.LogEHVI(
+            gpr_list=gprs_cpu, search_space=space, Y_train=Y_t,
+            n_qmc_samples=64, qmc_seed=3, **kwargs
+        )
+        np.testing.assert_allclose(
+            a_dev.eval_acqf_no_grad(cands), a_cpu.eval_acqf_no_grad(cands),
+            rtol=5e-3, atol=5e-3,
+        )
+        break  # append_running_data mutates the gprs; one configuration here
+
+    q_dev = acqf_mod.qLogEHVI(
+        gpr_list=gprs_dev, search_space=space, Y_train=Y_t,
+        normalized_params_of_running_trials=X_running, n_qmc_samples=64, qmc_seed=3,
+    )
+    q_cpu = acqf_mod.qLogEHVI(
+        gpr_list=gprs_cpu, search_space=space, Y_train=Y_t,
+        normalized_params_of_running_trials=X_running, n_qmc_samples=64, qmc_seed=3,
+    )
+    np.testing.assert_allclose(
+        q_dev.eval_acqf_no_grad(cands), q_cpu.eval_acqf_no_grad(cands),
+        rtol=5e-3, atol=5e-3,
+    )
