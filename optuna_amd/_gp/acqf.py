@@ -75,8 +75,8 @@ def logehvi(
     non_dominated_box_intervals: "torch.Tensor",
 ) -> "torch.Tensor":
     log_n = float(np.log(Y_post.shape[-2]))
-    diff = Y_post.unsqueeze(-2) - non_dominated_box_lower_bounds
-    diff.clamp_(min=torch.tensor(_EPS, dtype=torch.float64), max=non_dominated_box_intervals)
+    diff = (Y_post.unsqueeze(-2) - non_dominated_box_lower_bounds).clamp_min_(_EPS)
+    diff = torch.minimum(diff, non_dominated_box_intervals)
     return torch.special.logsumexp(diff.log().sum(dim=-1), dim=(-2, -1)) - log_n
 
 
@@ -85,8 +85,8 @@ def _per_sample_log_hvi(
     non_dominated_box_lower_bounds: "torch.Tensor",
     non_dominated_box_intervals: "torch.Tensor",
 ) -> "torch.Tensor":
-    diff = Y_post.unsqueeze(-2) - non_dominated_box_lower_bounds
-    diff.clamp_(min=torch.tensor(_EPS, dtype=torch.float64), max=non_dominated_box_intervals)
+    diff = (Y_post.unsqueeze(-2) - non_dominated_box_lower_bounds).clamp_min_(_EPS)
+    diff = torch.minimum(diff, non_dominated_box_intervals)
     return torch.special.logsumexp(diff.log().sum(dim=-1), dim=-1)
 
 
@@ -251,9 +251,10 @@ class LogPI(BaseAcquisitionFunc):
         self._stabilizing_noise = stabilizing_noise
         self._threshold = threshold
         if normalized_params_of_running_trials is not None:
-            # Kriging Believer: append running points at their posterior mean.
+            # Kriging Believer: append running points at their posterior mean
+            # (copy-on-write — the sampler's cached regressor stays pristine).
             X_running = torch.from_numpy(normalized_params_of_running_trials).to(gpr.device)
-            self._gpr.append_running_data(X_running, gpr.posterior(X_running)[0])
+            self._gpr = gpr.cloned_with_running(X_running, gpr.posterior(X_running)[0])
         super().__init__(gpr.length_scales, search_space, gpr.device)
 
     def eval_acqf(self, x: "torch.Tensor") -> "torch.Tensor":
@@ -399,8 +400,10 @@ class LogEHVI(BaseAcquisitionFunc):
         dev = gpr_list[0].device
         if normalized_params_of_running_trials is not None:
             X_running = torch.from_numpy(normalized_params_of_running_trials).to(dev)
-            for gpr in self._gpr_list:
-                gpr.append_running_data(X_running, gpr.posterior(X_running)[0])
+            self._gpr_list = [
+                g.cloned_with_running(X_running, g.posterior(X_running)[0])
+                for g in self._gpr_list
+            ]
         self._fixed_samples = sample_from_normal_sobol(
             dim=Y_train.shape[-1], n_samples=n_qmc_samples, seed=qmc_seed
         ).to(dev)

@@ -235,6 +235,25 @@ class GPRegressor:
             # Device path: a resident explicit inverse turns the two N×N
             # triangular solves per acquisition evaluation into one MFMA GEMM.
             self._cov_Y_Y_inv = torch.cholesky_inverse(self._cov_Y_Y_chol)
+        self._n_obs_at_fit = self._X_train.shape[0]
+
+    def cloned_with_running(
+        self, X_running: "torch.Tensor", y_running: "torch.Tensor"
+    ) -> "GPRegressor":
+        """Copy-on-write Kriging-Believer append.
+
+        Returns a shallow clone whose covariance state is extended with the
+        running rows, leaving ``self`` (which the sampler caches across
+        suggests for incremental updates) untouched. The clone drops the
+        explicit inverse — the extended system is transient, so the posterior
+        falls back to triangular solves there.
+        """
+        import copy as _copy
+
+        clone = _copy.copy(self)
+        clone._cov_Y_Y_inv = None
+        clone.append_running_data(X_running, y_running)
+        return clone
 
     def append_running_data(self, X_running: "torch.Tensor", y_running: "torch.Tensor") -> None:
         """Kriging-Believer append: extend the Cholesky with running-trial rows."""
@@ -249,6 +268,78 @@ class GPRegressor:
         self._cov_Y_Y_inv_Y = _solve_cholesky(self._cov_Y_Y_chol, self._y_all).squeeze(-1)
         if self._cov_Y_Y_inv is not None:
             self._cov_Y_Y_inv = torch.cholesky_inverse(self._cov_Y_Y_chol)
+
+    # Number of observations the hyperparameters were last fitted on (0 =
+    # never fitted); incremental updates grow the data but leave this alone.
+    _n_obs_at_fit = 0
+
+    def update_data(self, X_full: np.ndarray, y_full: np.ndarray) -> bool:
+        """Absorb appended observations + re-standardized targets WITHOUT a
+        hyperparameter refit.
+
+        The Cholesky factor is extended by the new rows (O(N²k)), the explicit
+        inverse gets the matching block update (Schur complement, O(N²k)
+        GEMMs), and alpha is re-solved for the full new target vector — the
+        targets are study-level standardized, so every entry shifts slightly
+        each suggest even though the data only appends. Total cost is a few
+        milliseconds on the MI355X versus ~1 s for a full refit at 5k obs.
+
+        Returns False (caller must refit) when ``X_full`` is not an exact
+        extension of the current training inputs or the regressor is in a
+        state this fast path does not cover.
+        """
+        if self._squared_X_diff is not None or self._cov_Y_Y_chol is None:
+            return False
+        n_old = self._X_train.shape[0]
+        if (
+            X_full.ndim != 2
+            or X_full.shape[0] < n_old
+            or X_full.shape[1] != self._X_train.shape[1]
+            or self._cov_Y_Y_chol.shape[-1] != n_old
+            or y_full.shape[0] != X_full.shape[0]
+        ):
+            return False
+        Xf = torch.from_numpy(np.ascontiguousarray(X_full)).to(self.device)
+        yf = torch.from_numpy(np.ascontiguousarray(y_full)).to(self.device)
+        if not torch.equal(Xf[:n_old], self._X_train):
+            return False
+
+        with torch.no_grad():
+            X_new = Xf[n_old:]
+            k = X_new.shape[0]
+            if k > 0:
+                k_nt = self.kernel(X_new, self._X_train)  # (k, n_old)
+                k_nn = self.kernel(X_new, X_new)
+                k_nn.diagonal().add_(self.noise_var)
+                if self._cov_Y_Y_inv is not None:
+                    # Block inverse via the Schur complement of the old block.
+                    Ainv = self._cov_Y_Y_inv
+                    B = k_nt.transpose(-1, -2)  # (n_old, k)
+                    AinvB = Ainv.matmul(B)
+                    S = k_nn - B.transpose(-1, -2).matmul(AinvB)
+                    Sinv = torch.cholesky_inverse(torch.linalg.cholesky(S))
+                    off = -AinvB.matmul(Sinv)  # (n_old, k)
+                    n_new_total = n_old + k
+                    inv = torch.empty(
+                        (n_new_total, n_new_total), dtype=torch.float64, device=self.device
+                    )
+                    inv[:n_old, :n_old] = Ainv - off.matmul(AinvB.transpose(-1, -2))
+                    inv[:n_old, n_old:] = off
+                    inv[n_old:, :n_old] = off.transpose(-1, -2)
+                    inv[n_old:, n_old:] = Sinv
+                    self._cov_Y_Y_inv = inv
+                self._cov_Y_Y_chol = _extend_cholesky(self._cov_Y_Y_chol, k_nt, k_nn)
+            self._X_train = Xf
+            self._X_all = Xf
+            self._y_train = yf.unsqueeze(-1)
+            self._y_all = self._y_train
+            if self._cov_Y_Y_inv is not None:
+                self._cov_Y_Y_inv_Y = self._cov_Y_Y_inv.mv(yf)
+            else:
+                self._cov_Y_Y_inv_Y = _solve_cholesky(
+                    self._cov_Y_Y_chol, self._y_train
+                ).squeeze(-1)
+        return True
 
     def posterior(
         self, x: "torch.Tensor", joint: bool = False
@@ -614,6 +705,23 @@ class ConditionalGPRegressor:
         return torch.cat([fantasy, samples.unsqueeze(-1)], dim=-1)
 
 
+def _incremental_update_applicable(prev: GPRegressor, n_now: int) -> bool:
+    """Refit cadence for large device-resident histories.
+
+    Hyperparameters move slowly once thousands of observations are in, so a
+    full L-BFGS refit (dozens of N³ factorizations) only runs after the data
+    grew by ~1% (at most 64 rows) since the last fit; in between, the cached
+    regressor absorbs new rows with O(N²) updates. Below the device-fit
+    threshold the reference's refit-every-suggest behavior is kept — the host
+    fit is cheap there.
+    """
+    n_fit = prev._n_obs_at_fit
+    if n_fit < GPRegressor._DEVICE_FIT_MIN_OBS:
+        return False
+    grown = n_now - n_fit
+    return 0 <= grown < min(64, max(1, n_fit // 100))
+
+
 def fit_kernel_params(
     X: np.ndarray,
     Y: np.ndarray,
@@ -625,6 +733,14 @@ def fit_kernel_params(
     gtol: float = 1e-2,
 ) -> GPRegressor:
     """Fit with warm-start params; retry once from defaults; fall back unfitted."""
+    if (
+        gpr_cache is not None
+        and log_prior is not None
+        and _incremental_update_applicable(gpr_cache, X.shape[0])
+        and gpr_cache.update_data(X, Y)
+    ):
+        return gpr_cache
+
     default_params = torch.ones(X.shape[1] + 2, dtype=torch.float64)
 
     def _fresh(params_src: GPRegressor | None) -> GPRegressor:

@@ -211,3 +211,71 @@ def test_large_history_skips_dense_sqdiff_and_still_fits() -> None:
     assert gpr._squared_X_diff is None
     mean, var = gpr.posterior(torch.from_numpy(X[:5]))
     assert torch.isfinite(mean).all() and (var >= 0).all()
+
+
+def test_update_data_matches_fresh_cache() -> None:
+    """Incremental extend (new rows + re-standardized targets) must reproduce
+    the from-scratch covariance state for the same hyperparameters."""
+    rng = np.random.RandomState(17)
+    N0, D = gp.GPRegressor._MAX_DENSE_SQDIFF_OBS + 4, 3
+    X0 = rng.rand(N0, D)
+    y0 = rng.randn(N0)
+    gpr = gp.fit_kernel_params(
+        X=X0,
+        Y=y0,
+        is_categorical=np.zeros(D, dtype=bool),
+        log_prior=prior.default_log_prior,
+        minimum_noise=prior.DEFAULT_MINIMUM_NOISE_VAR,
+        deterministic_objective=False,
+    )
+    X_full = np.vstack([X0, rng.rand(7, D)])
+    y_full = np.concatenate([y0, rng.randn(7)]) * 1.01 + 0.003  # drifted targets
+    assert gpr.update_data(X_full, y_full)
+    assert gpr._X_train.shape[0] == N0 + 7
+
+    ref = gp.GPRegressor(
+        is_categorical=torch.zeros(D, dtype=torch.bool),
+        X_train=torch.from_numpy(X_full),
+        y_train=torch.from_numpy(y_full),
+        inverse_squared_lengthscales=gpr.inverse_squared_lengthscales.clone(),
+        kernel_scale=gpr.kernel_scale.clone(),
+        noise_var=gpr.noise_var.clone(),
+    )
+    ref._cache_matrix()
+    x_eval = torch.from_numpy(rng.rand(11, D))
+    mean_u, var_u = gpr.posterior(x_eval)
+    mean_r, var_r = ref.posterior(x_eval)
+    torch.testing.assert_close(mean_u, mean_r, rtol=1e-7, atol=1e-9)
+    torch.testing.assert_close(var_u, var_r, rtol=1e-5, atol=1e-9)
+
+
+def test_update_data_rejects_non_extension() -> None:
+    rng = np.random.RandomState(19)
+    N0, D = gp.GPRegressor._MAX_DENSE_SQDIFF_OBS + 4, 3
+    X0 = rng.rand(N0, D)
+    gpr = gp.fit_kernel_params(
+        X=X0,
+        Y=rng.randn(N0),
+        is_categorical=np.zeros(D, dtype=bool),
+        log_prior=prior.default_log_prior,
+        minimum_noise=prior.DEFAULT_MINIMUM_NOISE_VAR,
+        deterministic_objective=False,
+    )
+    X_bad = X0.copy()
+    X_bad[5, 0] += 0.5  # mutated prefix — must force a refit
+    assert not gpr.update_data(X_bad, rng.randn(N0))
+    assert not gpr.update_data(X0[:-1], rng.randn(N0 - 1))  # shrunk
+
+
+def test_cloned_with_running_leaves_cache_pristine() -> None:
+    rng = np.random.RandomState(23)
+    X = rng.rand(50, 3)
+    y = rng.randn(50)
+    gpr = _make_gpr(X, y, fit=False)
+    chol_before = gpr._cov_Y_Y_chol.clone()
+    X_run = torch.from_numpy(rng.rand(4, 3))
+    clone = gpr.cloned_with_running(X_run, gpr.posterior(X_run)[0])
+    assert clone is not gpr
+    assert clone._X_all.shape[0] == 54
+    assert gpr._X_all.shape[0] == 50
+    torch.testing.assert_close(gpr._cov_Y_Y_chol, chol_before)
