@@ -59,9 +59,10 @@ class _TcpStoreLog:
     def __init__(self, store: Any) -> None:
         self._store = store
 
-    def append(self, records: list[dict[str, Any]]) -> int:
-        """Publish a batch; returns the end index of the reserved range so the
-        caller's read-back sync can skip the counter query."""
+    def append(self, records: list[dict[str, Any]]) -> tuple[int, int]:
+        """Publish a batch; returns the reserved (start, end) index range so
+        the caller can fill the gap below `start` from the store and apply its
+        own records locally — no read-back of what it just wrote."""
         n = len(records)
         # Reserve a contiguous index range, then publish the records — one
         # round trip for the whole batch when the store supports multi_set.
@@ -77,7 +78,7 @@ class _TcpStoreLog:
         else:
             for key, payload in zip(keys, payloads):
                 self._store.set(key, payload)
-        return end
+        return start, end
 
     def read_range(self, start: int, end: int) -> list[dict[str, Any]]:
         if end <= start:
@@ -114,8 +115,12 @@ class RcclStorage(BaseStorage):
         # trips from ~20 to 0.
         self._pending: list[dict[str, Any]] = []
         self._sync_ttl = float(os.environ.get("OPTUNA_AMD_RCCL_SYNC_TTL", "0.002"))
+        self._cur_ttl = self._sync_ttl
         self._last_sync_at = 0.0
+        self._snapshot_every = int(os.environ.get("OPTUNA_AMD_RCCL_SNAPSHOT_EVERY", "2048"))
+        self._last_snapshot_idx = 0
         with self._thread_lock:
+            self._join_from_snapshot()
             self._sync()
 
     @classmethod
@@ -201,12 +206,17 @@ class RcclStorage(BaseStorage):
         # Pending records must precede this one in the global order.
         batch = self._coalesce_pending() + [rec]
         self._pending = []
-        end = self._log.append(batch)
-        # Apply everything up to our own record without another counter query;
-        # records appended concurrently after `end` arrive on a later sync.
-        self._replay.apply_logs(
-            self._log.read_range(self._replay.log_number_read, end)
-        )
+        start, end = self._log.append(batch)
+        # Fill the gap below our reserved range from the store, then apply our
+        # own records from memory — the store is never asked for bytes this
+        # worker just wrote. Records appended concurrently after `end` arrive
+        # on a later sync.
+        if start > self._replay.log_number_read:
+            self._replay.apply_logs(
+                self._log.read_range(self._replay.log_number_read, start)
+            )
+        self._replay.apply_logs(batch)
+        self._maybe_publish_snapshot(end)
 
     def _defer_append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
         rec = {"op_code": int(op), "worker_id": self._replay.worker_id, **fields}
@@ -214,6 +224,53 @@ class RcclStorage(BaseStorage):
         # idempotently when it comes back through the log.
         self._replay.apply_logs([rec], advance=False)
         self._pending.append(rec)
+
+    # ---- snapshots (delta late-join) ------------------------------------------------
+
+    _SNAP_PTR_KEY = "optuna_amd/oplog/snap_latest"
+    _SNAP_KEY = "optuna_amd/oplog/snap/{idx}"
+
+    def _maybe_publish_snapshot(self, log_idx: int) -> None:
+        """Publish a pickled replay state when this worker's append crossed a
+        snapshot boundary, so late joiners replay only the log tail instead of
+        the whole history (JournalStorage snapshots every 100 studies; here the
+        boundary is every `_snapshot_every` log records — trials dominate)."""
+        import pickle
+
+        covered = self._replay.log_number_read
+        if covered - self._last_snapshot_idx < self._snapshot_every:
+            return
+        # The snapshot is stamped with the exact log index this state covers;
+        # concurrent publishers overwrite the pointer — last writer wins and
+        # the versioned payload key prevents torn reads.
+        payload = pickle.dumps(self._replay)
+        self._log._store.set(self._SNAP_KEY.format(idx=covered), payload)
+        self._log._store.set(self._SNAP_PTR_KEY, str(covered))
+        prev = self._last_snapshot_idx
+        self._last_snapshot_idx = covered
+        if prev and hasattr(self._log._store, "delete_key"):
+            self._log._store.delete_key(self._SNAP_KEY.format(idx=prev))
+
+    def _join_from_snapshot(self) -> None:
+        """Adopt the latest published snapshot (one RPC) and replay only the
+        tail. No-op when no snapshot exists."""
+        import pickle
+
+        store = self._log._store
+        try:
+            if not (hasattr(store, "check") and store.check([self._SNAP_PTR_KEY])):
+                return
+            covered = int(store.get(self._SNAP_PTR_KEY))
+            raw = store.get(self._SNAP_KEY.format(idx=covered))
+        except Exception:
+            return  # racing with a snapshot rotation: fall back to full replay
+        state = pickle.loads(raw)
+        state._worker_id_prefix = self._worker_id_prefix
+        state.log_number_read = covered
+        state.last_created_trial_id = -1
+        state.my_created_trial_ids = []
+        self._replay = state
+        self._last_snapshot_idx = covered
 
     def _sync(self, force: bool = False) -> None:
         """Pull unseen records into the local replay state.
@@ -226,10 +283,17 @@ class RcclStorage(BaseStorage):
         import time as _time
 
         now = _time.monotonic()
-        if not force and now - self._last_sync_at < self._sync_ttl:
+        if not force and now - self._last_sync_at < self._cur_ttl:
             return
         records = self._log.read_from(self._replay.log_number_read)
         self._replay.apply_logs(records)
+        # Adaptive cadence: idle logs back the poll rate off exponentially
+        # (sequencer round trips drop ~25x when nobody is writing); any
+        # activity snaps it back to the base TTL.
+        if records:
+            self._cur_ttl = self._sync_ttl
+        else:
+            self._cur_ttl = min(self._cur_ttl * 2.0, self._sync_ttl * 25.0)
         self._last_sync_at = _time.monotonic()
 
     # ---- studies --------------------------------------------------------------------
@@ -371,9 +435,13 @@ class RcclStorage(BaseStorage):
             before = len(self._replay.my_created_trial_ids)
             batch = self._pending + records
             self._pending = []
-            self._log.append(batch)
-            self._last_sync_at = 0.0
-            self._sync()
+            start, end = self._log.append(batch)
+            if start > self._replay.log_number_read:
+                self._replay.apply_logs(
+                    self._log.read_range(self._replay.log_number_read, start)
+                )
+            self._replay.apply_logs(batch)
+            self._maybe_publish_snapshot(end)
             return list(self._replay.my_created_trial_ids[before:])
 
     def set_trial_param(
