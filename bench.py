@@ -101,6 +101,13 @@ def main() -> None:
                 study_name="bench", storage=storage, sampler=make_sampler()
             )
         dist.barrier()
+        # Steady-state data plane: one all_gather of op batches per round —
+        # device tensors over RCCL/xGMI with the nccl backend, zero sequencer
+        # RPCs. The TCPStore handled bootstrap/populate above.
+        # OPTUNA_AMD_RCCL_MODE=sequencer falls back to the store-ordered mode.
+        if os.environ.get("OPTUNA_AMD_RCCL_MODE", "collective") == "collective":
+            storage.attach_collective_plane()
+        dist.barrier()
     else:
         study = optuna_amd.create_study(sampler=make_sampler())
         _populate(study, names, dists_def, N_HISTORY)
@@ -134,6 +141,9 @@ def main() -> None:
     if dist is not None:
         dist.barrier()
     t1 = time.perf_counter()
+
+    if world_size > 1 and getattr(storage, "_plane", None) is not None:
+        storage.collective_flush()  # propagate the final round's tells
 
     elapsed = t1 - t0
     if dist is not None:

@@ -119,6 +119,10 @@ class RcclStorage(BaseStorage):
         self._last_sync_at = 0.0
         self._snapshot_every = int(os.environ.get("OPTUNA_AMD_RCCL_SNAPSHOT_EVERY", "2048"))
         self._last_snapshot_idx = 0
+        # Collective (lockstep BSP) mode — see attach_collective_plane.
+        self._plane = None
+        self._plane_log: list[dict[str, Any]] = []
+        self._next_local_seq = 0
         with self._thread_lock:
             self._join_from_snapshot()
             self._sync()
@@ -202,6 +206,12 @@ class RcclStorage(BaseStorage):
         return out
 
     def _append(self, op: JournalOperation, fields: dict[str, Any]) -> None:
+        if self._plane is not None:
+            # Lockstep mode: every write buffers; batches move device-to-device
+            # in the next round's exchange (inside create_new_trial).
+            assert op != JournalOperation.CREATE_TRIAL
+            self._defer_append(op, fields)
+            return
         rec = {"op_code": int(op), "worker_id": self._replay.worker_id, **fields}
         # Pending records must precede this one in the global order.
         batch = self._coalesce_pending() + [rec]
@@ -229,6 +239,7 @@ class RcclStorage(BaseStorage):
 
     _SNAP_PTR_KEY = "optuna_amd/oplog/snap_latest"
     _SNAP_KEY = "optuna_amd/oplog/snap/{idx}"
+    _SNAP_CHUNK = 4 * 1024 * 1024
 
     def _maybe_publish_snapshot(self, log_idx: int) -> None:
         """Publish a pickled replay state when this worker's append crossed a
@@ -236,35 +247,49 @@ class RcclStorage(BaseStorage):
         the whole history (JournalStorage snapshots every 100 studies; here the
         boundary is every `_snapshot_every` log records — trials dominate)."""
         import pickle
+        import zlib
 
         covered = self._replay.log_number_read
         if covered - self._last_snapshot_idx < self._snapshot_every:
             return
         # The snapshot is stamped with the exact log index this state covers;
         # concurrent publishers overwrite the pointer — last writer wins and
-        # the versioned payload key prevents torn reads.
-        payload = pickle.dumps(self._replay)
-        self._log._store.set(self._SNAP_KEY.format(idx=covered), payload)
-        self._log._store.set(self._SNAP_PTR_KEY, str(covered))
+        # the versioned payload keys prevent torn reads. Compressed + chunked:
+        # TCPStore caps one payload at 8 MiB and a 10k-trial state pickles to
+        # ~11 MB raw.
+        payload = zlib.compress(pickle.dumps(self._replay), level=1)
+        chunks = [payload[i : i + self._SNAP_CHUNK] for i in range(0, len(payload), self._SNAP_CHUNK)] or [b""]
+        store = self._log._store
+        for k, chunk in enumerate(chunks):
+            store.set(self._SNAP_KEY.format(idx=covered) + f"/{k}", chunk)
+        store.set(self._SNAP_PTR_KEY, f"{covered}:{len(chunks)}")
         prev = self._last_snapshot_idx
         self._last_snapshot_idx = covered
-        if prev and hasattr(self._log._store, "delete_key"):
-            self._log._store.delete_key(self._SNAP_KEY.format(idx=prev))
+        if prev and hasattr(store, "delete_key"):
+            k = 0
+            while store.delete_key(self._SNAP_KEY.format(idx=prev) + f"/{k}"):
+                k += 1
 
     def _join_from_snapshot(self) -> None:
         """Adopt the latest published snapshot (one RPC) and replay only the
         tail. No-op when no snapshot exists."""
         import pickle
+        import zlib
 
         store = self._log._store
         try:
             if not (hasattr(store, "check") and store.check([self._SNAP_PTR_KEY])):
                 return
-            covered = int(store.get(self._SNAP_PTR_KEY))
-            raw = store.get(self._SNAP_KEY.format(idx=covered))
+            covered_s, _, n_chunks_s = store.get(self._SNAP_PTR_KEY).decode().partition(":")
+            covered = int(covered_s)
+            n_chunks = int(n_chunks_s or "1")
+            raw = b"".join(
+                store.get(self._SNAP_KEY.format(idx=covered) + f"/{k}")
+                for k in range(n_chunks)
+            )
         except Exception:
             return  # racing with a snapshot rotation: fall back to full replay
-        state = pickle.loads(raw)
+        state = pickle.loads(zlib.decompress(raw))
         state._worker_id_prefix = self._worker_id_prefix
         state.log_number_read = covered
         state.last_created_trial_id = -1
@@ -280,6 +305,8 @@ class RcclStorage(BaseStorage):
         their forced post-append sync or previewed when buffered), so read-side
         syncs are rate-limited to one sequencer round trip per TTL window.
         """
+        if self._plane is not None:
+            return  # replicas converge at the collective exchanges
         import time as _time
 
         now = _time.monotonic()
@@ -295,6 +322,76 @@ class RcclStorage(BaseStorage):
         else:
             self._cur_ttl = min(self._cur_ttl * 2.0, self._sync_ttl * 25.0)
         self._last_sync_at = _time.monotonic()
+
+    # ---- collective (lockstep BSP) mode ---------------------------------------------
+
+    # Rank-partitioned trial-id space for collectively-created trials; far above
+    # any sequencer-era positional id, so the two modes can coexist in one log.
+    _COLLECTIVE_ID_BASE = 1 << 40
+
+    def attach_collective_plane(self, group: Any = None, device: Any = None) -> None:
+        """Switch this replica into lockstep collective mode (RCCL/xGMI).
+
+        After attaching, writes buffer locally and move device-to-device in ONE
+        all_gather per round, triggered inside ``create_new_trial`` — so every
+        rank must call ask/tell the same number of times (the torchrun study
+        loop does). The TCPStore only serves bootstrap and late-join; the
+        steady-state path makes zero loopback RPCs. Call on every rank at the
+        same point (e.g. right after a barrier), with identical replay states
+        (force-synced here).
+        """
+        from optuna_amd.parallel.collective import CollectiveOpPlane
+
+        with self._thread_lock:
+            self._sync(force=True)
+            self._plane = CollectiveOpPlane(group=group, device=device)
+            self._next_local_seq = 0
+
+    def _collective_create(self, log_fields: dict[str, Any]) -> int:
+        plane = self._plane
+        assert plane is not None
+        seq = self._next_local_seq
+        self._next_local_seq += 1
+        trial_id = self._COLLECTIVE_ID_BASE + seq * plane.world + plane.rank
+        rec = {
+            "op_code": int(JournalOperation.CREATE_TRIAL),
+            "worker_id": self._replay.worker_id,
+            "trial_id": trial_id,
+            **log_fields,
+        }
+        batch = self._coalesce_pending() + [rec]
+        self._pending = []
+        # Merge order = (round, rank, local seq): identical on every replica,
+        # so numbering and last-writer-wins converge without a sequencer. Own
+        # records were already preview-applied at defer time (and the preview
+        # may include the trial's FINISH, after which a re-applied param write
+        # would be rejected) — so apply only the peers' batches plus our own
+        # CREATE, which is never previewed because its number is the merge
+        # position.
+        for r, rank_batch in enumerate(plane.exchange_records(batch)):
+            self._plane_log.extend(rank_batch)  # full merged log (to_journal)
+            if r == plane.rank:
+                rank_batch = [
+                    rec_
+                    for rec_ in rank_batch
+                    if rec_["op_code"] == int(JournalOperation.CREATE_TRIAL)
+                ]
+            self._replay.apply_logs(rank_batch, advance=False)
+        return trial_id
+
+    def collective_flush(self) -> None:
+        """Exchange buffered records without creating a trial (end-of-run
+        flush). Lockstep: every rank must call this at the same point."""
+        with self._thread_lock:
+            plane = self._plane
+            assert plane is not None, "collective_flush requires an attached plane"
+            batch = self._coalesce_pending()
+            self._pending = []
+            for r, rank_batch in enumerate(plane.exchange_records(batch)):
+                self._plane_log.extend(rank_batch)
+                if r == plane.rank:
+                    continue  # own records were preview-applied at defer time
+                self._replay.apply_logs(rank_batch, advance=False)
 
     # ---- studies --------------------------------------------------------------------
 
@@ -337,9 +434,14 @@ class RcclStorage(BaseStorage):
     def get_study_id_from_name(self, study_name: str) -> int:
         with self._thread_lock:
             self._sync()
-            for fs in self._replay.all_studies():
-                if fs.study_name == study_name:
-                    return fs._study_id
+            for force in (False, True):
+                for fs in self._replay.all_studies():
+                    if fs.study_name == study_name:
+                        return fs._study_id
+                if not force:
+                    # A TTL-rate-limited sync can miss a study created on a
+                    # peer moments ago; retry once with a forced sync.
+                    self._sync(force=True)
             raise KeyError(NOT_FOUND_MSG)
 
     def get_study_name_from_id(self, study_id: int) -> str:
@@ -409,6 +511,8 @@ class RcclStorage(BaseStorage):
     def create_new_trial(self, study_id: int, template_trial: FrozenTrial | None = None) -> int:
         log = self._create_trial_record(study_id, template_trial)
         with self._thread_lock:
+            if self._plane is not None:
+                return self._collective_create(log)
             self._append(JournalOperation.CREATE_TRIAL, log)
             self._sync()
             return self._replay.last_created_trial_id
@@ -495,6 +599,11 @@ class RcclStorage(BaseStorage):
             log["datetime_complete"] = _utcnow_iso()
         with self._thread_lock:
             if state == TrialState.RUNNING:
+                if self._plane is not None:
+                    raise RuntimeError(
+                        "WAITING->RUNNING claims (enqueue_trial) are not part of "
+                        "the lockstep collective mode; use the sequencer mode."
+                    )
                 self._sync()
                 existing = self._replay._trials.get(trial_id)
                 if existing is None:
@@ -572,6 +681,26 @@ class RcclStorage(BaseStorage):
         with self._thread_lock:
             self._sync(force=True)
             records = self._log.read_from(0)
+            plane_records = list(self._plane_log)
+        if plane_records:
+            # Collective-mode records carry explicit rank-partitioned trial
+            # ids; rewrite them to the journal's positional convention so the
+            # export replays under any JournalStorage (incl. the reference's).
+            n_creates = sum(
+                1 for r in records if r["op_code"] == int(JournalOperation.CREATE_TRIAL)
+            )
+            id_map: dict[int, int] = {}
+            rewritten = []
+            for rec in plane_records:
+                rec = dict(rec)
+                if rec["op_code"] == int(JournalOperation.CREATE_TRIAL):
+                    explicit = rec.pop("trial_id")
+                    id_map[explicit] = n_creates
+                    n_creates += 1
+                elif "trial_id" in rec:
+                    rec["trial_id"] = id_map.get(rec["trial_id"], rec["trial_id"])
+                rewritten.append(rec)
+            records = records + rewritten
         expanded: list[dict[str, Any]] = []
         for rec in records:
             if rec["op_code"] == int(JournalOperation.SET_TRIAL_PARAMS_BATCH):

@@ -567,7 +567,14 @@ class _ReplayState:
         study_id = log["study_id"]
         if not self._study_exists(study_id, log):
             return
-        trial_id = len(self._trials)
+        # Default ids are positional (global creation order, the journal
+        # convention). RcclStorage's collective plane pre-assigns
+        # rank-partitioned ids so a worker knows its trial id before the
+        # records are merged — those records carry the id explicitly and
+        # re-application is a no-op.
+        trial_id = log.get("trial_id", len(self._trials))
+        if trial_id in self._trials:
+            return
         distributions = {
             k: json_to_distribution(v) for k, v in log.get("distributions", {}).items()
         }
