@@ -404,6 +404,72 @@ __global__ void k_mix_logpdf_merge(const double* __restrict__ part_m,
     }
 }
 
+// Per-dim variant for independent-mode TPE (multi-objective default): one
+// workgroup per (sample, dim); out[s, d] = logsumexp_k(logw[k] + term(k, d)).
+// S·D blocks (24 candidates × D dims) with a thread-strided K loop — fills the
+// chip far better than the S-block joint kernel at small S, and one launch
+// replaces D independent 1-dim suggest round trips.
+__global__ void k_mix_logpdf_perdim(const double* __restrict__ x,  // (S, D)
+                                    const double* __restrict__ xedges,  // (S, 2D)
+                                    const double* __restrict__ c1,
+                                    const double* __restrict__ c2,
+                                    const double* __restrict__ c3,
+                                    const double* __restrict__ logw,
+                                    const double* __restrict__ steps,
+                                    const double* __restrict__ n_choices,
+                                    double cat_base, int64_t K, int64_t D,
+                                    double* __restrict__ out) {  // (S, D)
+    __shared__ double red_m[256];
+    __shared__ double red_s[256];
+    const int64_t s = blockIdx.x;
+    const int64_t d = blockIdx.y;
+    const double xs = x[s * D + d];
+    const double x2v = xs * xs;
+    const double xl = xedges[s * 2 * D + d];
+    const double xr = xedges[s * 2 * D + D + d];
+
+    double m = -INFINITY, acc = 0.0;
+    // Base-offset the per-dim coefficient/descriptor pointers so mix_term's
+    // d=0 indexing reads dim d's data against the scalar candidate values.
+    const double* c1d = c1 + (size_t)d * K;
+    const double* c2d = c2 + (size_t)d * K;
+    const double* c3d = c3 + (size_t)d * K;
+    for (int64_t k = threadIdx.x; k < K; k += blockDim.x) {
+        double t = logw[k] + mix_term(c1d, c2d, c3d, steps + d, n_choices + d,
+                                      cat_base, &xs, &x2v, &xl, &xr, K, k, 0);
+        if (t > m) {
+            acc = acc * exp(m - t) + 1.0;
+            m = t;
+        } else {
+            acc += exp(t - m);
+        }
+    }
+    red_m[threadIdx.x] = m;
+    red_s[threadIdx.x] = acc;
+    __syncthreads();
+    for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if (threadIdx.x < stride) {
+            double m2 = red_m[threadIdx.x + stride];
+            double s2 = red_s[threadIdx.x + stride];
+            double m1 = red_m[threadIdx.x];
+            double s1 = red_s[threadIdx.x];
+            if (m2 > m1) {
+                s1 = s1 * exp(m1 - m2) + s2;
+                m1 = m2;
+            } else if (m1 != -INFINITY) {
+                s1 = s1 + s2 * exp(m2 - m1);
+            }
+            red_m[threadIdx.x] = m1;
+            red_s[threadIdx.x] = s1;
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out[s * D + d] =
+            (red_m[0] == -INFINITY) ? -INFINITY : red_m[0] + log(red_s[0]);
+    }
+}
+
 // Chunk rows so S × n_chunks blocks ≈ 2 per CU; below one chunk the original
 // single-phase kernel is cheaper (no scratch round-trip).
 static constexpr int64_t MIX_CHUNK = 512;
@@ -1112,7 +1178,8 @@ class TpeDeviceHistory {
                               bool consider_endpoints, bool magic_clip,
                               const arr_f64& extras_raw,         // (L, D) or empty
                               const arr_f64& extras_sorted,      // (D, L)
-                              const arr_i32& extras_sorted_idx)  // (D, L)
+                              const arr_i32& extras_sorted_idx,  // (D, L)
+                              bool per_dim)
     {
         const bool resident_sorted = sorted_cols.empty() && sorted_valid_;
         const int64_t Nv =
@@ -1143,7 +1210,8 @@ class TpeDeviceHistory {
         // f64 slots: c1|c2|c3|logw|alow|ahigh|steps|x|xedges|out|lse|extras + i32.
         const size_t n_scratch = 2 * (size_t)mix_n_chunks(K) * S;
         const size_t n_extras_f64 = 2 * (size_t)L * D_;  // raw + per-dim sorted
-        size_t f64_total = 3 * n_c + K + 4 * D_ + 3 * (size_t)S * D_ + S +
+        const size_t out_elems = (size_t)S * (per_dim ? D_ : 1);
+        size_t f64_total = 3 * n_c + K + 4 * D_ + 3 * (size_t)S * D_ + out_elems +
                            n_scratch + n_extras_f64;
         size_t i32_doubles = ((size_t)Nv * D_ /*sorted*/ + n_ /*pos*/ +
                               2 * (size_t)Nv * D_ /*sub*/ +
@@ -1164,7 +1232,7 @@ class TpeDeviceHistory {
         double* d_x = d_nchoices + D_;
         double* d_xedges = d_x + (size_t)S * D_;
         double* d_out = d_xedges + 2 * (size_t)S * D_;
-        double* d_scratch = d_out + S;
+        double* d_scratch = d_out + out_elems;
         double* d_extras_raw = d_scratch + n_scratch;
         double* d_extras_sorted = d_extras_raw + (size_t)L * D_;
         int32_t* d_sorted =
@@ -1233,6 +1301,19 @@ class TpeDeviceHistory {
                                magic_clip ? 1 : 0, d_c1, d_c2, d_c3);
         }
         const double cat_base = prior_weight / (double)K;
+        if (per_dim) {
+            hipLaunchKernelGGL(k_mix_logpdf_perdim,
+                               dim3((unsigned)S, (unsigned)D_), dim3(256), 0, st,
+                               d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps,
+                               d_nchoices, cat_base, K, D_, d_out);
+            py::array_t<double> out({S, D_});
+            HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out,
+                                     (size_t)S * D_ * 8, hipMemcpyDeviceToHost,
+                                     st));
+            HIP_CHECK(hipStreamSynchronize(st));
+            HIP_CHECK(hipGetLastError());
+            return out;
+        }
         launch_mix_logpdf(st, d_x, d_xedges, d_c1, d_c2, d_c3, d_logw, d_steps,
                           d_nchoices, cat_base, K, D_, S, d_out, d_scratch);
         py::array_t<double> out(S);
@@ -1349,5 +1430,6 @@ PYBIND11_MODULE(_hipcore, m) {
              py::arg("magic_clip") = true,
              py::arg("extras_raw") = arr_f64(),
              py::arg("extras_sorted") = arr_f64(),
-             py::arg("extras_sorted_idx") = arr_i32());
+             py::arg("extras_sorted_idx") = arr_i32(),
+             py::arg("per_dim") = false);
 }

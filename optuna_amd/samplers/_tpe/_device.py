@@ -175,6 +175,7 @@ class _SpaceDeviceMirror:
         consider_magic_clip: bool,
         prior_weight: float = 1.0,
         extras: np.ndarray | None = None,
+        per_dim: bool = False,
     ) -> np.ndarray:
         """Score candidates against the resident table.
 
@@ -232,6 +233,7 @@ class _SpaceDeviceMirror:
             np.ascontiguousarray(xedges),
             consider_endpoints,
             consider_magic_clip,
+            per_dim=per_dim,
             **kwargs,
         )
 
@@ -245,16 +247,19 @@ def score_above_resident(
     consider_magic_clip: bool,
     prior_weight: float = 1.0,
     extras: np.ndarray | None = None,
+    per_dim: bool = False,
 ) -> np.ndarray:
     """log g(x) for candidates via the device-resident table (creates the mirror
-    on first use; attached to the host space cache so lifetimes match)."""
+    on first use; attached to the host space cache so lifetimes match).
+    ``per_dim=True`` returns the (S, D) per-dimension 1-D mixture log-pdfs
+    (independent-mode TPE) instead of the joint (S,) product score."""
     mirror = getattr(cache, "_device_mirror", None)
     if mirror is None:
         mirror = _SpaceDeviceMirror(cache.space)
         cache._device_mirror = mirror  # type: ignore[attr-defined]
     return mirror.score(
         cache, sel, weights, samples, consider_endpoints, consider_magic_clip,
-        prior_weight=prior_weight, extras=extras,
+        prior_weight=prior_weight, extras=extras, per_dim=per_dim,
     )
 
 

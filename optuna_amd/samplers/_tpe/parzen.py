@@ -238,10 +238,12 @@ class _ParzenEstimator:
                 - 0.5 * np.log(2 * np.pi)
                 - self._log_total_mass[:, cont]
             )
+            self._c3 = c3  # (K, Dc) — per-dim scoring (independent mode)
             self._c3_rowsum = c3.sum(axis=1)  # (K,)
         else:
             self._c1 = np.empty((len(self._weights), 0))
             self._c2 = np.empty((len(self._weights), 0))
+            self._c3 = np.empty((len(self._weights), 0))
             self._c3_rowsum = np.zeros(len(self._weights))
 
     # ---- fitting helpers ------------------------------------------------------------
@@ -426,6 +428,110 @@ class _ParzenEstimator:
             self._param_names
         ) else np.empty((0, 0))
         return self._log_pdf_array(x)
+
+    def sample_per_dim(
+        self, rng: np.random.RandomState, size: int
+    ) -> dict[str, np.ndarray]:
+        """Independent-mode draws: each dim picks its own mixture component.
+
+        The joint ``sample`` picks ONE component per draw shared by all dims (a
+        product-mixture draw); independent-mode TPE treats every dim as its own
+        1-D mixture, which is exactly a per-(draw, dim) component choice.
+        """
+        D = self.n_dims
+        active = np.column_stack(
+            [rng.choice(len(self._weights), p=self._weights, size=size) for _ in range(D)]
+        )
+        out = np.empty((size, D), dtype=np.float64)
+        num = self._numerical
+        Dn = num.mus.shape[1]
+        if Dn:
+            act_n = active[:, num.dim_indices]  # (S, Dn)
+            cols = np.arange(Dn)[np.newaxis, :]
+            mus = num.mus[act_n, cols]
+            sigmas = num.sigmas[act_n, cols]
+            a = (num.adapted_lows[np.newaxis, :] - mus) / sigmas
+            b = (num.adapted_highs[np.newaxis, :] - mus) / sigmas
+            vals = _tn.rvs(a, b, loc=mus, scale=sigmas, random_state=rng)
+            is_log = (num.kinds == KIND_LOG) | (num.kinds == KIND_LOG_DISC)
+            vals[:, is_log] = np.exp(vals[:, is_log])
+            has_step = num.steps > 0
+            if np.any(has_step):
+                lo = num.lows[has_step]
+                st = num.steps[has_step]
+                hi = num.highs[has_step]
+                vals[:, has_step] = np.clip(
+                    lo + np.round((vals[:, has_step] - lo) / st) * st, lo, hi
+                )
+            out[:, num.dim_indices] = vals
+        for cat in self._categoricals:
+            active_w = cat.weights[active[:, cat.dim_index], :]
+            q = rng.rand(size)
+            cum = np.cumsum(active_w, axis=-1)
+            cum[:, -1] = 1.0
+            out[:, cat.dim_index] = np.sum(cum < q[:, np.newaxis], axis=-1)
+        return {name: out[:, i] for i, name in enumerate(self._param_names)}
+
+    def log_pdf_per_dim(self, samples_dict: dict[str, np.ndarray]) -> np.ndarray:
+        """(S, D) of per-dim 1-D mixture log densities (independent mode)."""
+        x = np.column_stack([samples_dict[name] for name in self._param_names])
+        S = x.shape[0]
+        K = len(self._weights)
+        D = self.n_dims
+        out = np.empty((S, D), dtype=np.float64)
+        logw = np.log(self._weights)[np.newaxis, :]
+
+        def lse(lt: np.ndarray) -> np.ndarray:  # (S, K) -> (S,)
+            m = lt.max(axis=1)
+            m[np.isneginf(m)] = 0.0
+            with np.errstate(divide="ignore"):
+                return np.log(np.exp(lt - m[:, None]).sum(axis=1)) + m
+
+        num = self._numerical
+        Dn = num.mus.shape[1]
+        if Dn:
+            xv = x[:, num.dim_indices]
+            is_log = (num.kinds == KIND_LOG) | (num.kinds == KIND_LOG_DISC)
+            xv = np.where(is_log[np.newaxis, :], np.log(np.maximum(xv, EPS)), xv)
+            is_disc = num.steps > 0
+            cont = self._cont_mask
+            cont_cols = np.nonzero(cont)[0]
+            for jc, j in enumerate(cont_cols):
+                xj = xv[:, j][:, np.newaxis]  # (S, 1)
+                lt = (
+                    xj * xj * self._c1[:, jc][np.newaxis, :]
+                    + xj * self._c2[:, jc][np.newaxis, :]
+                    + self._c3[:, jc][np.newaxis, :]
+                    + logw
+                )
+                vals = lse(lt)
+                outside = (xv[:, j] < num.adapted_lows[j]) | (
+                    xv[:, j] > num.adapted_highs[j]
+                )
+                vals[outside] = -np.inf
+                out[:, num.dim_indices[j]] = vals
+            for j in np.nonzero(is_disc)[0]:
+                half = num.steps[j] / 2
+                xj_raw = x[:, num.dim_indices[j]]
+                if is_log[j]:
+                    left = np.log(xj_raw - half)
+                    right = np.log(xj_raw + half)
+                else:
+                    left = xj_raw - half
+                    right = xj_raw + half
+                m = num.mus[:, j]
+                sg = num.sigmas[:, j]
+                cell = _tn._log_gauss_mass(
+                    (left[:, np.newaxis] - m) / sg, (right[:, np.newaxis] - m) / sg
+                )
+                out[:, num.dim_indices[j]] = lse(
+                    cell - self._log_total_mass[:, j][np.newaxis, :] + logw
+                )
+        for cat in self._categoricals:
+            xi = x[:, cat.dim_index].astype(np.int64)
+            with np.errstate(divide="ignore"):
+                out[:, cat.dim_index] = lse(np.log(cat.weights.T[xi, :]) + logw)
+        return out
 
     # Array-level interfaces (used by the HIP dispatch and bench harness).
 

@@ -46,6 +46,9 @@ if TYPE_CHECKING:
 EPS = 1e-12
 _logger = _logging.get_logger(__name__)
 
+# Sentinel: the batched independent-mode path does not apply; sample per-dim.
+_BATCH_MISS = object()
+
 _RELATIVE_PARAMS_KEY = "tpe:relative_params"
 # RDB system-attr values must stay under 2046 chars; long payloads are chunked.
 _SYSTEM_ATTR_MAX_LENGTH = 2045
@@ -124,6 +127,10 @@ class TPESampler(BaseSampler):
         self._parzen_estimator_cls = _ParzenEstimator
         # Per-study incremental history mirrors (see _history.py).
         self._histories: dict[int, Any] = {}
+        # Per-trial batched prefetch for independent-mode sampling (the
+        # multi-objective default): ONE device round scores all dims at once;
+        # later per-dim requests of the same trial hit this cache.
+        self._indep_prefetch: tuple[int, dict[str, tuple[Any, Any]]] | None = None
         # n_jobs>1 runs suggests from worker threads; the history mirror is
         # stateful (capacity buffers, incrementally sorted indices), so its
         # read-modify-write cycle must be serialized. The reference's samplers
@@ -249,7 +256,55 @@ class TPESampler(BaseSampler):
                     )
                 )
 
+        batched = self._sample_independent_batched(
+            study, trial, param_name, param_distribution
+        )
+        if batched is not _BATCH_MISS:
+            return batched
         return self._sample(study, trial, {param_name: param_distribution})[param_name]
+
+    def _sample_independent_batched(
+        self,
+        study: "Study",
+        trial: FrozenTrial,
+        param_name: str,
+        param_distribution: BaseDistribution,
+    ) -> Any:
+        """Serve independent-mode requests from a per-trial all-dims batch.
+
+        Independent mode asks one dim at a time, but each ask re-reads the
+        history, rebuilds estimators and round-trips the device — D times per
+        trial. When the history's intersection space contains the requested
+        dim (same distribution) and every finished trial defines every dim of
+        that space (so the per-dim observation subsets coincide with the joint
+        subsets), all dims are sampled in one per-dim-batched pass and cached
+        for the trial. Falls back per-dim otherwise (_BATCH_MISS).
+        """
+        with self._history_lock:
+            pf = self._indep_prefetch
+            if pf is not None and pf[0] == trial._trial_id:
+                hit = pf[1].get(param_name)
+                if hit is not None and hit[1] == param_distribution:
+                    return hit[0]
+            space = {
+                name: dist
+                for name, dist in self._search_space.calculate(study).items()
+                if not dist.single()
+            }
+            if (
+                len(space) <= 1
+                or param_name not in space
+                or space[param_name] != param_distribution
+            ):
+                return _BATCH_MISS
+            ret = self._sample_locked(study, trial, space, per_dim=True)
+            if ret is None:
+                return _BATCH_MISS
+            self._indep_prefetch = (
+                trial._trial_id,
+                {n: (v, space[n]) for n, v in ret.items()},
+            )
+            return ret[param_name]
 
     # ---- the compute core -----------------------------------------------------------
 
@@ -293,8 +348,12 @@ class TPESampler(BaseSampler):
             return self._sample_locked(study, trial, search_space)
 
     def _sample_locked(
-        self, study: "Study", trial: FrozenTrial, search_space: dict[str, BaseDistribution]
-    ) -> dict[str, Any]:
+        self,
+        study: "Study",
+        trial: FrozenTrial,
+        search_space: dict[str, BaseDistribution],
+        per_dim: bool = False,
+    ) -> dict[str, Any] | None:
         from optuna_amd.samplers._tpe._history import _TpeHistory
 
         history = self._histories.get(study._study_id)
@@ -328,11 +387,19 @@ class TPESampler(BaseSampler):
         n = len(history)
         below_rows, above_rows = history.split(study, self._gamma(n))
 
-        obs_below, orders_below = history.observations(search_space, below_rows)
-
         from optuna_amd.samplers._tpe import _device
 
         above_sel = history.valid_rows(search_space, above_rows)
+        if per_dim:
+            # The batch is only equivalent to D independent 1-dim suggests
+            # when every trial of both subsets defines every dim — then the
+            # per-dim observation sets coincide with the joint ones.
+            below_sel_chk = history.valid_rows(search_space, below_rows)
+            if len(below_sel_chk) != len(below_rows) or len(above_sel) != len(
+                above_rows
+            ):
+                return None
+        obs_below, orders_below = history.observations(search_space, below_rows)
 
         # Constant-liar rows: other workers' RUNNING trials join the "above"
         # set (params shared via system attrs). As an (L, D) matrix they ride
@@ -383,7 +450,10 @@ class TPESampler(BaseSampler):
         mpe_below = self._build_mpe(
             study, search_space, obs_below, handle_below=True, orders=orders_below
         )
-        samples_below = mpe_below.sample(self._rng.rng, self._n_ei_candidates)
+        if per_dim:
+            samples_below = mpe_below.sample_per_dim(self._rng.rng, self._n_ei_candidates)
+        else:
+            samples_below = mpe_below.sample(self._rng.rng, self._n_ei_candidates)
 
         # Device path (K1+K2): the big "above" KDE is fit and scored against the
         # HBM-resident parameter table for every distribution type (continuous,
@@ -401,15 +471,34 @@ class TPESampler(BaseSampler):
                 self._parzen_estimator_parameters.consider_magic_clip,
                 prior_weight=self._parzen_estimator_parameters.prior_weight,
                 extras=liar_extras,
+                per_dim=per_dim,
             )
-            acq_func_vals = mpe_below.log_pdf(samples_below) - log_g
+            if per_dim:
+                acq_func_vals = mpe_below.log_pdf_per_dim(samples_below) - log_g
+            else:
+                acq_func_vals = mpe_below.log_pdf(samples_below) - log_g
         else:
             mpe_above = self._build_mpe(
                 study, search_space, obs_above, handle_below=False, orders=orders_above
             )
-            acq_func_vals = self._compute_acquisition_func(
-                samples_below, mpe_below, mpe_above
-            )
+            if per_dim:
+                acq_func_vals = mpe_below.log_pdf_per_dim(
+                    samples_below
+                ) - mpe_above.log_pdf_per_dim(samples_below)
+            else:
+                acq_func_vals = self._compute_acquisition_func(
+                    samples_below, mpe_below, mpe_above
+                )
+        if per_dim:
+            # Per-dim argmax: each dim independently keeps its own best of the
+            # S candidates — exactly what D separate 1-dim suggests would do.
+            ret = {}
+            for c, (param_name, dist) in enumerate(search_space.items()):
+                best = int(np.argmax(acq_func_vals[:, c]))
+                ret[param_name] = dist.to_external_repr(
+                    samples_below[param_name][best].item()
+                )
+            return ret
         ret = TPESampler._compare(samples_below, acq_func_vals)
 
         for param_name, dist in search_space.items():

@@ -856,3 +856,45 @@ def test_gp_incremental_update_block_inverse_matches_refit() -> None:
     # the cadence policy: growth < n_fit/100 (=9 here) → update, more → refit
     assert gp_mod._incremental_update_applicable(gpr, N0 + 5)
     assert not gp_mod._incremental_update_applicable(gpr, N0 + 50)
+
+
+def test_per_dim_device_score_matches_host(core) -> None:
+    """k_mix_logpdf_perdim vs the host per-dim estimator on the same mixture."""
+    from optuna_amd.samplers._tpe import _device
+    from optuna_amd.samplers._tpe._history import _SpaceCache
+    from optuna_amd.samplers._tpe.parzen import (
+        _ParzenEstimator,
+        _ParzenEstimatorParameters,
+    )
+    from optuna_amd.samplers._tpe.sampler import default_weights
+    from optuna_amd.testing.trials import _create_frozen_trial
+
+    rng = np.random.RandomState(21)
+    n, D = 900, 6
+    space = {f"x{i}": FloatDistribution(-4.0, 4.0) for i in range(D)}
+    mat = rng.uniform(-4, 4, size=(n, D))
+    trials = [
+        _create_frozen_trial(
+            number=r,
+            values=(float(rng.rand()),),
+            params={f"x{i}": float(mat[r, i]) for i in range(D)},
+            distributions=dict(space),
+        )
+        for r in range(n)
+    ]
+    cache = _SpaceCache(space)
+    cache.append(trials)
+    sel = np.arange(n)
+    w = np.append(default_weights(n), 1.0)
+    w = w / w.sum()
+    samples = {f"x{i}": rng.uniform(-4, 4, 24) for i in range(D)}
+    got = _device.score_above_resident(
+        cache, sel, w, samples, False, True, per_dim=True
+    )
+    assert got.shape == (24, D)
+
+    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, False)
+    obs = {f"x{i}": mat[:, i] for i in range(D)}
+    mpe = _ParzenEstimator(obs, space, params)
+    want = mpe.log_pdf_per_dim(samples)
+    np.testing.assert_allclose(got, want, rtol=1e-9, atol=1e-9)
