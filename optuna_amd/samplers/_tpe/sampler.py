@@ -129,8 +129,10 @@ class TPESampler(BaseSampler):
         self._histories: dict[int, Any] = {}
         # Per-trial batched prefetch for independent-mode sampling (the
         # multi-objective default): ONE device round scores all dims at once;
-        # later per-dim requests of the same trial hit this cache.
-        self._indep_prefetch: tuple[int, dict[str, tuple[Any, Any]]] | None = None
+        # later per-dim requests of the same trial hit this cache. Keyed by
+        # trial id (bounded) so n_jobs>1 threads interleaving trials don't
+        # evict each other's batch.
+        self._indep_prefetch: dict[int, dict[str, tuple[Any, Any]]] = {}
         # n_jobs>1 runs suggests from worker threads; the history mirror is
         # stateful (capacity buffers, incrementally sorted indices), so its
         # read-modify-write cycle must be serialized. The reference's samplers
@@ -281,9 +283,9 @@ class TPESampler(BaseSampler):
         for the trial. Falls back per-dim otherwise (_BATCH_MISS).
         """
         with self._history_lock:
-            pf = self._indep_prefetch
-            if pf is not None and pf[0] == trial._trial_id:
-                hit = pf[1].get(param_name)
+            pf = self._indep_prefetch.get(trial._trial_id)
+            if pf is not None:
+                hit = pf.get(param_name)
                 if hit is not None and hit[1] == param_distribution:
                     return hit[0]
             space = {
@@ -300,10 +302,11 @@ class TPESampler(BaseSampler):
             ret = self._sample_locked(study, trial, space, per_dim=True)
             if ret is None:
                 return _BATCH_MISS
-            self._indep_prefetch = (
-                trial._trial_id,
-                {n: (v, space[n]) for n, v in ret.items()},
-            )
+            if len(self._indep_prefetch) >= 64:
+                self._indep_prefetch.pop(next(iter(self._indep_prefetch)))
+            self._indep_prefetch[trial._trial_id] = {
+                n: (v, space[n]) for n, v in ret.items()
+            }
             return ret[param_name]
 
     # ---- the compute core -----------------------------------------------------------
