@@ -30,34 +30,44 @@ def _solve_hssp_2d(
     sorted_vals = rank_i_loss_vals[order]
     sorted_idx = rank_i_indices[order]
 
-    # Doubly-linked neighbor structure over the sorted front.
+    # Doubly-linked neighbor structure over the sorted front; contributions
+    # maintained as arrays so each greedy round is one vectorized argmax
+    # (k·n numpy ops instead of k·n Python-loop iterations).
     left = np.arange(-1, n - 1)
     right = np.arange(1, n + 1)
-    selected = np.zeros(n, dtype=bool)
+    x_of = sorted_vals[:, 0]
+    y_of = sorted_vals[:, 1]
 
-    def contrib(j: int) -> float:
-        x_right = reference_point[0] if right[j] >= n else sorted_vals[right[j], 0]
-        y_left = reference_point[1] if left[j] < 0 else sorted_vals[left[j], 1]
-        return float((x_right - sorted_vals[j, 0]) * (y_left - sorted_vals[j, 1]))
+    def right_x() -> np.ndarray:
+        out = np.where(right < n, x_of[np.minimum(right, n - 1)], reference_point[0])
+        return out
 
+    def left_y() -> np.ndarray:
+        out = np.where(left >= 0, y_of[np.maximum(left, 0)], reference_point[1])
+        return out
+
+    contribs = (right_x() - x_of) * (left_y() - y_of)
     chosen: list[int] = []
     for _ in range(subset_size):
-        best_j, best_c = -1, -np.inf
-        for j in range(n):
-            if selected[j]:
-                continue
-            c = contrib(j)
-            if c > best_c:
-                best_j, best_c = j, c
-        assert best_j >= 0
-        selected[best_j] = True
+        best_j = int(np.argmax(contribs))
+        assert np.isfinite(contribs[best_j])
         chosen.append(best_j)
-        # Splice out of the neighbor list: neighbors' contributions now extend
-        # over the removed point's span.
-        if left[best_j] >= 0:
-            right[left[best_j]] = right[best_j]
-        if right[best_j] < n:
-            left[right[best_j]] = left[best_j]
+        contribs[best_j] = -np.inf
+        lj, rj = left[best_j], right[best_j]
+        # Splice out: the neighbors' rectangles now extend over the removed
+        # point's span; only their two contributions change.
+        if lj >= 0:
+            right[lj] = rj
+            if contribs[lj] != -np.inf:
+                ry = reference_point[0] if rj >= n else x_of[rj]
+                ly = reference_point[1] if left[lj] < 0 else y_of[left[lj]]
+                contribs[lj] = (ry - x_of[lj]) * (ly - y_of[lj])
+        if rj < n:
+            left[rj] = lj
+            if contribs[rj] != -np.inf:
+                ry = reference_point[0] if right[rj] >= n else x_of[right[rj]]
+                ly = reference_point[1] if lj < 0 else y_of[lj]
+                contribs[rj] = (ry - x_of[rj]) * (ly - y_of[rj])
 
     return sorted_idx[np.asarray(chosen)]
 
