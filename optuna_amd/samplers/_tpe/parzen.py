@@ -496,20 +496,25 @@ class _ParzenEstimator:
             is_disc = num.steps > 0
             cont = self._cont_mask
             cont_cols = np.nonzero(cont)[0]
-            for jc, j in enumerate(cont_cols):
-                xj = xv[:, j][:, np.newaxis]  # (S, 1)
+            if len(cont_cols):
+                # All continuous dims at once: (S, K, Dc) of per-dim kernel
+                # log-densities, logsumexp over K per (sample, dim).
+                xc = xv[:, cont_cols]  # (S, Dc)
                 lt = (
-                    xj * xj * self._c1[:, jc][np.newaxis, :]
-                    + xj * self._c2[:, jc][np.newaxis, :]
-                    + self._c3[:, jc][np.newaxis, :]
-                    + logw
+                    xc[:, None, :] * xc[:, None, :] * self._c1[None, :, :]
+                    + xc[:, None, :] * self._c2[None, :, :]
+                    + self._c3[None, :, :]
+                    + logw[:, :, None]
                 )
-                vals = lse(lt)
-                outside = (xv[:, j] < num.adapted_lows[j]) | (
-                    xv[:, j] > num.adapted_highs[j]
+                m = lt.max(axis=1)  # (S, Dc)
+                m[np.isneginf(m)] = 0.0
+                with np.errstate(divide="ignore"):
+                    vals = np.log(np.exp(lt - m[:, None, :]).sum(axis=1)) + m
+                outside = (xc < num.adapted_lows[cont_cols]) | (
+                    xc > num.adapted_highs[cont_cols]
                 )
                 vals[outside] = -np.inf
-                out[:, num.dim_indices[j]] = vals
+                out[:, num.dim_indices[cont_cols]] = vals
             for j in np.nonzero(is_disc)[0]:
                 half = num.steps[j] / 2
                 xj_raw = x[:, num.dim_indices[j]]

@@ -237,13 +237,16 @@ class TPESampler(BaseSampler):
         param_distribution: BaseDistribution,
     ) -> Any:
         states = (TrialState.COMPLETE, TrialState.PRUNED)
-        trials = study._get_trials(deepcopy=False, states=states, use_cache=True)
-        if len(trials) < self._n_startup_trials:
+        # O(1) startup check: a full trial fetch here would run once per DIM
+        # per suggest (this is the per-dim entry point).
+        n_finished = study._storage.get_n_trials(study._study_id, states)
+        if n_finished < self._n_startup_trials:
             return self._random_sampler.sample_independent(
                 study, trial, param_name, param_distribution
             )
 
         if self._warn_independent_sampling and self._is_multivariate(study):
+            trials = study._get_trials(deepcopy=False, states=states, use_cache=True)
             if any(param_name in t.params for t in trials):
                 _logger.warning(
                     _INDEPENDENT_SAMPLING_WARNING_TEMPLATE.format(
