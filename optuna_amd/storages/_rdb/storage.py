@@ -726,7 +726,10 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
         self, session: sa_orm.Session, trial: models.TrialModel
     ) -> FrozenTrial:
         params = (
-            session.query(models.TrialParamModel).filter_by(trial_id=trial.trial_id).all()
+            session.query(models.TrialParamModel)
+            .filter_by(trial_id=trial.trial_id)
+            .order_by(models.TrialParamModel.param_id)  # insertion order
+            .all()
         )
         values = (
             session.query(models.TrialValueModel).filter_by(trial_id=trial.trial_id).all()

@@ -329,3 +329,301 @@ class StorageTestCase:
         trial = storage.get_trial(trial_id)
         assert isinstance(trial.datetime_start, datetime)
         assert isinstance(trial.datetime_complete, datetime)
+
+    # ---- attribute round-trips / isolation ------------------------------------------
+
+    _FLOAT_ATTRS = {
+        "zero": 0.0,
+        "neg": -1.5,
+        "nan": float("nan"),
+        "pinf": float("inf"),
+        "ninf": float("-inf"),
+    }
+
+    @staticmethod
+    def _same_float(a: float, b: float) -> bool:
+        return (math.isnan(a) and math.isnan(b)) or a == b
+
+    def test_study_user_attrs_float_round_trip(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        for k, v in self._FLOAT_ATTRS.items():
+            storage.set_study_user_attr(study_id, k, v)
+        got = storage.get_study_user_attrs(study_id)
+        for k, v in self._FLOAT_ATTRS.items():
+            assert self._same_float(got[k], v), k
+
+    def test_study_system_attrs_float_round_trip(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        for k, v in self._FLOAT_ATTRS.items():
+            storage.set_study_system_attr(study_id, k, v)
+        got = storage.get_study_system_attrs(study_id)
+        for k, v in self._FLOAT_ATTRS.items():
+            assert self._same_float(got[k], v), k
+
+    def test_trial_attrs_float_round_trip(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        for k, v in self._FLOAT_ATTRS.items():
+            storage.set_trial_user_attr(trial_id, "u" + k, v)
+            storage.set_trial_system_attr(trial_id, "s" + k, v)
+        t = storage.get_trial(trial_id)
+        for k, v in self._FLOAT_ATTRS.items():
+            assert self._same_float(t.user_attrs["u" + k], v)
+            assert self._same_float(t.system_attrs["s" + k], v)
+
+    def test_user_and_system_attrs_are_separate_namespaces(
+        self, storage: BaseStorage
+    ) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        storage.set_study_user_attr(study_id, "k", "user")
+        storage.set_study_system_attr(study_id, "k", "system")
+        assert storage.get_study_user_attrs(study_id)["k"] == "user"
+        assert storage.get_study_system_attrs(study_id)["k"] == "system"
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_user_attr(trial_id, "k", "tu")
+        storage.set_trial_system_attr(trial_id, "k", "ts")
+        t = storage.get_trial(trial_id)
+        assert t.user_attrs["k"] == "tu" and t.system_attrs["k"] == "ts"
+
+    def test_attr_overwrite_keeps_latest(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        for i in range(5):
+            storage.set_study_user_attr(study_id, "k", i)
+        assert storage.get_study_user_attrs(study_id)["k"] == 4
+        trial_id = storage.create_new_trial(study_id)
+        for i in range(5):
+            storage.set_trial_system_attr(trial_id, "k", i)
+        assert storage.get_trial(trial_id).system_attrs["k"] == 4
+
+    def test_attrs_support_json_values(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        payload = {"list": [1, 2, "x"], "nested": {"a": None, "b": True}}
+        storage.set_study_user_attr(study_id, "json", payload)
+        assert storage.get_study_user_attrs(study_id)["json"] == payload
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_user_attr(trial_id, "json", payload)
+        assert storage.get_trial(trial_id).user_attrs["json"] == payload
+
+    # ---- trial lifecycle details ----------------------------------------------------
+
+    def test_new_trial_fields_are_empty(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        t = storage.get_trial(storage.create_new_trial(study_id))
+        assert t.state == TrialState.RUNNING
+        assert t.params == {} and t.distributions == {}
+        assert t.user_attrs == {} and t.system_attrs == {}
+        assert t.intermediate_values == {}
+        assert t.values is None
+        assert t.datetime_start is not None and t.datetime_complete is None
+
+    def test_trial_numbers_are_per_study(self, storage: BaseStorage) -> None:
+        s1 = storage.create_new_study(MINIMIZE, study_name="pertrial-a")
+        s2 = storage.create_new_study(MINIMIZE, study_name="pertrial-b")
+        for expected in range(3):
+            t1 = storage.create_new_trial(s1)
+            t2 = storage.create_new_trial(s2)
+            assert storage.get_trial_number_from_id(t1) == expected
+            assert storage.get_trial_number_from_id(t2) == expected
+
+    def test_template_trial_all_fields(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        dist = FloatDistribution(-2.0, 2.0)
+        template = create_trial(
+            state=TrialState.COMPLETE,
+            value=0.25,
+            params={"x": 1.5},
+            distributions={"x": dist},
+            user_attrs={"u": 1},
+            system_attrs={"s": 2},
+            intermediate_values={0: 0.1, 3: 0.3},
+        )
+        t = storage.get_trial(storage.create_new_trial(study_id, template))
+        assert t.state == TrialState.COMPLETE
+        assert t.value == 0.25
+        assert t.params == {"x": 1.5}
+        assert t.distributions == {"x": dist}
+        assert t.user_attrs == {"u": 1}
+        assert t.system_attrs == {"s": 2}
+        assert t.intermediate_values == {0: 0.1, 3: 0.3}
+
+    def test_set_trial_param_distribution_compat_enforced(
+        self, storage: BaseStorage
+    ) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        t1 = storage.create_new_trial(study_id)
+        storage.set_trial_param(t1, "x", 0.5, FloatDistribution(0, 1))
+        t2 = storage.create_new_trial(study_id)
+        with pytest.raises(ValueError):
+            storage.set_trial_param(
+                t2, "x", 0.0, CategoricalDistribution(choices=(0.0, 1.0))
+            )
+
+    def test_get_trial_param_internal_repr(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        cat = CategoricalDistribution(choices=("a", "b", "c"))
+        storage.set_trial_param(trial_id, "c", 2, cat)
+        storage.set_trial_param(trial_id, "f", 0.125, FloatDistribution(0, 1))
+        assert storage.get_trial_param(trial_id, "c") == 2
+        assert storage.get_trial_param(trial_id, "f") == 0.125
+        t = storage.get_trial(trial_id)
+        assert t.params["c"] == "c"  # external repr
+
+    def test_params_preserve_insertion_order(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        names = ["z", "a", "m", "b"]
+        for i, n in enumerate(names):
+            storage.set_trial_param(trial_id, n, float(i), FloatDistribution(0, 10))
+        assert list(storage.get_trial(trial_id).params.keys()) == names
+
+    def test_values_length_any(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(
+            [StudyDirection.MINIMIZE, StudyDirection.MINIMIZE, StudyDirection.MINIMIZE]
+        )
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (1.0, 2.0, 3.0))
+        assert storage.get_trial(trial_id).values == [1.0, 2.0, 3.0]
+
+    def test_fail_and_prune_states(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        fail_id = storage.create_new_trial(study_id)
+        assert storage.set_trial_state_values(fail_id, TrialState.FAIL)
+        assert storage.get_trial(fail_id).state == TrialState.FAIL
+        prune_id = storage.create_new_trial(study_id)
+        assert storage.set_trial_state_values(prune_id, TrialState.PRUNED, (7.0,))
+        pruned = storage.get_trial(prune_id)
+        assert pruned.state == TrialState.PRUNED and pruned.value == 7.0
+
+    def test_intermediate_value_overwrite_same_step(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_intermediate_value(trial_id, 1, 10.0)
+        storage.set_trial_intermediate_value(trial_id, 1, 20.0)
+        assert storage.get_trial(trial_id).intermediate_values == {1: 20.0}
+
+    def test_updates_rejected_on_every_finished_state(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        for state in (TrialState.COMPLETE, TrialState.FAIL, TrialState.PRUNED):
+            trial_id = storage.create_new_trial(study_id)
+            storage.set_trial_state_values(
+                trial_id, state, (0.0,) if state != TrialState.FAIL else None
+            )
+            with pytest.raises(UpdateFinishedTrialError):
+                storage.set_trial_param(trial_id, "x", 0.0, FloatDistribution(0, 1))
+            with pytest.raises(UpdateFinishedTrialError):
+                storage.set_trial_user_attr(trial_id, "k", 1)
+            with pytest.raises(UpdateFinishedTrialError):
+                storage.set_trial_intermediate_value(trial_id, 0, 0.0)
+
+    def test_check_trial_is_updatable(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.check_trial_is_updatable(trial_id, TrialState.RUNNING)
+        storage.check_trial_is_updatable(trial_id, TrialState.WAITING)
+        with pytest.raises(UpdateFinishedTrialError):
+            storage.check_trial_is_updatable(trial_id, TrialState.COMPLETE)
+
+    # ---- reads: isolation, filters, summaries ---------------------------------------
+
+    def test_get_all_trials_returns_all_studies_separately(
+        self, storage: BaseStorage
+    ) -> None:
+        s1 = storage.create_new_study(MINIMIZE, study_name="sep-a")
+        s2 = storage.create_new_study(MINIMIZE, study_name="sep-b")
+        for _ in range(2):
+            storage.create_new_trial(s1)
+        storage.create_new_trial(s2)
+        assert len(storage.get_all_trials(s1)) == 2
+        assert len(storage.get_all_trials(s2)) == 1
+
+    def test_get_all_trials_mutation_does_not_leak(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_user_attr(trial_id, "k", [1, 2])
+        fetched = storage.get_all_trials(study_id, deepcopy=True)
+        fetched[0].user_attrs["k"].append(3)
+        assert storage.get_trial(trial_id).user_attrs["k"] == [1, 2]
+
+    def test_get_all_trials_empty_state_filter(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        storage.create_new_trial(study_id)
+        assert storage.get_all_trials(study_id, states=()) == []
+
+    def test_get_n_trials_state_combinations(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        per_state = {
+            TrialState.COMPLETE: 3,
+            TrialState.FAIL: 2,
+            TrialState.RUNNING: 1,
+        }
+        for state, count in per_state.items():
+            for _ in range(count):
+                trial_id = storage.create_new_trial(study_id)
+                if state != TrialState.RUNNING:
+                    storage.set_trial_state_values(
+                        trial_id,
+                        state,
+                        (0.0,) if state == TrialState.COMPLETE else None,
+                    )
+        assert storage.get_n_trials(study_id) == 6
+        assert storage.get_n_trials(study_id, TrialState.COMPLETE) == 3
+        assert storage.get_n_trials(study_id, (TrialState.COMPLETE, TrialState.FAIL)) == 5
+
+    def test_best_trial_ignores_running_and_failed(self, storage: BaseStorage) -> None:
+        # NaN objective values are rejected at the tell() layer (reference
+        # behavior), so the storage contract only covers finite/inf values.
+        study_id = storage.create_new_study(MINIMIZE)
+        storage.create_new_trial(study_id)  # RUNNING
+        failed = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(failed, TrialState.FAIL)
+        t2 = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(t2, TrialState.COMPLETE, (5.0,))
+        assert storage.get_best_trial(study_id).value == 5.0
+
+    def test_best_trial_multi_objective_raises(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(
+            [StudyDirection.MINIMIZE, StudyDirection.MINIMIZE]
+        )
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (0.0, 1.0))
+        with pytest.raises(RuntimeError):
+            storage.get_best_trial(study_id)
+
+    def test_get_all_studies_reflects_attrs(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE, study_name="summary-attrs")
+        storage.set_study_user_attr(study_id, "u", 1)
+        frozen = {fs._study_id: fs for fs in storage.get_all_studies()}[study_id]
+        assert frozen.study_name == "summary-attrs"
+        assert frozen.user_attrs.get("u") == 1
+        assert frozen.directions == [StudyDirection.MINIMIZE]
+
+    def test_concurrent_study_attr_writes(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+
+        def writer(offset: int) -> None:
+            for i in range(10):
+                storage.set_study_user_attr(study_id, f"k{offset}-{i}", i)
+
+        threads = [threading.Thread(target=writer, args=(j,)) for j in range(4)]
+        for th in threads:
+            th.start()
+        for th in threads:
+            th.join()
+        attrs = storage.get_study_user_attrs(study_id)
+        assert len([k for k in attrs if k.startswith("k")]) == 40
+
+    def test_concurrent_trial_system_attr_writes(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(study_id)
+
+        def writer(offset: int) -> None:
+            for i in range(10):
+                storage.set_trial_system_attr(trial_id, f"k{offset}-{i}", i)
+
+        threads = [threading.Thread(target=writer, args=(j,)) for j in range(4)]
+        for th in threads:
+            th.start()
+        for th in threads:
+            th.join()
+        assert len(storage.get_trial(trial_id).system_attrs) == 40
