@@ -231,13 +231,27 @@ class PedAnovaImportanceEvaluator(BaseImportanceEvaluator):
         *,
         target: Callable[[FrozenTrial], float] | None = None,
     ) -> dict[str, float]:
+        completed = study.get_trials(deepcopy=False, states=(TrialState.COMPLETE,))
+        if len(completed) == 0:
+            raise ValueError(
+                "Cannot evaluate parameter importances without completed trials."
+            )
         if params is None:
             # All parameters appearing in completed trials, incl. conditional ones.
             seen: dict[str, None] = {}
-            for t in study.get_trials(deepcopy=False, states=(TrialState.COMPLETE,)):
+            for t in completed:
                 for name in t.params:
                     seen.setdefault(name)
             params = list(seen)
+        else:
+            # PedAnova accepts conditional params, but each requested one must
+            # appear in at least one completed trial.
+            for p_ in params:
+                if not any(p_ in t.params for t in completed):
+                    raise ValueError(
+                        "Study must contain completed trials with all specified "
+                        f"parameters. Missing: {p_!r}."
+                    )
         trials = _get_filtered_trials(study, target)
         if len(trials) <= 1:
             warnings.warn(

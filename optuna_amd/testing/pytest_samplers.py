@@ -134,3 +134,160 @@ class MultiObjectiveSamplerTestCase:
             study.optimize(objective, n_trials=self.n_trials)
         assert len(study.trials) == self.n_trials
         assert len(study.best_trials) >= 1
+
+
+class ExtendedSamplerTestCase:
+    """Behavioral cases beyond the basics: NaN objectives, single/dynamic
+    spaces, mixed-distribution objectives, partial fixing, reproducibility in
+    another process. Mix into a sampler's test class alongside
+    BasicSamplerTestCase."""
+
+    n_trials: int = 10
+
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        raise NotImplementedError
+
+    def test_nan_objective_then_recovery(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        # A NaN objective marks the trial FAIL; the sampler must keep working.
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=sampler_factory(7))
+
+            def objective(trial: optuna_amd.Trial) -> float:
+                x = trial.suggest_float("x", 0, 1)
+                if trial.number % 3 == 1:
+                    return float("nan")
+                return x
+
+            study.optimize(objective, n_trials=self.n_trials, catch=())
+        states = [t.state for t in study.trials]
+        assert optuna_amd.trial.TrialState.FAIL in states
+        assert optuna_amd.trial.TrialState.COMPLETE in states
+        assert study.best_trial.value is not None
+
+    def test_single_value_distributions(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        # low == high collapses to the single value for every dist kind.
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=sampler_factory(5))
+
+            def objective(trial: optuna_amd.Trial) -> float:
+                f = trial.suggest_float("f", 3.25, 3.25)
+                i = trial.suggest_int("i", 7, 7)
+                c = trial.suggest_categorical("c", ("only",))
+                assert f == 3.25 and i == 7 and c == "only"
+                return trial.suggest_float("x", 0, 1)
+
+            study.optimize(objective, n_trials=self.n_trials)
+        assert all(t.params["i"] == 7 for t in study.trials)
+
+    def test_single_parameter_objective(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=sampler_factory(2))
+            study.optimize(lambda t: t.suggest_float("x", -1, 1) ** 2, n_trials=self.n_trials)
+        assert len(study.trials) == self.n_trials
+        assert 0 <= study.best_value <= 1
+
+    def test_mixed_distribution_objective(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=sampler_factory(9))
+
+            def objective(trial: optuna_amd.Trial) -> float:
+                total = trial.suggest_float("f", -1, 1)
+                total += trial.suggest_float("flog", 1e-3, 10, log=True) * 0
+                total += trial.suggest_float("fstep", 0, 1, step=0.25)
+                total += trial.suggest_int("i", 0, 16)
+                total += trial.suggest_int("ilog", 1, 64, log=True) * 0
+                total += {"a": 0.0, "b": 1.0}[trial.suggest_categorical("c", ("a", "b"))]
+                return total
+
+            study.optimize(objective, n_trials=self.n_trials)
+        for t in study.trials:
+            assert t.params["fstep"] in [0.0, 0.25, 0.5, 0.75, 1.0]
+            assert isinstance(t.params["i"], int)
+
+    def test_dynamic_value_range(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        # The range of "x" shifts between trials — samplers must tolerate it
+        # (the storage records a widened/compatible distribution).
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=sampler_factory(11))
+
+            def objective(trial: optuna_amd.Trial) -> float:
+                width = 1.0 + (trial.number % 3)
+                return trial.suggest_float("x", -width, width) ** 2
+
+            study.optimize(objective, n_trials=self.n_trials)
+        assert len(study.trials) == self.n_trials
+
+    def test_partial_fixed_wrapper(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        from optuna_amd.samplers import PartialFixedSampler
+
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            base = sampler_factory(3)
+            study = optuna_amd.create_study(
+                sampler=PartialFixedSampler({"y": 0.5}, base)
+            )
+
+            def objective(trial: optuna_amd.Trial) -> float:
+                x = trial.suggest_float("x", -1, 1)
+                y = trial.suggest_float("y", -1, 1)
+                return x * x + y
+
+            study.optimize(objective, n_trials=self.n_trials)
+        assert all(t.params["y"] == 0.5 for t in study.trials)
+
+    def test_reproducible_across_processes(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        # Same seed in a spawned subprocess must yield the identical sequence
+        # (no hidden process-local entropy).
+        import multiprocessing
+
+        factory = sampler_factory
+        ctx = multiprocessing.get_context("spawn")
+        queue = ctx.Queue()
+        proc = ctx.Process(
+            target=_child_param_sequence, args=(type(self), queue)
+        )
+        proc.start()
+        child = queue.get(timeout=120)
+        proc.join(timeout=60)
+        here = _run_param_sequence(factory, self.n_trials)
+        assert child == here
+
+
+def _run_param_sequence(
+    factory: Callable[[int], BaseSampler], n_trials: int
+) -> list[dict[str, Any]]:
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        study = optuna_amd.create_study(sampler=factory(1234))
+        study.optimize(
+            lambda t: t.suggest_float("x", -1, 1) + t.suggest_int("i", 0, 9),
+            n_trials=n_trials,
+        )
+    return [t.params for t in study.trials]
+
+
+def _child_param_sequence(case_cls: type, queue: Any) -> None:
+    case = case_cls()
+    # Re-derive the factory from the fixture definition in the subclass.
+    factory = case.__class__.sampler_factory.__wrapped__(case)  # type: ignore[attr-defined]
+    queue.put(_run_param_sequence(factory, case_cls.n_trials))

@@ -9,23 +9,24 @@ import optuna_amd
 from optuna_amd.samplers import BaseSampler
 from optuna_amd.testing.pytest_samplers import (
     BasicSamplerTestCase,
+    ExtendedSamplerTestCase,
     MultiObjectiveSamplerTestCase,
 )
 
 
-class TestRandomSampler(BasicSamplerTestCase, MultiObjectiveSamplerTestCase):
+class TestRandomSampler(BasicSamplerTestCase, ExtendedSamplerTestCase, MultiObjectiveSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.RandomSampler(seed=seed)
 
 
-class TestTPESampler(BasicSamplerTestCase, MultiObjectiveSamplerTestCase):
+class TestTPESampler(BasicSamplerTestCase, ExtendedSamplerTestCase, MultiObjectiveSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.TPESampler(seed=seed, n_startup_trials=3)
 
 
-class TestTPEMultivariate(BasicSamplerTestCase):
+class TestTPEMultivariate(BasicSamplerTestCase, ExtendedSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.TPESampler(
@@ -33,7 +34,7 @@ class TestTPEMultivariate(BasicSamplerTestCase):
         )
 
 
-class TestCmaEsSampler(BasicSamplerTestCase):
+class TestCmaEsSampler(BasicSamplerTestCase, ExtendedSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.CmaEsSampler(
@@ -87,7 +88,7 @@ class TestGPSampler(BasicSamplerTestCase):
         )
 
 
-class TestNSGAIISampler(BasicSamplerTestCase, MultiObjectiveSamplerTestCase):
+class TestNSGAIISampler(BasicSamplerTestCase, ExtendedSamplerTestCase, MultiObjectiveSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.NSGAIISampler(seed=seed, population_size=4)
@@ -99,7 +100,7 @@ class TestNSGAIIISampler(MultiObjectiveSamplerTestCase):
         return lambda seed: optuna_amd.samplers.NSGAIIISampler(seed=seed, population_size=4)
 
 
-class TestQMCSampler(BasicSamplerTestCase):
+class TestQMCSampler(BasicSamplerTestCase, ExtendedSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.QMCSampler(
