@@ -867,10 +867,19 @@ class _VersionManager:
                 )
 
     def upgrade(self) -> None:
-        with _create_scoped_session(self.scoped_session) as session:
+        """Migrate an older-schema database in place to SCHEMA_VERSION 12.
+
+        The alembic chain the reference ships (v0.9.0.a … v3.2.0.a) is
+        re-implemented as shape-driven steps in ``upgrade.py`` — idempotent,
+        and correct from any entry version.
+        """
+        from optuna_amd.storages._rdb.upgrade import upgrade_to_v12
+
+        applied = upgrade_to_v12(self.engine)
+        if applied:
+            _logger.info(f"Applied schema migrations: {', '.join(applied)}")
+        with _create_scoped_session(self.scoped_session, True) as session:
             vi = session.query(models.VersionInfoModel).one_or_none()
-            if vi is not None and vi.schema_version != models.SCHEMA_VERSION:
-                raise NotImplementedError(
-                    "Automatic upgrade from pre-v12 schemas is not supported by this "
-                    "build; export the study from the original library first."
-                )
+            if vi is not None:
+                vi.schema_version = models.SCHEMA_VERSION
+                vi.library_version = __version__
