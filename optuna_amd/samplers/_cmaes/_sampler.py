@@ -268,6 +268,7 @@ class CmaEsSampler(BaseSampler):
                 seed=self._cma_rng.rng.randint(1, 2**31 - 2),
                 n_max_resampling=10 * n_dimension,
                 population_size=self._popsize,
+                lr_adapt=self._lr_adapt,
             )
 
         if self._x0 is None:

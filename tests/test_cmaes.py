@@ -285,3 +285,36 @@ def test_experimental_argument_warnings(kwargs) -> None:
 
     with pytest.warns(ExperimentalWarning):
         optuna_amd.samplers.CmaEsSampler(**kwargs)
+
+
+def test_lr_adapt_actually_adapts() -> None:
+    """LRA-CMA: learning rates must move off 1.0 on a noisy objective and the
+    sampler round-trips the adapted state through system attrs."""
+    import numpy as np
+
+    import optuna_amd
+    from optuna_amd.samplers._cmaes._core import CMA
+
+    rng = np.random.RandomState(3)
+    opt = CMA(mean=np.full(6, 2.0), sigma=1.5, seed=5, lr_adapt=True)
+    for _ in range(80):
+        sols = []
+        for _ in range(opt.population_size):
+            x = opt.ask()
+            f = float(np.sum(x * x) + 2.0 * rng.randn())
+            sols.append((x, f))
+        opt.tell(sols)
+    assert opt._eta_m != 1.0 or opt._eta_c != 1.0
+
+    # End-to-end through the sampler (state pickled into system attrs).
+    import warnings
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        sampler = optuna_amd.samplers.CmaEsSampler(seed=0, lr_adapt=True, n_startup_trials=1)
+        study = optuna_amd.create_study(sampler=sampler)
+        study.optimize(
+            lambda t: sum(t.suggest_float(f"x{i}", -3, 3) ** 2 for i in range(4)),
+            n_trials=25,
+        )
+    assert len(study.trials) == 25
