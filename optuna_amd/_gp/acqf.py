@@ -405,8 +405,8 @@ class LogEHVI(BaseAcquisitionFunc):
                 for g in self._gpr_list
             ]
         self._fixed_samples = sample_from_normal_sobol(
-            dim=Y_train.shape[-1], n_samples=n_qmc_samples, seed=qmc_seed
-        ).to(dev)
+            dim=Y_train.shape[-1], n_samples=n_qmc_samples, seed=qmc_seed, device=dev
+        )
         ref_point = _get_reference_point(Y_train)
         box_lower, box_upper = _get_boxes(Y_train, ref_point)
         self._box_lower = box_lower.to(dev)

@@ -658,8 +658,11 @@ class ConditionalGPRegressor:
         X_running = X_running.to(gpr.device)
         self._X_running = X_running
         fixed_samples = sample_from_normal_sobol(
-            dim=X_running.shape[0] + 1, n_samples=n_qmc_samples, seed=qmc_seed
-        ).to(gpr.device)
+            dim=X_running.shape[0] + 1,
+            n_samples=n_qmc_samples,
+            seed=qmc_seed,
+            device=gpr.device,
+        )
         self._fixed_samples_x = fixed_samples[..., -1]
         self._stabilizing_noise = stabilizing_noise
         with torch.no_grad():

@@ -279,3 +279,32 @@ def test_cloned_with_running_leaves_cache_pristine() -> None:
     assert clone._X_all.shape[0] == 54
     assert gpr._X_all.shape[0] == 50
     torch.testing.assert_close(gpr._cov_Y_Y_chol, chol_before)
+
+
+def test_sobol_uniform_matches_scipy_unscrambled_structure() -> None:
+    """Our torch Sobol reproduces scipy's unscrambled sequence when the
+    digital shift is removed (same Joe-Kuo direction numbers, Gray order)."""
+    from scipy.stats import qmc as scipy_qmc
+
+    from optuna_amd._gp import qmc
+
+    dim, n = 5, 64
+    u = qmc.sobol_uniform(dim, n, seed=0)
+    shift = np.random.RandomState(0).randint(0, 1 << qmc._BITS, size=dim).astype(np.int64)
+    x_int = (u.numpy() * (1 << qmc._BITS) - 0.5).round().astype(np.int64)
+    unscrambled = (x_int ^ shift).astype(np.float64) / (1 << qmc._BITS)
+    ref = scipy_qmc.Sobol(dim, scramble=False, bits=qmc._BITS).random(n)
+    np.testing.assert_allclose(unscrambled, ref, atol=1e-9)
+
+
+def test_sobol_normal_moments_and_determinism() -> None:
+    from optuna_amd._gp import qmc
+
+    a = qmc.sample_from_normal_sobol(8, 1024, seed=3)
+    b = qmc.sample_from_normal_sobol(8, 1024, seed=3)
+    torch.testing.assert_close(a, b)
+    c = qmc.sample_from_normal_sobol(8, 1024, seed=4)
+    assert not torch.equal(a, c)
+    assert abs(a.mean().item()) < 0.02
+    assert abs(a.std().item() - 1.0) < 0.02
+    assert torch.isfinite(a).all()
