@@ -209,7 +209,7 @@ def test_large_history_skips_dense_sqdiff_and_still_fits() -> None:
         deterministic_objective=False,
     )
     assert gpr._squared_X_diff is None
-    mean, var = gpr.posterior(torch.from_numpy(X[:5]))
+    mean, var = gpr.posterior(torch.from_numpy(X[:5]).to(gpr.device))
     assert torch.isfinite(mean).all() and (var >= 0).all()
 
 
@@ -233,16 +233,17 @@ def test_update_data_matches_fresh_cache() -> None:
     assert gpr.update_data(X_full, y_full)
     assert gpr._X_train.shape[0] == N0 + 7
 
+    dev = gpr.device  # cuda on a GPU box, cpu otherwise
     ref = gp.GPRegressor(
-        is_categorical=torch.zeros(D, dtype=torch.bool),
-        X_train=torch.from_numpy(X_full),
-        y_train=torch.from_numpy(y_full),
+        is_categorical=torch.zeros(D, dtype=torch.bool).to(dev),
+        X_train=torch.from_numpy(X_full).to(dev),
+        y_train=torch.from_numpy(y_full).to(dev),
         inverse_squared_lengthscales=gpr.inverse_squared_lengthscales.clone(),
         kernel_scale=gpr.kernel_scale.clone(),
         noise_var=gpr.noise_var.clone(),
     )
     ref._cache_matrix()
-    x_eval = torch.from_numpy(rng.rand(11, D))
+    x_eval = torch.from_numpy(rng.rand(11, D)).to(dev)
     mean_u, var_u = gpr.posterior(x_eval)
     mean_r, var_r = ref.posterior(x_eval)
     torch.testing.assert_close(mean_u, mean_r, rtol=1e-7, atol=1e-9)
