@@ -1399,6 +1399,216 @@ class TpeDeviceHistory {
     bool sorted_valid_ = false;
 };
 
+
+// ---------------------------------------------------------------------------
+// K6a: exact 3-D hypervolume (minimization, ref-dominated region).
+//
+// Points sorted ascending by x. HV = sum_t slab_x(t) * A_t where A_t is the
+// area in (y, z) dominated by the first t+1 points. A_t is evaluated by a
+// y-ascending sweep that maintains the running min z over the points of the
+// prefix — one THREAD per prefix t, with the y-sorted stream staged through
+// LDS tiles so the O(N^2) scan reads each point once per 256-thread block.
+// ---------------------------------------------------------------------------
+__global__ void k_hv3d(const double* __restrict__ xs,   // (N) ascending
+                       const double* __restrict__ yv,   // (N) y ascending
+                       const double* __restrict__ zv,   // (N) z in y-order
+                       const int32_t* __restrict__ rk,  // (N) x-rank in y-order
+                       int64_t N, double ref_x, double ref_y, double ref_z,
+                       double* __restrict__ out) {  // (N) per-prefix volumes
+    __shared__ double t_y[256];
+    __shared__ double t_z[256];
+    __shared__ int32_t t_r[256];
+    const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const double slab =
+        (t < N) ? ((t + 1 < N ? xs[t + 1] : ref_x) - xs[t]) : 0.0;
+    double minz = ref_z;
+    double area = 0.0;
+    for (int64_t base = 0; base < N; base += 256) {
+        const int64_t m = min((int64_t)256, N - base);
+        __syncthreads();
+        if (threadIdx.x < m) {
+            t_y[threadIdx.x] = yv[base + threadIdx.x];
+            t_z[threadIdx.x] = zv[base + threadIdx.x];
+            t_r[threadIdx.x] = rk[base + threadIdx.x];
+        }
+        // the y-interval right edge needs the NEXT point's y
+        __syncthreads();
+        for (int64_t q = 0; q < m; ++q) {
+            if (t_r[q] <= t) minz = fmin(minz, t_z[q]);
+            const int64_t g = base + q;
+            const double y_next = (g + 1 < N) ? (q + 1 < m ? t_y[q + 1] : yv[g + 1]) : ref_y;
+            area += (y_next - t_y[q]) * (ref_z - minz);
+        }
+    }
+    if (t < N) out[t] = slab * area;
+}
+
+// 3-D hypervolume of x-lexsorted points (all strictly inside ref).
+double hv3d(const arr_f64& pts, double ref_x, double ref_y, double ref_z) {
+    if (pts.ndim() != 2 || pts.shape(1) != 3)
+        throw std::runtime_error("hv3d: pts must be (N, 3)");
+    const int64_t N = pts.shape(0);
+    if (N == 0) return 0.0;
+    // Host prep: y-order view (N log N — negligible next to the N^2 sweep).
+    std::vector<double> xs(N), yv(N), zv(N);
+    std::vector<int32_t> rk(N);
+    std::vector<int64_t> order(N);
+    const double* P = pts.data();
+    for (int64_t i = 0; i < N; ++i) {
+        xs[i] = P[i * 3];
+        order[i] = i;
+    }
+    std::sort(order.begin(), order.end(), [&](int64_t a, int64_t b) {
+        return P[a * 3 + 1] < P[b * 3 + 1];
+    });
+    for (int64_t i = 0; i < N; ++i) {
+        yv[i] = P[order[i] * 3 + 1];
+        zv[i] = P[order[i] * 3 + 2];
+        rk[i] = (int32_t)order[i];
+    }
+    hipStream_t st = g_ws.get_stream();
+    const size_t need = 4 * (size_t)N + (size_t)(N + 1) / 2 + 8;
+    double* base = g_ws.ensure(need);
+    double* d_xs = base;
+    double* d_yv = d_xs + N;
+    double* d_zv = d_yv + N;
+    double* d_out = d_zv + N;
+    int32_t* d_rk = reinterpret_cast<int32_t*>(d_out + N);
+    g_ws.begin_uploads();
+    g_ws.h2d(d_xs, xs.data(), N * 8, st);
+    g_ws.h2d(d_yv, yv.data(), N * 8, st);
+    g_ws.h2d(d_zv, zv.data(), N * 8, st);
+    g_ws.h2d(d_rk, rk.data(), N * 4, st);
+    const int64_t blocks = (N + 255) / 256;
+    hipLaunchKernelGGL(k_hv3d, dim3((unsigned)blocks), dim3(256), 0, st, d_xs,
+                       d_yv, d_zv, d_rk, N, ref_x, ref_y, ref_z, d_out);
+    std::vector<double> host_out(N);
+    HIP_CHECK(hipMemcpyAsync(host_out.data(), d_out, N * 8,
+                             hipMemcpyDeviceToHost, st));
+    HIP_CHECK(hipStreamSynchronize(st));
+    HIP_CHECK(hipGetLastError());
+    double hv = 0.0;
+    for (double v : host_out) hv += v;
+    return hv;
+}
+
+// ---------------------------------------------------------------------------
+// K6b: batched greedy-HSSP contributions, 3 objectives. One workgroup per
+// candidate: the limited set max(c, s_j) over the selected points lives in
+// LDS, gets x- and y-sorted by odd-even transposition, and 64 lanes split the
+// per-prefix staircase sweep. contribution(c | S) = incl(c) - HV3D(limited).
+// ---------------------------------------------------------------------------
+#define HSSP_MAX_SEL 256
+
+__global__ void k_hssp3d_contrib(const double* __restrict__ cand,  // (n, 3)
+                                 const double* __restrict__ sel,   // (k, 3)
+                                 int64_t n, int64_t k, double ref_x,
+                                 double ref_y, double ref_z,
+                                 double* __restrict__ out) {  // (n)
+    __shared__ double lx[HSSP_MAX_SEL];
+    __shared__ double ly[HSSP_MAX_SEL];
+    __shared__ double lz[HSSP_MAX_SEL];
+    __shared__ double sy[HSSP_MAX_SEL];
+    __shared__ double sz[HSSP_MAX_SEL];
+    __shared__ int32_t sr[HSSP_MAX_SEL];
+    __shared__ double red[64];
+
+    const int64_t c = blockIdx.x;
+    const double cx = cand[c * 3], cy = cand[c * 3 + 1], cz = cand[c * 3 + 2];
+    const double incl = (ref_x - cx) * (ref_y - cy) * (ref_z - cz);
+    if (k == 0) {
+        if (threadIdx.x == 0) out[c] = incl;
+        return;
+    }
+    for (int64_t j = threadIdx.x; j < k; j += blockDim.x) {
+        lx[j] = fmax(cx, sel[j * 3]);
+        ly[j] = fmax(cy, sel[j * 3 + 1]);
+        lz[j] = fmax(cz, sel[j * 3 + 2]);
+    }
+    __syncthreads();
+    // Odd-even transposition sort of (lx, ly, lz) by lx — k passes.
+    for (int64_t pass = 0; pass < k; ++pass) {
+        const int64_t start = pass & 1;
+        for (int64_t j = start + 2 * (int64_t)threadIdx.x; j + 1 < k;
+             j += 2 * blockDim.x) {
+            if (lx[j] > lx[j + 1]) {
+                double a = lx[j]; lx[j] = lx[j + 1]; lx[j + 1] = a;
+                a = ly[j]; ly[j] = ly[j + 1]; ly[j + 1] = a;
+                a = lz[j]; lz[j] = lz[j + 1]; lz[j + 1] = a;
+            }
+        }
+        __syncthreads();
+    }
+    // y-order copy with x-rank payload.
+    for (int64_t j = threadIdx.x; j < k; j += blockDim.x) {
+        sy[j] = ly[j];
+        sz[j] = lz[j];
+        sr[j] = (int32_t)j;
+    }
+    __syncthreads();
+    for (int64_t pass = 0; pass < k; ++pass) {
+        const int64_t start = pass & 1;
+        for (int64_t j = start + 2 * (int64_t)threadIdx.x; j + 1 < k;
+             j += 2 * blockDim.x) {
+            if (sy[j] > sy[j + 1]) {
+                double a = sy[j]; sy[j] = sy[j + 1]; sy[j + 1] = a;
+                a = sz[j]; sz[j] = sz[j + 1]; sz[j + 1] = a;
+                int32_t r = sr[j]; sr[j] = sr[j + 1]; sr[j + 1] = r;
+            }
+        }
+        __syncthreads();
+    }
+    // Per-prefix staircase sweep, lanes strided over prefixes.
+    double total = 0.0;
+    for (int64_t t = threadIdx.x; t < k; t += blockDim.x) {
+        const double slab = (t + 1 < k ? lx[t + 1] : ref_x) - lx[t];
+        if (slab <= 0.0) continue;
+        double minz = ref_z;
+        double area = 0.0;
+        for (int64_t q = 0; q < k; ++q) {
+            if (sr[q] <= t) minz = fmin(minz, sz[q]);
+            const double y_next = (q + 1 < k) ? sy[q + 1] : ref_y;
+            area += (y_next - sy[q]) * (ref_z - minz);
+        }
+        total += slab * area;
+    }
+    red[threadIdx.x] = total;
+    __syncthreads();
+    for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if ((int)threadIdx.x < stride) red[threadIdx.x] += red[threadIdx.x + stride];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) out[c] = incl - red[0];
+}
+
+// Per-candidate greedy-HSSP contributions against the current selected set.
+py::array_t<double> hssp3d_contrib(const arr_f64& cand, const arr_f64& sel,
+                                   double ref_x, double ref_y, double ref_z) {
+    if (cand.ndim() != 2 || cand.shape(1) != 3)
+        throw std::runtime_error("hssp3d_contrib: cand must be (n, 3)");
+    const int64_t n = cand.shape(0);
+    const int64_t k = sel.ndim() == 2 ? sel.shape(0) : 0;
+    if (k > HSSP_MAX_SEL)
+        throw std::runtime_error("hssp3d_contrib: selected set too large");
+    py::array_t<double> out(n);
+    if (n == 0) return out;
+    hipStream_t st = g_ws.get_stream();
+    double* base = g_ws.ensure((size_t)(n + k) * 3 + n + 8);
+    double* d_cand = base;
+    double* d_sel = d_cand + (size_t)n * 3;
+    double* d_out = d_sel + (size_t)k * 3;
+    g_ws.begin_uploads();
+    g_ws.h2d(d_cand, cand.data(), (size_t)n * 3 * 8, st);
+    if (k > 0) g_ws.h2d(d_sel, sel.data(), (size_t)k * 3 * 8, st);
+    hipLaunchKernelGGL(k_hssp3d_contrib, dim3((unsigned)n), dim3(64), 0, st,
+                       d_cand, d_sel, n, k, ref_x, ref_y, ref_z, d_out);
+    HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, n * 8,
+                             hipMemcpyDeviceToHost, st));
+    HIP_CHECK(hipStreamSynchronize(st));
+    HIP_CHECK(hipGetLastError());
+    return out;
+}
+
 PYBIND11_MODULE(_hipcore, m) {
     m.doc() = "optuna_amd MI355X (gfx950) HIP kernels: TPE parzen fit + mixture "
               "log-pdf, truncnorm device library";
@@ -1409,6 +1619,10 @@ PYBIND11_MODULE(_hipcore, m) {
     m.def("truncnorm_logpdf", &truncnorm_logpdf);
     m.def("nondomination_rank", &nondomination_rank, py::arg("vals"),
           py::arg("n_below"));
+    m.def("hv3d", &hv3d, py::arg("pts"), py::arg("ref_x"), py::arg("ref_y"),
+          py::arg("ref_z"));
+    m.def("hssp3d_contrib", &hssp3d_contrib, py::arg("cand"), py::arg("sel"),
+          py::arg("ref_x"), py::arg("ref_y"), py::arg("ref_z"));
     m.def("kde_logpdf", &kde_logpdf, py::arg("obs"), py::arg("sorted_pos"),
           py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("steps"),
           py::arg("n_choices"), py::arg("prior_weight"),

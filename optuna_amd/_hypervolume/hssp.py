@@ -1,5 +1,10 @@
 """Greedy hypervolume subset selection (HSSP) with the (1−1/e) guarantee.
 
+3-objective fronts above a size threshold run the K6b HIP path: every greedy
+round evaluates ALL candidates' exclusive contributions in one kernel launch
+(one workgroup per candidate, selected set staged in LDS), replacing the
+host's lazy-heap sequence of WFG evaluations.
+
 2-D uses an exact O(k·n) incremental rectangle update; N-D uses greedy selection
 with lazily-updated submodular contributions (a candidate's cached contribution
 only shrinks as the selected set grows, so the max-heap order can be trusted
@@ -88,6 +93,12 @@ def _solve_hssp(
         return rank_i_indices[order[:subset_size]]
     if rank_i_loss_vals.shape[1] == 2:
         return _solve_hssp_2d(rank_i_loss_vals, rank_i_indices, subset_size, reference_point)
+    if rank_i_loss_vals.shape[1] == 3 and len(rank_i_loss_vals) >= _DEVICE_HSSP_MIN_ROWS:
+        device_choice = _solve_hssp_3d_device(
+            rank_i_loss_vals, rank_i_indices, subset_size, reference_point
+        )
+        if device_choice is not None:
+            return device_choice
 
     n = len(rank_i_loss_vals)
     selected_mask = np.zeros(n, dtype=bool)
@@ -119,3 +130,36 @@ def _solve_hssp(
             heapq.heappush(heap, (-c, stamp, j))
 
     return rank_i_indices[np.asarray(chosen)]
+
+
+_DEVICE_HSSP_MIN_ROWS = 512
+
+
+def _solve_hssp_3d_device(
+    vals: np.ndarray,
+    indices: np.ndarray,
+    subset_size: int,
+    reference_point: np.ndarray,
+) -> np.ndarray | None:
+    """Exact (non-lazy) greedy via the K6b contributions kernel, or None."""
+    from optuna_amd import _hip
+
+    core = _hip.get()
+    if core is None or not core.available() or subset_size > 256:
+        return None
+    cand = np.ascontiguousarray(vals, dtype=np.float64)
+    rx, ry, rz = (float(v) for v in reference_point)
+    taken = np.zeros(len(vals), dtype=bool)
+    chosen: list[int] = []
+    for _ in range(subset_size):
+        sel = (
+            np.ascontiguousarray(vals[chosen], dtype=np.float64)
+            if chosen
+            else np.empty((0, 3), dtype=np.float64)
+        )
+        contribs = np.asarray(core.hssp3d_contrib(cand, sel, rx, ry, rz))
+        contribs[taken] = -np.inf
+        j = int(np.argmax(contribs))
+        taken[j] = True
+        chosen.append(j)
+    return indices[np.asarray(chosen)]

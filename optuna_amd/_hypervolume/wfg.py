@@ -129,5 +129,31 @@ def compute_hypervolume(
     if reference_point.shape[0] == 2:
         return _compute_2d(sorted_pareto_sols, reference_point)
     if reference_point.shape[0] == 3:
+        if len(sorted_pareto_sols) >= _DEVICE_HV3D_MIN_ROWS:
+            dev = _hv3d_device(sorted_pareto_sols, reference_point)
+            if dev is not None:
+                return dev
         return _compute_3d(sorted_pareto_sols, reference_point)
     return _compute_hv(sorted_pareto_sols, reference_point)
+
+
+# The K6a kernel's O(N^2) prefix sweep beats the host Python loop from a few
+# hundred points on; launch latency dominates below this.
+_DEVICE_HV3D_MIN_ROWS = 512
+
+
+def _hv3d_device(sorted_pts: np.ndarray, reference_point: np.ndarray) -> float | None:
+    """K6a HIP path: per-prefix staircase sweep on the MI355X (or None)."""
+    from optuna_amd import _hip
+
+    core = _hip.get()
+    if core is None or not core.available():
+        return None
+    return float(
+        core.hv3d(
+            np.ascontiguousarray(sorted_pts, dtype=np.float64),
+            float(reference_point[0]),
+            float(reference_point[1]),
+            float(reference_point[2]),
+        )
+    )
