@@ -37,7 +37,7 @@ def main() -> None:
     parser.add_argument("--dims", type=int, default=20)
     parser.add_argument(
         "--suite",
-        choices=("tpe", "random", "gp", "cmaes", "nsgaii", "motpe"),
+        choices=("tpe", "random", "gp", "cmaes", "nsgaii", "motpe", "motpe3"),
         default="tpe",
         help="BASELINE.json config to run (default: the headline TPE config)",
     )
@@ -241,6 +241,48 @@ def _run_alt_suite(args: argparse.Namespace) -> None:
             study.tell(t, float(np.sum(x**2)))
 
         config = {"model": "CmaEsSampler", "dims": dims, "parallelism": "single"}
+    elif args.suite == "motpe3":
+        # 3-objective MO-TPE: the non-domination split runs the K6 rank kernel
+        # and the boundary-rank tie-break runs the K6b greedy-HSSP kernel.
+        n_hist = max(args.history, 6000)
+        sampler = optuna_amd.samplers.TPESampler(seed=0, n_startup_trials=10)
+        study = optuna_amd.create_study(
+            directions=["minimize"] * 3, sampler=sampler
+        )
+        names = [f"x{i}" for i in range(10)]
+        dists = {n: optuna_amd.distributions.FloatDistribution(0.0, 1.0) for n in names}
+        pm = rng.uniform(0, 1, size=(n_hist, 10))
+        import math as _math
+
+        def _dtlz2(row):
+            g = float(np.sum((row[2:] - 0.5) ** 2))
+            f1 = (1 + g) * _math.cos(row[0] * _math.pi / 2) * _math.cos(row[1] * _math.pi / 2)
+            f2 = (1 + g) * _math.cos(row[0] * _math.pi / 2) * _math.sin(row[1] * _math.pi / 2)
+            f3 = (1 + g) * _math.sin(row[0] * _math.pi / 2)
+            return [f1, f2, f3]
+
+        study.add_trials(
+            [
+                optuna_amd.create_trial(
+                    params={n: float(pm[r, i]) for i, n in enumerate(names)},
+                    distributions=dists,
+                    values=_dtlz2(pm[r]),
+                )
+                for r in range(n_hist)
+            ]
+        )
+
+        def one_step() -> None:
+            t = study.ask()
+            x = np.array([t.suggest_float(n, 0, 1) for n in names])
+            study.tell(t, _dtlz2(x))
+
+        config = {
+            "model": "MO-TPE (3-objective DTLZ2), K6 rank + K6b greedy-HSSP",
+            "history_trials": n_hist,
+            "dims": 10,
+            "parallelism": "single",
+        }
     elif args.suite == "motpe":
         # Multi-objective TPE at a large history: the non-domination split runs
         # through the K6 dominance-bitmatrix kernel on a GPU box (>=4096 rows).
@@ -292,8 +334,28 @@ def _run_alt_suite(args: argparse.Namespace) -> None:
             f3 = (1 + g) * _math.sin(x[0] * _math.pi / 2)
             study.tell(t, (f1, f2, f3))
 
+        # Config 4 names "WFG hypervolume": track the population front's
+        # exact hypervolume once per generation (K6a on device).
+        from optuna_amd._hypervolume import compute_hypervolume
+        from optuna_amd.study._multi_objective import _is_pareto_front
+
+        hv_state = {"count": 0, "last_hv": 0.0}
+        base_step = one_step
+
+        def one_step() -> None:
+            base_step()
+            hv_state["count"] += 1
+            if hv_state["count"] % 50 == 0:
+                vals = np.array(
+                    [t.values for t in study.get_trials(deepcopy=False) if t.values]
+                )
+                ref = vals.max(axis=0) * 1.1
+                uniq = np.unique(vals, axis=0)
+                front = uniq[_is_pareto_front(uniq, assume_unique_lexsorted=True)]
+                hv_state["last_hv"] = compute_hypervolume(front, ref, assume_pareto=True)
+
         config = {
-            "model": "NSGAIISampler DTLZ2 3-objective",
+            "model": "NSGAIISampler DTLZ2 3-objective + per-generation WFG hypervolume",
             "population_size": 50,
             "parallelism": "single",
         }
