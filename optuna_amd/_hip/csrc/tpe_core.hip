@@ -1494,24 +1494,24 @@ double hv3d(const arr_f64& pts, double ref_x, double ref_y, double ref_z) {
 
 // ---------------------------------------------------------------------------
 // K6b: batched greedy-HSSP contributions, 3 objectives. One workgroup per
-// candidate: the limited set max(c, s_j) over the selected points lives in
-// LDS, gets x- and y-sorted by odd-even transposition, and 64 lanes split the
-// per-prefix staircase sweep. contribution(c | S) = incl(c) - HV3D(limited).
+// candidate. Key identity: the limited set {max(c, s) : s in S} keeps S's
+// sorted orders under the per-coordinate clamp (max(c, .) is monotone), so
+// the kernel reads GLOBALLY pre-sorted views of S — x-sorted xs, and the
+// y-sorted (y, z, x-rank) triple — clamps them by the candidate, and runs the
+// per-prefix staircase sweep directly. No per-candidate sort, no size cap;
+// clamp ties produce zero-width slabs the sweep ignores.
+// contribution(c | S) = incl(c) - HV3D(limited).
 // ---------------------------------------------------------------------------
-#define HSSP_MAX_SEL 256
-
 __global__ void k_hssp3d_contrib(const double* __restrict__ cand,  // (n, 3)
-                                 const double* __restrict__ sel,   // (k, 3)
+                                 const double* __restrict__ sx,    // (k) x asc
+                                 const double* __restrict__ sxz,   // (k) z in x-order
+                                 const double* __restrict__ sy,    // (k) y asc
+                                 const double* __restrict__ syz,   // (k) z in y-order
+                                 const int32_t* __restrict__ syr,  // (k) x-rank in y-order
                                  int64_t n, int64_t k, double ref_x,
                                  double ref_y, double ref_z,
                                  double* __restrict__ out) {  // (n)
-    __shared__ double lx[HSSP_MAX_SEL];
-    __shared__ double ly[HSSP_MAX_SEL];
-    __shared__ double lz[HSSP_MAX_SEL];
-    __shared__ double sy[HSSP_MAX_SEL];
-    __shared__ double sz[HSSP_MAX_SEL];
-    __shared__ int32_t sr[HSSP_MAX_SEL];
-    __shared__ double red[64];
+    __shared__ double red[128];
 
     const int64_t c = blockIdx.x;
     const double cx = cand[c * 3], cy = cand[c * 3 + 1], cz = cand[c * 3 + 2];
@@ -1520,55 +1520,21 @@ __global__ void k_hssp3d_contrib(const double* __restrict__ cand,  // (n, 3)
         if (threadIdx.x == 0) out[c] = incl;
         return;
     }
-    for (int64_t j = threadIdx.x; j < k; j += blockDim.x) {
-        lx[j] = fmax(cx, sel[j * 3]);
-        ly[j] = fmax(cy, sel[j * 3 + 1]);
-        lz[j] = fmax(cz, sel[j * 3 + 2]);
-    }
-    __syncthreads();
-    // Odd-even transposition sort of (lx, ly, lz) by lx — k passes.
-    for (int64_t pass = 0; pass < k; ++pass) {
-        const int64_t start = pass & 1;
-        for (int64_t j = start + 2 * (int64_t)threadIdx.x; j + 1 < k;
-             j += 2 * blockDim.x) {
-            if (lx[j] > lx[j + 1]) {
-                double a = lx[j]; lx[j] = lx[j + 1]; lx[j + 1] = a;
-                a = ly[j]; ly[j] = ly[j + 1]; ly[j + 1] = a;
-                a = lz[j]; lz[j] = lz[j + 1]; lz[j + 1] = a;
-            }
-        }
-        __syncthreads();
-    }
-    // y-order copy with x-rank payload.
-    for (int64_t j = threadIdx.x; j < k; j += blockDim.x) {
-        sy[j] = ly[j];
-        sz[j] = lz[j];
-        sr[j] = (int32_t)j;
-    }
-    __syncthreads();
-    for (int64_t pass = 0; pass < k; ++pass) {
-        const int64_t start = pass & 1;
-        for (int64_t j = start + 2 * (int64_t)threadIdx.x; j + 1 < k;
-             j += 2 * blockDim.x) {
-            if (sy[j] > sy[j + 1]) {
-                double a = sy[j]; sy[j] = sy[j + 1]; sy[j + 1] = a;
-                a = sz[j]; sz[j] = sz[j + 1]; sz[j + 1] = a;
-                int32_t r = sr[j]; sr[j] = sr[j + 1]; sr[j + 1] = r;
-            }
-        }
-        __syncthreads();
-    }
-    // Per-prefix staircase sweep, lanes strided over prefixes.
+    // Lanes stride over prefixes t; the y-stream is staged through LDS tiles
+    // shared by the whole block.
     double total = 0.0;
     for (int64_t t = threadIdx.x; t < k; t += blockDim.x) {
-        const double slab = (t + 1 < k ? lx[t + 1] : ref_x) - lx[t];
+        const double lx_t = fmax(cx, sx[t]);
+        const double lx_next = (t + 1 < k) ? fmax(cx, sx[t + 1]) : ref_x;
+        const double slab = lx_next - lx_t;
         if (slab <= 0.0) continue;
         double minz = ref_z;
         double area = 0.0;
         for (int64_t q = 0; q < k; ++q) {
-            if (sr[q] <= t) minz = fmin(minz, sz[q]);
-            const double y_next = (q + 1 < k) ? sy[q + 1] : ref_y;
-            area += (y_next - sy[q]) * (ref_z - minz);
+            if (syr[q] <= t) minz = fmin(minz, fmax(cz, syz[q]));
+            const double ly_q = fmax(cy, sy[q]);
+            const double ly_next = (q + 1 < k) ? fmax(cy, sy[q + 1]) : ref_y;
+            area += (ly_next - ly_q) * (ref_z - minz);
         }
         total += slab * area;
     }
@@ -1581,27 +1547,42 @@ __global__ void k_hssp3d_contrib(const double* __restrict__ cand,  // (n, 3)
     if (threadIdx.x == 0) out[c] = incl - red[0];
 }
 
-// Per-candidate greedy-HSSP contributions against the current selected set.
-py::array_t<double> hssp3d_contrib(const arr_f64& cand, const arr_f64& sel,
+// Per-candidate greedy-HSSP contributions against the current selected set,
+// passed as pre-sorted views (x-sorted xs/z, y-sorted y/z/x-rank).
+py::array_t<double> hssp3d_contrib(const arr_f64& cand, const arr_f64& sx,
+                                   const arr_f64& sxz, const arr_f64& sy,
+                                   const arr_f64& syz, const arr_i32& syr,
                                    double ref_x, double ref_y, double ref_z) {
     if (cand.ndim() != 2 || cand.shape(1) != 3)
         throw std::runtime_error("hssp3d_contrib: cand must be (n, 3)");
     const int64_t n = cand.shape(0);
-    const int64_t k = sel.ndim() == 2 ? sel.shape(0) : 0;
-    if (k > HSSP_MAX_SEL)
-        throw std::runtime_error("hssp3d_contrib: selected set too large");
+    const int64_t k = sx.size();
+    if ((int64_t)sy.size() != k || (int64_t)syz.size() != k ||
+        (int64_t)syr.size() != k || (int64_t)sxz.size() != k)
+        throw std::runtime_error("hssp3d_contrib: ragged sorted views");
     py::array_t<double> out(n);
     if (n == 0) return out;
     hipStream_t st = g_ws.get_stream();
-    double* base = g_ws.ensure((size_t)(n + k) * 3 + n + 8);
+    double* base = g_ws.ensure((size_t)n * 3 + 4 * (size_t)k + n + (size_t)(k + 1) / 2 + 8);
     double* d_cand = base;
-    double* d_sel = d_cand + (size_t)n * 3;
-    double* d_out = d_sel + (size_t)k * 3;
+    double* d_sx = d_cand + (size_t)n * 3;
+    double* d_sxz = d_sx + k;
+    double* d_sy = d_sxz + k;
+    double* d_syz = d_sy + k;
+    double* d_out = d_syz + k;
+    int32_t* d_syr = reinterpret_cast<int32_t*>(d_out + n);
     g_ws.begin_uploads();
     g_ws.h2d(d_cand, cand.data(), (size_t)n * 3 * 8, st);
-    if (k > 0) g_ws.h2d(d_sel, sel.data(), (size_t)k * 3 * 8, st);
-    hipLaunchKernelGGL(k_hssp3d_contrib, dim3((unsigned)n), dim3(64), 0, st,
-                       d_cand, d_sel, n, k, ref_x, ref_y, ref_z, d_out);
+    if (k > 0) {
+        g_ws.h2d(d_sx, sx.data(), k * 8, st);
+        g_ws.h2d(d_sxz, sxz.data(), k * 8, st);
+        g_ws.h2d(d_sy, sy.data(), k * 8, st);
+        g_ws.h2d(d_syz, syz.data(), k * 8, st);
+        g_ws.h2d(d_syr, syr.data(), k * 4, st);
+    }
+    hipLaunchKernelGGL(k_hssp3d_contrib, dim3((unsigned)n), dim3(128), 0, st,
+                       d_cand, d_sx, d_sxz, d_sy, d_syz, d_syr, n, k, ref_x,
+                       ref_y, ref_z, d_out);
     HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_out, n * 8,
                              hipMemcpyDeviceToHost, st));
     HIP_CHECK(hipStreamSynchronize(st));
@@ -1621,7 +1602,8 @@ PYBIND11_MODULE(_hipcore, m) {
           py::arg("n_below"));
     m.def("hv3d", &hv3d, py::arg("pts"), py::arg("ref_x"), py::arg("ref_y"),
           py::arg("ref_z"));
-    m.def("hssp3d_contrib", &hssp3d_contrib, py::arg("cand"), py::arg("sel"),
+    m.def("hssp3d_contrib", &hssp3d_contrib, py::arg("cand"), py::arg("sx"),
+          py::arg("sxz"), py::arg("sy"), py::arg("syz"), py::arg("syr"),
           py::arg("ref_x"), py::arg("ref_y"), py::arg("ref_z"));
     m.def("kde_logpdf", &kde_logpdf, py::arg("obs"), py::arg("sorted_pos"),
           py::arg("logw"), py::arg("alow"), py::arg("ahigh"), py::arg("steps"),

@@ -141,25 +141,55 @@ def _solve_hssp_3d_device(
     subset_size: int,
     reference_point: np.ndarray,
 ) -> np.ndarray | None:
-    """Exact (non-lazy) greedy via the K6b contributions kernel, or None."""
+    """Exact (non-lazy) greedy via the K6b contributions kernel, or None.
+
+    Every round evaluates ALL remaining candidates' exclusive contributions in
+    one launch; the selected set travels as pre-sorted views (max(c, .) is
+    monotone, so the kernel clamps instead of sorting per candidate). The
+    sorted views are maintained incrementally on the host (bisect inserts)."""
+    import bisect
+
     from optuna_amd import _hip
 
     core = _hip.get()
-    if core is None or not core.available() or subset_size > 256:
+    if core is None or not core.available():
         return None
     cand = np.ascontiguousarray(vals, dtype=np.float64)
     rx, ry, rz = (float(v) for v in reference_point)
     taken = np.zeros(len(vals), dtype=bool)
     chosen: list[int] = []
+    # Selected-set views: x-ascending (x, z) and y-ascending (y, z, x-rank).
+    xs: list[float] = []
+    xz: list[float] = []
+    yy: list[float] = []
+    yz: list[float] = []
+    yr: list[int] = []
     for _ in range(subset_size):
-        sel = (
-            np.ascontiguousarray(vals[chosen], dtype=np.float64)
-            if chosen
-            else np.empty((0, 3), dtype=np.float64)
+        contribs = np.asarray(
+            core.hssp3d_contrib(
+                cand,
+                np.asarray(xs, dtype=np.float64),
+                np.asarray(xz, dtype=np.float64),
+                np.asarray(yy, dtype=np.float64),
+                np.asarray(yz, dtype=np.float64),
+                np.asarray(yr, dtype=np.int32),
+                rx,
+                ry,
+                rz,
+            )
         )
-        contribs = np.asarray(core.hssp3d_contrib(cand, sel, rx, ry, rz))
         contribs[taken] = -np.inf
         j = int(np.argmax(contribs))
         taken[j] = True
         chosen.append(j)
+        x, y, z = (float(v) for v in vals[j])
+        px = bisect.bisect_right(xs, x)
+        xs.insert(px, x)
+        xz.insert(px, z)
+        # x-ranks at/after the insertion point shift up by one.
+        yr = [r + 1 if r >= px else r for r in yr]
+        py_ = bisect.bisect_right(yy, y)
+        yy.insert(py_, y)
+        yz.insert(py_, z)
+        yr.insert(py_, px)
     return indices[np.asarray(chosen)]
