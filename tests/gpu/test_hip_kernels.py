@@ -944,12 +944,12 @@ def test_hssp3d_device_matches_host_greedy(core) -> None:
     dev = hssp._solve_hssp_3d_device(vals, idx, k, ref)
     assert dev is not None
 
-    old = hssp._DEVICE_HSSP_MIN_ROWS
-    hssp._DEVICE_HSSP_MIN_ROWS = 10**9
+    old = hssp._DEVICE_HSSP_MIN_WORK
+    hssp._DEVICE_HSSP_MIN_WORK = 10**12
     try:
         host = hssp._solve_hssp(vals, idx, k, ref)
     finally:
-        hssp._DEVICE_HSSP_MIN_ROWS = old
+        hssp._DEVICE_HSSP_MIN_WORK = old
 
     hv_dev = wfg.compute_hypervolume(vals[dev], ref)
     hv_host = wfg.compute_hypervolume(vals[host], ref)
