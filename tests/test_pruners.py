@@ -224,3 +224,96 @@ def test_wilcoxon_pruner() -> None:
         WilcoxonPruner(p_threshold=2.0)
     with pytest.raises(ValueError):
         WilcoxonPruner(n_startup_steps=-1)
+
+
+def test_hyperband_bracket_study_attribute_whitelist() -> None:
+    """The per-bracket filtered study view exposes only what samplers need;
+    everything else must AttributeError (reference _BracketStudy contract)."""
+    import optuna_amd
+
+    pruner = optuna_amd.pruners.HyperbandPruner(
+        min_resource=1, max_resource=16, reduction_factor=2
+    )
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.RandomSampler(seed=0), pruner=pruner
+    )
+    pruner._try_initialization(study)
+    view = pruner._create_bracket_study(study, 0)
+
+    for allowed in ("get_trials", "direction", "_storage", "_study_id", "pruner", "study_name"):
+        getattr(view, allowed)
+    view.get_trials(deepcopy=False)
+    assert view._bracket_id == 0
+
+    for forbidden in ("optimize", "set_user_attr", "user_attrs", "system_attrs",
+                      "trials_dataframe"):
+        with pytest.raises(AttributeError):
+            getattr(view, forbidden)
+
+
+def test_hyperband_max_resource_auto_detected() -> None:
+    import optuna_amd
+
+    pruner = optuna_amd.pruners.HyperbandPruner(min_resource=1, reduction_factor=2)
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.RandomSampler(seed=0), pruner=pruner
+    )
+    n_reports = 9
+
+    def objective(trial):
+        for i in range(n_reports):
+            trial.report(1.0, i)
+            if trial.should_prune():
+                raise optuna_amd.TrialPruned()
+        return 1.0
+
+    study.optimize(objective, n_trials=12)
+    assert pruner._max_resource == n_reports
+
+
+def test_sha_rung_promotion_thresholds() -> None:
+    """Promotion beyond rung r requires a value within the top 1/rf of the
+    competitors at that rung (async SHA semantics)."""
+    import optuna_amd
+
+    pruner = optuna_amd.pruners.SuccessiveHalvingPruner(
+        min_resource=1, reduction_factor=2, min_early_stopping_rate=0
+    )
+    study = optuna_amd.create_study(pruner=pruner)
+
+    for i in range(7):
+        t = study.ask()
+        t.report(0.1 * (i + 1), step=7)
+        pruner.prune(study, study._storage.get_trial(t._trial_id))
+
+    def rungs_of(value: float) -> set[str]:
+        t = study.ask()
+        t.report(value, step=7)
+        pruner.prune(study, study._storage.get_trial(t._trial_id))
+        attrs = study._storage.get_trial(t._trial_id).system_attrs
+        return {k for k in attrs if k.startswith("completed_rung_")}
+
+    # 7th-from-bottom: completes rung 0 only.
+    assert rungs_of(0.75) == {"completed_rung_0"}
+    # 3rd-from-bottom: promoted once.
+    assert "completed_rung_1" in rungs_of(0.25)
+    # best-so-far: promoted twice.
+    r = rungs_of(0.05)
+    assert "completed_rung_2" in r and "completed_rung_3" not in r
+
+
+def test_sha_first_trial_never_pruned() -> None:
+    import optuna_amd
+
+    pruner = optuna_amd.pruners.SuccessiveHalvingPruner(
+        min_resource=1, reduction_factor=2, min_early_stopping_rate=0
+    )
+    study = optuna_amd.create_study(pruner=pruner)
+    t = study.ask()
+    for i in range(10):
+        t.report(1.0, step=i)
+        assert not t.should_prune()
+    attrs = study._storage.get_trial(t._trial_id).system_attrs
+    for rung in range(4):
+        assert f"completed_rung_{rung}" in attrs
+    assert "completed_rung_4" not in attrs

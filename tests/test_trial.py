@@ -170,3 +170,106 @@ def test_deprecated_suggest_aliases() -> None:
         t.suggest_loguniform("b", 1e-3, 1)
     with pytest.warns(FutureWarning):
         t.suggest_discrete_uniform("c", 0, 1, 0.5)
+
+
+def test_suggest_precedence_enqueued_over_sampler() -> None:
+    """_suggest precedence: cached > fixed (enqueued) > single > relative >
+    independent (reference trial/_trial.py:620-651)."""
+    import warnings
+
+    import optuna_amd
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        study = optuna_amd.create_study(
+            sampler=optuna_amd.samplers.RandomSampler(seed=0)
+        )
+        study.enqueue_trial({"x": 0.123})
+        got = {}
+
+        def objective(trial):
+            got[trial.number] = trial.suggest_float("x", 0, 1)
+            return got[trial.number]
+
+        study.optimize(objective, n_trials=2)
+    assert got[0] == 0.123  # enqueued fixed param wins over the sampler
+    assert got[1] != 0.123  # next trial samples freely
+
+
+def test_suggest_precedence_fixed_out_of_range_warns_but_returns() -> None:
+    import warnings
+
+    import optuna_amd
+
+    study = optuna_amd.create_study(sampler=optuna_amd.samplers.RandomSampler(seed=0))
+    study.enqueue_trial({"x": 7.0})  # outside [0, 1]
+
+    def objective(trial):
+        with warnings.catch_warnings(record=True) as caught:
+            warnings.simplefilter("always")
+            v = trial.suggest_float("x", 0, 1)
+        assert any("out of range" in str(w.message) for w in caught)
+        return v
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        study.optimize(objective, n_trials=1)
+    assert study.trials[0].params["x"] == 7.0
+
+
+def test_suggest_precedence_cached_over_everything() -> None:
+    import optuna_amd
+
+    study = optuna_amd.create_study(sampler=optuna_amd.samplers.RandomSampler(seed=0))
+    t = study.ask()
+    first = t.suggest_float("x", 0, 1)
+    again = t.suggest_float("x", 0, 1)
+    assert first == again  # same trial re-ask returns the cached value
+
+
+def test_suggest_single_value_beats_sampler() -> None:
+    import optuna_amd
+
+    class ExplodingSampler(optuna_amd.samplers.RandomSampler):
+        def sample_independent(self, *a, **k):  # pragma: no cover
+            raise AssertionError("single() distributions never reach the sampler")
+
+    study = optuna_amd.create_study(sampler=ExplodingSampler(seed=0))
+    t = study.ask()
+    assert t.suggest_float("x", 2.5, 2.5) == 2.5
+
+
+def test_waiting_claim_race_under_threads() -> None:
+    """N threads pop the WAITING queue concurrently: every enqueued trial is
+    claimed exactly once (the storage CAS settles races)."""
+    import threading
+
+    import optuna_amd
+
+    study = optuna_amd.create_study(sampler=optuna_amd.samplers.RandomSampler(seed=0))
+    n_queued = 24
+    for i in range(n_queued):
+        study.enqueue_trial({"x": float(i)})
+
+    claimed: list[float] = []
+    lock = threading.Lock()
+
+    def worker() -> None:
+        while True:
+            t = study.ask()
+            x = t.suggest_float("x", -1e9, 1e9)
+            with lock:
+                fresh = len(claimed) < n_queued
+                if fresh:
+                    claimed.append(x)
+            study.tell(t, 0.0)
+            if not fresh:
+                return
+
+    threads = [threading.Thread(target=worker) for _ in range(6)]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join()
+    # every enqueued value claimed exactly once
+    assert sorted(claimed[:n_queued]) == [float(i) for i in range(n_queued)]
