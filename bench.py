@@ -37,13 +37,13 @@ def main() -> None:
     parser.add_argument("--dims", type=int, default=20)
     parser.add_argument(
         "--suite",
-        choices=("tpe", "random", "gp", "cmaes", "nsgaii", "motpe", "motpe3"),
+        choices=("tpe", "random", "gp", "cmaes", "cmaes_hb", "nsgaii", "motpe", "motpe3"),
         default="tpe",
         help="BASELINE.json config to run (default: the headline TPE config)",
     )
     args = parser.parse_args()
 
-    if args.suite != "tpe":
+    if args.suite not in ("tpe", "cmaes_hb"):
         _run_alt_suite(args)
         return
 
@@ -75,15 +75,26 @@ def main() -> None:
         )
         dist.init_process_group(backend=backend)
 
-    D = args.dims
-    N_HISTORY = args.history
+    cmaes_hb = args.suite == "cmaes_hb"
+    D = max(args.dims, 100) if cmaes_hb else args.dims
+    N_HISTORY = 0 if cmaes_hb else args.history
     names = [f"x{i}" for i in range(D)]
     dists_def = {n: optuna_amd.distributions.FloatDistribution(-5.0, 5.0) for n in names}
 
-    def make_sampler() -> "optuna_amd.samplers.TPESampler":
+    def make_sampler():
+        if cmaes_hb:
+            # Config 5: CMA-ES 100-dim; Hyperband prunes on reported steps.
+            return optuna_amd.samplers.CmaEsSampler(seed=42, n_startup_trials=1)
         return optuna_amd.samplers.TPESampler(
             seed=42 + rank, n_startup_trials=10, constant_liar=(world_size > 1)
         )
+
+    def make_pruner():
+        if cmaes_hb:
+            return optuna_amd.pruners.HyperbandPruner(
+                min_resource=1, max_resource=8, reduction_factor=2
+            )
+        return None
 
     # ---- storage / study setup ------------------------------------------------------
     if world_size > 1:
@@ -92,7 +103,8 @@ def main() -> None:
         storage = RcclStorage.from_env()
         if rank == 0:
             study = optuna_amd.create_study(
-                study_name="bench", storage=storage, sampler=make_sampler()
+                study_name="bench", storage=storage, sampler=make_sampler(),
+                pruner=make_pruner(),
             )
             _populate(study, names, dists_def, N_HISTORY)
         dist.barrier()
@@ -100,6 +112,8 @@ def main() -> None:
             study = optuna_amd.load_study(
                 study_name="bench", storage=storage, sampler=make_sampler()
             )
+            if make_pruner() is not None:
+                study.pruner = make_pruner()
         dist.barrier()
         # Steady-state data plane: one all_gather of op batches per round —
         # device tensors over RCCL/xGMI with the nccl backend, zero sequencer
@@ -109,7 +123,7 @@ def main() -> None:
             storage.attach_collective_plane()
         dist.barrier()
     else:
-        study = optuna_amd.create_study(sampler=make_sampler())
+        study = optuna_amd.create_study(sampler=make_sampler(), pruner=make_pruner())
         _populate(study, names, dists_def, N_HISTORY)
 
     rng = np.random.RandomState(1234 + rank)
@@ -119,8 +133,21 @@ def main() -> None:
         x = np.empty(D)
         for i, n in enumerate(names):
             x[i] = trial.suggest_float(n, -5.0, 5.0)
-        # Synthetic objective (shifted sphere + noise), no model download needed.
         value = float(np.sum((x - 1.0) ** 2) + 0.01 * rng.randn())
+        if cmaes_hb:
+            # Report a shrinking intermediate curve; Hyperband prunes weak
+            # trials at the SHA rungs.
+            pruned = False
+            for step in range(8):
+                trial.report(value * (1.0 + 1.0 / (step + 1)), step)
+                if trial.should_prune():
+                    study.tell(trial, state=optuna_amd.trial.TrialState.PRUNED)
+                    pruned = True
+                    break
+            if not pruned:
+                study.tell(trial, value)
+            return
+        # Synthetic objective (shifted sphere + noise), no model download needed.
         study.tell(trial, value)
 
     # ---- warmup ---------------------------------------------------------------------
@@ -157,8 +184,13 @@ def main() -> None:
         from optuna_amd import _hip
 
         value = n_gpus * args.steps / elapsed
+        metric_name = (
+            "cmaes_hb sampler suggest()/sec (100-dim CMA-ES + Hyperband)"
+            if cmaes_hb
+            else "sampler suggest()/sec at 10k-trial history, 20-dim space"
+        )
         result = {
-            "metric": "sampler suggest()/sec at 10k-trial history, 20-dim space",
+            "metric": metric_name,
             "value": value,
             "unit": "suggest/s",
             "n_gpus": n_gpus,
@@ -169,11 +201,15 @@ def main() -> None:
             "scaling": "weak",
             # Reference measured locally at 3.74 suggest/s on this pool's CPU
             # for this exact config (BASELINE.md "Measured locally").
-            "vs_baseline": value / 3.74,
+            "vs_baseline": None if cmaes_hb else value / 3.74,
             "dtype": "fp64",
             "data": "synthetic",
             "config": {
-                "model": "TPESampler multivariate Parzen-KDE + EI",
+                "model": (
+                    "CmaEsSampler 100-dim + HyperbandPruner"
+                    if cmaes_hb
+                    else "TPESampler multivariate Parzen-KDE + EI"
+                ),
                 "history_trials": N_HISTORY,
                 "dims": D,
                 "n_ei_candidates": 24,
