@@ -955,3 +955,56 @@ def test_hssp3d_device_matches_host_greedy(core) -> None:
     hv_host = wfg.compute_hypervolume(vals[host], ref)
     np.testing.assert_allclose(hv_dev, hv_host, rtol=1e-9)
     np.testing.assert_array_equal(np.sort(dev), np.sort(host))
+
+
+def test_collective_plane_nccl_device_tensors() -> None:
+    """The collective plane's exchange must work with the nccl(=RCCL) backend
+    and device tensors — the path every rank runs in the driver's multi-GPU
+    weak-scaling bench (world_size=1 here: degenerate but full code path)."""
+    import os
+
+    import torch
+    import torch.distributed as dist
+
+    assert torch.cuda.is_available()
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29771")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        from optuna_amd.parallel.collective import CollectiveOpPlane
+
+        plane = CollectiveOpPlane()
+        assert plane._device.type == "cuda"
+        payloads = plane.exchange_bytes(b"hello-xgmi")
+        assert payloads == [b"hello-xgmi"]
+        records = plane.exchange_records([{"op_code": 5, "x": 1.5}])
+        assert records == [[{"op_code": 5, "x": 1.5}]]
+        assert plane.n_rounds == 2
+
+        # And through the storage layer end-to-end.
+        import datetime
+
+        from torch.distributed import TCPStore
+
+        import optuna_amd
+        from optuna_amd.storages._rccl import RcclStorage
+
+        store = TCPStore(
+            "127.0.0.1", 29772, 1, is_master=True,
+            timeout=datetime.timedelta(seconds=30),
+        )
+        storage = RcclStorage(store, worker_label="solo")
+        study = optuna_amd.create_study(
+            study_name="nccl1", storage=storage,
+            sampler=optuna_amd.samplers.RandomSampler(seed=0),
+        )
+        storage.attach_collective_plane()
+        for _ in range(3):
+            t = study.ask()
+            t.suggest_float("x", 0, 1)
+            study.tell(t, 0.5)
+        storage.collective_flush()
+        trials = storage.get_all_trials(study._study_id, deepcopy=False)
+        assert sum(t.state.name == "COMPLETE" for t in trials) == 3
+    finally:
+        dist.destroy_process_group()
