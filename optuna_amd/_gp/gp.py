@@ -519,6 +519,7 @@ class GPRegressor:
         minimum_noise: float,
         deterministic_objective: bool,
         gtol: float,
+        maxiter: int | None = None,
     ) -> "GPRegressor":
         import scipy.optimize
 
@@ -554,14 +555,18 @@ class GPRegressor:
                     raw_params, X, y, minimum_noise, deterministic_objective
                 )
 
+            opts = {"gtol": gtol}
+            if maxiter is not None:
+                opts["maxiter"] = maxiter
             res = scipy.optimize.minimize(
                 loss_func_closed,
                 initial_raw_params,
                 jac=True,
                 method="l-bfgs-b",
-                options={"gtol": gtol},
+                options=opts,
             )
-            if not res.success:
+            hit_cap = maxiter is not None and "ITERATIONS REACHED LIMIT" in str(res.message)
+            if not res.success and not hit_cap:
                 raise RuntimeError(f"Optimization failed: {res.message}")
             raw_opt = torch.from_numpy(res.x).to(X.device)
             self.inverse_squared_lengthscales = torch.exp(raw_opt[:n_params])
@@ -767,11 +772,19 @@ def fit_kernel_params(
     error = None
     for src in [gpr_cache, None]:
         try:
+            # Warm starts re-enter near the optimum; at device-fit sizes every
+            # L-BFGS iteration is an N^3 factorization, so cap the polish.
+            warm_cap = (
+                16
+                if src is not None and X.shape[0] >= GPRegressor._DEVICE_FIT_MIN_OBS
+                else None
+            )
             return _fresh(src)._fit_kernel_params(
                 log_prior=log_prior,
                 minimum_noise=minimum_noise,
                 deterministic_objective=deterministic_objective,
                 gtol=gtol,
+                maxiter=warm_cap,
             )
         except RuntimeError as e:
             error = e
