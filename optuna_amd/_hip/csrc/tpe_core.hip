@@ -1590,6 +1590,110 @@ py::array_t<double> hssp3d_contrib(const arr_f64& cand, const arr_f64& sx,
     return out;
 }
 
+
+// Masked argmax over the contributions, one block; the winner is marked taken
+// so the next round's kernel skips it — only the 8-byte index crosses the bus
+// per greedy round.
+__global__ void k_hssp3d_argmax(const double* __restrict__ vals, int64_t n,
+                                uint8_t* __restrict__ taken,
+                                int64_t* __restrict__ out_idx) {
+    __shared__ double red_v[256];
+    __shared__ int64_t red_i[256];
+    double best = -INFINITY;
+    int64_t bi = -1;
+    for (int64_t i = threadIdx.x; i < n; i += blockDim.x) {
+        if (!taken[i] && vals[i] > best) {
+            best = vals[i];
+            bi = i;
+        }
+    }
+    red_v[threadIdx.x] = best;
+    red_i[threadIdx.x] = bi;
+    __syncthreads();
+    for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if ((int)threadIdx.x < stride) {
+            if (red_v[threadIdx.x + stride] > red_v[threadIdx.x] ||
+                (red_v[threadIdx.x + stride] == red_v[threadIdx.x] &&
+                 red_i[threadIdx.x + stride] >= 0 &&
+                 (red_i[threadIdx.x] < 0 ||
+                  red_i[threadIdx.x + stride] < red_i[threadIdx.x]))) {
+                red_v[threadIdx.x] = red_v[threadIdx.x + stride];
+                red_i[threadIdx.x] = red_i[threadIdx.x + stride];
+            }
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out_idx[0] = red_i[0];
+        if (red_i[0] >= 0) taken[red_i[0]] = 1;
+    }
+}
+
+// Greedy-HSSP driver state: candidates and the taken-mask stay resident; each
+// round re-uploads only the (small, growing) sorted selected-set views and
+// pulls back ONE index.
+class Hssp3dSession {
+  public:
+    Hssp3dSession(const arr_f64& cand, double rx, double ry, double rz)
+        : n_(cand.shape(0)), rx_(rx), ry_(ry), rz_(rz) {
+        if (cand.ndim() != 2 || cand.shape(1) != 3)
+            throw std::runtime_error("Hssp3dSession: cand must be (n, 3)");
+        hipStream_t st = g_ws.get_stream();
+        HIP_CHECK(hipMalloc(&d_cand_, (size_t)n_ * 3 * 8));
+        HIP_CHECK(hipMalloc(&d_out_, (size_t)n_ * 8 + 8 + n_));
+        d_idx_ = reinterpret_cast<int64_t*>(d_out_ + n_);
+        d_taken_ = reinterpret_cast<uint8_t*>(d_idx_ + 1);
+        HIP_CHECK(hipMemcpyAsync(d_cand_, cand.data(), (size_t)n_ * 3 * 8,
+                                 hipMemcpyHostToDevice, st));
+        HIP_CHECK(hipMemsetAsync(d_taken_, 0, n_, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+    }
+    ~Hssp3dSession() {
+        if (d_cand_) (void)hipFree(d_cand_);
+        if (d_out_) (void)hipFree(d_out_);
+    }
+    Hssp3dSession(const Hssp3dSession&) = delete;
+    Hssp3dSession& operator=(const Hssp3dSession&) = delete;
+
+    int64_t round(const arr_f64& sx, const arr_f64& sxz, const arr_f64& sy,
+                  const arr_f64& syz, const arr_i32& syr) {
+        const int64_t k = sx.size();
+        hipStream_t st = g_ws.get_stream();
+        double* base = g_ws.ensure(4 * (size_t)k + (size_t)(k + 1) / 2 + 8);
+        double* d_sx = base;
+        double* d_sxz = d_sx + k;
+        double* d_sy = d_sxz + k;
+        double* d_syz = d_sy + k;
+        int32_t* d_syr = reinterpret_cast<int32_t*>(d_syz + k);
+        g_ws.begin_uploads();
+        if (k > 0) {
+            g_ws.h2d(d_sx, sx.data(), k * 8, st);
+            g_ws.h2d(d_sxz, sxz.data(), k * 8, st);
+            g_ws.h2d(d_sy, sy.data(), k * 8, st);
+            g_ws.h2d(d_syz, syz.data(), k * 8, st);
+            g_ws.h2d(d_syr, syr.data(), k * 4, st);
+        }
+        hipLaunchKernelGGL(k_hssp3d_contrib, dim3((unsigned)n_), dim3(128), 0,
+                           st, d_cand_, d_sx, d_sxz, d_sy, d_syz, d_syr, n_, k,
+                           rx_, ry_, rz_, d_out_);
+        hipLaunchKernelGGL(k_hssp3d_argmax, dim3(1), dim3(256), 0, st, d_out_,
+                           n_, d_taken_, d_idx_);
+        int64_t idx = -1;
+        HIP_CHECK(hipMemcpyAsync(&idx, d_idx_, 8, hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        HIP_CHECK(hipGetLastError());
+        return idx;
+    }
+
+  private:
+    int64_t n_;
+    double rx_, ry_, rz_;
+    double* d_cand_ = nullptr;
+    double* d_out_ = nullptr;
+    int64_t* d_idx_ = nullptr;
+    uint8_t* d_taken_ = nullptr;
+};
+
 PYBIND11_MODULE(_hipcore, m) {
     m.doc() = "optuna_amd MI355X (gfx950) HIP kernels: TPE parzen fit + mixture "
               "log-pdf, truncnorm device library";
@@ -1602,6 +1706,11 @@ PYBIND11_MODULE(_hipcore, m) {
           py::arg("n_below"));
     m.def("hv3d", &hv3d, py::arg("pts"), py::arg("ref_x"), py::arg("ref_y"),
           py::arg("ref_z"));
+    py::class_<Hssp3dSession>(m, "Hssp3dSession")
+        .def(py::init<const arr_f64&, double, double, double>(), py::arg("cand"),
+             py::arg("ref_x"), py::arg("ref_y"), py::arg("ref_z"))
+        .def("round", &Hssp3dSession::round, py::arg("sx"), py::arg("sxz"),
+             py::arg("sy"), py::arg("syz"), py::arg("syr"));
     m.def("hssp3d_contrib", &hssp3d_contrib, py::arg("cand"), py::arg("sx"),
           py::arg("sxz"), py::arg("sy"), py::arg("syz"), py::arg("syr"),
           py::arg("ref_x"), py::arg("ref_y"), py::arg("ref_z"));

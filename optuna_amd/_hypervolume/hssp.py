@@ -164,7 +164,10 @@ def _solve_hssp_3d_device(
         return None
     cand = np.ascontiguousarray(vals, dtype=np.float64)
     rx, ry, rz = (float(v) for v in reference_point)
-    taken = np.zeros(len(vals), dtype=bool)
+    # Candidates + taken-mask stay device-resident for the whole greedy; per
+    # round only the growing sorted views go up and ONE winner index comes
+    # back (in-kernel masked argmax).
+    session = core.Hssp3dSession(cand, rx, ry, rz)
     chosen: list[int] = []
     # Selected-set views: x-ascending (x, z) and y-ascending (y, z, x-rank).
     xs: list[float] = []
@@ -173,22 +176,16 @@ def _solve_hssp_3d_device(
     yz: list[float] = []
     yr: list[int] = []
     for _ in range(subset_size):
-        contribs = np.asarray(
-            core.hssp3d_contrib(
-                cand,
+        j = int(
+            session.round(
                 np.asarray(xs, dtype=np.float64),
                 np.asarray(xz, dtype=np.float64),
                 np.asarray(yy, dtype=np.float64),
                 np.asarray(yz, dtype=np.float64),
                 np.asarray(yr, dtype=np.int32),
-                rx,
-                ry,
-                rz,
             )
         )
-        contribs[taken] = -np.inf
-        j = int(np.argmax(contribs))
-        taken[j] = True
+        assert j >= 0
         chosen.append(j)
         x, y, z = (float(v) for v in vals[j])
         px = bisect.bisect_right(xs, x)
