@@ -1629,6 +1629,93 @@ __global__ void k_hssp3d_argmax(const double* __restrict__ vals, int64_t n,
     }
 }
 
+// Device-side selected-set insertion: after the argmax picked winner w, put
+// cand[w] into the x-sorted (x, z) and y-sorted (y, z, x-rank) views and bump
+// the device k counter. One block; the O(k) shifts are a few hundred
+// elements. Keeping the views on device lets the whole greedy run as a
+// launch train with a single host sync at the end.
+__global__ void k_hssp3d_insert(const double* __restrict__ cand,
+                                const int64_t* __restrict__ idx,
+                                double* __restrict__ sx, double* __restrict__ sxz,
+                                double* __restrict__ sy, double* __restrict__ syz,
+                                int32_t* __restrict__ syr,
+                                int64_t* __restrict__ k_ptr,
+                                int64_t* __restrict__ chosen, int64_t round) {
+    if (threadIdx.x != 0) return;  // serial: k is a few hundred at most
+    const int64_t w = idx[0];
+    chosen[round] = w;
+    if (w < 0) return;
+    const double x = cand[w * 3], y = cand[w * 3 + 1], z = cand[w * 3 + 2];
+    const int64_t k = k_ptr[0];
+    int64_t px = 0;
+    while (px < k && sx[px] <= x) ++px;
+    for (int64_t i = k; i > px; --i) {
+        sx[i] = sx[i - 1];
+        sxz[i] = sxz[i - 1];
+    }
+    sx[px] = x;
+    sxz[px] = z;
+    int64_t py = 0;
+    while (py < k && sy[py] <= y) ++py;
+    for (int64_t i = k; i > py; --i) {
+        sy[i] = sy[i - 1];
+        syz[i] = syz[i - 1];
+        syr[i] = syr[i - 1];
+    }
+    // existing x-ranks at/after the insertion point shift up by one
+    for (int64_t i = 0; i <= k; ++i) {
+        if (i != py && syr[i] >= (int32_t)px) syr[i] += 1;
+    }
+    sy[py] = y;
+    syz[py] = z;
+    syr[py] = (int32_t)px;
+    k_ptr[0] = k + 1;
+}
+
+// Contribution kernel variant reading k from device memory (launch-train mode).
+__global__ void k_hssp3d_contrib_dk(const double* __restrict__ cand,
+                                    const double* __restrict__ sx,
+                                    const double* __restrict__ sxz,
+                                    const double* __restrict__ sy,
+                                    const double* __restrict__ syz,
+                                    const int32_t* __restrict__ syr,
+                                    int64_t n, const int64_t* __restrict__ k_ptr,
+                                    double ref_x, double ref_y, double ref_z,
+                                    double* __restrict__ out) {
+    __shared__ double red[128];
+    const int64_t k = k_ptr[0];
+    const int64_t c = blockIdx.x;
+    const double cx = cand[c * 3], cy = cand[c * 3 + 1], cz = cand[c * 3 + 2];
+    const double incl = (ref_x - cx) * (ref_y - cy) * (ref_z - cz);
+    if (k == 0) {
+        if (threadIdx.x == 0) out[c] = incl;
+        return;
+    }
+    double total = 0.0;
+    for (int64_t t = threadIdx.x; t < k; t += blockDim.x) {
+        const double lx_t = fmax(cx, sx[t]);
+        const double lx_next = (t + 1 < k) ? fmax(cx, sx[t + 1]) : ref_x;
+        const double slab = lx_next - lx_t;
+        if (slab <= 0.0) continue;
+        double minz = ref_z;
+        double area = 0.0;
+        for (int64_t q = 0; q < k; ++q) {
+            if (syr[q] <= t) minz = fmin(minz, fmax(cz, syz[q]));
+            const double ly_q = fmax(cy, sy[q]);
+            const double ly_next = (q + 1 < k) ? fmax(cy, sy[q + 1]) : ref_y;
+            area += (ly_next - ly_q) * (ref_z - minz);
+        }
+        total += slab * area;
+    }
+    red[threadIdx.x] = total;
+    __syncthreads();
+    for (int stride = blockDim.x / 2; stride > 0; stride >>= 1) {
+        if ((int)threadIdx.x < stride) red[threadIdx.x] += red[threadIdx.x + stride];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) out[c] = incl - red[0];
+}
+
 // Greedy-HSSP driver state: candidates and the taken-mask stay resident; each
 // round re-uploads only the (small, growing) sorted selected-set views and
 // pulls back ONE index.
@@ -1654,6 +1741,39 @@ class Hssp3dSession {
     }
     Hssp3dSession(const Hssp3dSession&) = delete;
     Hssp3dSession& operator=(const Hssp3dSession&) = delete;
+
+    // Launch-train greedy: subset_size rounds of contrib→argmax→insert
+    // enqueued back-to-back with device-resident views, ONE host sync.
+    py::array_t<int64_t> run(int64_t subset_size) {
+        if (subset_size > n_) subset_size = n_;
+        hipStream_t st = g_ws.get_stream();
+        const size_t kcap = (size_t)subset_size + 1;
+        double* base = g_ws.ensure(4 * kcap + kcap / 2 + subset_size + 4);
+        double* d_sx = base;
+        double* d_sxz = d_sx + kcap;
+        double* d_sy = d_sxz + kcap;
+        double* d_syz = d_sy + kcap;
+        int32_t* d_syr = reinterpret_cast<int32_t*>(d_syz + kcap);
+        int64_t* d_k = reinterpret_cast<int64_t*>(d_syr + ((kcap + 1) & ~size_t(1)));
+        int64_t* d_chosen = d_k + 1;
+        HIP_CHECK(hipMemsetAsync(d_k, 0, 8, st));
+        for (int64_t r = 0; r < subset_size; ++r) {
+            hipLaunchKernelGGL(k_hssp3d_contrib_dk, dim3((unsigned)n_),
+                               dim3(128), 0, st, d_cand_, d_sx, d_sxz, d_sy,
+                               d_syz, d_syr, n_, d_k, rx_, ry_, rz_, d_out_);
+            hipLaunchKernelGGL(k_hssp3d_argmax, dim3(1), dim3(256), 0, st,
+                               d_out_, n_, d_taken_, d_idx_);
+            hipLaunchKernelGGL(k_hssp3d_insert, dim3(1), dim3(64), 0, st,
+                               d_cand_, d_idx_, d_sx, d_sxz, d_sy, d_syz,
+                               d_syr, d_k, d_chosen, r);
+        }
+        py::array_t<int64_t> out(subset_size);
+        HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_chosen, subset_size * 8,
+                                 hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        HIP_CHECK(hipGetLastError());
+        return out;
+    }
 
     int64_t round(const arr_f64& sx, const arr_f64& sxz, const arr_f64& sy,
                   const arr_f64& syz, const arr_i32& syr) {
@@ -1710,7 +1830,8 @@ PYBIND11_MODULE(_hipcore, m) {
         .def(py::init<const arr_f64&, double, double, double>(), py::arg("cand"),
              py::arg("ref_x"), py::arg("ref_y"), py::arg("ref_z"))
         .def("round", &Hssp3dSession::round, py::arg("sx"), py::arg("sxz"),
-             py::arg("sy"), py::arg("syz"), py::arg("syr"));
+             py::arg("sy"), py::arg("syz"), py::arg("syr"))
+        .def("run", &Hssp3dSession::run, py::arg("subset_size"));
     m.def("hssp3d_contrib", &hssp3d_contrib, py::arg("cand"), py::arg("sx"),
           py::arg("sxz"), py::arg("sy"), py::arg("syz"), py::arg("syr"),
           py::arg("ref_x"), py::arg("ref_y"), py::arg("ref_z"));

@@ -155,8 +155,6 @@ def _solve_hssp_3d_device(
     one launch; the selected set travels as pre-sorted views (max(c, .) is
     monotone, so the kernel clamps instead of sorting per candidate). The
     sorted views are maintained incrementally on the host (bisect inserts)."""
-    import bisect
-
     from optuna_amd import _hip
 
     core = _hip.get()
@@ -164,37 +162,11 @@ def _solve_hssp_3d_device(
         return None
     cand = np.ascontiguousarray(vals, dtype=np.float64)
     rx, ry, rz = (float(v) for v in reference_point)
-    # Candidates + taken-mask stay device-resident for the whole greedy; per
-    # round only the growing sorted views go up and ONE winner index comes
-    # back (in-kernel masked argmax).
+    # The whole greedy runs as one device launch train: candidates, taken
+    # mask, sorted selected-set views and the per-round winner all stay on
+    # device; contrib -> masked argmax -> sorted insert repeat subset_size
+    # times and a single sync brings back the chosen indices.
     session = core.Hssp3dSession(cand, rx, ry, rz)
-    chosen: list[int] = []
-    # Selected-set views: x-ascending (x, z) and y-ascending (y, z, x-rank).
-    xs: list[float] = []
-    xz: list[float] = []
-    yy: list[float] = []
-    yz: list[float] = []
-    yr: list[int] = []
-    for _ in range(subset_size):
-        j = int(
-            session.round(
-                np.asarray(xs, dtype=np.float64),
-                np.asarray(xz, dtype=np.float64),
-                np.asarray(yy, dtype=np.float64),
-                np.asarray(yz, dtype=np.float64),
-                np.asarray(yr, dtype=np.int32),
-            )
-        )
-        assert j >= 0
-        chosen.append(j)
-        x, y, z = (float(v) for v in vals[j])
-        px = bisect.bisect_right(xs, x)
-        xs.insert(px, x)
-        xz.insert(px, z)
-        # x-ranks at/after the insertion point shift up by one.
-        yr = [r + 1 if r >= px else r for r in yr]
-        py_ = bisect.bisect_right(yy, y)
-        yy.insert(py_, y)
-        yz.insert(py_, z)
-        yr.insert(py_, px)
-    return indices[np.asarray(chosen)]
+    chosen = np.asarray(session.run(int(subset_size)))
+    assert (chosen >= 0).all()
+    return indices[chosen]
