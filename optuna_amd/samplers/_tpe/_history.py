@@ -133,9 +133,20 @@ class _SpaceCache:
             vals = self.params[new_rows, c]
             if m == 1:
                 # side="right": equal values keep insertion (= row) order.
+                # The tail shift moves right, which numpy handles in place for
+                # fp64/int32 strided slices only via an explicit reversed copy;
+                # one reusable scratch avoids two 80 KB allocations per tell.
                 i = int(np.searchsorted(vb[:n], vals[0], side="right"))
-                vb[i + 1 : n + 1] = vb[i:n].copy()
-                rb[i + 1 : n + 1] = rb[i:n].copy()
+                tail = n - i
+                sf, si = self._shift_scratch or (None, None)
+                if sf is None or len(sf) < tail:
+                    sf = np.empty(max(1024, 2 * tail), dtype=np.float64)
+                    si = np.empty(max(1024, 2 * tail), dtype=np.int32)
+                    self._shift_scratch = (sf, si)
+                sf[:tail] = vb[i:n]
+                vb[i + 1 : n + 1] = sf[:tail]
+                si[:tail] = rb[i:n]
+                rb[i + 1 : n + 1] = si[:tail]
                 vb[i] = vals[0]
                 rb[i] = new_rows[0]
                 log_pos[0, c] = i
@@ -155,6 +166,8 @@ class _SpaceCache:
                 log_rows[:, c] = rows_sorted
         self._insert_log.append((log_pos, log_rows))
         self._n_sorted = n + m
+
+    _shift_scratch: tuple[np.ndarray, np.ndarray] | None = None
 
 
 class _TpeHistory:
