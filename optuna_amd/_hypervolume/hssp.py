@@ -140,7 +140,6 @@ def _solve_hssp(
 # (MO-TPE's 3-objective gamma boundary: ~350 candidates x ~200 selections
 # measured ~1 s/suggest on host vs ~50 ms on device).
 _DEVICE_HSSP_MIN_WORK = 8192
-_DEVICE_HSSP_MIN_ROWS = 512  # kept for tests pinning the old symbol
 
 
 def _solve_hssp_3d_device(

@@ -400,6 +400,13 @@ class RcclStorage(BaseStorage):
     ) -> int:
         study_name = study_name or DEFAULT_STUDY_NAME_PREFIX + str(uuid.uuid4())
         with self._thread_lock:
+            if self._plane is not None:
+                # Study ids are positional in the replay; concurrent creates on
+                # different ranks would collide before the merge. Create
+                # studies during bootstrap (sequencer mode), then attach.
+                raise RuntimeError(
+                    "create studies before attaching the collective plane"
+                )
             self._append(
                 JournalOperation.CREATE_STUDY,
                 {"study_name": study_name, "directions": [int(d) for d in directions]},
