@@ -291,3 +291,70 @@ def _child_param_sequence(case_cls: type, queue: Any) -> None:
     # Re-derive the factory from the fixture definition in the subclass.
     factory = case.__class__.sampler_factory.__wrapped__(case)  # type: ignore[attr-defined]
     queue.put(_run_param_sequence(factory, case_cls.n_trials))
+
+
+class RelativeSamplerTestCase:
+    """For samplers with a real relative stage (TPE-multivariate, GP, CMA-ES):
+    the relative search space must be inferred and sampled within bounds over
+    numerical, categorical and mixed spaces."""
+
+    n_trials: int = 14
+
+    @pytest.fixture
+    def sampler_factory(self) -> Callable[[int], BaseSampler]:
+        raise NotImplementedError
+
+    def _run(self, factory: Callable[[int], BaseSampler], objective) -> "optuna_amd.Study":
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=factory(17))
+            study.optimize(objective, n_trials=self.n_trials)
+        return study
+
+    def test_relative_numerical(self, sampler_factory: Callable[[int], BaseSampler]) -> None:
+        def objective(trial: optuna_amd.Trial) -> float:
+            a = trial.suggest_float("a", -2.0, 2.0)
+            b = trial.suggest_float("b", 1e-2, 1e2, log=True)
+            c = trial.suggest_int("c", 0, 20)
+            assert -2.0 <= a <= 2.0 and 1e-2 <= b <= 1e2 and 0 <= c <= 20
+            return a * a + abs(np_log10(b)) + c
+
+        study = self._run(sampler_factory, objective)
+        assert len(study.trials) == self.n_trials
+        # After startup, the relative stage must actually provide params (the
+        # suggest flow records identical distributions for every trial).
+        last = study.trials[-1]
+        assert set(last.params) == {"a", "b", "c"}
+
+    def test_relative_categorical_mixed(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        def objective(trial: optuna_amd.Trial) -> float:
+            x = trial.suggest_float("x", 0.0, 1.0)
+            k = trial.suggest_categorical("k", ("lo", "mid", "hi"))
+            bump = {"lo": 0.0, "mid": 0.3, "hi": 0.9}[k]
+            return (x - bump) ** 2
+
+        study = self._run(sampler_factory, objective)
+        assert all(t.params["k"] in ("lo", "mid", "hi") for t in study.trials)
+
+    def test_relative_params_with_n_jobs(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> None:
+        with warnings.catch_warnings():
+            warnings.simplefilter("ignore")
+            study = optuna_amd.create_study(sampler=sampler_factory(23))
+            study.optimize(
+                lambda t: t.suggest_float("x", -1, 1) ** 2
+                + (t.suggest_float("y", -1, 1) - 0.3) ** 2,
+                n_trials=16,
+                n_jobs=4,
+            )
+        assert len(study.trials) == 16
+        assert all(set(t.params) == {"x", "y"} for t in study.trials)
+
+
+def np_log10(v: float) -> float:
+    import math
+
+    return math.log10(v)

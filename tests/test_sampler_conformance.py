@@ -11,6 +11,7 @@ from optuna_amd.testing.pytest_samplers import (
     BasicSamplerTestCase,
     ExtendedSamplerTestCase,
     MultiObjectiveSamplerTestCase,
+    RelativeSamplerTestCase,
 )
 
 
@@ -26,7 +27,7 @@ class TestTPESampler(BasicSamplerTestCase, ExtendedSamplerTestCase, MultiObjecti
         return lambda seed: optuna_amd.samplers.TPESampler(seed=seed, n_startup_trials=3)
 
 
-class TestTPEMultivariate(BasicSamplerTestCase, ExtendedSamplerTestCase):
+class TestTPEMultivariate(BasicSamplerTestCase, ExtendedSamplerTestCase, RelativeSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.TPESampler(
@@ -34,7 +35,7 @@ class TestTPEMultivariate(BasicSamplerTestCase, ExtendedSamplerTestCase):
         )
 
 
-class TestCmaEsSampler(BasicSamplerTestCase, ExtendedSamplerTestCase):
+class TestCmaEsSampler(BasicSamplerTestCase, ExtendedSamplerTestCase, RelativeSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.CmaEsSampler(
@@ -78,7 +79,7 @@ class TestCmaEsSamplerSeparable(BasicSamplerTestCase):
         return make
 
 
-class TestGPSampler(BasicSamplerTestCase):
+class TestGPSampler(BasicSamplerTestCase, RelativeSamplerTestCase):
     n_trials = 6
 
     @pytest.fixture
