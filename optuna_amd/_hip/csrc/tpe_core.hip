@@ -1640,10 +1640,12 @@ __global__ void k_hssp3d_insert(const double* __restrict__ cand,
                                 double* __restrict__ sy, double* __restrict__ syz,
                                 int32_t* __restrict__ syr,
                                 int64_t* __restrict__ k_ptr,
-                                int64_t* __restrict__ chosen, int64_t round) {
+                                int64_t* __restrict__ chosen,
+                                int64_t* __restrict__ round_ptr) {
     if (threadIdx.x != 0) return;  // serial: k is a few hundred at most
     const int64_t w = idx[0];
-    chosen[round] = w;
+    chosen[round_ptr[0]] = w;
+    round_ptr[0] += 1;
     if (w < 0) return;
     const double x = cand[w * 3], y = cand[w * 3 + 1], z = cand[w * 3 + 2];
     const int64_t k = k_ptr[0];
@@ -1748,25 +1750,36 @@ class Hssp3dSession {
         if (subset_size > n_) subset_size = n_;
         hipStream_t st = g_ws.get_stream();
         const size_t kcap = (size_t)subset_size + 1;
-        double* base = g_ws.ensure(4 * kcap + kcap / 2 + subset_size + 4);
+        double* base = g_ws.ensure(4 * kcap + kcap / 2 + subset_size + 6);
         double* d_sx = base;
         double* d_sxz = d_sx + kcap;
         double* d_sy = d_sxz + kcap;
         double* d_syz = d_sy + kcap;
         int32_t* d_syr = reinterpret_cast<int32_t*>(d_syz + kcap);
         int64_t* d_k = reinterpret_cast<int64_t*>(d_syr + ((kcap + 1) & ~size_t(1)));
-        int64_t* d_chosen = d_k + 1;
-        HIP_CHECK(hipMemsetAsync(d_k, 0, 8, st));
-        for (int64_t r = 0; r < subset_size; ++r) {
-            hipLaunchKernelGGL(k_hssp3d_contrib_dk, dim3((unsigned)n_),
-                               dim3(128), 0, st, d_cand_, d_sx, d_sxz, d_sy,
-                               d_syz, d_syr, n_, d_k, rx_, ry_, rz_, d_out_);
-            hipLaunchKernelGGL(k_hssp3d_argmax, dim3(1), dim3(256), 0, st,
-                               d_out_, n_, d_taken_, d_idx_);
-            hipLaunchKernelGGL(k_hssp3d_insert, dim3(1), dim3(64), 0, st,
-                               d_cand_, d_idx_, d_sx, d_sxz, d_sy, d_syz,
-                               d_syr, d_k, d_chosen, r);
-        }
+        int64_t* d_round = d_k + 1;
+        int64_t* d_chosen = d_round + 1;
+        HIP_CHECK(hipMemsetAsync(d_k, 0, 16, st));
+        // Every round launches the same three kernels with identical args
+        // (k and the round index live in device memory), so one captured
+        // hipGraph replays the whole train at graph-launch cost.
+        hipGraph_t graph = nullptr;
+        hipGraphExec_t exec = nullptr;
+        HIP_CHECK(hipStreamBeginCapture(st, hipStreamCaptureModeThreadLocal));
+        hipLaunchKernelGGL(k_hssp3d_contrib_dk, dim3((unsigned)n_), dim3(128),
+                           0, st, d_cand_, d_sx, d_sxz, d_sy, d_syz, d_syr, n_,
+                           d_k, rx_, ry_, rz_, d_out_);
+        hipLaunchKernelGGL(k_hssp3d_argmax, dim3(1), dim3(256), 0, st, d_out_,
+                           n_, d_taken_, d_idx_);
+        hipLaunchKernelGGL(k_hssp3d_insert, dim3(1), dim3(64), 0, st, d_cand_,
+                           d_idx_, d_sx, d_sxz, d_sy, d_syz, d_syr, d_k,
+                           d_chosen, d_round);
+        HIP_CHECK(hipStreamEndCapture(st, &graph));
+        HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
+        for (int64_t r = 0; r < subset_size; ++r)
+            HIP_CHECK(hipGraphLaunch(exec, st));
+        (void)hipGraphExecDestroy(exec);
+        (void)hipGraphDestroy(graph);
         py::array_t<int64_t> out(subset_size);
         HIP_CHECK(hipMemcpyAsync(out.mutable_data(), d_chosen, subset_size * 8,
                                  hipMemcpyDeviceToHost, st));
