@@ -1642,36 +1642,78 @@ __global__ void k_hssp3d_insert(const double* __restrict__ cand,
                                 int64_t* __restrict__ k_ptr,
                                 int64_t* __restrict__ chosen,
                                 int64_t* __restrict__ round_ptr) {
-    if (threadIdx.x != 0) return;  // serial: k is a few hundred at most
-    const int64_t w = idx[0];
-    chosen[round_ptr[0]] = w;
-    round_ptr[0] += 1;
+    // One block. The O(k) tail shifts run in parallel: every thread stages
+    // its strided elements in registers, barriers, then writes them one slot
+    // to the right (read-before-write is global across the block).
+    __shared__ int64_t s_w, s_px, s_py, s_k;
+    if (threadIdx.x == 0) {
+        const int64_t w = idx[0];
+        s_w = w;
+        chosen[round_ptr[0]] = w;
+        round_ptr[0] += 1;
+        const int64_t k = k_ptr[0];
+        s_k = k;
+        if (w >= 0) {
+            const double x = cand[w * 3], y = cand[w * 3 + 1];
+            int64_t px = 0;
+            while (px < k && sx[px] <= x) ++px;
+            int64_t py = 0;
+            while (py < k && sy[py] <= y) ++py;
+            s_px = px;
+            s_py = py;
+            k_ptr[0] = k + 1;
+        }
+    }
+    __syncthreads();
+    const int64_t w = s_w;
     if (w < 0) return;
+    const int64_t k = s_k, px = s_px, py = s_py;
     const double x = cand[w * 3], y = cand[w * 3 + 1], z = cand[w * 3 + 2];
-    const int64_t k = k_ptr[0];
-    int64_t px = 0;
-    while (px < k && sx[px] <= x) ++px;
-    for (int64_t i = k; i > px; --i) {
-        sx[i] = sx[i - 1];
-        sxz[i] = sxz[i - 1];
+
+    constexpr int MAXLOC = 32;  // supports k up to 32 * blockDim
+    double rx_v[MAXLOC], rz_v[MAXLOC], ry_v[MAXLOC], ryz_v[MAXLOC];
+    int32_t rr_v[MAXLOC];
+    int nx = 0, ny = 0;
+    for (int64_t i = px + threadIdx.x; i < k; i += blockDim.x) {
+        rx_v[nx] = sx[i];
+        rz_v[nx] = sxz[i];
+        ++nx;
     }
-    sx[px] = x;
-    sxz[px] = z;
-    int64_t py = 0;
-    while (py < k && sy[py] <= y) ++py;
-    for (int64_t i = k; i > py; --i) {
-        sy[i] = sy[i - 1];
-        syz[i] = syz[i - 1];
-        syr[i] = syr[i - 1];
+    for (int64_t i = py + threadIdx.x; i < k; i += blockDim.x) {
+        ry_v[ny] = sy[i];
+        ryz_v[ny] = syz[i];
+        // pre-bump the x-ranks shifting right of the x insertion point
+        int32_t r = syr[i];
+        rr_v[ny] = (r >= (int32_t)px) ? r + 1 : r;
+        ++ny;
     }
-    // existing x-ranks at/after the insertion point shift up by one
-    for (int64_t i = 0; i <= k; ++i) {
-        if (i != py && syr[i] >= (int32_t)px) syr[i] += 1;
+    // ranks BEFORE py also need the bump; handle them in-place (no shift)
+    __syncthreads();
+    for (int64_t i = threadIdx.x; i < py; i += blockDim.x) {
+        int32_t r = syr[i];
+        if (r >= (int32_t)px) syr[i] = r + 1;
     }
-    sy[py] = y;
-    syz[py] = z;
-    syr[py] = (int32_t)px;
-    k_ptr[0] = k + 1;
+    {
+        int j = 0;
+        for (int64_t i = px + threadIdx.x; i < k; i += blockDim.x, ++j) {
+            sx[i + 1] = rx_v[j];
+            sxz[i + 1] = rz_v[j];
+        }
+        j = 0;
+        for (int64_t i = py + threadIdx.x; i < k; i += blockDim.x, ++j) {
+            sy[i + 1] = ry_v[j];
+            syz[i + 1] = ryz_v[j];
+            syr[i + 1] = rr_v[j];
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        sx[px] = x;
+        sxz[px] = z;
+        sy[py] = y;
+        syz[py] = z;
+        syr[py] = (int32_t)px;
+    }
 }
 
 // Contribution kernel variant reading k from device memory (launch-train mode).

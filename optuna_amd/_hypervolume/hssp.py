@@ -157,7 +157,9 @@ def _solve_hssp_3d_device(
     from optuna_amd import _hip
 
     core = _hip.get()
-    if core is None or not core.available():
+    if core is None or not core.available() or subset_size > 2000:
+        # 2000 = the insert kernel's register-staging bound (32 slots x 64
+        # lanes); greedy subsets beyond it fall back to the host path.
         return None
     cand = np.ascontiguousarray(vals, dtype=np.float64)
     rx, ry, rz = (float(v) for v in reference_point)
