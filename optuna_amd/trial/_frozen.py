@@ -249,45 +249,20 @@ class FrozenTrial(BaseTrial):
     def values(self, v: Sequence[float] | None) -> None:
         self._values = None if v is None else list(v)
 
-    @property
-    def datetime_start(self) -> datetime.datetime | None:
-        return self._datetime_start
+    # Plain pass-through fields of the record (readable AND assignable — the
+    # storages' copy-on-write updates rebind them wholesale).
+    def _record_field(slot: str) -> property:  # noqa: N805 — class-body helper
+        return property(
+            lambda self: getattr(self, slot),
+            lambda self, value: setattr(self, slot, value),
+        )
 
-    @datetime_start.setter
-    def datetime_start(self, value: datetime.datetime | None) -> None:
-        self._datetime_start = value
-
-    @property
-    def params(self) -> dict[str, Any]:
-        return self._params
-
-    @params.setter
-    def params(self, params: dict[str, Any]) -> None:
-        self._params = params
-
-    @property
-    def distributions(self) -> dict[str, BaseDistribution]:
-        return self._distributions
-
-    @distributions.setter
-    def distributions(self, value: dict[str, BaseDistribution]) -> None:
-        self._distributions = value
-
-    @property
-    def user_attrs(self) -> dict[str, Any]:
-        return self._user_attrs
-
-    @user_attrs.setter
-    def user_attrs(self, value: dict[str, Any]) -> None:
-        self._user_attrs = value
-
-    @property
-    def system_attrs(self) -> dict[str, Any]:
-        return self._system_attrs
-
-    @system_attrs.setter
-    def system_attrs(self, value: dict[str, Any]) -> None:
-        self._system_attrs = value
+    datetime_start = _record_field("_datetime_start")
+    params = _record_field("_params")
+    distributions = _record_field("_distributions")
+    user_attrs = _record_field("_user_attrs")
+    system_attrs = _record_field("_system_attrs")
+    del _record_field
 
     @property
     def constraints(self) -> dict[int, float]:
