@@ -72,26 +72,25 @@ class NSGAIIISampler(BaseGASampler):
         self._rng = LazyRandomState(seed)
         self._constraints_func = constraints_func
         self._search_space = IntersectionSearchSpace()
-        self._elite_population_selection_strategy = (
-            elite_population_selection_strategy
-            or NSGAIIIElitePopulationSelectionStrategy(
+        if elite_population_selection_strategy is None:
+            elite_population_selection_strategy = NSGAIIIElitePopulationSelectionStrategy(
+                rng=self._rng,
                 population_size=population_size,
-                constraints_func=constraints_func,
                 reference_points=reference_points,
                 dividing_parameter=dividing_parameter,
-                rng=self._rng,
+                constraints_func=constraints_func,
             )
-        )
+        self._elite_population_selection_strategy = elite_population_selection_strategy
         self._child_generation_strategy = (
             child_generation_strategy
             or NSGAIIChildGenerationStrategy(
+                rng=self._rng,
+                constraints_func=constraints_func,
+                crossover=crossover,
                 crossover_prob=crossover_prob,
+                swapping_prob=swapping_prob,
                 mutation=mutation,
                 mutation_prob=mutation_prob,
-                swapping_prob=swapping_prob,
-                crossover=crossover,
-                constraints_func=constraints_func,
-                rng=self._rng,
             )
         )
         self._after_trial_strategy = after_trial_strategy or NSGAIIAfterTrialStrategy(
