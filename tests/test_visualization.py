@@ -182,3 +182,95 @@ def test_mo_requires_target(mo_study) -> None:
         mo_study, target=lambda t: t.values[0], target_name="obj0"
     )
     assert fig is not None
+
+
+def test_contour_info(study) -> None:
+    from optuna_amd.visualization._contour import _get_contour_info
+
+    info = _get_contour_info(study)
+    assert info.sorted_params == ["c", "lg", "x"]
+    n = len(info.sorted_params)
+    assert len(info.sub_plot_infos) == n and all(len(r) == n for r in info.sub_plot_infos)
+    # off-diagonal cells pair distinct params
+    for i, row in enumerate(info.sub_plot_infos):
+        for j, cell in enumerate(row):
+            if i != j:
+                assert cell.xaxis.name != cell.yaxis.name
+
+
+def test_parallel_coordinate_info(study) -> None:
+    from optuna_amd.visualization._parallel_coordinate import (
+        _get_parallel_coordinate_info,
+    )
+
+    info = _get_parallel_coordinate_info(study)
+    assert len(info.dims_params) >= 2
+    assert info.dim_objective is not None
+
+
+def test_rank_info(study) -> None:
+    from optuna_amd.visualization._rank import _get_rank_info
+
+    info = _get_rank_info(study, params=None, target=None, target_name="Objective Value")
+    assert len(info.params) >= 2
+
+
+def test_param_importances_info(study) -> None:
+    from optuna_amd.visualization._param_importances import _get_importances_info
+    from optuna_amd.importance import PedAnovaImportanceEvaluator
+
+    info = _get_importances_info(
+        study, PedAnovaImportanceEvaluator(), params=None, target=None,
+        target_name="Objective Value",
+    )
+    assert set(info.param_names) <= {"x", "lg", "c"}
+    assert len(info.importance_values) == len(info.param_names)
+
+
+def test_contour_with_categorical_and_log(study) -> None:
+    import optuna_amd
+    from optuna_amd.visualization import plot_contour
+
+    fig = plot_contour(study, params=["x", "lg"])
+    assert fig is not None
+
+
+def test_edf_multiple_studies() -> None:
+    import optuna_amd
+    from optuna_amd.visualization._edf import _get_edf_info
+
+    studies = []
+    for seed in (1, 2):
+        s = optuna_amd.create_study(
+            sampler=optuna_amd.samplers.RandomSampler(seed=seed)
+        )
+        s.optimize(lambda t: t.suggest_float("x", 0, 1) ** 2, n_trials=6)
+        studies.append(s)
+    info = _get_edf_info(studies)
+    assert len(info.lines) == 2
+    for line in info.lines:
+        assert (np.diff(line.y_values) >= 0).all()  # EDFs are nondecreasing
+
+
+def test_timeline_states_and_order(study) -> None:
+    from optuna_amd.visualization._timeline import _get_timeline_info
+
+    info = _get_timeline_info(study)
+    assert len(info.bars) == len(study.trials)
+    for bar in info.bars:
+        assert bar.complete >= bar.start
+
+
+def test_terminator_improvement_info() -> None:
+    import optuna_amd
+    from optuna_amd.visualization._terminator_improvement import (
+        _get_improvement_info,
+    )
+
+    s = optuna_amd.create_study(sampler=optuna_amd.samplers.RandomSampler(seed=3))
+    s.optimize(
+        lambda t: t.suggest_float("a", 0, 1) + t.suggest_float("b", 0, 1),
+        n_trials=6,
+    )
+    info = _get_improvement_info(s)
+    assert len(info.trial_numbers) == 6
