@@ -718,8 +718,8 @@ def _incremental_update_applicable(prev: GPRegressor, n_now: int) -> bool:
 
     Hyperparameters move slowly once thousands of observations are in, so a
     full L-BFGS refit (dozens of N³ factorizations) only runs after the data
-    grew by ~1% (at most 64 rows) since the last fit; in between, the cached
-    regressor absorbs new rows with O(N²) updates. Below the device-fit
+    grew by ~2.5% (at most 128 rows) since the last fit; in between, the
+    cached regressor absorbs new rows with O(N²) updates. Below the device-fit
     threshold the reference's refit-every-suggest behavior is kept — the host
     fit is cheap there.
     """
@@ -727,7 +727,7 @@ def _incremental_update_applicable(prev: GPRegressor, n_now: int) -> bool:
     if n_fit < GPRegressor._DEVICE_FIT_MIN_OBS:
         return False
     grown = n_now - n_fit
-    return 0 <= grown < min(64, max(1, n_fit // 100))
+    return 0 <= grown < min(128, max(1, n_fit // 40))
 
 
 def fit_kernel_params(

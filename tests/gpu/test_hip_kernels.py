@@ -853,7 +853,7 @@ def test_gp_incremental_update_block_inverse_matches_refit() -> None:
     mean_r, var_r = ref.posterior(x_eval)
     torch.testing.assert_close(mean_u, mean_r, rtol=1e-6, atol=1e-8)
     torch.testing.assert_close(var_u, var_r, rtol=1e-4, atol=1e-8)
-    # the cadence policy: growth < n_fit/100 (=9 here) → update, more → refit
+    # the cadence policy: growth < n_fit/40 (=22 here) → update, more → refit
     assert gp_mod._incremental_update_applicable(gpr, N0 + 5)
     assert not gp_mod._incremental_update_applicable(gpr, N0 + 50)
 
