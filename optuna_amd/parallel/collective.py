@@ -78,11 +78,26 @@ class CollectiveOpPlane:
         self.n_rounds += 1
         return [bytes(recv[r][: lens[r]].cpu().numpy().tobytes()) for r in range(self.world)]
 
+    # Payloads above this size ride compressed (CMA-ES generations carry
+    # ~160 KB of hex-pickled optimizer state per rank; JSON+hex deflates ~3x).
+    _COMPRESS_MIN = 4096
+
     def exchange_records(
         self, records: list[dict[str, Any]]
     ) -> list[list[dict[str, Any]]]:
         """Exchange op-record batches; returns one batch per rank, rank order."""
-        payloads = self.exchange_bytes(
-            json.dumps(records, separators=(",", ":")).encode()
-        )
-        return [json.loads(p) if p else [] for p in payloads]
+        import zlib
+
+        raw = json.dumps(records, separators=(",", ":")).encode()
+        if len(raw) >= self._COMPRESS_MIN:
+            raw = b"Z" + zlib.compress(raw, level=1)
+        else:
+            raw = b"J" + raw
+        out: list[list[dict[str, Any]]] = []
+        for p in self.exchange_bytes(raw):
+            if not p:
+                out.append([])
+                continue
+            body = zlib.decompress(p[1:]) if p[:1] == b"Z" else p[1:]
+            out.append(json.loads(body) if body else [])
+        return out
