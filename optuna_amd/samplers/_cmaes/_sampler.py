@@ -15,6 +15,7 @@ with the reference's ask/tell protocol: raw samples stored per trial under the
 """
 from __future__ import annotations
 
+import copy
 import pickle
 from typing import TYPE_CHECKING, Any, Sequence
 
@@ -225,8 +226,13 @@ class CmaEsSampler(BaseSampler):
                 optimizer_str[i : i + _SYSTEM_ATTR_MAX_LENGTH],
             )
 
+    _restore_cache: tuple[tuple[int, int], CMA] | None = None
+
     def _restore_optimizer(self, completed_trials: list[FrozenTrial]) -> CMA | None:
-        # Scan backwards: the newest stored state wins.
+        # Scan backwards: the newest stored state wins. The ~160 KB hex pickle
+        # only changes at generation boundaries, so the decoded optimizer is
+        # cached keyed by (source trial id, payload length) — every other ask
+        # skips hex-decode + unpickle (milliseconds per suggest at 100 dims).
         for trial in reversed(completed_trials):
             chunks = [
                 (key, value)
@@ -237,7 +243,13 @@ class CmaEsSampler(BaseSampler):
                 continue
             chunks.sort(key=lambda kv: int(kv[0].rsplit(":", 1)[1]))
             optimizer_str = "".join(v for _, v in chunks)
-            return pickle.loads(bytes.fromhex(optimizer_str))
+            cache_key = (trial._trial_id, len(optimizer_str))
+            cached = self._restore_cache
+            if cached is not None and cached[0] == cache_key:
+                return copy.deepcopy(cached[1])
+            optimizer = pickle.loads(bytes.fromhex(optimizer_str))
+            self._restore_cache = (cache_key, copy.deepcopy(optimizer))
+            return optimizer
         return None
 
     # ---- initialization -------------------------------------------------------------
