@@ -99,6 +99,7 @@ def test_collective_lockstep_two_ranks() -> None:
         for r in range(2)
     ]
     for p in procs:
+        p.daemon = True
         p.start()
     results = {}
     for _ in range(2):
@@ -107,6 +108,9 @@ def test_collective_lockstep_two_ranks() -> None:
         results[rank] = (ids, table, n_rounds)
     for p in procs:
         p.join(timeout=60)
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
 
     ids0, table0, _ = results[0]
     ids1, table1, _ = results[1]
@@ -140,6 +144,7 @@ def test_collective_mode_to_journal_roundtrip(tmp_path) -> None:
         for r in range(2)
     ]
     for p in procs:
+        p.daemon = True
         p.start()
     results = {}
     for _ in range(2):
@@ -148,6 +153,9 @@ def test_collective_mode_to_journal_roundtrip(tmp_path) -> None:
         results[rank] = payload
     for p in procs:
         p.join(timeout=60)
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
 
     import optuna_amd
     from optuna_amd.storages.journal import JournalFileBackend, JournalStorage

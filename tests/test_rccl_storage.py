@@ -99,9 +99,13 @@ def test_two_process_shared_study() -> None:
         ctx.Process(target=_worker, args=(rank, 2, port, queue)) for rank in range(2)
     ]
     for p in procs:
+        p.daemon = True
         p.start()
     for p in procs:
         p.join(timeout=180)
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
     results = [queue.get(timeout=10) for _ in range(2)]
     for rank, n, payload in results:
         assert n != -1, f"rank {rank} failed:\n{payload}"
@@ -126,8 +130,13 @@ def _claim_worker(rank: int, world_size: int, port: int, trial_id: int, queue) -
             store.set("trial_id", str(tid))
         tid = int(store.get("trial_id"))
         store.add("barrier", 1)
+        import time as _time
+
+        deadline = _time.monotonic() + 60
         while store.add("barrier", 0) < world_size:
-            pass
+            _time.sleep(0.005)
+            if _time.monotonic() > deadline:
+                raise TimeoutError("barrier timeout")
         won = storage.set_trial_state_values(tid, TrialState.RUNNING)
         queue.put((rank, bool(won)))
         store.add("exit", 1)
@@ -153,9 +162,13 @@ def test_waiting_claim_cas_across_processes() -> None:
         for rank in range(world)
     ]
     for p in procs:
+        p.daemon = True
         p.start()
     for p in procs:
         p.join(timeout=120)
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
     results = [queue.get(timeout=10) for _ in range(world)]
     wins = []
     for rank, won in results:
