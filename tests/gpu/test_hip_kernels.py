@@ -1008,3 +1008,49 @@ def test_collective_plane_nccl_device_tensors() -> None:
         assert sum(t.state.name == "COMPLETE" for t in trials) == 3
     finally:
         dist.destroy_process_group()
+
+
+def test_mo_tpe_uses_per_dim_device_path(core, monkeypatch) -> None:
+    """Multi-objective TPE must score BOTH the below and above KDEs through
+    the per-dim resident-table kernel (one batched round per suggest)."""
+    import warnings
+
+    from optuna_amd.samplers._tpe import _device as device_mod
+
+    calls = {"per_dim": 0, "joint": 0}
+    orig = device_mod.score_above_resident
+
+    def spy(*args, **kwargs):
+        calls["per_dim" if kwargs.get("per_dim") else "joint"] += 1
+        return orig(*args, **kwargs)
+
+    monkeypatch.setattr(device_mod, "score_above_resident", spy)
+    warnings.simplefilter("ignore")
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(2)
+    names = [f"x{i}" for i in range(6)]
+    dists = {n: FloatDistribution(0.0, 1.0) for n in names}
+    study = optuna_amd.create_study(
+        directions=["minimize", "minimize"],
+        sampler=optuna_amd.samplers.TPESampler(seed=0, n_startup_trials=5),
+    )
+    pm = rng.uniform(0, 1, size=(3000, 6))
+    study.add_trials(
+        [
+            optuna_amd.create_trial(
+                params={n: float(pm[r, i]) for i, n in enumerate(names)},
+                distributions=dists,
+                values=[float(pm[r, 0]), float(1 - pm[r, 1])],
+            )
+            for r in range(3000)
+        ]
+    )
+
+    def objective(trial):
+        x = [trial.suggest_float(n, 0, 1) for n in names]
+        return x[0], 1 - x[1]
+
+    study.optimize(objective, n_trials=3)
+    # below (gamma=300 >= 128) + above per suggest, D dims served per batch
+    assert calls["per_dim"] >= 6
+    assert calls["joint"] == 0
