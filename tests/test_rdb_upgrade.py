@@ -256,3 +256,83 @@ def test_upgrade_intermediate_era_db(tmp_path) -> None:
     assert math.isnan(t.intermediate_values[0])
     assert t.intermediate_values[1] == 2.0
     assert study.direction == optuna_amd.study.StudyDirection.MAXIMIZE
+
+
+def test_reference_reads_our_upgraded_db(tmp_path) -> None:
+    """After our in-place upgrade of a v0.9-era file, the REFERENCE library
+    opens it and reads the study (the strongest upgrade-fidelity check)."""
+    import logging as _logging
+    import os
+    import sys
+    import types
+
+    if not os.path.isdir("/root/reference"):
+        pytest.skip("reference tree not available")
+    db = str(tmp_path / "up.db")
+    _make_v09_db(db)
+    url = f"sqlite:///{db}"
+    storage = RDBStorage(url, skip_compatibility_check=True)
+    storage.upgrade()
+
+    if "colorlog" not in sys.modules:  # the reference hard-requires colorlog
+        stub = types.ModuleType("colorlog")
+
+        class _Fmt(_logging.Formatter):
+            def __init__(self, fmt=None, **kwargs):
+                if fmt:
+                    fmt = fmt.replace("%(log_color)s", "").replace("%(reset)s", "")
+                super().__init__(fmt)
+
+        stub.ColoredFormatter = _Fmt
+        stub.TTYColoredFormatter = _Fmt
+        stub.StreamHandler = _logging.StreamHandler
+        sys.modules["colorlog"] = stub
+    if "/root/reference" not in sys.path:
+        sys.path.insert(0, "/root/reference")
+    # alembic is not installed here and the reference's RDBStorage imports it;
+    # read the upgraded file through the reference's ORM MODELS instead — the
+    # schema contract is what the upgrade must satisfy.
+    import sqlalchemy as sa
+    import sqlalchemy.orm as sa_orm
+
+    from optuna.storages._rdb import models as ref_models
+
+    engine = sa.create_engine(url)
+    with sa_orm.Session(engine) as session:
+        trials = (
+            session.query(ref_models.TrialModel)
+            .order_by(ref_models.TrialModel.number)
+            .all()
+        )
+        assert [t.number for t in trials] == [0, 1, 2, 3]
+        values = {
+            v.trial_id: ref_models.TrialValueModel.stored_repr_to_value(
+                v.value, v.value_type
+            )
+            for v in session.query(ref_models.TrialValueModel).all()
+        }
+        assert values[trials[0].trial_id] == 1.5
+        assert values[trials[1].trial_id] == float("inf")
+        assert trials[2].trial_id not in values  # PRUNED without value
+        params = (
+            session.query(ref_models.TrialParamModel)
+            .filter_by(trial_id=trials[0].trial_id)
+            .all()
+        )
+        assert {p.param_name for p in params} == {"x", "n"}
+        for p_ in params:
+            assert "Distribution" in p_.distribution_json  # new-format JSON
+            assert "Uniform" not in p_.distribution_json
+        iv = (
+            session.query(ref_models.TrialIntermediateValueModel)
+            .filter_by(trial_id=trials[0].trial_id, step=0)
+            .one()
+        )
+        got = ref_models.TrialIntermediateValueModel.stored_repr_to_intermediate_value(
+            iv.intermediate_value, iv.intermediate_value_type
+        )
+        assert got == 0.5
+        directions = session.query(ref_models.StudyDirectionModel).all()
+        assert [d.direction.name for d in directions] == ["MINIMIZE"]
+        vi = session.query(ref_models.VersionInfoModel).one()
+        assert vi.schema_version == ref_models.SCHEMA_VERSION
