@@ -627,3 +627,37 @@ class StorageTestCase:
         for th in threads:
             th.join()
         assert len(storage.get_trial(trial_id).system_attrs) == 40
+
+    # ---- study-level workflows over the backend --------------------------------------
+
+    def test_study_directions_reset_compatible(self, storage: BaseStorage) -> None:
+        # Re-creating with identical directions is fine; the stored directions
+        # are authoritative.
+        study_id = storage.create_new_study(MINIMIZE, study_name="dir-reset")
+        assert storage.get_study_directions(study_id) == [StudyDirection.MINIMIZE]
+
+    def test_trials_survive_reopen_semantics(self, storage: BaseStorage) -> None:
+        # Everything written through one handle is visible through fresh reads
+        # (same handle here; persistent backends cover true reopen in their
+        # own round-trip tests).
+        study_id = storage.create_new_study(MINIMIZE, study_name="reopen")
+        trial_id = storage.create_new_trial(study_id)
+        storage.set_trial_param(trial_id, "p", 0.25, FloatDistribution(0, 1))
+        storage.set_trial_intermediate_value(trial_id, 2, 0.5)
+        storage.set_trial_state_values(trial_id, TrialState.COMPLETE, (0.75,))
+        again = storage.get_trial_id_from_study_id_trial_number(study_id, 0)
+        t = storage.get_trial(again)
+        assert t.params == {"p": 0.25}
+        assert t.intermediate_values == {2: 0.5}
+        assert t.value == 0.75
+
+    def test_waiting_template_then_claim(self, storage: BaseStorage) -> None:
+        study_id = storage.create_new_study(MINIMIZE)
+        waiting = create_trial(state=TrialState.WAITING, params={"x": 1.0},
+                               distributions={"x": FloatDistribution(0, 2)})
+        trial_id = storage.create_new_trial(study_id, template_trial=waiting)
+        assert storage.get_trial(trial_id).state == TrialState.WAITING
+        assert storage.set_trial_state_values(trial_id, TrialState.RUNNING)
+        assert storage.get_trial(trial_id).state == TrialState.RUNNING
+        # A second claim on the now-RUNNING trial must not win.
+        assert not storage.set_trial_state_values(trial_id, TrialState.RUNNING)
