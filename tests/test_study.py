@@ -260,3 +260,43 @@ def test_trials_dataframe() -> None:
     assert len(df) == 3
     assert "params_x" in df.columns
     assert "value" in df.columns
+
+
+@pytest.mark.parametrize("mode", ["sqlite", "journal"])
+def test_copy_study_across_backends(mode, tmp_path) -> None:
+    """copy_study moves a full study (params, values, attrs) between storages."""
+    import warnings
+
+    from optuna_amd.testing.storages import StorageSupplier
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        src = optuna_amd.create_study(study_name="copy-src")
+        src.set_user_attr("tag", "v")
+        src.optimize(lambda t: t.suggest_float("x", 0, 1) ** 2, n_trials=5)
+        with StorageSupplier(mode) as dst_storage:
+            optuna_amd.copy_study(
+                from_study_name="copy-src",
+                from_storage=src._storage,
+                to_storage=dst_storage,
+            )
+            copied = optuna_amd.load_study(study_name="copy-src", storage=dst_storage)
+            assert [t.value for t in copied.trials] == [t.value for t in src.trials]
+            assert copied.user_attrs["tag"] == "v"
+
+
+def test_get_all_study_summaries_over_rdb(tmp_path) -> None:
+    import warnings
+
+    from optuna_amd.testing.storages import StorageSupplier
+
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        with StorageSupplier("sqlite") as storage:
+            for name in ("sum-a", "sum-b"):
+                s = optuna_amd.create_study(study_name=name, storage=storage)
+                s.optimize(lambda t: t.suggest_float("x", 0, 1), n_trials=2)
+            summaries = optuna_amd.get_all_study_summaries(storage)
+            names = sorted(sm.study_name for sm in summaries)
+            assert names == ["sum-a", "sum-b"]
+            assert all(sm.n_trials == 2 for sm in summaries)
