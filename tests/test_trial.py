@@ -275,3 +275,46 @@ def test_waiting_claim_race_under_threads() -> None:
         th.join()
     # every enqueued value claimed exactly once
     assert sorted(claimed[:n_queued]) == [float(i) for i in range(n_queued)]
+
+
+def test_frozen_trial_equality_and_repr() -> None:
+    import optuna_amd
+    from optuna_amd.distributions import FloatDistribution
+
+    t1 = optuna_amd.create_trial(
+        value=1.0, params={"x": 0.5}, distributions={"x": FloatDistribution(0, 1)}
+    )
+    t2 = optuna_amd.create_trial(
+        value=1.0, params={"x": 0.5}, distributions={"x": FloatDistribution(0, 1)}
+    )
+    t2._trial_id = t1._trial_id
+    t2.number = t1.number
+    t2.datetime_start = t1.datetime_start
+    t2.datetime_complete = t1.datetime_complete
+    assert t1 == t2
+    t2.params = {"x": 0.7}
+    assert t1 != t2
+    assert "x" in repr(t1)
+
+
+def test_frozen_trial_value_values_exclusive() -> None:
+    import optuna_amd
+
+    with pytest.raises(ValueError):
+        optuna_amd.create_trial(value=1.0, values=[1.0, 2.0])
+    t = optuna_amd.create_trial(values=[1.0, 2.0])
+    assert t.values == [1.0, 2.0]
+    with pytest.raises(RuntimeError):
+        _ = t.value  # multi-objective trials expose .values only
+
+
+def test_frozen_trial_duration_and_last_step() -> None:
+    import datetime as dt
+
+    import optuna_amd
+
+    t = optuna_amd.create_trial(value=0.0, intermediate_values={0: 1.0, 7: 2.0})
+    assert t.last_step == 7
+    t.datetime_start = dt.datetime(2026, 1, 1, 0, 0, 0)
+    t.datetime_complete = dt.datetime(2026, 1, 1, 0, 1, 30)
+    assert t.duration == dt.timedelta(seconds=90)
