@@ -1054,3 +1054,49 @@ def test_mo_tpe_uses_per_dim_device_path(core, monkeypatch) -> None:
     # below (gamma=300 >= 128) + above per suggest, D dims served per batch
     assert calls["per_dim"] >= 6
     assert calls["joint"] == 0
+
+
+def test_mo3_split_routes_hssp_to_device(core, monkeypatch) -> None:
+    """The 3-objective split's boundary-rank tie-break must go through the K6b
+    device greedy (a silent host fallback would still pass the goldens)."""
+    import warnings
+
+    from optuna_amd._hypervolume import hssp
+
+    calls = {"device": 0}
+    orig = hssp._solve_hssp_3d_device
+
+    def spy(*args, **kwargs):
+        out = orig(*args, **kwargs)
+        if out is not None:
+            calls["device"] += 1
+        return out
+
+    monkeypatch.setattr(hssp, "_solve_hssp_3d_device", spy)
+    warnings.simplefilter("ignore")
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(4)
+    names = [f"x{i}" for i in range(5)]
+    dists = {n: FloatDistribution(0.0, 1.0) for n in names}
+    study = optuna_amd.create_study(
+        directions=["minimize"] * 3,
+        sampler=optuna_amd.samplers.TPESampler(seed=0, n_startup_trials=5),
+    )
+    pm = rng.uniform(0, 1, size=(4000, 5))
+    study.add_trials(
+        [
+            optuna_amd.create_trial(
+                params={n: float(pm[r, i]) for i, n in enumerate(names)},
+                distributions=dists,
+                values=[float(pm[r, 0]), float(pm[r, 1]), float(1 - pm[r, 2])],
+            )
+            for r in range(4000)
+        ]
+    )
+
+    def objective(trial):
+        x = [trial.suggest_float(n, 0, 1) for n in names]
+        return x[0], x[1], 1 - x[2]
+
+    study.optimize(objective, n_trials=2)
+    assert calls["device"] >= 1
