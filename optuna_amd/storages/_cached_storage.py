@@ -168,6 +168,13 @@ class _CachedStorage(BaseStorage, BaseHeartbeat):
                     return copy.deepcopy(trial)
         return self._backend.get_trial(trial_id)
 
+    def get_n_trials(
+        self, study_id: int, state: "tuple[TrialState, ...] | TrialState | None" = None
+    ) -> int:
+        # Counts come straight from the backend (one COUNT query on RDB) — no
+        # cache materialization, always fresh.
+        return self._backend.get_n_trials(study_id, state)
+
     def get_all_trials(
         self,
         study_id: int,

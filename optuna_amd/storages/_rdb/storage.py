@@ -4,8 +4,8 @@ Parity: reference ``optuna/storages/_rdb/storage.py`` (RDBStorage :106, scoped
 sessions + commit/rollback :73, row-lock trial numbering :456-594, heartbeat
 :1041-1093, version manager :1096). The on-disk schema is v12-compatible
 (see models.py), so databases interoperate with the reference; schema versioning
-is handled by a lightweight version manager instead of alembic (upgrade of older
-schemas is not supported in this build and raises with a clear message).
+is handled by a lightweight version manager instead of alembic — pre-v12 files
+upgrade in place through the shape-driven chain in ``upgrade.py``.
 """
 from __future__ import annotations
 
@@ -541,6 +541,22 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
         with _create_scoped_session(self.scoped_session) as session:
             trial = self._get_trial_model(session, trial_id)
             return self._build_frozen_trial(session, trial)
+
+    def get_n_trials(
+        self, study_id: int, state: "tuple[TrialState, ...] | TrialState | None" = None
+    ) -> int:
+        """One COUNT query instead of materializing every trial row (the
+        samplers poll this once per suggest as their startup/delta check)."""
+        if isinstance(state, TrialState):
+            state = (state,)
+        with _create_scoped_session(self.scoped_session) as session:
+            self._get_study(session, study_id)
+            q = session.query(sqlalchemy.func.count(models.TrialModel.trial_id)).filter(
+                models.TrialModel.study_id == study_id
+            )
+            if state is not None:
+                q = q.filter(models.TrialModel.state.in_(list(state)))
+            return int(q.scalar() or 0)
 
     def get_all_trials(
         self,
