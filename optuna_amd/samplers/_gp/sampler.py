@@ -118,6 +118,7 @@ class GPSampler(BaseSampler):
         self._log_prior = prior.default_log_prior
         self._minimum_noise: float = prior.DEFAULT_MINIMUM_NOISE_VAR
         self._gprs_cache_list: list[gp.GPRegressor] | None = None
+        self._norm_history_cache: dict[str, Any] | None = None
         self._constraints_gprs_cache_list: list[gp.GPRegressor] | None = None
         if deterministic_objective:
             from optuna_amd._experimental import warn_experimental_argument
@@ -240,6 +241,51 @@ class GPSampler(BaseSampler):
                 )
         return params
 
+    def _normalized_history(
+        self,
+        study: "Study",
+        completed_trials: list[FrozenTrial],
+        search_space: dict[str, BaseDistribution],
+        internal_search_space: "gp_search_space.SearchSpace",
+    ) -> tuple[np.ndarray, np.ndarray]:
+        """(normalized params, raw values) with an incremental per-study cache.
+
+        Finished trials are immutable and the history is append-only, so at
+        5k observations the O(N·D) per-dim dict walks and the values gather
+        only run over the delta (a prefix-id comparison guards exactness; any
+        mismatch — new space, deleted study, out-of-order view — rebuilds).
+        """
+        key = tuple(search_space.items())
+        ids = [t._trial_id for t in completed_trials]
+        cache = self._norm_history_cache
+        if (
+            cache is not None
+            and cache["study"] == study._study_id
+            and cache["key"] == key
+            and len(ids) >= len(cache["ids"])
+            and ids[: len(cache["ids"])] == cache["ids"]
+        ):
+            n_old = len(cache["ids"])
+            if len(ids) > n_old:
+                new_trials = completed_trials[n_old:]
+                cache["X"] = np.vstack(
+                    [cache["X"], internal_search_space.get_normalized_params(new_trials)]
+                )
+                cache["Y"] = np.vstack(
+                    [cache["Y"], np.array([t.values for t in new_trials], dtype=float)]
+                )
+                cache["ids"] = ids
+        else:
+            cache = {
+                "study": study._study_id,
+                "key": key,
+                "ids": ids,
+                "X": internal_search_space.get_normalized_params(completed_trials),
+                "Y": np.array([t.values for t in completed_trials], dtype=float),
+            }
+            self._norm_history_cache = cache
+        return cache["X"], cache["Y"]
+
     def _fit_objective_gps(
         self,
         X: np.ndarray,
@@ -276,7 +322,9 @@ class GPSampler(BaseSampler):
         search_space: dict[str, BaseDistribution],
     ) -> dict[str, Any]:
         internal_search_space = gp_search_space.SearchSpace(search_space)
-        normalized_params = internal_search_space.get_normalized_params(completed_trials)
+        normalized_params, raw_values = self._normalized_history(
+            study, completed_trials, search_space, internal_search_space
+        )
         X_running = (
             internal_search_space.get_normalized_params(
                 running_trials, [_get_params(t) for t in running_trials]
@@ -287,9 +335,7 @@ class GPSampler(BaseSampler):
         signs = np.array(
             [-1.0 if d == StudyDirection.MINIMIZE else 1.0 for d in study.directions]
         )
-        standardized_score_vals, _, _ = _standardize_values(
-            signs * np.array([t.values for t in completed_trials])
-        )
+        standardized_score_vals, _, _ = _standardize_values(signs * raw_values)
 
         n_objectives = standardized_score_vals.shape[-1]
         gprs_list = self._fit_objective_gps(
