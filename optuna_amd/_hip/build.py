@@ -52,6 +52,7 @@ def build(force: bool = False, verbose: bool = True) -> Path:
         f"-I{pybind11.get_include()}",
         f"-I{py_include}",
         str(_SRC),
+        "-lrocblas",
         "-o",
         str(out),
     ]
