@@ -173,6 +173,8 @@ class GPRegressor:
         Used by the sampler to run acquisition evaluations on the MI355X once
         the history is large enough that posterior GEMMs dominate.
         """
+        x_aliased = self._X_all is self._X_train
+        y_aliased = self._y_all is self._y_train
         for name in (
             "_is_categorical",
             "_X_train",
@@ -190,6 +192,11 @@ class GPRegressor:
             t = getattr(self, name)
             if t is not None:
                 setattr(self, name, t.to(device))
+        # "no running rows appended" is tracked by identity; restore it.
+        if x_aliased:
+            self._X_all = self._X_train
+        if y_aliased:
+            self._y_all = self._y_train
         return self
 
     def kernel(
