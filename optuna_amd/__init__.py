@@ -36,6 +36,7 @@ from optuna_amd.version import __version__  # noqa: F401
 
 
 __all__ = [
+    "__version__",
     "FixedTrial",
     "FrozenTrial",
     "MaxTrialsCallback",

@@ -9,6 +9,7 @@ from optuna_amd.samplers._tpe.sampler import TPESampler
 
 
 __all__ = [
+    "nsgaii",
     "BaseGASampler",
     "BaseSampler",
     "BruteForceSampler",

@@ -131,6 +131,7 @@ def __getattr__(name: str):  # lazy to avoid import cycles / optional deps
 
 __all__ = [
     "BaseStorage",
+    "_CachedStorage",
     "InMemoryStorage",
     "RDBStorage",
     "JournalStorage",
