@@ -1147,3 +1147,48 @@ def test_fused_logei_matches_torch_path() -> None:
     np.testing.assert_allclose(f_tail, r_tail, rtol=1e-6)
     ft, gt = far.eval_acqf_batched_with_grad(cands[:4].copy())
     assert np.isfinite(ft).all() and np.isfinite(gt).all()
+
+
+def test_gp_sampler_uses_fused_logei(core, monkeypatch) -> None:
+    """Config-3-shaped GPSampler suggests must run acquisition through the
+    fused K5 session (not the torch op-by-op path)."""
+    import warnings
+
+    import torch
+
+    from optuna_amd._gp import acqf as acqf_mod
+
+    counts = {"fused": 0}
+    orig = acqf_mod.LogEI._fused_session
+
+    def spy(self):
+        out = orig(self)
+        if out is not None:
+            counts["fused"] += 1
+        return out
+
+    monkeypatch.setattr(acqf_mod.LogEI, "_fused_session", spy)
+    warnings.simplefilter("ignore")
+    optuna_amd.logging.set_verbosity(optuna_amd.logging.WARNING)
+    rng = np.random.RandomState(3)
+    names = [f"x{i}" for i in range(8)]
+    dists = {n: FloatDistribution(-3.0, 3.0) for n in names}
+    study = optuna_amd.create_study(
+        sampler=optuna_amd.samplers.GPSampler(seed=0, n_startup_trials=5)
+    )
+    pm = rng.uniform(-3, 3, size=(700, 8))
+    study.add_trials(
+        [
+            optuna_amd.create_trial(
+                params={n: float(pm[r, i]) for i, n in enumerate(names)},
+                distributions=dists,
+                value=float(np.sum((pm[r] - 0.5) ** 2)),
+            )
+            for r in range(700)
+        ]
+    )
+    study.optimize(
+        lambda t: sum((t.suggest_float(n, -3, 3) - 0.5) ** 2 for n in names),
+        n_trials=2,
+    )
+    assert counts["fused"] > 0
