@@ -122,3 +122,50 @@ def test_eq_hash() -> None:
     assert FloatDistribution(0, 1) == FloatDistribution(0, 1)
     assert hash(FloatDistribution(0, 1)) == hash(FloatDistribution(0, 1))
     assert FloatDistribution(0, 1) != FloatDistribution(0, 2)
+
+
+def test_float_log_and_step_mutually_exclusive() -> None:
+    with pytest.raises(ValueError):
+        FloatDistribution(1.0, 10.0, log=True, step=0.5)
+
+
+def test_int_log_and_step_mutually_exclusive() -> None:
+    with pytest.raises(ValueError):
+        IntDistribution(1, 10, log=True, step=2)
+
+
+def test_contains_edges() -> None:
+    d = FloatDistribution(-1.0, 1.0)
+    assert d._contains(-1.0) and d._contains(1.0)
+    assert not d._contains(-1.0000001) and not d._contains(1.0000001)
+    di = IntDistribution(2, 8, step=3)  # {2, 5, 8}
+    assert di._contains(2) and di._contains(8)
+    assert not di._contains(9)
+    c = CategoricalDistribution(("a", "b"))
+    assert c._contains(0) and c._contains(1) and not c._contains(2)
+
+
+def test_repr_round_trip() -> None:
+    for d in (
+        FloatDistribution(-2.0, 3.5),
+        FloatDistribution(1e-4, 1e2, log=True),
+        FloatDistribution(0.0, 1.0, step=0.25),
+        IntDistribution(-3, 12),
+        IntDistribution(1, 512, log=True),
+        CategoricalDistribution(("x", 1, None)),
+    ):
+        assert eval(repr(d)) == d  # noqa: S307 — repr is the constructor form
+
+
+def test_int_to_external_repr_is_int() -> None:
+    d = IntDistribution(0, 10)
+    v = d.to_external_repr(3.0)
+    assert isinstance(v, int) and v == 3
+
+
+def test_single_variants() -> None:
+    assert FloatDistribution(2.0, 2.0).single()
+    assert IntDistribution(5, 5).single()
+    assert CategoricalDistribution(("only",)).single()
+    assert not FloatDistribution(0.0, 1e-9).single()
+    assert FloatDistribution(1.0, 1.4, step=0.5).single()  # only one grid point
