@@ -19,15 +19,28 @@ def _free_port() -> int:
 
 
 def _make_store(port: int, world_size: int, is_master: bool):
+    import time
+
     from torch.distributed import TCPStore
 
-    return TCPStore(
-        "127.0.0.1",
-        port,
-        world_size,
-        is_master=is_master,
-        timeout=datetime.timedelta(seconds=60),
-    )
+    # The free-port probe races other tests/processes; the master retries a
+    # failed bind a few times before giving up.
+    last: Exception | None = None
+    for _ in range(5):
+        try:
+            return TCPStore(
+                "127.0.0.1",
+                port,
+                world_size,
+                is_master=is_master,
+                timeout=datetime.timedelta(seconds=60),
+            )
+        except (RuntimeError, OSError) as e:
+            last = e
+            if not is_master:
+                raise
+            time.sleep(0.5)
+    raise RuntimeError(f"TCPStore bind failed on {port}: {last}")
 
 
 class TestRcclStorageConformance(StorageTestCase):
