@@ -133,7 +133,13 @@ class FloatDistribution(BaseDistribution):
     def single(self) -> bool:
         if self.step is None:
             return self.low == self.high
-        return self.high - self.low < self.step
+        if self.low == self.high:
+            return True
+        # Exact decimal arithmetic: float subtraction would call e.g.
+        # [0.2, 0.3] step 0.1 single (0.3-0.2 = 0.0999...).
+        return (
+            decimal.Decimal(str(self.high)) - decimal.Decimal(str(self.low))
+        ) < decimal.Decimal(str(self.step))
 
     def _contains(self, param_value_in_internal_repr: float) -> bool:
         value = param_value_in_internal_repr
@@ -144,6 +150,22 @@ class FloatDistribution(BaseDistribution):
 
     def to_external_repr(self, param_value_in_internal_repr: float) -> float:
         return float(param_value_in_internal_repr)
+
+    def to_internal_repr(self, param_value_in_external_repr: float) -> float:
+        try:
+            internal = float(param_value_in_external_repr)
+        except (ValueError, TypeError) as e:
+            raise ValueError(
+                f"'{param_value_in_external_repr}' is not a valid type. "
+                "float-castable value is expected."
+            ) from e
+        if math.isnan(internal):
+            raise ValueError(f"`{param_value_in_external_repr}` is invalid value.")
+        if self.log and internal <= 0.0:
+            raise ValueError(
+                f"`{param_value_in_external_repr}` is invalid value for the case log=True."
+            )
+        return internal
 
 
 class IntDistribution(BaseDistribution):
@@ -184,13 +206,23 @@ class IntDistribution(BaseDistribution):
 
     def to_internal_repr(self, param_value_in_external_repr: int) -> float:
         try:
-            return float(param_value_in_external_repr)
+            internal = float(param_value_in_external_repr)
         except (TypeError, ValueError) as e:
             raise ValueError(
-                f"'{param_value_in_external_repr}' is not a valid value for IntDistribution."
+                f"'{param_value_in_external_repr}' is not a valid type. "
+                "float-castable value is expected."
             ) from e
+        if math.isnan(internal):
+            raise ValueError(f"`{param_value_in_external_repr}` is invalid value.")
+        if self.log and internal <= 0.0:
+            raise ValueError(
+                f"`{param_value_in_external_repr}` is invalid value for the case log=True."
+            )
+        return internal
 
     def single(self) -> bool:
+        if self.log:
+            return self.low == self.high
         return self.high - self.low < self.step
 
     def _contains(self, param_value_in_internal_repr: float) -> bool:
@@ -237,6 +269,21 @@ class CategoricalDistribution(BaseDistribution):
         index = int(param_value_in_internal_repr)
         return 0 <= index < len(self.choices)
 
+    def __eq__(self, other: Any) -> bool:
+        # Choice-wise comparison: tuple equality would miss NaN choices
+        # (nan != nan), but two NaN choices ARE the same category.
+        if not isinstance(other, BaseDistribution):
+            return NotImplemented
+        if type(self) is not type(other):
+            return False
+        if len(self.choices) != len(other.choices):
+            return False
+        return all(
+            _categorical_choice_equal(a, b) for a, b in zip(self.choices, other.choices)
+        )
+
+    __hash__ = BaseDistribution.__hash__
+
 
 # --------------------------------------------------------------------------------------
 # JSON codec (RDB / journal compatibility format)
@@ -256,6 +303,19 @@ _LEGACY_DECODERS = {
 def json_to_distribution(json_str: str) -> BaseDistribution:
     parsed = json.loads(json_str)
     if "name" not in parsed:
+        # Abbreviated format {"type": "float"|"int"|"categorical", ...}
+        # (reference distributions.py:589-607 — written by some ecosystem
+        # tools and older exports).
+        kind = parsed.get("type")
+        if kind == "categorical":
+            return CategoricalDistribution(parsed["choices"])
+        if kind in ("float", "int"):
+            kwargs = {
+                k: parsed[k] for k in ("low", "high", "log", "step") if k in parsed
+            }
+            if kind == "float":
+                return FloatDistribution(**kwargs)
+            return IntDistribution(**kwargs)
         raise ValueError(f"Invalid distribution JSON: {json_str}.")
     name = parsed["name"]
     attributes = parsed["attributes"]
@@ -278,25 +338,143 @@ def distribution_to_json(dist: BaseDistribution) -> str:
     return json.dumps({"name": type(dist).__name__, "attributes": attributes})
 
 
+def _categorical_choice_equal(
+    a: CategoricalChoiceType, b: CategoricalChoiceType
+) -> bool:
+    """Equality that treats two NaN choices as the same category."""
+    both_nan = (
+        isinstance(a, float)
+        and isinstance(b, float)
+        and math.isnan(a)
+        and math.isnan(b)
+    )
+    return bool(a == b) or both_nan
+
+
 def check_distribution_compatibility(
     dist_old: BaseDistribution, dist_new: BaseDistribution
 ) -> None:
     """Raise ValueError if the two distributions cannot share stored parameter values.
 
-    Same-class requirement; categorical choices must match exactly
+    Same-class requirement; log configuration must match for numerical
+    distributions; categorical choices must match exactly
     (reference distributions.py:623-654).
     """
     if dist_old.__class__ is not dist_new.__class__:
-        raise ValueError(
-            f"Cannot set different distribution kind to the same parameter name: "
-            f"{dist_old} != {dist_new}."
-        )
+        raise ValueError("Cannot set different distribution kind to the same parameter name.")
+    if isinstance(dist_old, (FloatDistribution, IntDistribution)):
+        assert isinstance(dist_new, (FloatDistribution, IntDistribution))
+        if dist_old.log != dist_new.log:
+            raise ValueError("Cannot set different log configuration to the same parameter name.")
     if isinstance(dist_old, CategoricalDistribution):
         assert isinstance(dist_new, CategoricalDistribution)
-        if dist_old.choices != dist_new.choices:
+        if dist_old != dist_new:
             raise ValueError(
                 CategoricalDistribution.__name__ + " does not support dynamic value space."
             )
+
+
+class UniformDistribution(FloatDistribution):
+    """Deprecated alias of ``FloatDistribution`` (pre-v3 API)."""
+
+    def __init__(self, low: float, high: float) -> None:
+        super().__init__(low=low, high=high)
+
+    def _asdict(self) -> dict[str, Any]:
+        d = dict(self.__dict__)
+        d.pop("log", None)
+        d.pop("step", None)
+        return d
+
+
+class LogUniformDistribution(FloatDistribution):
+    """Deprecated alias of ``FloatDistribution(log=True)`` (pre-v3 API)."""
+
+    def __init__(self, low: float, high: float) -> None:
+        super().__init__(low=low, high=high, log=True)
+
+    def _asdict(self) -> dict[str, Any]:
+        d = dict(self.__dict__)
+        d.pop("log", None)
+        d.pop("step", None)
+        return d
+
+
+class DiscreteUniformDistribution(FloatDistribution):
+    """Deprecated alias of ``FloatDistribution(step=q)`` (pre-v3 API)."""
+
+    def __init__(self, low: float, high: float, q: float) -> None:
+        super().__init__(low=low, high=high, step=q)
+
+    @property
+    def q(self) -> float:
+        assert self.step is not None
+        return self.step
+
+    def _asdict(self) -> dict[str, Any]:
+        d = dict(self.__dict__)
+        d.pop("log", None)
+        d["q"] = d.pop("step")
+        return d
+
+
+class IntUniformDistribution(IntDistribution):
+    """Deprecated alias of ``IntDistribution`` (pre-v3 API)."""
+
+    def __init__(self, low: int, high: int, step: int = 1) -> None:
+        super().__init__(low=low, high=high, step=step)
+
+    def _asdict(self) -> dict[str, Any]:
+        d = dict(self.__dict__)
+        d.pop("log", None)
+        return d
+
+
+class IntLogUniformDistribution(IntDistribution):
+    """Deprecated alias of ``IntDistribution(log=True)`` (pre-v3 API)."""
+
+    def __init__(self, low: int, high: int, step: int = 1) -> None:
+        super().__init__(low=low, high=high, log=True)
+
+    def _asdict(self) -> dict[str, Any]:
+        d = dict(self.__dict__)
+        d.pop("log", None)
+        return d
+
+
+def _convert_old_distribution_to_new_distribution(
+    distribution: BaseDistribution,
+    suppress_warning: bool = False,
+) -> BaseDistribution:
+    """Map a pre-v3 alias onto the current class (reference :709-790)."""
+    new_distribution: BaseDistribution
+    if isinstance(distribution, UniformDistribution):
+        new_distribution = FloatDistribution(distribution.low, distribution.high)
+    elif isinstance(distribution, LogUniformDistribution):
+        new_distribution = FloatDistribution(
+            distribution.low, distribution.high, log=True
+        )
+    elif isinstance(distribution, DiscreteUniformDistribution):
+        new_distribution = FloatDistribution(
+            distribution.low, distribution.high, step=distribution.q
+        )
+    elif isinstance(distribution, IntUniformDistribution):
+        new_distribution = IntDistribution(
+            distribution.low, distribution.high, step=distribution.step
+        )
+    elif isinstance(distribution, IntLogUniformDistribution):
+        new_distribution = IntDistribution(
+            distribution.low, distribution.high, log=True, step=distribution.step
+        )
+    else:
+        new_distribution = distribution
+    if new_distribution != distribution and not suppress_warning:
+        warnings.warn(
+            f"{distribution} is deprecated and internally converted to "
+            f"{new_distribution}.",
+            FutureWarning,
+        )
+    return new_distribution
 
 
 def _get_single_value(dist: BaseDistribution) -> Any:

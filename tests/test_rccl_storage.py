@@ -166,28 +166,39 @@ def _claim_worker(rank: int, world_size: int, port: int, trial_id: int, queue) -
 
 def test_waiting_claim_cas_across_processes() -> None:
     """Exactly one of N processes wins the WAITING→RUNNING claim."""
-    port = _free_port()
     ctx = multiprocessing.get_context("spawn")
-    queue = ctx.Manager().Queue()
     world = 3
-    procs = [
-        ctx.Process(target=_claim_worker, args=(rank, world, port, 0, queue))
-        for rank in range(world)
-    ]
-    for p in procs:
-        p.daemon = True
-        p.start()
-    for p in procs:
-        p.join(timeout=120)
-    for p in procs:
-        if p.is_alive():
-            p.terminate()
-    results = [queue.get(timeout=10) for _ in range(world)]
-    wins = []
-    for rank, won in results:
-        assert isinstance(won, bool), f"rank {rank} failed:\n{won}"
-        wins.append(won)
-    assert sum(wins) == 1
+    last_error = ""
+    # Under a loaded full-suite run the freshly probed port can be stolen by
+    # another test before the master binds it; retry the whole round on a new
+    # port instead of failing on infrastructure.
+    for _attempt in range(3):
+        port = _free_port()
+        queue = ctx.Manager().Queue()
+        procs = [
+            ctx.Process(target=_claim_worker, args=(rank, world, port, 0, queue))
+            for rank in range(world)
+        ]
+        for p in procs:
+            p.daemon = True
+            p.start()
+        for p in procs:
+            p.join(timeout=120)
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+        results = []
+        while len(results) < world:
+            try:
+                results.append(queue.get(timeout=10))
+            except Exception:
+                break
+        wins = [won for _rank, won in results if isinstance(won, bool)]
+        if len(wins) == world:
+            assert sum(wins) == 1
+            return
+        last_error = "\n".join(str(w) for _r, w in results if not isinstance(w, bool))
+    pytest.fail(f"CAS workers kept failing:\n{last_error}")
 
 
 def test_rccl_journal_checkpoint_roundtrip(tmp_path) -> None:
