@@ -73,8 +73,14 @@ __all__ = [
 
 
 def __getattr__(name: str):
+    """Lazy submodule access: ``optuna_amd.<submodule>`` works without an
+    explicit import (matching the reference package, whose heavyweight
+    subpackages are reachable as attributes)."""
     import importlib
 
-    if name in ("importance", "visualization", "artifacts", "terminator", "cli", "integration"):
+    if name.startswith("__"):
+        raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
+    try:
         return importlib.import_module(f"optuna_amd.{name}")
-    raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
+    except ModuleNotFoundError:
+        raise AttributeError(f"module {__name__!r} has no attribute {name!r}") from None

@@ -103,13 +103,19 @@ def compute_hypervolume(
 ) -> float:
     """Hypervolume dominated by ``loss_vals`` w.r.t. ``reference_point`` (minimize).
 
-    Points not strictly better than the reference point in every objective
+    Every point must dominate or equal the reference point coordinatewise;
+    anything else (including NaN anywhere — NaN comparisons are False) is a
+    ``ValueError``. Points merely equal to the reference on some axis
     contribute nothing and are dropped.
     """
+    if not np.all(loss_vals <= reference_point):
+        raise ValueError(
+            "All points must be coordinatewise <= the reference point "
+            "(NaN in either input also fails this check)."
+        )
     if not np.all(np.isfinite(reference_point)):
-        # An infinite reference point yields an infinite volume as long as at
-        # least one point exists.
-        return float("inf") if loss_vals.shape[0] > 0 else 0.0
+        # A +inf reference coordinate makes the dominated volume infinite.
+        return float("inf")
     assert loss_vals.shape[1] == reference_point.shape[0]
 
     within = np.all(loss_vals < reference_point, axis=-1)

@@ -141,29 +141,55 @@ class _SearchSpaceTransform:
     def _transform_numerical_param(
         self, value: float, distribution: BaseDistribution
     ) -> float:
-        assert isinstance(distribution, (FloatDistribution, IntDistribution))
-        if distribution.log and self._transform_log:
-            return math.log(value)
-        return float(value)
+        return _transform_numerical_param(value, distribution, self._transform_log)
 
     def _untransform_numerical_param(
         self, trans_value: float, distribution: BaseDistribution
     ) -> float | int:
-        assert isinstance(distribution, (FloatDistribution, IntDistribution))
-        if distribution.log and self._transform_log:
-            value = math.exp(trans_value)
-        else:
-            value = trans_value
+        return _untransform_numerical_param(trans_value, distribution, self._transform_log)
 
-        if isinstance(distribution, FloatDistribution):
-            if distribution.step is not None:
-                # Round onto the grid then clip (grid widened by step/2 in bounds).
-                k = round((value - distribution.low) / distribution.step)
-                value = distribution.low + k * distribution.step
-            value = float(min(max(value, distribution.low), distribution.high))
-            return value
-        else:
-            k = round((value - distribution.low) / distribution.step)
-            int_value = int(distribution.low + k * distribution.step)
-            int_value = min(max(int_value, distribution.low), distribution.high)
-            return int_value
+
+def _transform_numerical_param(
+    param: int | float, distribution: BaseDistribution, transform_log: bool
+) -> float:
+    assert isinstance(distribution, (FloatDistribution, IntDistribution))
+    if distribution.log and transform_log:
+        return math.log(param)
+    return float(param)
+
+
+def _untransform_numerical_param(
+    trans_param: float, distribution: BaseDistribution, transform_log: bool
+) -> int | float:
+    """Map one transformed axis value back into the distribution's domain.
+
+    Semantics match the reference (_transform.py:269-306): plain/log floats are
+    capped strictly below ``high`` (``nextafter``) so downstream half-open-range
+    code never sees the boundary, stepped domains snap onto the grid with
+    half-up rounding (NOT banker's rounding — a draw exactly between two grid
+    points goes up, giving every grid point equal width), and log ints round in
+    linear space after exponentiation.
+    """
+    assert isinstance(distribution, (FloatDistribution, IntDistribution))
+    if isinstance(distribution, FloatDistribution):
+        if distribution.log:
+            value = math.exp(trans_param) if transform_log else trans_param
+            if not distribution.single():
+                value = float(min(value, np.nextafter(distribution.high, -math.inf)))
+            return float(value)
+        if distribution.step is not None:
+            k = math.floor((trans_param - distribution.low) / distribution.step + 0.5)
+            value = distribution.low + k * distribution.step
+            return float(min(max(value, distribution.low), distribution.high))
+        if distribution.single():
+            return float(trans_param)
+        return float(min(trans_param, np.nextafter(distribution.high, -math.inf)))
+    else:
+        if distribution.log:
+            if not transform_log:
+                return int(trans_param)
+            int_value = int(round(math.exp(trans_param)))
+            return min(max(int_value, distribution.low), distribution.high)
+        k = math.floor((trans_param - distribution.low) / distribution.step + 0.5)
+        int_value = int(distribution.low + k * distribution.step)
+        return min(max(int_value, distribution.low), distribution.high)

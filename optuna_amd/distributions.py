@@ -471,7 +471,7 @@ def _convert_old_distribution_to_new_distribution(
     if new_distribution != distribution and not suppress_warning:
         warnings.warn(
             f"{distribution} is deprecated and internally converted to "
-            f"{new_distribution}.",
+            f"{new_distribution}. See https://github.com/optuna/optuna/issues/2941.",
             FutureWarning,
         )
     return new_distribution

@@ -66,9 +66,29 @@ def _configure_library_root_logger() -> None:
         root.propagate = False
 
 
+def _reset_library_root_logger() -> None:
+    """Undo ``_configure_library_root_logger`` (test/teardown hook)."""
+    global _default_handler
+    with _lock:
+        if _default_handler is None:
+            return
+        root = _get_library_root_logger()
+        root.removeHandler(_default_handler)
+        root.setLevel(logging.NOTSET)
+        _default_handler = None
+
+
 def get_logger(name: str) -> logging.Logger:
-    """Return a logger under the library namespace with the default handler set up."""
+    """Return a logger under the library namespace with the default handler set up.
+
+    Names under the reference package's namespace (``optuna`` / ``optuna.*``)
+    are remapped onto this library's namespace so drop-in callers keep
+    participating in :func:`set_verbosity` / propagation controls.
+    """
     _configure_library_root_logger()
+    lib = _get_library_name()
+    if name == "optuna" or name.startswith("optuna."):
+        name = lib + name[len("optuna"):]
     return logging.getLogger(name)
 
 

@@ -8,6 +8,7 @@ from typing import TYPE_CHECKING
 
 import numpy as np
 
+from optuna_amd._experimental import experimental_class
 from optuna_amd.pruners._base import BasePruner
 from optuna_amd.study._study_direction import StudyDirection
 from optuna_amd.trial import FrozenTrial
@@ -17,6 +18,7 @@ if TYPE_CHECKING:
     from optuna_amd.study import Study
 
 
+@experimental_class("2.8.0")
 class PatientPruner(BasePruner):
     """Allow the wrapped pruner to act only after ``patience`` steps without improvement."""
 

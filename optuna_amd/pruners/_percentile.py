@@ -120,7 +120,10 @@ class PercentilePruner(BasePruner):
             return False
 
         completed_trials = study.get_trials(deepcopy=False, states=(TrialState.COMPLETE,))
-        if len(completed_trials) < self._n_startup_trials:
+        n_trials = len(completed_trials)
+        if n_trials == 0:
+            return False
+        if n_trials < self._n_startup_trials:
             return False
 
         direction = study.direction

@@ -20,7 +20,7 @@ def _check_value(value: Any) -> float:
     try:
         return float(value)
     except (TypeError, ValueError) as e:
-        raise ValueError(
+        raise TypeError(
             f"The `value` argument is of type '{type(value).__name__}' but supposed to be a "
             "float."
         ) from e

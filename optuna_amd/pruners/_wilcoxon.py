@@ -15,6 +15,7 @@ from typing import TYPE_CHECKING
 
 import numpy as np
 
+from optuna_amd._experimental import experimental_class
 from optuna_amd.pruners._base import BasePruner
 from optuna_amd.study._study_direction import StudyDirection
 from optuna_amd.trial import FrozenTrial
@@ -24,6 +25,7 @@ if TYPE_CHECKING:
     from optuna_amd.study import Study
 
 
+@experimental_class("3.6.0")
 class WilcoxonPruner(BasePruner):
     """Prune when a signed-rank test concludes the trial is worse than the best one."""
 

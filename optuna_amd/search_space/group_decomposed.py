@@ -33,12 +33,14 @@ class _SearchSpaceGroup:
             if not overlap:
                 next_spaces.append(space)
                 continue
-            # Split the existing group into (inside-overlap, outside-overlap) parts.
+            # Split the existing group into (inside-overlap, outside-overlap)
+            # parts — intersection first, remainder after, keeping the group's
+            # position in the list.
             inside = {k: v for k, v in space.items() if k in overlap}
             outside = {k: v for k, v in space.items() if k not in overlap}
+            next_spaces.append(inside)
             if outside:
                 next_spaces.append(outside)
-            next_spaces.append(inside)
             dist_keys -= overlap
         if dist_keys:
             next_spaces.append({k: distributions[k] for k in distributions if k in dist_keys})

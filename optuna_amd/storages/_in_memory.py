@@ -264,7 +264,10 @@ class InMemoryStorage(BaseStorage):
         trial = self._get_trial(trial_id)
         if trial.state != TrialState.COMPLETE:
             return
-        assert trial.value is not None
+        if trial.value is None:
+            # Legal via direct storage calls (study.tell() validates, the raw
+            # set_trial_state_values() API does not require values).
+            return
         if math.isnan(trial.value):
             return  # NaN never becomes the incumbent
         incumbent_id = info.best_trial_id

@@ -5,12 +5,39 @@ Parity: reference ``optuna/study/_constrained_optimization.py``
 """
 from __future__ import annotations
 
-from typing import Sequence
+import warnings
+from typing import Any, Sequence
 
 from optuna_amd.trial import FrozenTrial
 
 
 _CONSTRAINTS_KEY = "constraints"
+
+
+def _get_constraints_from_system_attrs(system_attrs: dict[str, Any]) -> dict[str, float]:
+    """Constraint values from both storage formats.
+
+    The legacy format is one list under ``"constraints"`` (keys become their
+    positional index as a string); the current format is one attr per
+    constraint under ``"constraints:<key>"``, which wins on collision.
+    """
+    constraints: dict[str, float] = {}
+    legacy = system_attrs.get(_CONSTRAINTS_KEY)
+    if legacy is not None:
+        for i, v in enumerate(legacy):
+            constraints[str(i)] = v
+    prefix = _CONSTRAINTS_KEY + ":"
+    for attr_key, value in system_attrs.items():
+        if attr_key.startswith(prefix):
+            key = attr_key[len(prefix):]
+            if key in constraints:
+                warnings.warn("Overwrite an old format constraint.")
+            constraints[key] = value
+    return constraints
+
+
+def _is_constrained_optimization(trials: Sequence[FrozenTrial]) -> bool:
+    return any(len(t.constraints) > 0 for t in trials)
 
 
 def _get_constraints(trial: FrozenTrial) -> list[float] | None:
@@ -21,10 +48,10 @@ def _get_constraints(trial: FrozenTrial) -> list[float] | None:
 
 
 def _is_feasible(trial: FrozenTrial) -> bool:
-    constraints = _get_constraints(trial)
-    if constraints is None:
+    constraints = _get_constraints_from_system_attrs(trial.system_attrs)
+    if not constraints:
         return True
-    return all(c <= 0.0 for c in constraints)
+    return all(c <= 0.0 for c in constraints.values())
 
 
 def _get_feasible_trials(trials: Sequence[FrozenTrial]) -> list[FrozenTrial]:

@@ -126,8 +126,12 @@ class Trial(BaseTrial):
     @property
     def relative_params(self) -> dict[str, Any]:
         if self._relative_params is None:
+            from optuna_amd.pruners import _filter_study
+
             trial = self._get_latest_trial()
-            study = self.study
+            # Under HyperbandPruner the sampler must only observe sibling
+            # trials of this trial's bracket.
+            study = _filter_study(self.study, trial)
             self._relative_search_space = study.sampler.infer_relative_search_space(study, trial)
             self._relative_params = study.sampler.sample_relative(
                 study, trial, self._relative_search_space
@@ -263,9 +267,12 @@ class Trial(BaseTrial):
         elif self._is_relative_param(name, distribution):
             param_value = self._relative_params[name]  # type: ignore[index]
         else:
-            study = self.study
+            from optuna_amd.pruners import _filter_study
+
+            latest = self._get_latest_trial()
+            study = _filter_study(self.study, latest)
             param_value = study.sampler.sample_independent(
-                study, self._get_latest_trial(), name, distribution
+                study, latest, name, distribution
             )
 
         param_value_in_internal_repr = distribution.to_internal_repr(param_value)

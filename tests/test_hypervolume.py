@@ -33,8 +33,15 @@ def test_hv_dominated_points_ignored() -> None:
     assert compute_hypervolume(pts, ref) == pytest.approx(4.0)
 
 
-def test_hv_point_outside_ref() -> None:
+def test_hv_point_outside_ref_raises() -> None:
     pts = np.array([[1.0, 5.0], [2.0, 2.0]])
+    ref = np.array([4.0, 4.0])
+    with pytest.raises(ValueError):
+        compute_hypervolume(pts, ref)
+
+
+def test_hv_point_on_ref_boundary_contributes_zero() -> None:
+    pts = np.array([[1.0, 4.0], [2.0, 2.0]])
     ref = np.array([4.0, 4.0])
     assert compute_hypervolume(pts, ref) == pytest.approx(4.0)
 

@@ -109,7 +109,7 @@ def test_set_constraint() -> None:
     t.set_constraint([-1.0, 0.5])
     frozen = study._storage.get_trial(t._trial_id)
     assert frozen.system_attrs["constraints"] == [-1.0, 0.5]
-    assert frozen.constraints == {0: -1.0, 1: 0.5}
+    assert frozen.constraints == {"0": -1.0, "1": 0.5}
 
 
 def test_fixed_trial() -> None:
