@@ -76,20 +76,14 @@ def _adjust_discrete_high(low: float, high: float, step: float) -> float:
     d_r = d_high - d_low
     if d_r % d_step != 0:
         # Largest grid point low + k*step ≤ high, in exact decimal arithmetic
-        # (reference distributions.py:661-675; a quantize to the step's decimal
-        # places would corrupt non-decimal lows). float() rounds the exact grid
-        # point, so iterate to a fixed point — otherwise a JSON round-trip of
-        # the stored (rounded) high can renormalize to a smaller grid point.
+        # (reference distributions.py:660-675; a quantize to the step's decimal
+        # places would corrupt non-decimal lows). Note the float() rounding of
+        # the exact grid point means a re-ingested high may renormalize one
+        # more grid point down for irrational steps — the reference behaves
+        # identically, and stability there matters less than landing on the
+        # true k on first construction.
         k = d_r // d_step
         adjusted = float(k * d_step + d_low)
-        while (
-            k > 0
-            and (decimal.Decimal(str(adjusted)) - d_low) % d_step != 0
-        ):
-            k -= 1
-            adjusted = float(k * d_step + d_low)
-        if k == 0:
-            adjusted = float(low)
         warnings.warn(
             f"The distribution is specified by [{low}, {high}] and step={step}, but the range "
             f"is not divisible by `step`. It will be replaced by [{low}, {adjusted}]."

@@ -60,7 +60,10 @@ class FixedTrial(BaseTrial):
         value = self._params[name]
         param_value_in_internal_repr = distribution.to_internal_repr(value)
         if not distribution._contains(param_value_in_internal_repr):
-            raise ValueError(
+            import warnings
+
+            # The fixed value is still returned (matching the reference).
+            warnings.warn(
                 f"The value {value} of the parameter '{name}' is out of "
                 f"the range of the distribution {distribution}."
             )
@@ -79,6 +82,31 @@ class FixedTrial(BaseTrial):
 
     def set_system_attr(self, key: str, value: Any) -> None:
         self._system_attrs[key] = value
+
+    @property
+    def constraints(self) -> dict[str, float]:
+        """Constraint values as {key: value}; feasible iff all <= 0."""
+        from optuna_amd.study._constrained_optimization import (
+            _get_constraints_from_system_attrs,
+        )
+
+        return _get_constraints_from_system_attrs(self._system_attrs)
+
+    def set_constraint(self, key: str, value: Any) -> None:
+        """Record one named constraint value (feasible iff ≤ 0)."""
+        from optuna_amd.trial._frozen import _checked_constraint_value
+
+        value = _checked_constraint_value(key, value)
+        attr_key = f"constraints:{key}"
+        if attr_key in self._system_attrs:
+            import warnings
+
+            warnings.warn(
+                f"The constraint value is ignored because this constraint `key={key!r}` "
+                "is already set."
+            )
+            return
+        self._system_attrs[attr_key] = value
 
     @property
     def params(self) -> dict[str, Any]:

@@ -28,6 +28,22 @@ from optuna_amd.trial._state import TrialState
 
 _logger = _logging.get_logger(__name__)
 
+
+def _checked_constraint_value(key: Any, value: Any) -> float:
+    """Validate one constraint value: float-castable (TypeError otherwise),
+    never NaN (ValueError)."""
+    try:
+        value = float(value)
+    except (TypeError, ValueError):
+        raise TypeError(
+            f"The `value` argument is of type '{type(value)}' but supposed to be a float."
+        ) from None
+    if math.isnan(value):
+        raise ValueError(
+            f"Attempted to set a constraint for {key!r}, but NaN is not allowed."
+        )
+    return value
+
 # Value types that deepcopy may share (immutable or treated as such everywhere).
 _ATOMIC_TYPES = (int, float, str, bool, type(None), datetime.datetime, TrialState)
 
@@ -164,7 +180,8 @@ class FrozenTrial(BaseTrial):
         value = self._params[name]
         param_value_in_internal_repr = distribution.to_internal_repr(value)
         if not distribution._contains(param_value_in_internal_repr):
-            raise ValueError(
+            # The stored value is still replayed (matching the reference).
+            warnings.warn(
                 f"The value {value} of the parameter '{name}' is out of "
                 f"the range of the distribution {distribution}."
             )
@@ -291,16 +308,7 @@ class FrozenTrial(BaseTrial):
     def set_constraint(self, key: str, value: Any) -> None:
         """Record one named constraint value (feasible iff ≤ 0); float-castable
         values accepted, NaN rejected, repeated keys ignored with a warning."""
-        try:
-            value = float(value)
-        except (TypeError, ValueError):
-            raise TypeError(
-                f"The `value` argument is of type '{type(value)}' but supposed to be a float."
-            ) from None
-        if math.isnan(value):
-            raise ValueError(
-                f"Attempted to set a constraint for {key!r}, but NaN is not allowed."
-            )
+        value = _checked_constraint_value(key, value)
         attr_key = f"constraints:{key}"
         if attr_key in self._system_attrs:
             warnings.warn(

@@ -12,6 +12,7 @@ import datetime
 import warnings
 from typing import TYPE_CHECKING, Any, Sequence
 
+from optuna_amd import distributions as _distributions_mod
 from optuna_amd import logging as _logging
 from optuna_amd.distributions import (
     BaseDistribution,
@@ -23,8 +24,11 @@ from optuna_amd.distributions import (
     check_distribution_compatibility,
 )
 from optuna_amd.trial._base import BaseTrial
-from optuna_amd.trial._frozen import FrozenTrial
+from optuna_amd.trial._frozen import FrozenTrial, _checked_constraint_value
 from optuna_amd.trial._state import TrialState
+
+
+_UNSET = object()
 
 
 if TYPE_CHECKING:
@@ -237,14 +241,48 @@ class Trial(BaseTrial):
         self.storage.set_trial_system_attr(self._trial_id, key, value)
         self._cached_frozen_trial.system_attrs[key] = value
 
-    def set_constraint(self, constraints: Sequence[float]) -> None:
-        """Store constraint values for constrained samplers (``_CONSTRAINTS_KEY`` convention).
+    @property
+    def constraints(self) -> dict[str, float]:
+        """Constraint values as ``{key: value}``; feasible iff all ≤ 0."""
+        from optuna_amd.study._constrained_optimization import (
+            _get_constraints_from_system_attrs,
+        )
+
+        return _get_constraints_from_system_attrs(
+            self.storage.get_trial_system_attrs(self._trial_id)
+        )
+
+    def set_constraint(self, key: Any, value: Any = _UNSET) -> None:
+        """Record one named constraint value (feasible iff ≤ 0).
+
+        ``set_constraint("cost", 1.5)`` writes the per-key system attr
+        ``constraints:cost``; the single-argument legacy form
+        ``set_constraint([c0, c1, ...])`` writes the whole list under
+        ``constraints``. Both are visible through ``FrozenTrial.constraints``.
 
         Parity: reference trial/_trial.py:778-813.
         """
         from optuna_amd.study._constrained_optimization import _CONSTRAINTS_KEY
 
-        self.storage.set_trial_system_attr(self._trial_id, _CONSTRAINTS_KEY, list(constraints))
+        if value is _UNSET:
+            if isinstance(key, str) or not isinstance(key, Sequence):
+                raise TypeError(
+                    "set_constraint requires (key, value), or a single sequence of "
+                    "constraint values (legacy form)."
+                )
+            self.storage.set_trial_system_attr(self._trial_id, _CONSTRAINTS_KEY, list(key))
+            return
+
+        value = _checked_constraint_value(key, value)
+        constraint_key = f"{_CONSTRAINTS_KEY}:{key}"
+        if constraint_key in self.storage.get_trial_system_attrs(self._trial_id):
+            warnings.warn(
+                f"The constraint value is ignored because this constraint `key={key!r}` "
+                "is already set."
+            )
+            return
+        self.storage.set_trial_system_attr(self._trial_id, constraint_key, value)
+        self._cached_frozen_trial.system_attrs[constraint_key] = value
 
     # ---- the suggest core -----------------------------------------------------------
 
@@ -254,8 +292,10 @@ class Trial(BaseTrial):
         trial = self._cached_frozen_trial
 
         if name in trial.distributions:
-            # Already suggested in this trial: verify compatibility, replay.
+            # Already suggested in this trial: verify compatibility, replay the
+            # first call's value (warning if the spec quietly changed).
             check_distribution_compatibility(trial.distributions[name], distribution)
+            self._check_distribution(name, distribution)
             return trial.distributions[name].to_external_repr(
                 storage.get_trial_param(trial_id, name)
             )
@@ -263,7 +303,9 @@ class Trial(BaseTrial):
         if self._is_fixed_param(name, distribution):
             param_value = self._fixed_params[name]
         elif distribution.single():
-            param_value = _get_single_value(distribution)
+            # Via the module (not the local binding) so test doubles patched on
+            # optuna_amd.distributions are honored.
+            param_value = _distributions_mod._get_single_value(distribution)
         elif self._is_relative_param(name, distribution):
             param_value = self._relative_params[name]  # type: ignore[index]
         else:
@@ -281,6 +323,16 @@ class Trial(BaseTrial):
         trial.distributions[name] = distribution
         return param_value
 
+    def _check_distribution(self, name: str, distribution: BaseDistribution) -> None:
+        old_distribution = self._cached_frozen_trial.distributions.get(name, distribution)
+        if old_distribution != distribution:
+            warnings.warn(
+                f'Inconsistent parameter values for distribution with name "{name}"! '
+                "The values of the first call are used and later calls are ignored. "
+                f"Using these values: {old_distribution._asdict()}",
+                RuntimeWarning,
+            )
+
     def _is_fixed_param(self, name: str, distribution: BaseDistribution) -> bool:
         if name not in self._fixed_params:
             return False
@@ -292,7 +344,9 @@ class Trial(BaseTrial):
                 f"Fixed parameter '{name}' with value {param_value} is out of range "
                 f"for distribution {distribution}."
             )
-        return contained
+        # The fixed value is used even when out of range (matching the
+        # reference: an enqueued/fixed param always wins, with a warning).
+        return True
 
     def _is_relative_param(self, name: str, distribution: BaseDistribution) -> bool:
         if name not in self.relative_params:

@@ -119,8 +119,9 @@ def test_fixed_trial() -> None:
     assert t.suggest_int("i", 0, 10) == 3
     with pytest.raises(ValueError):
         t.suggest_float("missing", 0, 1)
-    with pytest.raises(ValueError):
-        t.suggest_float("x", 0.6, 1.0)  # out of range
+    with pytest.warns(UserWarning):
+        # Out of range: warned, but the fixed value is still returned.
+        assert t.suggest_float("x", 0.6, 1.0) == 0.5
     assert not t.should_prune()
     assert t.params == {"x": 0.5, "c": "b", "i": 3}
 
@@ -214,9 +215,9 @@ def test_suggest_precedence_fixed_out_of_range_warns_but_returns() -> None:
     with warnings.catch_warnings():
         warnings.simplefilter("ignore")
         study.optimize(objective, n_trials=1)
-    # Reference semantics: an out-of-range fixed param warns and falls through
-    # to the sampler — the enqueued 7.0 is NOT used.
-    assert 0.0 <= study.trials[0].params["x"] <= 1.0
+    # Reference semantics: an out-of-range fixed param warns but is still used
+    # (the enqueued value always wins over the sampler).
+    assert study.trials[0].params["x"] == 7.0
 
 
 def test_suggest_precedence_cached_over_everything() -> None:
