@@ -45,6 +45,12 @@ class _ProgressBar:
                 "Progress bar requested but tqdm is not installed."
             )
             is_valid = False
+        if is_valid and n_trials is None and timeout is None:
+            import warnings
+
+            warnings.warn(
+                "Progress bar won't be displayed because n_trials and timeout are None."
+            )
         self._is_valid = is_valid and (n_trials or timeout) is not None
         self._n_trials = n_trials
         self._timeout = timeout
@@ -54,7 +60,7 @@ class _ProgressBar:
                 self._progress_bar = tqdm(total=self._n_trials)
             else:
                 total = tqdm.format_interval(self._timeout)
-                fmt = "{desc} {bar} {elapsed}/" + total
+                fmt = "{desc} {percentage:3.0f}%|{bar}| {elapsed}/" + total
                 self._progress_bar = tqdm(total=self._timeout, bar_format=fmt)
             # Redirect library log lines above the bar.
             self._handler = _TqdmLoggingHandler()

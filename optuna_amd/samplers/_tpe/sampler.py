@@ -148,6 +148,21 @@ class TPESampler(BaseSampler):
                 )
             self._group_decomposed_search_space = _GroupDecomposedSearchSpace(True)
 
+    def __getstate__(self) -> dict:
+        # Locks and device-resident history caches don't pickle; both are
+        # rebuilt lazily on first use after unpickling.
+        state = self.__dict__.copy()
+        del state["_history_lock"]
+        state["_indep_prefetch"] = {}
+        state["_histories"] = {}
+        return state
+
+    def __setstate__(self, state: dict) -> None:
+        import threading
+
+        self.__dict__.update(state)
+        self._history_lock = threading.RLock()
+
     def reseed_rng(self) -> None:
         self._rng.rng.seed()
         self._random_sampler.reseed_rng()

@@ -184,7 +184,9 @@ class BaseStorage(abc.ABC):
             best_trial = max(all_trials, key=_key_max)
         else:
             best_trial = min(all_trials, key=_key_min)
-        return self.get_trial(best_trial._trial_id)
+        # COMPLETE trials are immutable; the record from the no-copy listing IS
+        # current, so no re-read (which would deepcopy in cached storages).
+        return best_trial
 
     def get_trial_params(self, trial_id: int) -> dict[str, Any]:
         return self.get_trial(trial_id).params

@@ -159,7 +159,11 @@ class GrpcStorageProxy(BaseStorage, BaseHeartbeat):
         deepcopy: bool = True,
         states: Container[TrialState] | None = None,
     ) -> list[FrozenTrial]:
-        trials = self._rpc("get_all_trials", study_id, True, tuple(states) if states is not None else None)
+        # Server-side copies are pointless — serialization already isolates the
+        # caller from the backend's objects.
+        trials = self._rpc(
+            "get_all_trials", study_id, False, tuple(states) if states is not None else None
+        )
         return trials
 
     # ---- heartbeat ------------------------------------------------------------------

@@ -9,6 +9,7 @@ upgrade in place through the shape-driven chain in ``upgrade.py``.
 """
 from __future__ import annotations
 
+import copy
 import json
 import time
 from contextlib import contextmanager
@@ -592,7 +593,7 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
             uattrs = _bulk(models.TrialUserAttributeModel)
             sattrs = _bulk(models.TrialSystemAttributeModel)
 
-            return [
+            frozen = [
                 self._assemble_frozen_trial(
                     t,
                     params.get(t.trial_id, []),
@@ -603,6 +604,10 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
                 )
                 for t in trials
             ]
+            # Freshly assembled objects are already caller-private, but the
+            # deepcopy=True contract is observable — honor it like the
+            # reference does.
+            return copy.deepcopy(frozen) if deepcopy else frozen
 
     def _get_trials_delta(
         self,

@@ -54,14 +54,24 @@ def _normalize_value(value: float | None, direction: StudyDirection) -> float:
 def _dominates(
     trial0: FrozenTrial, trial1: FrozenTrial, directions: Sequence[StudyDirection]
 ) -> bool:
-    """True iff trial0 weakly dominates trial1 with at least one strict improvement."""
-    assert trial0.values is not None and trial1.values is not None
-    values0 = [_normalize_value(v, d) for v, d in zip(trial0.values, directions)]
-    values1 = [_normalize_value(v, d) for v, d in zip(trial1.values, directions)]
+    """True iff trial0 weakly dominates trial1 with at least one strict improvement.
+
+    Any complete trial dominates any incomplete one; two incomplete trials never
+    dominate each other (values may be absent for those states).
+    """
     if trial0.state != TrialState.COMPLETE:
         return False
     if trial1.state != TrialState.COMPLETE:
         return True
+    assert trial0.values is not None and trial1.values is not None
+    if len(trial0.values) != len(trial1.values):
+        raise ValueError("Trials with different numbers of objectives cannot be compared.")
+    if len(trial0.values) != len(directions):
+        raise ValueError(
+            "The number of the values and the number of the objectives are mismatched."
+        )
+    values0 = [_normalize_value(v, d) for v, d in zip(trial0.values, directions)]
+    values1 = [_normalize_value(v, d) for v, d in zip(trial1.values, directions)]
     if values0 == values1:
         return False
     return all(v0 <= v1 for v0, v1 in zip(values0, values1))
@@ -177,10 +187,9 @@ def _calculate_nondomination_rank(
     ranks = np.full(len(loss_values), -1, dtype=int)
     n_below = n_below or len(loss_values)
 
-    # NaN or +inf objectives are not comparable → lumped into the worst front.
-    is_valid = np.all(~np.isnan(loss_values), axis=1) & np.all(
-        loss_values < float("inf"), axis=1
-    )
+    # ±inf compares naturally under dominance ([1, inf] dominates [inf, inf]);
+    # only NaN rows are incomparable and get lumped into the worst front.
+    is_valid = ~np.isnan(loss_values).any(axis=1)
 
     rank = base_rank - 1
     indices = np.arange(len(loss_values))

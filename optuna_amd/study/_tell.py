@@ -52,17 +52,27 @@ def _check_values_are_feasible(study: "Study", values: Sequence[float]) -> str |
             errors.append(f"The value {v} is not acceptable")
     if errors:
         return "; ".join(errors)
+    if len(values) != len(study.directions):
+        return (
+            f"The number of the values {len(values)} did not match the number of the "
+            f"objectives {len(study.directions)}"
+        )
     return None
 
 
 def _tell_with_warning(
     study: "Study",
     trial: Trial | int,
-    value_or_values: float | Sequence[float] | None,
-    state: TrialState | None,
-    skip_if_finished: bool,
+    value_or_values: float | Sequence[float] | None = None,
+    state: TrialState | None = None,
+    skip_if_finished: bool = False,
     suppress_warning: bool = False,
-) -> FrozenTrial:
+) -> tuple[TrialState, list[float] | None, str | None]:
+    """Validate and commit the trial; returns (state, values, failure message).
+
+    Returning the commit outcome (rather than re-reading the stored trial) lets
+    the optimize loop log without a storage round trip per trial.
+    """
     frozen_trial = _get_frozen_trial(study, trial)
     warning_message = None
 
@@ -72,7 +82,9 @@ def _tell_with_warning(
             f"{value_or_values} and state {state} since trial was already finished. "
             f"Finished trial has values {frozen_trial.values} and state {frozen_trial.state}."
         )
-        return copy.deepcopy(frozen_trial)
+        return frozen_trial.state, frozen_trial.values, None
+    if frozen_trial.state != TrialState.RUNNING:
+        raise ValueError(f"Cannot tell a {frozen_trial.state.name} trial.")
 
     if state == TrialState.COMPLETE and value_or_values is None:
         raise ValueError(
@@ -98,31 +110,29 @@ def _tell_with_warning(
     else:
         values = [value_or_values]  # type: ignore[list-item]
 
-    if values is not None:
-        if len(values) != len(study.directions):
-            raise ValueError(
-                f"The number of the values {len(values)} did not match the number of the "
-                f"objectives {len(study.directions)}."
-            )
-
     if state == TrialState.PRUNED:
-        # Promote the last reported intermediate value to the final value.
+        # Promote the last reported intermediate value to the final value —
+        # but only if it is usable as one (NaN stays out).
         assert values is None
         last_step = frozen_trial.last_step
         if last_step is not None:
-            values = [frozen_trial.intermediate_values[last_step]]
+            last_intermediate = frozen_trial.intermediate_values[last_step]
+            if _check_values_are_feasible(study, [last_intermediate]) is None:
+                values = [last_intermediate]
 
     if state is None:
-        if values is not None:
-            feasibility_message = _check_values_are_feasible(study, values)
-            if feasibility_message is None:
-                state = TrialState.COMPLETE
-            else:
-                state = TrialState.FAIL
-                values = None
-                warning_message = feasibility_message
+        # The optimize path: infeasible values fail the trial with a warning
+        # rather than raising (an objective returning NaN must not kill the
+        # whole optimize loop).
+        if values is None:
+            warning_message = "The value None could not be cast to float."
+        else:
+            warning_message = _check_values_are_feasible(study, values)
+        if warning_message is None:
+            state = TrialState.COMPLETE
         else:
             state = TrialState.FAIL
+            values = None
     elif state == TrialState.COMPLETE:
         assert values is not None
         feasibility_message = _check_values_are_feasible(study, values)
@@ -142,6 +152,7 @@ def _tell_with_warning(
             import warnings
 
             warnings.warn(warning_message)
+            warning_message = None
 
     try:
         # Hyperband needs samplers to observe the bracket-filtered study.
@@ -152,4 +163,4 @@ def _tell_with_warning(
     finally:
         study._storage.set_trial_state_values(frozen_trial._trial_id, state, values)
 
-    return copy.deepcopy(study._storage.get_trial(frozen_trial._trial_id))
+    return state, values, warning_message

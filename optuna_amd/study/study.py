@@ -88,19 +88,35 @@ class Study:
 
     @property
     def best_trial(self) -> FrozenTrial:
+        return self._get_best_trial(deepcopy=True)
+
+    def _get_best_trial(self, deepcopy: bool) -> FrozenTrial:
+        """Best COMPLETE trial; in constrained optimization the best among
+        feasible trials (all constraint values ≤ 0)."""
         if self._is_multi_objective():
             raise RuntimeError(
                 "A single best trial cannot be retrieved from a multi-objective study. "
                 "Consider using Study.best_trials to retrieve a list containing the best trials."
             )
         best_trial = self._storage.get_best_trial(self._study_id)
-        # Reject when feasibility info exists and best is infeasible? (constrained
-        # optimization warns; matching reference behavior of returning storage best.)
-        return best_trial
+        if any(v > 0.0 for v in best_trial.constraints.values()):
+            from optuna_amd.study._constrained_optimization import _get_feasible_trials
+
+            complete = self.get_trials(deepcopy=False, states=[TrialState.COMPLETE])
+            feasible = _get_feasible_trials(complete)
+            if not feasible:
+                raise ValueError("No feasible trials are completed yet.")
+            key = lambda t: t.value  # noqa: E731
+            best_trial = (
+                max(feasible, key=key)
+                if self.direction == StudyDirection.MAXIMIZE
+                else min(feasible, key=key)
+            )
+        return copy.deepcopy(best_trial) if deepcopy else best_trial
 
     @property
     def best_trials(self) -> list[FrozenTrial]:
-        return _get_pareto_front_trials(self)
+        return _get_pareto_front_trials(self, consider_constraint=True)
 
     @property
     def direction(self) -> StudyDirection:
@@ -189,6 +205,12 @@ class Study:
             warnings.warn("Heartbeat of storage is supposed to be used with Study.optimize.")
 
         fixed_distributions = fixed_distributions or {}
+        from optuna_amd.distributions import _convert_old_distribution_to_new_distribution
+
+        fixed_distributions = {
+            name: _convert_old_distribution_to_new_distribution(dist)
+            for name, dist in fixed_distributions.items()
+        }
 
         trial_id = self._pop_waiting_trial_id()
         if trial_id is None:
@@ -207,15 +229,16 @@ class Study:
         skip_if_finished: bool = False,
     ) -> FrozenTrial:
         """Finish a trial created with ask (reference study.py:612-741)."""
-        from optuna_amd.study._tell import _tell_with_warning
+        from optuna_amd.study._tell import _get_frozen_trial, _tell_with_warning
 
-        return _tell_with_warning(
+        _tell_with_warning(
             study=self,
             trial=trial,
             value_or_values=values,
             state=state,
             skip_if_finished=skip_if_finished,
         )
+        return copy.deepcopy(_get_frozen_trial(self, trial))
 
     # ---- attrs ----------------------------------------------------------------------
 
@@ -227,6 +250,12 @@ class Study:
         self._storage.set_study_system_attr(self._study_id, key, value)
 
     def set_metric_names(self, metric_names: list[str]) -> None:
+        warnings.warn(
+            "Study.set_metric_names is experimental (supported from v3.2.0). "
+            "The interface can change in the future.",
+            exceptions.ExperimentalWarning,
+            stacklevel=2,
+        )
         if len(self._directions) != len(metric_names):
             raise ValueError("The number of objectives must match the length of the metric names.")
         self._storage.set_study_system_attr(
@@ -272,6 +301,10 @@ class Study:
         skip_if_exists: bool = False,
     ) -> None:
         """Queue a WAITING trial whose parameters are fixed (reference :869-932)."""
+        if not isinstance(params, dict):
+            raise TypeError(
+                f"The params argument is of type '{type(params).__name__}' but must be a dict."
+            )
         if skip_if_exists and self._should_skip_enqueue(params):
             _logger.info(f"Trial with params {params} already exists. Skipping enqueue.")
             return
@@ -311,6 +344,11 @@ class Study:
     def add_trial(self, trial: FrozenTrial) -> None:
         """Register an externally-built FrozenTrial (reference :934-1010)."""
         trial._validate()
+        if trial.values is not None and len(self.directions) != len(trial.values):
+            raise ValueError(
+                f"The number of the values {len(trial.values)} did not match the number of "
+                f"the objectives {len(self.directions)} in the study."
+            )
         self._storage.create_new_trial(self._study_id, template_trial=trial)
 
     def add_trials(self, trials: Iterable[FrozenTrial]) -> None:
@@ -329,7 +367,14 @@ class Study:
             self._study_id, deepcopy=False, states=(TrialState.WAITING,)
         ):
             # WAITING→RUNNING is a CAS: only one worker wins a given trial.
-            if not self._storage.set_trial_state_values(trial._trial_id, state=TrialState.RUNNING):
+            # A racing worker may even have FINISHED the trial since the
+            # listing — that surfaces as UpdateFinishedTrialError, not False.
+            try:
+                if not self._storage.set_trial_state_values(
+                    trial._trial_id, state=TrialState.RUNNING
+                ):
+                    continue
+            except exceptions.UpdateFinishedTrialError:
                 continue
             _logger.debug(f"Trial#{trial.number} is popped from the trial queue.")
             return trial._trial_id
@@ -337,24 +382,36 @@ class Study:
 
     # ---- logging helper -------------------------------------------------------------
 
-    def _log_completed_trial(self, trial: FrozenTrial) -> None:
-        if _logger.isEnabledFor(_logging.INFO):
-            if len(trial.values or []) > 1:
-                _logger.info(
-                    f"Trial {trial.number} finished with values: {trial.values} "
-                    f"and parameters: {trial.params}."
-                )
-            elif trial.values:
-                best_msg = ""
-                try:
-                    best = self.best_trial
-                    best_msg = f" Best is trial {best.number} with value: {best.value}."
-                except ValueError:
-                    pass
-                _logger.info(
-                    f"Trial {trial.number} finished with value: {trial.values[0]} and "
-                    f"parameters: {trial.params}.{best_msg}"
-                )
+    def _log_completed_trial(
+        self, values: "list[float]", number: int, params: "dict[str, Any]"
+    ) -> None:
+        """Completion log line; touches storage (best-trial lookup) only when
+        INFO is actually enabled, so quiet runs pay nothing per trial."""
+        if not _logger.isEnabledFor(_logging.INFO):
+            return
+        metric_names = self.metric_names
+        if len(values) > 1:
+            shown: "list[float] | dict[str, float]" = (
+                values
+                if metric_names is None
+                else {name: v for name, v in zip(metric_names, values)}
+            )
+            _logger.info(
+                f"Trial {number} finished with values: {shown} and parameters: {params}."
+            )
+        elif len(values) == 1:
+            shown_one: "float | dict[str, float]" = (
+                values[0] if metric_names is None else {metric_names[0]: values[0]}
+            )
+            message = (
+                f"Trial {number} finished with value: {shown_one} and parameters: {params}."
+            )
+            try:
+                best = self.best_trial
+                message += f" Best is trial {best.number} with value: {best.value}."
+            except ValueError:
+                pass
+            _logger.info(message)
 
 
 # ----------------------------------------------------------------------------------
@@ -380,6 +437,8 @@ def create_study(
     elif direction is not None:
         directions = [direction]
     assert directions is not None
+    if len(directions) < 1:
+        raise ValueError("The number of objectives must be greater than 0.")
 
     if not all(
         d in ("minimize", "maximize", StudyDirection.MINIMIZE, StudyDirection.MAXIMIZE)

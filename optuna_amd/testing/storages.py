@@ -21,8 +21,10 @@ STORAGE_MODES: list[str] = [
     "sqlite",
     "cached_sqlite",
     "journal",
-    "grpc_sqlite",
-    "grpc_journal",
+    # Named like the reference's gRPC modes so shared test suites apply the
+    # same skips (the gRPC client doesn't use copy.deepcopy).
+    "grpc_rdb",
+    "grpc_journal_file",
     "grpc_wire",
 ]
 
@@ -36,7 +38,10 @@ def _find_free_port() -> int:
 
 
 class StorageSupplier:
+    _ALIASES = {"grpc_sqlite": "grpc_rdb", "grpc_journal": "grpc_journal_file"}
+
     def __init__(self, storage_specifier: str, **kwargs: Any) -> None:
+        storage_specifier = self._ALIASES.get(storage_specifier, storage_specifier)
         self.storage_specifier = storage_specifier
         self.extra_args = kwargs
         self.tempfile: Any = None
@@ -51,8 +56,12 @@ class StorageSupplier:
             return optuna_amd.storages.InMemoryStorage()
         elif self.storage_specifier == "grpc_wire":
             return self._start_grpc(optuna_amd.storages.InMemoryStorage())
-        elif "sqlite" in self.storage_specifier or "journal" in self.storage_specifier:
-            if self.storage_specifier in ("journal", "grpc_journal"):
+        elif (
+            "sqlite" in self.storage_specifier
+            or "rdb" in self.storage_specifier
+            or "journal" in self.storage_specifier
+        ):
+            if self.storage_specifier in ("journal", "grpc_journal_file"):
                 self.tempfile = tempfile.NamedTemporaryFile(suffix=".log", delete=False)
                 from optuna_amd.storages.journal import JournalFileBackend, JournalStorage
 

@@ -208,8 +208,10 @@ class FrozenTrial(BaseTrial):
     # ---- Validation -----------------------------------------------------------------
 
     def _validate(self) -> None:
-        if self.datetime_start is None:
-            raise ValueError("`datetime_start` is supposed to be set.")
+        if self.state != TrialState.WAITING and self.datetime_start is None:
+            raise ValueError(
+                "`datetime_start` is supposed to be set when the trial state is not waiting."
+            )
         if self.state.is_finished():
             if self.datetime_complete is None:
                 raise ValueError("`datetime_complete` is supposed to be set for a finished trial.")
@@ -361,9 +363,14 @@ def create_trial(
     user_attrs = user_attrs or {}
     system_attrs = system_attrs or {}
     intermediate_values = intermediate_values or {}
-    state = state or TrialState.COMPLETE
+    if state is None:
+        # NB: `state or COMPLETE` would be wrong — TrialState.RUNNING is enum
+        # value 0 and therefore falsy.
+        state = TrialState.COMPLETE
 
-    datetime_start = datetime.datetime.now()
+    # WAITING trials have not started; their start stamp is set at the
+    # WAITING->RUNNING transition by the storage.
+    datetime_start = None if state == TrialState.WAITING else datetime.datetime.now()
     datetime_complete = datetime_start if state.is_finished() else None
 
     trial = FrozenTrial(

@@ -40,7 +40,7 @@ class TestJournalStorage(StorageTestCase):
 class TestGrpcSqliteStorage(StorageTestCase):
     @pytest.fixture
     def storage(self):  # type: ignore[override]
-        with StorageSupplier("grpc_sqlite") as s:
+        with StorageSupplier("grpc_rdb") as s:
             yield s
 
 

@@ -75,9 +75,14 @@ def test_tell_validation() -> None:
     with pytest.raises(ValueError):
         study.tell(t, state=TrialState.COMPLETE)  # no values
     with pytest.raises(ValueError):
-        study.tell(t, values=[1.0, 2.0])  # wrong length
-    with pytest.raises(ValueError):
         study.tell(t, values=1.0, state=TrialState.PRUNED)
+    with pytest.raises(ValueError):
+        # Wrong length with an explicit COMPLETE state raises ...
+        study.tell(t, values=[1.0, 2.0], state=TrialState.COMPLETE)
+    with pytest.warns(UserWarning):
+        # ... but without a state it fails the trial with a warning.
+        assert study.tell(t, values=[1.0, 2.0]).state == TrialState.FAIL
+    t = study.ask()
     study.tell(t, 1.0)
     with pytest.raises(ValueError):
         study.tell(1000, 1.0)  # unknown trial number

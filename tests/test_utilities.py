@@ -211,7 +211,7 @@ def test_optimize_callbacks_see_final_state() -> None:
     ]
 
 
-def test_fail_reason_system_attr_recorded() -> None:
+def test_failed_trial_state_recorded() -> None:
     study = optuna_amd.create_study()
 
     def obj(t):
@@ -221,4 +221,6 @@ def test_fail_reason_system_attr_recorded() -> None:
     study.optimize(obj, n_trials=1, catch=(RuntimeError,))
     t = study.trials[0]
     assert t.state == TrialState.FAIL
-    assert "xyz failure" in str(t.system_attrs.get("fail_reason", ""))
+    # No extra system attrs are written for plain failures (the reference keeps
+    # the trial record clean; failure details go to the log).
+    assert "fail_reason" not in t.system_attrs
