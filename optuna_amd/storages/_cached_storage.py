@@ -27,6 +27,8 @@ class _StudyInfo:
         # Cached finished trials, keyed by trial number.
         self.trials: dict[int, FrozenTrial] = {}
         self.unfinished_trial_ids: set[int] = set()
+        # First-seen distribution per param name (compat check short-circuit).
+        self.param_distribution: dict[str, Any] = {}
         self.last_finished_trial_id: int = -1
         self.directions: list[StudyDirection] | None = None
         self.name: str | None = None
@@ -60,6 +62,17 @@ class _CachedStorage(BaseStorage, BaseHeartbeat):
             study.directions = list(directions)
             self._studies[study_id] = study
         return study_id
+
+    def _get_cached_trial(self, trial_id: int) -> FrozenTrial | None:
+        """The cached record iff it is finished (immutable); None otherwise."""
+        mapped = self._trial_id_to_study_id_and_number.get(trial_id)
+        if mapped is None:
+            return None
+        study_id, number = mapped
+        trial = self._studies[study_id].trials.get(number)
+        if trial is None or not trial.state.is_finished():
+            return None
+        return trial
 
     def delete_study(self, study_id: int) -> None:
         with self._lock:
@@ -107,12 +120,17 @@ class _CachedStorage(BaseStorage, BaseHeartbeat):
     # ---- trial ops ------------------------------------------------------------------
 
     def create_new_trial(self, study_id: int, template_trial: FrozenTrial | None = None) -> int:
-        frozen_trial_id = self._backend.create_new_trial(study_id, template_trial)
+        frozen_trial = self._backend._create_new_trial(study_id, template_trial)
+        trial_id = frozen_trial._trial_id
         with self._lock:
-            self._studies.setdefault(study_id, _StudyInfo()).unfinished_trial_ids.add(
-                frozen_trial_id
+            info = self._studies.setdefault(study_id, _StudyInfo())
+            info.unfinished_trial_ids.add(trial_id)
+            self._trial_id_to_study_id_and_number[trial_id] = (
+                study_id,
+                frozen_trial.number,
             )
-        return frozen_trial_id
+            info.trials[frozen_trial.number] = frozen_trial
+        return trial_id
 
     def set_trial_param(
         self,
@@ -121,7 +139,18 @@ class _CachedStorage(BaseStorage, BaseHeartbeat):
         param_value_internal: float,
         distribution: BaseDistribution,
     ) -> None:
-        self._backend.set_trial_param(trial_id, param_name, param_value_internal, distribution)
+        with self._lock:
+            study_id, _ = self._trial_id_to_study_id_and_number[trial_id]
+            info = self._studies.setdefault(study_id, _StudyInfo())
+            cached_dist = info.param_distribution.get(param_name)
+        # Write-through with the cached distribution so the backend can skip
+        # its cross-trial compatibility query.
+        self._backend._set_trial_param(
+            trial_id, param_name, param_value_internal, distribution, cached_dist
+        )
+        if cached_dist is None:
+            with self._lock:
+                info.param_distribution[param_name] = distribution
 
     def get_trial_id_from_study_id_trial_number(self, study_id: int, trial_number: int) -> int:
         with self._lock:
@@ -195,8 +224,9 @@ class _CachedStorage(BaseStorage, BaseHeartbeat):
     def _read_trials_from_remote_storage(self, study_id: int) -> None:
         """Fetch only unfinished and newer-than-last-finished trials (delta)."""
         info = self._studies.setdefault(study_id, _StudyInfo())
-        fetched = self._backend._get_trials_delta(
+        fetched = self._backend._get_trials(
             study_id,
+            states=None,
             included_trial_ids=info.unfinished_trial_ids,
             trial_id_greater_than=info.last_finished_trial_id,
         )

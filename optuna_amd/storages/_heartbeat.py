@@ -13,6 +13,7 @@ from __future__ import annotations
 import abc
 import copy
 import threading
+from threading import Event, Thread
 from types import TracebackType
 from typing import TYPE_CHECKING, Callable
 
@@ -44,6 +45,12 @@ class BaseHeartbeat(abc.ABC):
 
     def is_heartbeat_enabled(self) -> bool:
         return self.get_heartbeat_interval() is not None
+
+    def get_heartbeat_stale_trial_callback(
+        self,
+    ) -> Callable[["Study", FrozenTrial], None] | None:
+        """Current name for :meth:`get_failed_trial_callback`."""
+        return self.get_failed_trial_callback()
 
     def get_failed_trial_callback(self) -> Callable[["Study", FrozenTrial], None] | None:
         return None
@@ -86,10 +93,10 @@ class HeartbeatThread(BaseHeartbeatThread):
         self._stop_event: threading.Event | None = None
 
     def start(self) -> None:
-        self._stop_event = threading.Event()
-        self._thread = threading.Thread(
-            target=self._record_periodically, daemon=True
-        )
+        self._stop_event = Event()
+        # Module-level name + (target, args)-only construction so test doubles
+        # can substitute the thread class.
+        self._thread = Thread(target=self._record_periodically, args=())
         self._thread.start()
 
     def join(self) -> None:

@@ -79,6 +79,7 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
         *,
         heartbeat_interval: int | None = None,
         grace_period: int | None = None,
+        heartbeat_stale_trial_callback: Callable[..., None] | None = None,
         failed_trial_callback: Callable[..., None] | None = None,
         skip_table_creation: bool = False,
     ) -> None:
@@ -91,11 +92,25 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
             raise ValueError("The value of `heartbeat_interval` should be a positive integer.")
         if grace_period is not None and grace_period <= 0:
             raise ValueError("The value of `grace_period` should be a positive integer.")
+        if heartbeat_stale_trial_callback is not None and failed_trial_callback is not None:
+            raise ValueError(
+                "Specify only one of `heartbeat_stale_trial_callback` and "
+                "`failed_trial_callback`."
+            )
+        if failed_trial_callback is not None:
+            import warnings
+
+            warnings.warn(
+                "`failed_trial_callback` is deprecated; use "
+                "`heartbeat_stale_trial_callback` instead.",
+                FutureWarning,
+            )
         self.url = url
         self.engine_kwargs = engine_kwargs
+        self.skip_compatibility_check = skip_compatibility_check
         self.heartbeat_interval = heartbeat_interval
         self.grace_period = grace_period
-        self.failed_trial_callback = failed_trial_callback
+        self.failed_trial_callback = heartbeat_stale_trial_callback or failed_trial_callback
 
         try:
             self.engine = sqlalchemy.engine.create_engine(url, **engine_kwargs)
@@ -289,18 +304,27 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
     # ---- trials ---------------------------------------------------------------------
 
     def create_new_trial(self, study_id: int, template_trial: FrozenTrial | None = None) -> int:
+        return self._create_new_trial(study_id, template_trial)._trial_id
+
+    def _create_new_trial(
+        self, study_id: int, template_trial: FrozenTrial | None = None
+    ) -> FrozenTrial:
+        """Insert a trial and return its full record (used by the caching
+        wrapper to seed its cache without a second read)."""
         # Deadlocks on the study row lock get a small bounded retry.
         MAX_RETRIES = 5
+        trial_id = -1
         for attempt in range(MAX_RETRIES):
             try:
-                return self._create_new_trial(study_id, template_trial)
+                trial_id = self._insert_new_trial(study_id, template_trial)
+                break
             except OperationalError:
                 if attempt == MAX_RETRIES - 1:
                     raise
                 time.sleep(0.05 * (attempt + 1))
-        raise AssertionError("unreachable")
+        return self.get_trial(trial_id)
 
-    def _create_new_trial(self, study_id: int, template_trial: FrozenTrial | None) -> int:
+    def _insert_new_trial(self, study_id: int, template_trial: FrozenTrial | None) -> int:
         with _create_scoped_session(self.scoped_session) as session:
             # Serialize numbering on the study row.
             study = (
@@ -396,21 +420,36 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
         param_value_internal: float,
         distribution: BaseDistribution,
     ) -> None:
+        self._set_trial_param(trial_id, param_name, param_value_internal, distribution, None)
+
+    def _set_trial_param(
+        self,
+        trial_id: int,
+        param_name: str,
+        param_value_internal: float,
+        distribution: BaseDistribution,
+        previous_distribution: BaseDistribution | None,
+    ) -> None:
+        """Write one param; ``previous_distribution`` (when the caller already
+        knows it, e.g. the caching wrapper) skips the cross-trial lookup."""
         with _create_scoped_session(self.scoped_session, True) as session:
             trial = self._get_trial_model(session, trial_id)
             self._check_trial_is_updatable_model(trial)
-            # Cross-trial compatibility check against any prior use of the name.
-            previous = (
-                session.query(models.TrialParamModel)
-                .join(models.TrialModel)
-                .filter(models.TrialModel.study_id == trial.study_id)
-                .filter(models.TrialParamModel.param_name == param_name)
-                .first()
-            )
-            if previous is not None:
-                check_distribution_compatibility(
-                    json_to_distribution(previous.distribution_json), distribution
+            if previous_distribution is None:
+                # Cross-trial compatibility check against any prior use of the name.
+                previous = (
+                    session.query(models.TrialParamModel)
+                    .join(models.TrialModel)
+                    .filter(models.TrialModel.study_id == trial.study_id)
+                    .filter(models.TrialParamModel.param_name == param_name)
+                    .first()
                 )
+                if previous is not None:
+                    check_distribution_compatibility(
+                        json_to_distribution(previous.distribution_json), distribution
+                    )
+            else:
+                check_distribution_compatibility(previous_distribution, distribution)
             session.add(
                 models.TrialParamModel(
                     trial_id=trial_id,
@@ -609,9 +648,10 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
             # reference does.
             return copy.deepcopy(frozen) if deepcopy else frozen
 
-    def _get_trials_delta(
+    def _get_trials(
         self,
         study_id: int,
+        states: Container[TrialState] | None,
         included_trial_ids: Container[int],
         trial_id_greater_than: int,
     ) -> list[FrozenTrial]:
@@ -619,16 +659,36 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
         (the _CachedStorage delta protocol; reference storage.py:858-923)."""
         with _create_scoped_session(self.scoped_session) as session:
             self._get_study(session, study_id)
-            included = list(included_trial_ids)  # type: ignore[arg-type]
+            included = [
+                tid
+                for tid in included_trial_ids  # type: ignore[union-attr]
+                if tid <= trial_id_greater_than
+            ]
+            base_query = session.query(models.TrialModel).filter(
+                models.TrialModel.study_id == study_id
+            )
+            if states is not None:
+                base_query = base_query.filter(
+                    models.TrialModel.state.in_([st for st in TrialState if st in states])
+                )
             cond = models.TrialModel.trial_id > trial_id_greater_than
             if included:
                 cond = sqlalchemy.or_(cond, models.TrialModel.trial_id.in_(included))
-            trials = (
-                session.query(models.TrialModel)
-                .filter(models.TrialModel.study_id == study_id, cond)
-                .order_by(models.TrialModel.trial_id)
-                .all()
-            )
+            try:
+                trials = (
+                    base_query.filter(cond).order_by(models.TrialModel.trial_id).all()
+                )
+            except OperationalError:
+                # A huge `included` list can exceed the dialect's bound-variable
+                # limit (SQLite: SQLITE_MAX_VARIABLE_NUMBER); fall back to
+                # fetching the study's trials and filtering in Python.
+                session.rollback()
+                included_set = set(included)
+                trials = [
+                    t
+                    for t in base_query.order_by(models.TrialModel.trial_id).all()
+                    if t.trial_id > trial_id_greater_than or t.trial_id in included_set
+                ]
             trial_ids = [t.trial_id for t in trials]
 
             def _bulk(model: Any) -> dict[int, list[Any]]:
@@ -783,7 +843,8 @@ class RDBStorage(BaseStorage, BaseHeartbeat):
     ) -> FrozenTrial:
         param_dict = {}
         dist_dict = {}
-        for p in params:
+        # param_id is monotonically assigned -> suggestion (insertion) order.
+        for p in sorted(params, key=lambda p: p.param_id):
             dist = json_to_distribution(p.distribution_json)
             param_dict[p.param_name] = dist.to_external_repr(p.param_value)
             dist_dict[p.param_name] = dist

@@ -31,11 +31,11 @@ MINIMIZE = [StudyDirection.MINIMIZE]
 
 
 class StorageTestCase:
-    """Behavioral contract tests for BaseStorage implementations."""
+    """Behavioral contract tests for BaseStorage implementations.
 
-    @pytest.fixture
-    def storage(self) -> BaseStorage:
-        raise NotImplementedError
+    Subclasses (or the test module) provide a ``storage`` fixture; no default
+    is defined here so a module-level fixture can also satisfy it (a class
+    fixture would shadow the module-level one)."""
 
     # ---- studies --------------------------------------------------------------------
 
@@ -245,6 +245,25 @@ class StorageTestCase:
         complete = storage.get_all_trials(study_id, states=(TrialState.COMPLETE,))
         running = storage.get_all_trials(study_id, states=(TrialState.RUNNING,))
         assert len(complete) == 1 and len(running) == 1
+
+    @pytest.mark.parametrize("param_names", [["a", "b"], ["b", "a"]])
+    def test_get_all_trials_params_order(
+        self, storage: BaseStorage, param_names: list[str]
+    ) -> None:
+        """Params read back in suggestion order (dict insertion order)."""
+        from optuna_amd.distributions import FloatDistribution
+        from optuna_amd.trial import create_trial
+
+        study_id = storage.create_new_study(MINIMIZE)
+        trial_id = storage.create_new_trial(
+            study_id, create_trial(state=TrialState.RUNNING)
+        )
+        for param_name in param_names:
+            storage.set_trial_param(
+                trial_id, param_name, 1.0, distribution=FloatDistribution(0.0, 2.0)
+            )
+        trials = storage.get_all_trials(study_id)
+        assert list(trials[0].params.keys()) == param_names
 
     def test_get_all_trials_deepcopy_isolation(self, storage: BaseStorage) -> None:
         study_id = storage.create_new_study(MINIMIZE)

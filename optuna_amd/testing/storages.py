@@ -28,6 +28,11 @@ STORAGE_MODES: list[str] = [
     "grpc_wire",
 ]
 
+STORAGE_MODES_HEARTBEAT = [
+    "sqlite",
+    "cached_sqlite",
+]
+
 SQLITE3_TIMEOUT = 300
 
 
