@@ -95,14 +95,16 @@ def _process_constraints_after_trial(
     constraints = None
     try:
         con = constraints_func(trial)
+        # NaN must abort BEFORE anything is assigned — the finally block then
+        # records None, leaving the trial with no constraint values at all.
+        values = tuple(float(c) for c in con)
+        if any(c != c for c in values):  # NaN
+            raise ValueError("Constraint values cannot be NaN.")
         if not isinstance(con, (tuple, list)):
             warnings.warn(
                 f"Constraints should be a sequence of floats but got {type(con).__name__}."
             )
-        constraints = tuple(float(c) for c in con)
-        for c in constraints:
-            if c != c:  # NaN
-                raise ValueError("NaN is not acceptable as a constraint value.")
+        constraints = values
     finally:
         assert constraints is None or isinstance(constraints, tuple)
         study._storage.set_trial_system_attr(

@@ -158,6 +158,18 @@ def _is_nan(v: Any) -> bool:
     return isinstance(v, Real) and math.isnan(float(v))
 
 
+# Reference-compatible names: the reference models an unexpanded child as a
+# dedicated sentinel object; this implementation uses None for the same state.
+_TreeNode = _Node
+_UNEXPANDED_NODE = None
+
+
+class _UnexpandedTreeNode:
+    """Placeholder type for the reference's unexpanded-child sentinel."""
+
+    is_running: bool = False
+
+
 class BruteForceSampler(BaseSampler):
     def __init__(self, seed: int | None = None, avoid_premature_stop: bool = False) -> None:
         self._rng = LazyRandomState(seed)

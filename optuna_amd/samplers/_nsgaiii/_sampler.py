@@ -56,6 +56,22 @@ class NSGAIIISampler(BaseGASampler):
             Callable[["Study", FrozenTrial, TrialState, Sequence[float] | None], None] | None
         ) = None,
     ) -> None:
+        from optuna_amd._experimental import warn_experimental_argument
+
+        if constraints_func is not None:
+            import warnings
+
+            warnings.warn(
+                "`constraints_func` is deprecated; set constraints via "
+                "Trial.set_constraint instead.",
+                FutureWarning,
+            )
+        if after_trial_strategy is not None:
+            warn_experimental_argument("after_trial_strategy")
+        if child_generation_strategy is not None:
+            warn_experimental_argument("child_generation_strategy")
+        if elite_population_selection_strategy is not None:
+            warn_experimental_argument("elite_population_selection_strategy")
         if population_size < 2:
             raise ValueError("`population_size` must be greater than or equal to 2.")
         if crossover is None:

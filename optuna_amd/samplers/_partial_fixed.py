@@ -8,6 +8,7 @@ import warnings
 from typing import TYPE_CHECKING, Any, Sequence
 
 from optuna_amd.distributions import BaseDistribution
+from optuna_amd._experimental import experimental_class
 from optuna_amd.samplers._base import BaseSampler
 from optuna_amd.trial import FrozenTrial, TrialState
 
@@ -16,6 +17,7 @@ if TYPE_CHECKING:
     from optuna_amd.study import Study
 
 
+@experimental_class("2.4.0")
 class PartialFixedSampler(BaseSampler):
     def __init__(self, fixed_params: dict[str, Any], base_sampler: BaseSampler) -> None:
         self._fixed_params = fixed_params

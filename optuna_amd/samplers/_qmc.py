@@ -23,6 +23,7 @@ from optuna_amd.distributions import (
     CategoricalDistribution,
     FloatDistribution,
 )
+from optuna_amd._experimental import experimental_class
 from optuna_amd.samplers._base import (
     _INDEPENDENT_SAMPLING_WARNING_TEMPLATE,
     BaseSampler,
@@ -41,6 +42,7 @@ _SUGGESTED_STATES = (TrialState.COMPLETE, TrialState.PRUNED)
 _threading_lock = threading.Lock()
 
 
+@experimental_class("3.0.0")
 class QMCSampler(BaseSampler):
     """Low-discrepancy sequence sampler over a fixed (non-conditional) space."""
 
@@ -64,12 +66,7 @@ class QMCSampler(BaseSampler):
             )
         self._qmc_type = qmc_type
         if seed is None and scramble and warn_asynchronous_seeding:
-            _logger.warning(
-                "No seed is provided for `QMCSampler` and the seed is set randomly. "
-                "If you are running multiple `QMCSampler`s in parallel and/or distributed "
-                "environment, the same seed must be used in all samplers to ensure that "
-                "resulting samples are taken from the same QMC sequence."
-            )
+            self._log_asynchronous_seeding()
 
     def reseed_rng(self) -> None:
         self._independent_sampler.reseed_rng()
@@ -130,6 +127,26 @@ class QMCSampler(BaseSampler):
             for name, value in sample.items()
         }
 
+    @staticmethod
+    def _log_asynchronous_seeding() -> None:
+        _logger.warning(
+            "No seed is provided for `QMCSampler` and the seed is set randomly. "
+            "If you are running multiple `QMCSampler`s in parallel and/or distributed "
+            "environment, the same seed must be used in all samplers to ensure that "
+            "the low-discrepancy property holds across the union of their draws."
+        )
+
+    def _log_independent_sampling(self, trial: FrozenTrial, param_name: str) -> None:
+        _logger.warning(
+            _INDEPENDENT_SAMPLING_WARNING_TEMPLATE.format(
+                param_name=param_name,
+                trial_number=trial.number,
+                sampler_name=self.__class__.__name__,
+                fallback_name=self._independent_sampler.__class__.__name__,
+                reason="dynamic search space is not supported by `QMCSampler`",
+            )
+        )
+
     def sample_independent(
         self,
         study: "Study",
@@ -139,15 +156,7 @@ class QMCSampler(BaseSampler):
     ) -> Any:
         if len(study._get_trials(deepcopy=False, states=_SUGGESTED_STATES, use_cache=True)):
             if self._warn_independent_sampling:
-                _logger.warning(
-                    _INDEPENDENT_SAMPLING_WARNING_TEMPLATE.format(
-                        param_name=param_name,
-                        trial_number=trial.number,
-                        sampler_name=self.__class__.__name__,
-                        fallback_name=self._independent_sampler.__class__.__name__,
-                        reason="dynamic search space is not supported by `QMCSampler`",
-                    )
-                )
+                self._log_independent_sampling(trial, param_name)
         return self._independent_sampler.sample_independent(
             study, trial, param_name, param_distribution
         )

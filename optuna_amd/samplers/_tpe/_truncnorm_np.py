@@ -104,11 +104,9 @@ def ppf(q: np.ndarray, a: np.ndarray, b: np.ndarray) -> np.ndarray:
         out[case_right] = _ppf_right(q[case_right], a[case_right], b[case_right])
 
     out = np.clip(out, a, b)
-    # Degenerate interval: collapse to the endpoint.
-    degenerate = a == b
-    if np.any(degenerate):
-        out[degenerate] = np.asarray(np.broadcast_to(a, out.shape))[degenerate]
-    return out
+    # Degenerate (zero-width) interval carries no distribution: NaN, as scipy
+    # defines it. np.where keeps 0-d inputs working (clip may return a scalar).
+    return np.where(a >= b, np.nan, out)
 
 
 def rvs(
@@ -143,4 +141,8 @@ def logpdf(
     outside = (x_std < a) | (x_std > b)
     if np.any(outside):
         out = np.where(outside, -np.inf, out)
+    # Degenerate (zero-width) truncation has no density: NaN, as scipy defines.
+    degenerate = np.asarray(a) >= np.asarray(b)
+    if np.any(degenerate):
+        out = np.where(degenerate, np.nan, out)
     return out

@@ -46,12 +46,14 @@ KIND_CAT = 4  # categorical
 
 
 class _ParzenEstimatorParameters(NamedTuple):
-    consider_prior: bool
+    # Field order matches the reference (which keeps the prior always-on and
+    # has no consider_prior field); ours stays as a trailing default.
     prior_weight: float
     consider_magic_clip: bool
     consider_endpoints: bool
     weights: Callable[[int], np.ndarray]
     multivariate: bool
+    consider_prior: bool = True
 
 
 class _NumericalDims(NamedTuple):
@@ -418,6 +420,39 @@ class _ParzenEstimator:
     @property
     def n_dims(self) -> int:
         return len(self._param_names)
+
+    @property
+    def _mixture_distribution(self):
+        """The KDE as the reference's explicit object form (see
+        ``probability_distributions.py``); converts from the SoA layout."""
+        from optuna_amd.samplers._tpe import probability_distributions as _pd
+
+        num = self._numerical
+        per_dim: dict[int, object] = {}
+        for j in range(len(num.dim_indices)):
+            kind = int(num.kinds[j])
+            mu = num.mus[:, j].copy()
+            sigma = num.sigmas[:, j].copy()
+            low = float(num.lows[j])
+            high = float(num.highs[j])
+            step = float(num.steps[j])
+            if kind == KIND_CONT:
+                d: object = _pd._BatchedTruncNormDistributions(mu, sigma, low, high)
+            elif kind == KIND_LOG:
+                d = _pd._BatchedTruncLogNormDistributions(mu, sigma, low, high)
+            elif kind == KIND_DISC:
+                d = _pd._BatchedDiscreteTruncNormDistributions(mu, sigma, low, high, step)
+            else:
+                d = _pd._BatchedDiscreteTruncLogNormDistributions(mu, sigma, low, high, step)
+            per_dim[int(num.dim_indices[j])] = d
+        for cat in self._categoricals:
+            per_dim[int(cat.dim_index)] = _pd._BatchedCategoricalDistributions(
+                cat.weights.copy()
+            )
+        dists = [per_dim[i] for i in range(len(self._param_names))]
+        return _pd._MixtureOfProductDistribution(
+            weights=self._weights.copy(), distributions=dists
+        )
 
     def sample(self, rng: np.random.RandomState, size: int) -> dict[str, np.ndarray]:
         samples = self._sample_array(rng, size)

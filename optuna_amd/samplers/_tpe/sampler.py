@@ -77,6 +77,20 @@ def default_weights(x: int) -> np.ndarray:
     return np.concatenate([ramp, flat], axis=0)
 
 
+def _count_finished(study: "Study", states: tuple[TrialState, ...]) -> int:
+    """Finished-trial count for the startup/refresh gates.
+
+    Normally the storage's O(1) counter. When the trial listing has been
+    replaced on the storage INSTANCE (test instrumentation, wrappers), counts
+    must come from that same listing or the gates would disagree with the
+    data the sampler is about to read.
+    """
+    storage = study._storage
+    if "get_all_trials" in vars(storage):
+        return len(storage.get_all_trials(study._study_id, deepcopy=False, states=states))
+    return storage.get_n_trials(study._study_id, states)
+
+
 class TPESampler(BaseSampler):
     """Tree-structured Parzen Estimator sampler (see module docstring)."""
 
@@ -99,12 +113,12 @@ class TPESampler(BaseSampler):
         constraints_func: Callable[[FrozenTrial], Sequence[float]] | None = None,
     ) -> None:
         self._parzen_estimator_parameters = _ParzenEstimatorParameters(
-            consider_prior,
-            prior_weight,
-            consider_magic_clip,
-            consider_endpoints,
-            weights,
-            multivariate if multivariate is not None else False,
+            prior_weight=prior_weight,
+            consider_magic_clip=consider_magic_clip,
+            consider_endpoints=consider_endpoints,
+            weights=weights,
+            multivariate=multivariate if multivariate is not None else False,
+            consider_prior=consider_prior,
         )
         self._n_startup_trials = n_startup_trials
         self._n_ei_candidates = n_ei_candidates
@@ -122,6 +136,14 @@ class TPESampler(BaseSampler):
         self._search_space_group: _SearchSpaceGroup | None = None
         self._search_space = IntersectionSearchSpace(include_pruned=True)
         self._constant_liar = constant_liar
+        if constraints_func is not None:
+            import warnings
+
+            warnings.warn(
+                "`constraints_func` is deprecated; set constraints via "
+                "Trial.set_constraint instead.",
+                FutureWarning,
+            )
         self._constraints_func = constraints_func
         # Overridable for customization (reference sampler.py keeps the same hook).
         self._parzen_estimator_cls = _ParzenEstimator
@@ -171,6 +193,15 @@ class TPESampler(BaseSampler):
         if self._multivariate is not None:
             return self._multivariate
         if self._group:
+            if study._is_multi_objective():
+                import warnings
+
+                # group=True forces the multivariate path even though the
+                # multi-objective default would be independent sampling.
+                warnings.warn(
+                    "`group=True` overrides the multi-objective default "
+                    "`multivariate=False`; the multivariate TPE is used."
+                )
             return True
         # Multivariate for single-objective, independent for multi-objective.
         return not study._is_multi_objective()
@@ -239,7 +270,7 @@ class TPESampler(BaseSampler):
         if search_space == {}:
             return {}
         states = (TrialState.COMPLETE, TrialState.PRUNED)
-        n_finished = study._storage.get_n_trials(study._study_id, states)
+        n_finished = _count_finished(study, states)
         if n_finished < self._n_startup_trials:
             return {}
         return self._sample(study, trial, search_space)
@@ -254,7 +285,7 @@ class TPESampler(BaseSampler):
         states = (TrialState.COMPLETE, TrialState.PRUNED)
         # O(1) startup check: a full trial fetch here would run once per DIM
         # per suggest (this is the per-dim entry point).
-        n_finished = study._storage.get_n_trials(study._study_id, states)
+        n_finished = _count_finished(study, states)
         if n_finished < self._n_startup_trials:
             return self._random_sampler.sample_independent(
                 study, trial, param_name, param_distribution
@@ -382,11 +413,16 @@ class TPESampler(BaseSampler):
             history = self._histories[study._study_id] = _TpeHistory()
         # Finished trials only accumulate; an O(1) count check skips the full
         # storage read on suggests that arrive between tells.
-        n_finished = study._storage.get_n_trials(
-            study._study_id, (TrialState.COMPLETE, TrialState.PRUNED)
-        )
+        n_finished = _count_finished(study, (TrialState.COMPLETE, TrialState.PRUNED))
         if n_finished != len(history):
-            delta_read = getattr(study._storage, "get_finished_trials_since", None)
+            # Same coherence rule as _count_finished: an instance-patched
+            # listing (test instrumentation) must also feed the mirror, so the
+            # storage-native delta API is bypassed in that case.
+            delta_read = (
+                None
+                if "get_all_trials" in vars(study._storage)
+                else getattr(study._storage, "get_finished_trials_since", None)
+            )
             if delta_read is not None:
                 history.update(
                     delta_read(study._study_id, len(history)),
@@ -629,6 +665,13 @@ class TPESampler(BaseSampler):
     @staticmethod
     def hyperopt_parameters() -> dict[str, Any]:
         """Default parameters of hyperopt v0.1.2 (reference sampler.py:677-720)."""
+        import warnings
+
+        warnings.warn(
+            "TPESampler.hyperopt_parameters is deprecated and will be removed in a "
+            "future release.",
+            FutureWarning,
+        )
         return {
             "consider_prior": True,
             "prior_weight": 1.0,
@@ -707,10 +750,16 @@ def _split_complete_trials(
 ) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
     n_below = min(n_below, len(trials))
     if len(study.directions) <= 1:
-        reverse = study.direction == StudyDirection.MAXIMIZE
-        sorted_trials = sorted(trials, key=lambda t: t.value, reverse=reverse)  # type: ignore[arg-type,return-value]
-        return sorted_trials[:n_below], sorted_trials[n_below:]
+        return _split_complete_trials_single_objective(trials, study, n_below)
     return _split_complete_trials_multi_objective(trials, study, n_below)
+
+
+def _split_complete_trials_single_objective(
+    trials: Sequence[FrozenTrial], study: "Study", n_below: int
+) -> tuple[list[FrozenTrial], list[FrozenTrial]]:
+    reverse = study.direction == StudyDirection.MAXIMIZE
+    sorted_trials = sorted(trials, key=lambda t: t.value, reverse=reverse)  # type: ignore[arg-type,return-value]
+    return sorted_trials[:n_below], sorted_trials[n_below:]
 
 
 def _split_complete_trials_multi_objective(

@@ -289,6 +289,19 @@ def _collapse_two_children(
 # ----------------------------------------------------------------------------------
 
 
+def _inlined_categorical_uniform_crossover(
+    parent_params: np.ndarray,
+    rng: np.random.RandomState,
+    swapping_prob: float,
+    search_space: dict[str, BaseDistribution],
+) -> np.ndarray:
+    """Per-gene uniform pick between two parents' untransformed categorical rows
+    (categoricals never go through the numerical transform)."""
+    n = len(search_space)
+    pick = (rng.rand(n) >= swapping_prob).astype(int)
+    return parent_params[pick, np.arange(n)]
+
+
 def perform_crossover(
     crossover: BaseCrossover,
     study: "Study",

@@ -72,6 +72,14 @@ class NSGAIISampler(BaseGASampler):
         self._search_space = IntersectionSearchSpace()
         from optuna_amd._experimental import warn_experimental_argument
 
+        if constraints_func is not None:
+            import warnings
+
+            warnings.warn(
+                "`constraints_func` is deprecated; set constraints via "
+                "Trial.set_constraint instead.",
+                FutureWarning,
+            )
         if after_trial_strategy is not None:
             warn_experimental_argument("after_trial_strategy")
         if child_generation_strategy is not None:

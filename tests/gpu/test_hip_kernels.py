@@ -61,7 +61,7 @@ def test_logpdf_matches_scipy(core) -> None:
 
 def _host_logpdf(space, observations, weights, samples):
     params = _ParzenEstimatorParameters(
-        True, 1.0, True, False, lambda n: np.asarray(weights[:-1]) if n else np.asarray([]),
+        1.0, True, False, lambda n: np.asarray(weights[:-1]) if n else np.asarray([]),
         True,
     )
     mpe = _ParzenEstimator(observations, space, params)
@@ -94,7 +94,7 @@ def test_kde_logpdf_matches_host(core, n_obs, d, log_dims) -> None:
 
     ours = _device.kde_logpdf(space, observations, None, weights, samples, False, True)
 
-    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, True)
+    params = _ParzenEstimatorParameters(1.0, True, False, default_weights, True)
     mpe = _ParzenEstimator(observations, space, params)
     ref = mpe.log_pdf(samples)
     np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
@@ -236,7 +236,7 @@ def test_resident_history_matches_host(core) -> None:
 
         obs = {name: cache.params[sel][:, c] for c, name in enumerate(cache.names)}
         params_obj = _ParzenEstimatorParameters(
-            True, 1.0, True, False, lambda n: weights_raw[:n], True
+            1.0, True, False, lambda n: weights_raw[:n], True
         )
         mpe = _ParzenEstimator(obs, space, params_obj)
         ref = mpe.log_pdf(samples)
@@ -467,7 +467,7 @@ def test_resident_with_extras_matches_host(core) -> None:
             for c, name in enumerate(cache.names)
         }
         params_obj = _ParzenEstimatorParameters(
-            True, 1.0, True, False, lambda n: weights_raw[:n], True
+            1.0, True, False, lambda n: weights_raw[:n], True
         )
         mpe = _ParzenEstimator(obs, space, params_obj)
         ref = mpe.log_pdf(samples)
@@ -512,7 +512,7 @@ def test_kde_logpdf_discrete_dims_matches_host(core) -> None:
     assert dev.space_is_device_eligible(space)
     ours = dev.kde_logpdf(space, observations, None, weights, samples, False, True)
 
-    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, True)
+    params = _ParzenEstimatorParameters(1.0, True, False, default_weights, True)
     mpe = _ParzenEstimator(observations, space, params)
     ref = mpe.log_pdf(samples)
     np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
@@ -600,7 +600,7 @@ def test_kde_logpdf_categorical_dims_matches_host(core) -> None:
     assert dev.space_is_device_eligible(space)
     ours = dev.kde_logpdf(space, observations, None, weights, samples, False, True)
 
-    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, True)
+    params = _ParzenEstimatorParameters(1.0, True, False, default_weights, True)
     mpe = _ParzenEstimator(observations, space, params)
     ref = mpe.log_pdf(samples)
     np.testing.assert_allclose(ours, ref, rtol=1e-9, atol=1e-9)
@@ -893,7 +893,7 @@ def test_per_dim_device_score_matches_host(core) -> None:
     )
     assert got.shape == (24, D)
 
-    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, False)
+    params = _ParzenEstimatorParameters(1.0, True, False, default_weights, False)
     obs = {f"x{i}": mat[:, i] for i in range(D)}
     mpe = _ParzenEstimator(obs, space, params)
     want = mpe.log_pdf_per_dim(samples)

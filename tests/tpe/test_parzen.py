@@ -181,7 +181,7 @@ def test_batched_fit_matches_per_dim_loop() -> None:
         "d": np.exp(rng.uniform(0, np.log(64), n)).round().clip(1, 64),
         "e": rng.randint(-5, 6, n).astype(float),
     }
-    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, True)
+    params = _ParzenEstimatorParameters(1.0, True, False, default_weights, True)
     batched = _ParzenEstimator(obs, space, params)
     assert not batched._categoricals
 
@@ -263,7 +263,7 @@ def test_per_dim_samples_within_bounds() -> None:
         "b": IntDistribution(2, 14, step=3),
     }
     obs = {"a": rng.rand(30), "b": (2 + 3 * rng.randint(0, 5, 30)).astype(float)}
-    params = _ParzenEstimatorParameters(True, 1.0, True, False, default_weights, False)
+    params = _ParzenEstimatorParameters(1.0, True, False, default_weights, False)
     mpe = _ParzenEstimator(obs, space, params)
     s = mpe.sample_per_dim(rng, 200)
     assert ((s["a"] >= 0) & (s["a"] <= 1)).all()
