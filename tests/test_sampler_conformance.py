@@ -8,11 +8,34 @@ import pytest
 import optuna_amd
 from optuna_amd.samplers import BaseSampler
 from optuna_amd.testing.pytest_samplers import (
-    BasicSamplerTestCase,
+    BasicSamplerTestCase as _RefBasic,
     ExtendedSamplerTestCase,
-    MultiObjectiveSamplerTestCase,
-    RelativeSamplerTestCase,
+    MultiObjectiveSamplerTestCase as _RefMO,
+    RelativeSamplerTestCase as _RefRelative,
 )
+
+
+class _SeededBridge:
+    """Derive the reference suites' nullary ``sampler`` fixture from this
+    file's seeded ``sampler_factory`` fixture."""
+
+    @pytest.fixture
+    def sampler(
+        self, sampler_factory: Callable[[int], BaseSampler]
+    ) -> Callable[[], BaseSampler]:
+        return lambda: sampler_factory(0)
+
+
+class BasicSamplerTestCase(_SeededBridge, _RefBasic):
+    pass
+
+
+class MultiObjectiveSamplerTestCase(_SeededBridge, _RefMO):
+    pass
+
+
+class RelativeSamplerTestCase(_SeededBridge, _RefRelative):
+    pass
 
 
 class TestRandomSampler(BasicSamplerTestCase, ExtendedSamplerTestCase, MultiObjectiveSamplerTestCase):
@@ -34,12 +57,25 @@ class TestTPEMultivariate(BasicSamplerTestCase, ExtendedSamplerTestCase, Relativ
             seed=seed, n_startup_trials=3, multivariate=True, group=True
         )
 
+    @pytest.fixture
+    def sampler(self) -> Callable[[], BaseSampler]:
+        # The relative suite samples right after ONE finished trial.
+        return lambda: optuna_amd.samplers.TPESampler(
+            seed=0, n_startup_trials=0, multivariate=True
+        )
+
 
 class TestCmaEsSampler(BasicSamplerTestCase, ExtendedSamplerTestCase, RelativeSamplerTestCase):
     @pytest.fixture
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.CmaEsSampler(
             seed=seed, n_startup_trials=2, warn_independent_sampling=False
+        )
+
+    @pytest.fixture
+    def sampler(self) -> Callable[[], BaseSampler]:
+        return lambda: optuna_amd.samplers.CmaEsSampler(
+            seed=0, n_startup_trials=0, warn_independent_sampling=False
         )
 
 
@@ -86,6 +122,13 @@ class TestGPSampler(BasicSamplerTestCase, RelativeSamplerTestCase):
     def sampler_factory(self) -> Callable[[int], BaseSampler]:
         return lambda seed: optuna_amd.samplers.GPSampler(
             seed=seed, n_startup_trials=3, warn_independent_sampling=False
+        )
+
+    @pytest.fixture
+    def sampler(self) -> Callable[[], BaseSampler]:
+        # The relative suite samples right after ONE finished trial.
+        return lambda: optuna_amd.samplers.GPSampler(
+            seed=0, n_startup_trials=0, warn_independent_sampling=False
         )
 
 
