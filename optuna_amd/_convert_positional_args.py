@@ -13,11 +13,30 @@ from typing import Any, Callable, Sequence, TypeVar
 FT = TypeVar("FT", bound=Callable[..., Any])
 
 
+def _validate_version(version: str) -> None:
+    parts = version.split(".")
+    if len(parts) < 2 or not all(p.split("a")[0].split("b")[0].isdigit() for p in parts if p):
+        raise ValueError(f"Invalid version string: {version!r}.")
+
+
 def convert_positional_args(
     *,
     previous_positional_arg_names: Sequence[str],
+    deprecated_version: str | None = None,
+    removed_version: str | None = None,
     warning_stacklevel: int = 2,
 ) -> Callable[[FT], FT]:
+    if deprecated_version is not None or removed_version is not None:
+        if deprecated_version is None:
+            raise ValueError(
+                "deprecated_version must not be None when removed_version is specified."
+            )
+        if removed_version is None:
+            raise ValueError(
+                "removed_version must not be None when deprecated_version is specified."
+            )
+        _validate_version(deprecated_version)
+        _validate_version(removed_version)
     def decorator(func: FT) -> FT:
         sig = signature(func)
         kwonly = {

@@ -180,3 +180,11 @@ def round_one_normalized_param(
         (raw - bounds[0] + 0.5 * step) // step * step + bounds[0], bounds[0], bounds[1]
     )
     return normalize_one_param(raw, scale_type, bounds, step)
+
+
+# Reference-private names for the same functions.
+_normalize_one_param = normalize_one_param
+_unnormalize_one_param = unnormalize_one_param
+_round_one_normalized_param = round_one_normalized_param
+
+_ScaleType = ScaleType

@@ -50,6 +50,13 @@ def get_param_importances(
     if not isinstance(evaluator, BaseImportanceEvaluator):
         raise TypeError("Evaluator must be a subclass of BaseImportanceEvaluator.")
 
+    from optuna_amd.trial import TrialState
+
+    if not study.get_trials(deepcopy=False, states=(TrialState.COMPLETE,)):
+        # Nothing to attribute yet: an empty mapping, not an error (matching
+        # the reference; evaluators may still raise for explicit `params`).
+        return {}
+
     res = evaluator.evaluate(study, params=params, target=target)
     if normalize:
         s = sum(res.values())

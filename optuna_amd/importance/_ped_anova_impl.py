@@ -176,6 +176,10 @@ class PedAnovaImportanceEvaluator(BaseImportanceEvaluator):
         self._prior_weight = 1.0
         self._min_n_trials_in_regime = 2
 
+    def _get_top_quantile_trials(self, study, trials, quantile, target):
+        """Reference-public name of :meth:`_top_quantile_trials`."""
+        return self._top_quantile_trials(study, trials, quantile, target)
+
     def _top_quantile_trials(
         self,
         study: "Study",
@@ -233,9 +237,8 @@ class PedAnovaImportanceEvaluator(BaseImportanceEvaluator):
     ) -> dict[str, float]:
         completed = study.get_trials(deepcopy=False, states=(TrialState.COMPLETE,))
         if len(completed) == 0:
-            raise ValueError(
-                "Cannot evaluate parameter importances without completed trials."
-            )
+            # No attributable signal yet (matching the reference).
+            return {}
         if params is None:
             # All parameters appearing in completed trials, incl. conditional ones.
             seen: dict[str, None] = {}
