@@ -85,10 +85,15 @@ def test_target_callable() -> None:
     assert max(imp, key=imp.get) == "small"  # type: ignore[arg-type]
 
 
-def test_no_completed_trials_raises() -> None:
+def test_no_completed_trials_returns_empty() -> None:
     study = optuna_amd.create_study()
-    with pytest.raises(ValueError):
-        get_param_importances(study, evaluator=MeanDecreaseImpurityImportanceEvaluator())
+    study.add_trial(
+        optuna_amd.create_trial(
+            state=optuna_amd.trial.TrialState.RUNNING,
+        )
+    )
+    # Reference semantics: nothing to attribute yet -> empty mapping.
+    assert optuna_amd.importance.get_param_importances(study) == {}
 
 
 def test_pedanova_conditional_params() -> None:
