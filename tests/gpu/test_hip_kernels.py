@@ -560,7 +560,7 @@ def test_resident_discrete_matches_host(core) -> None:
 
     obs = {name: cache.params[sel][:, c] for c, name in enumerate(cache.names)}
     params_obj = _ParzenEstimatorParameters(
-        True, 1.0, True, False, lambda n: weights_raw[:n], True
+        1.0, True, False, lambda n: weights_raw[:n], True
     )
     mpe = _ParzenEstimator(obs, space, params_obj)
     ref = mpe.log_pdf(samples)
